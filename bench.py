@@ -1,0 +1,237 @@
+#!/usr/bin/env python3
+"""bench.py — BASELINE.json's headline benchmark on MI355X.
+
+Metric: "GiB/s encode+CRC throughput, RS(k+m) per stripe, 1/2/4/8 MI355X
+vs CPU".  A step is one pass of the hot path over one resident stripe
+batch: RS encode of every stripe (one fused HIP kernel over the batch)
+followed by crc32block framing of all k+m shards (the blobnode write path,
+datafile.go:342).  Default N=1 workload is BASELINE configs[1]:
+RS(6+3), 8 MiB shards, 1024 stripes, plus the CRC leg the metric names.
+
+`value` = whole-job SOURCE throughput: k·S·stripes·ranks / elapsed (GiB/s).
+The klauspost "total shard size" convention (vendor README.md:443) is
+reported alongside in config.total_shard_gib_s.
+
+Multi-GPU: stripes are independent; each rank owns its own
+batch (weak scaling), RCCL is used only for the start/stop barrier and the
+result allgather.  Inputs are synthetic seeded uniform bytes resident in
+HBM before the timed region.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+GIB = float(1 << 30)
+HBM_PEAK = 8.0e12  # B/s, MI355X spec (MI355X_MICROARCH.md chip parameters)
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=4)
+    p.add_argument("--warmup", type=int, default=2)
+    p.add_argument("--stripes", type=int, default=1024)
+    p.add_argument("--shard-mib", type=int, default=8)
+    p.add_argument("--codemode", default="EC6P3")
+    p.add_argument("--no-crc", action="store_true",
+                   help="EC encode only (config label form)")
+    p.add_argument("--cpu-sample-stripes", type=int, default=24)
+    p.add_argument("--skip-cpu-baseline", action="store_true")
+    return p.parse_args()
+
+
+def hbm_traffic_lookup(workload):
+    """Per-launch HBM bytes measured by rocprofv3 --pmc (separate passes,
+    gfx950 FETCH_SIZE correction per MI355X_MICROARCH.md §HBM), committed
+    under profiles/.  Null when no measurement exists for this workload."""
+    path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                        "profiles", "hbm_traffic.json")
+    try:
+        with open(path) as f:
+            d = json.load(f)
+        return d.get(workload, {}).get("bytes_per_launch")
+    except (OSError, ValueError):
+        return None
+
+
+def cpu_baseline_leg(t, shard_len, nstripes_sample, with_crc):
+    """Oracle (nibble-table algorithm, OpenMP all host cores) timed on the
+    GPU box's CPUs — the reported baseline, kind='port'."""
+    import numpy as np
+    from oracle import pyoracle as po
+    cores = po.threads_avail()
+    rng = np.random.default_rng(1)
+    stripes = []
+    for _ in range(nstripes_sample):
+        st = [rng.integers(0, 256, shard_len, dtype=np.uint8) for _ in range(t.N)]
+        st += [np.zeros(shard_len, np.uint8) for _ in range(t.M + t.L)]
+        stripes.append(st)
+    if with_crc:
+        enc_sz = po.crc32b_encode_size(shard_len, 65536)
+        dsts = [[np.zeros(enc_sz, np.uint8) for _ in st] for st in stripes]
+    t0 = time.perf_counter()
+    po.rs_encode_mt(t.N, t.M, stripes)
+    if with_crc:
+        import ctypes
+        L = po.lib()
+        flatd = [d for ds in dsts for d in ds]
+        flats = [s for st in stripes for s in st]
+        pd = (ctypes.POINTER(ctypes.c_uint8) * len(flatd))(
+            *[d.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)) for d in flatd])
+        ps = (ctypes.POINTER(ctypes.c_uint8) * len(flats))(
+            *[s.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)) for s in flats])
+        L.orc_crc32b_encode_mt(pd, ps, shard_len, 65536, len(flats), 0)
+    el = time.perf_counter() - t0
+    src_gib = t.N * shard_len * nstripes_sample / GIB
+    return {
+        "value": round(src_gib / el, 3),
+        "unit": "GiB/s",
+        "cores": cores,
+        "kind": "port",
+        "sample": "%d stripes RS(%d+%d) %d MiB shards%s, oracle AVX2 "
+                  "nibble-table + OpenMP, %.1fs" % (
+                      nstripes_sample, t.N, t.M, shard_len >> 20,
+                      "" if not with_crc else " + crc32block framing", el),
+    }
+
+
+def main():
+    args = parse_args()
+    import numpy as np
+    import torch
+
+    from cubefs_amd import codemode, crc32block, dist, ec
+    from cubefs_amd.runtime import lib
+
+    rank, world = dist.env_rank_world()
+    d = dist.init_process_group()
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    torch.cuda.set_device(local_rank)
+    dev = torch.device("cuda", local_rank)
+
+    t = codemode.get_tactic(args.codemode)
+    S = args.shard_mib << 20
+    ns = args.stripes
+    with_crc = not args.no_crc
+
+    enc = ec.Encoder(t, device=local_rank)
+    # run everything on torch's current stream so HIP events bracket the
+    # kernels (torch.cuda.Event sees only torch's stream)
+    cur = torch.cuda.current_stream(dev).cuda_stream
+    lib().gfrs_set_stream(enc._ctx, cur)
+    codec = None
+    if with_crc:
+        codec = crc32block.Codec(device=local_rank)
+        lib().gfrs_set_stream(codec._ctx, cur)
+
+    # synthetic input resident in HBM (seeded per SURVEY.md §8d)
+    g = torch.Generator(device=dev)
+    g.manual_seed(0xB10B5703 ^ rank)
+    batch = torch.empty((ns, t.total, S), dtype=torch.uint8, device=dev)
+    batch[:, :t.N].random_(0, 256, generator=g)
+    framed = None
+    enc_sz = crc32block.encode_size(S)
+    if with_crc:
+        framed = torch.empty((ns * t.total, enc_sz), dtype=torch.uint8,
+                             device=dev)
+
+    flat = batch.view(ns * t.total, S)
+
+    def step(events=None):
+        if events:
+            events[0].record()
+        enc.encode_batch(batch)
+        if events:
+            events[1].record()
+        if with_crc:
+            codec.encode_batch(framed, flat)
+
+    # warmup
+    for _ in range(args.warmup):
+        step()
+    torch.cuda.synchronize(dev)
+    dist.barrier()
+
+    evs = [(torch.cuda.Event(enable_timing=True),
+            torch.cuda.Event(enable_timing=True)) for _ in range(args.steps)]
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(evs[i])
+    torch.cuda.synchronize(dev)
+    elapsed = time.perf_counter() - t0
+    dist.barrier()
+
+    # MAX over ranks
+    if d is not None:
+        te = torch.tensor([elapsed], device=dev)
+        import torch.distributed as td
+        td.all_reduce(te, op=td.ReduceOp.MAX)
+        elapsed = float(te.item())
+
+    enc_ms = [a.elapsed_time(b) for a, b in evs]
+    rec = {"rank": rank, "stripes": ns, "src_bytes": t.N * S * ns * args.steps}
+    recs = dist.allgather_records(rec)
+
+    if rank == 0:
+        total_src = sum(r["src_bytes"] for r in recs)
+        value = total_src / GIB / elapsed
+        total_shard = total_src / t.N * t.total / GIB / elapsed
+        workload = "RS(%d+%d)%s, %d MiB shards, %d stripes/GPU" % (
+            t.N, t.M, "+crc32block" if with_crc else "", args.shard_mib, ns)
+        # roofline of the dominant kernel (rs_apply over the whole batch,
+        # one launch per step): algorithmic bytes = (k+m)·S·stripes
+        alg_bytes = float(t.total * S * ns)
+        avg_enc_s = (sum(enc_ms) / len(enc_ms)) / 1e3
+        achieved = alg_bytes / avg_enc_s
+        traffic = hbm_traffic_lookup(workload)
+        cpu_base = None
+        if not args.skip_cpu_baseline and world == 1:
+            cpu_base = cpu_baseline_leg(t, S, args.cpu_sample_stripes, with_crc)
+        out = {
+            "metric": "GiB/s encode+CRC throughput, RS(k+m) per stripe, "
+                      "1/2/4/8 MI355X vs CPU",
+            "value": round(value, 2),
+            "unit": "GiB/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed * 1e3 / args.steps, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "u8",
+            "data": "synthetic",
+            "config": {
+                "workload": workload,
+                "codemode": "EC%dP%d" % (t.N, t.M),
+                "shard_size": S,
+                "stripes_per_gpu": ns,
+                "crc_included": with_crc,
+                "block_len": 65536,
+                "value_convention": "source GiB/s = k*S*stripes/t",
+                "total_shard_gib_s": round(total_shard, 2),
+                "rs_kernel_ms": round(sum(enc_ms) / len(enc_ms), 3),
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": round(achieved / 1e9, 1),
+                "peak": round(HBM_PEAK / 1e9, 1),
+                "unit": "GB/s",
+                "frac": round(achieved / HBM_PEAK, 4),
+                "traffic": traffic,
+            },
+            "cpu_baseline": cpu_base,
+        }
+        print(json.dumps(out))
+
+    if d is not None:
+        import torch.distributed as td
+        td.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,853 @@
+/* cubefs_amd/csrc/gfrs_host.cpp — C-ABI host runtime of the MI355X EC/CRC
+ * engine (include/gfrs.h).
+ *
+ * Mirrors the blobstore/common/ec host semantics:
+ *   ec.NewEncoder / encoder / lrcEncoder     (encoder.go:78-112, lrcencoder.go)
+ *   Encode / Verify / Reconstruct[Data]      (encoder.go:114-151)
+ *   LRC layering: global RS(n,m) + per-AZ RS((n+m)/az, l/az)
+ *                                            (lrcencoder.go:35-82,133-207)
+ *   inversionTree decode-matrix cache keyed by the missing-index set
+ *                                            (inversion_tree.go:16-164) —
+ *                                            here a bitmask-keyed map
+ *   codemode local stripe layout             (codemode.go:301-318)
+ *
+ * GPU-only by design: no CPU compute fallback exists.  When no HIP device
+ * is visible every compute call fails with GFRS_ERR_NO_GPU.
+ */
+#include "../../include/gfrs.h"
+#include "gfrs_internal.h"
+
+#include <cstdarg>
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+#include <map>
+#include <mutex>
+#include <set>
+#include <string>
+#include <vector>
+
+namespace gfrs {
+
+static thread_local std::string g_err;
+
+static void seterr(const char *fmt, ...) {
+  char buf[512];
+  va_list ap;
+  va_start(ap, fmt);
+  vsnprintf(buf, sizeof(buf), fmt, ap);
+  va_end(ap);
+  g_err = buf;
+}
+
+static int hip_fail(const char *what, hipError_t e) {
+  seterr("%s: %s", what, hipGetErrorString(e));
+  return GFRS_ERR_HIP;
+}
+
+#define HIP_TRY(call)                                    \
+  do {                                                   \
+    hipError_t e_ = (call);                              \
+    if (e_ != hipSuccess) return hip_fail(#call, e_);    \
+  } while (0)
+
+/* per-device one-time CRC table init */
+static std::mutex g_dev_mu;
+static std::set<int> g_dev_inited;
+
+static int ensure_device_init(int device) {
+  std::lock_guard<std::mutex> lk(g_dev_mu);
+  if (g_dev_inited.count(device)) return GFRS_OK;
+  HIP_TRY(hipSetDevice(device));
+  int rc = crc_device_init_current();
+  if (rc != 0) {
+    seterr("crc_device_init_current failed");
+    return GFRS_ERR_HIP;
+  }
+  g_dev_inited.insert(device);
+  return GFRS_OK;
+}
+
+/* Grow-only device buffer. */
+struct DevBuf {
+  void *p = nullptr;
+  size_t cap = 0;
+  int ensure(size_t n) {
+    if (n <= cap) return GFRS_OK;
+    if (p) hipFree(p);
+    p = nullptr;
+    cap = 0;
+    HIP_TRY(hipMalloc(&p, n));
+    cap = n;
+    return GFRS_OK;
+  }
+  ~DevBuf() {
+    if (p) hipFree(p);
+  }
+};
+
+struct PinBuf {
+  void *p = nullptr;
+  size_t cap = 0;
+  int ensure(size_t n) {
+    if (n <= cap) return GFRS_OK;
+    if (p) hipHostFree(p);
+    p = nullptr;
+    cap = 0;
+    HIP_TRY(hipHostMalloc(&p, n));
+    cap = n;
+    return GFRS_OK;
+  }
+  ~PinBuf() {
+    if (p) hipHostFree(p);
+  }
+};
+
+/* An uploaded coding plan: index lists + per-coefficient nibble tables. */
+struct DevPlan {
+  DevBuf in_idx, out_idx, tabs;
+  int k = 0, nout = 0;
+  int upload(const std::vector<int32_t> &in, const std::vector<int32_t> &out,
+             const std::vector<uint8_t> &rows /* nout*k coefficients */,
+             hipStream_t s) {
+    const GfTables &t = gft();
+    k = int(in.size());
+    nout = int(out.size());
+    std::vector<uint8_t> tb(size_t(nout) * k * 32);
+    for (int r = 0; r < nout; r++)
+      for (int c = 0; c < k; c++) {
+        uint8_t coef = rows[size_t(r) * k + c];
+        memcpy(&tb[(size_t(r) * k + c) * 32], t.lo[coef], 16);
+        memcpy(&tb[(size_t(r) * k + c) * 32 + 16], t.hi[coef], 16);
+      }
+    int rc;
+    if ((rc = in_idx.ensure(in.size() * 4)) != GFRS_OK) return rc;
+    if ((rc = out_idx.ensure(out.size() * 4)) != GFRS_OK) return rc;
+    if ((rc = tabs.ensure(tb.size())) != GFRS_OK) return rc;
+    HIP_TRY(hipMemcpyAsync(in_idx.p, in.data(), in.size() * 4,
+                           hipMemcpyHostToDevice, s));
+    HIP_TRY(hipMemcpyAsync(out_idx.p, out.data(), out.size() * 4,
+                           hipMemcpyHostToDevice, s));
+    HIP_TRY(hipMemcpyAsync(tabs.p, tb.data(), tb.size(),
+                           hipMemcpyHostToDevice, s));
+    HIP_TRY(hipStreamSynchronize(s)); /* plans are built once, reused many times */
+    return GFRS_OK;
+  }
+};
+
+struct gfrs_ctx_impl {
+  gfrs_tactic t{};
+  int total = 0;   /* n+m+l */
+  int local_n = 0; /* (n+m)/az when l>0 */
+  int local_m = 0; /* l/az */
+  int device = 0;
+  hipStream_t own_stream = nullptr;
+  hipStream_t stream = nullptr;
+  std::mutex mu;
+
+  std::vector<uint8_t> enc_matrix;   /* (n+m)×n */
+  std::vector<uint8_t> local_matrix; /* (local_n+local_m)×local_n */
+  DevPlan enc_plan;                  /* global encode */
+  std::vector<DevPlan *> local_enc;  /* per-AZ local encode */
+  std::map<uint64_t, DevPlan *> dec_cache; /* missing-bitmask → data decode */
+  std::map<uint64_t, DevPlan *> par_cache; /* missing-bitmask → parity rows */
+
+  DevBuf ptr_buf;   /* pointer tables for pointer-mode launches */
+  DevBuf fail_buf;  /* verify flags / crc bad blocks */
+  DevBuf stage_dev; /* host-mode staging */
+  PinBuf stage_pin;
+
+  ~gfrs_ctx_impl() {
+    for (auto *p : local_enc) delete p;
+    for (auto &kv : dec_cache) delete kv.second;
+    for (auto &kv : par_cache) delete kv.second;
+    if (own_stream) hipStreamDestroy(own_stream);
+  }
+};
+
+struct StreamGuard {
+  int prev = -1;
+  explicit StreamGuard(const gfrs_ctx_impl *c) {
+    hipGetDevice(&prev);
+    hipSetDevice(c->device);
+  }
+  ~StreamGuard() {
+    if (prev >= 0) hipSetDevice(prev);
+  }
+};
+
+/* local stripe global indices for one AZ (codemode.go:301-318) */
+static std::vector<int32_t> local_stripe(const gfrs_tactic &t, int az_idx) {
+  int ln = t.n / t.az_count, lm = t.m / t.az_count, ll = t.l / t.az_count;
+  std::vector<int32_t> idx;
+  idx.reserve(ln + lm + ll);
+  for (int i = 0; i < ln; i++) idx.push_back(az_idx * ln + i);
+  for (int i = 0; i < lm; i++) idx.push_back(t.n + az_idx * lm + i);
+  for (int i = 0; i < ll; i++) idx.push_back(t.n + t.m + az_idx * ll + i);
+  return idx;
+}
+
+static bool tactic_valid(const gfrs_tactic *t) {
+  /* codemode.go:291-299 IsValid (EC portion) */
+  if (!t) return false;
+  if (t->n <= 0 || t->m <= 0 || t->l < 0 || t->az_count <= 0) return false;
+  if (t->n % t->az_count || t->m % t->az_count || t->l % t->az_count)
+    return false;
+  if (t->n + t->m > 256) return false;
+  if (t->l > 0 && (t->n + t->m) / t->az_count + t->l / t->az_count > 256)
+    return false;
+  return true;
+}
+
+}  // namespace gfrs
+
+using namespace gfrs;
+
+extern "C" {
+
+const char *gfrs_last_error(void) { return g_err.c_str(); }
+const char *gfrs_version(void) { return "gfrs 0.1.0 (gfx950)"; }
+
+int gfrs_device_count(void) {
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+  return n;
+}
+
+int gfrs_buffer_sizes(const gfrs_tactic *t, int64_t data_size,
+                      int64_t *shard_size, int64_t *ec_data_size,
+                      int64_t *ec_size) {
+  /* buf.go:67-133 */
+  if (!t || t->n <= 0) return GFRS_ERR_INVALID_CODEMODE;
+  if (data_size <= 0) return GFRS_ERR_SHORT_DATA;
+  int64_t ss = (data_size + t->n - 1) / t->n;
+  if (ss < t->min_shard_size) ss = t->min_shard_size;
+  *shard_size = ss;
+  *ec_data_size = ss * t->n;
+  *ec_size = ss * (t->n + t->m + t->l);
+  return GFRS_OK;
+}
+
+gfrs_ctx *gfrs_create(const gfrs_tactic *t, int device) {
+  if (!tactic_valid(t)) {
+    seterr("invalid code mode tactic");
+    return nullptr;
+  }
+  int ndev = gfrs_device_count();
+  if (ndev == 0) {
+    seterr("no HIP device visible (the gfrs engine has no CPU fallback)");
+    return nullptr;
+  }
+  if (device < 0) hipGetDevice(&device);
+  if (device >= ndev) {
+    seterr("device %d out of range (%d visible)", device, ndev);
+    return nullptr;
+  }
+  if (ensure_device_init(device) != GFRS_OK) return nullptr;
+
+  auto *c = new gfrs_ctx_impl();
+  c->t = *t;
+  c->total = t->n + t->m + t->l;
+  c->device = device;
+
+  int prev = -1;
+  hipGetDevice(&prev);
+  hipSetDevice(device);
+  bool ok = hipStreamCreate(&c->own_stream) == hipSuccess;
+  c->stream = c->own_stream;
+
+  /* global encode matrix (reedsolomon.go:220-244) */
+  if (ok) {
+    c->enc_matrix.resize(size_t(t->n + t->m) * t->n);
+    ok = gf_build_matrix(t->n, t->n + t->m, c->enc_matrix.data());
+  }
+  if (ok) {
+    std::vector<int32_t> in(t->n), out(t->m);
+    for (int i = 0; i < t->n; i++) in[i] = i;
+    for (int i = 0; i < t->m; i++) out[i] = t->n + i;
+    std::vector<uint8_t> rows(c->enc_matrix.begin() + size_t(t->n) * t->n,
+                              c->enc_matrix.end());
+    ok = c->enc_plan.upload(in, out, rows, c->stream) == GFRS_OK;
+  }
+  /* local engines (encoder.go:92-104) */
+  if (ok && t->l > 0) {
+    c->local_n = (t->n + t->m) / t->az_count;
+    c->local_m = t->l / t->az_count;
+    c->local_matrix.resize(size_t(c->local_n + c->local_m) * c->local_n);
+    ok = gf_build_matrix(c->local_n, c->local_n + c->local_m,
+                         c->local_matrix.data());
+    if (ok) {
+      std::vector<uint8_t> lrows(
+          c->local_matrix.begin() + size_t(c->local_n) * c->local_n,
+          c->local_matrix.end());
+      for (int az = 0; az < t->az_count && ok; az++) {
+        auto idx = local_stripe(*t, az);
+        std::vector<int32_t> in(idx.begin(), idx.begin() + c->local_n);
+        std::vector<int32_t> out(idx.begin() + c->local_n, idx.end());
+        auto *p = new DevPlan();
+        ok = p->upload(in, out, lrows, c->stream) == GFRS_OK;
+        c->local_enc.push_back(p);
+      }
+    }
+  }
+  if (prev >= 0) hipSetDevice(prev);
+  if (!ok) {
+    seterr("gfrs_create: init failed (%s)", g_err.c_str());
+    delete c;
+    return nullptr;
+  }
+  return reinterpret_cast<gfrs_ctx *>(c);
+}
+
+void gfrs_destroy(gfrs_ctx *ctx) { delete reinterpret_cast<gfrs_ctx_impl *>(ctx); }
+
+int gfrs_set_stream(gfrs_ctx *ctx, void *hip_stream) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  std::lock_guard<std::mutex> lk(c->mu);
+  c->stream = hip_stream ? reinterpret_cast<hipStream_t>(hip_stream)
+                         : c->own_stream;
+  return GFRS_OK;
+}
+
+int gfrs_synchronize(gfrs_ctx *ctx) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  StreamGuard g(c);
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  return GFRS_OK;
+}
+
+int gfrs_encode_matrix(gfrs_ctx *ctx, uint8_t *out) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  memcpy(out, c->enc_matrix.data(), c->enc_matrix.size());
+  return GFRS_OK;
+}
+
+int gfrs_compute_encode_matrix(int k, int total, uint8_t *out) {
+  if (!gf_build_matrix(k, total, out)) {
+    seterr("encode matrix build failed (k=%d total=%d)", k, total);
+    return GFRS_ERR_SINGULAR;
+  }
+  return GFRS_OK;
+}
+
+int gfrs_probe_perm(void) {
+  if (gfrs_device_count() == 0) return GFRS_ERR_NO_GPU;
+  return probe_perm_device();
+}
+
+}  /* extern "C" (continued in this file below) */
+
+/* ---------------- internal helpers for the compute entry points ------- */
+
+namespace gfrs {
+
+/* Upload a pointer table for pointer-mode launches (stream-ordered, so
+ * reuse of the scratch buffer is safe across calls on one stream). */
+static int upload_ptrs(gfrs_ctx_impl *c, void *const *shards, int nshards) {
+  int rc = c->ptr_buf.ensure(size_t(nshards) * 8);
+  if (rc != GFRS_OK) return rc;
+  /* small, use pinned staging for async copy */
+  if ((rc = c->stage_pin.ensure(size_t(nshards) * 8)) != GFRS_OK) return rc;
+  memcpy(c->stage_pin.p, shards, size_t(nshards) * 8);
+  HIP_TRY(hipMemcpyAsync(c->ptr_buf.p, c->stage_pin.p, size_t(nshards) * 8,
+                         hipMemcpyHostToDevice, c->stream));
+  return GFRS_OK;
+}
+
+/* decode plan for a missing pattern (bitmask over all shards of the
+ * engine's k+m space).  Mirrors inversion_tree caching. */
+static int get_decode_plan(gfrs_ctx_impl *c, int k, int m,
+                           const std::vector<uint8_t> &matrix,
+                           const std::vector<int> &engine_idx /* global ids */,
+                           const std::vector<uint8_t> &present,
+                           DevPlan **out_plan) {
+  uint64_t key = 0;
+  for (int i = 0; i < k + m; i++)
+    if (!present[i]) key |= 1ull << i;
+  /* separate cache namespaces for global vs local engines: tag with size */
+  key = key * 1000003ull + uint64_t(engine_idx[0]) * 31 + k;
+  auto it = c->dec_cache.find(key);
+  if (it != c->dec_cache.end()) {
+    *out_plan = it->second;
+    return GFRS_OK;
+  }
+  /* build: first k valid rows (reedsolomon.go:1453-1466) */
+  std::vector<int> valid;
+  for (int i = 0; i < k + m && int(valid.size()) < k; i++)
+    if (present[i]) valid.push_back(i);
+  if (int(valid.size()) < k) return GFRS_ERR_TOO_FEW_SHARDS;
+  std::vector<uint8_t> sub(size_t(k) * k), inv(size_t(k) * k);
+  for (int r = 0; r < k; r++)
+    memcpy(&sub[size_t(r) * k], &matrix[size_t(valid[r]) * k], k);
+  if (!gf_invert(sub.data(), k, inv.data())) return GFRS_ERR_SINGULAR;
+
+  std::vector<int32_t> in, out;
+  std::vector<uint8_t> rows;
+  for (int r = 0; r < k; r++) in.push_back(engine_idx[valid[r]]);
+  for (int i = 0; i < k; i++)
+    if (!present[i]) {
+      out.push_back(engine_idx[i]);
+      rows.insert(rows.end(), &inv[size_t(i) * k], &inv[size_t(i) * k + k]);
+    }
+  auto *p = new DevPlan();
+  int rc = p->upload(in, out, rows, c->stream);
+  if (rc != GFRS_OK) {
+    delete p;
+    return rc;
+  }
+  c->dec_cache[key] = p;
+  *out_plan = p;
+  return GFRS_OK;
+}
+
+/* parity-regeneration plan for missing parity rows */
+static int get_parity_plan(gfrs_ctx_impl *c, int k, int m,
+                           const std::vector<uint8_t> &matrix,
+                           const std::vector<int> &engine_idx,
+                           const std::vector<uint8_t> &present,
+                           DevPlan **out_plan) {
+  uint64_t key = 0;
+  for (int i = 0; i < k + m; i++)
+    if (!present[i]) key |= 1ull << i;
+  key = key * 999983ull + uint64_t(engine_idx[0]) * 31 + k;
+  auto it = c->par_cache.find(key);
+  if (it != c->par_cache.end()) {
+    *out_plan = it->second;
+    return GFRS_OK;
+  }
+  std::vector<int32_t> in, out;
+  std::vector<uint8_t> rows;
+  for (int i = 0; i < k; i++) in.push_back(engine_idx[i]);
+  for (int i = k; i < k + m; i++)
+    if (!present[i]) {
+      out.push_back(engine_idx[i]);
+      rows.insert(rows.end(), &matrix[size_t(i) * k],
+                  &matrix[size_t(i) * k + k]);
+    }
+  if (out.empty()) {
+    *out_plan = nullptr;
+    return GFRS_OK;
+  }
+  auto *p = new DevPlan();
+  int rc = p->upload(in, out, rows, c->stream);
+  if (rc != GFRS_OK) {
+    delete p;
+    return rc;
+  }
+  c->par_cache[key] = p;
+  *out_plan = p;
+  return GFRS_OK;
+}
+
+/* Run one engine's reconstruct over pointer-mode shards.
+ * engine_idx maps engine-local 0..k+m-1 to global shard slots. */
+static int reconstruct_engine(gfrs_ctx_impl *c, int k, int m,
+                              const std::vector<uint8_t> &matrix,
+                              const std::vector<int> &engine_idx,
+                              const std::vector<uint8_t> &present,
+                              size_t shard_len, int nstripes, int data_only,
+                              uint64_t strided_base, uint64_t stripe_stride,
+                              int nptr_or_0) {
+  int npresent = 0, dpresent = 0;
+  for (int i = 0; i < k + m; i++)
+    if (present[i]) {
+      npresent++;
+      if (i < k) dpresent++;
+    }
+  if (npresent == k + m || (data_only && dpresent == k)) return GFRS_OK;
+  if (npresent < k) return GFRS_ERR_TOO_FEW_SHARDS;
+
+  DevPlan *dec = nullptr;
+  int rc = GFRS_OK;
+  bool have_missing_data = dpresent < k;
+  if (have_missing_data) {
+    rc = get_decode_plan(c, k, m, matrix, engine_idx, present, &dec);
+    if (rc != GFRS_OK) return rc;
+    if (nptr_or_0 > 0)
+      launch_rs_apply(reinterpret_cast<const uint64_t *>(c->ptr_buf.p),
+                      nptr_or_0, (const int32_t *)dec->in_idx.p, dec->k,
+                      (const int32_t *)dec->out_idx.p, dec->nout,
+                      (const uint8_t *)dec->tabs.p, shard_len, nstripes,
+                      c->stream);
+    else
+      launch_rs_apply_strided(strided_base, stripe_stride,
+                              (const int32_t *)dec->in_idx.p, dec->k,
+                              (const int32_t *)dec->out_idx.p, dec->nout,
+                              (const uint8_t *)dec->tabs.p, shard_len,
+                              nstripes, c->stream);
+  }
+  if (!data_only) {
+    DevPlan *par = nullptr;
+    rc = get_parity_plan(c, k, m, matrix, engine_idx, present, &par);
+    if (rc != GFRS_OK) return rc;
+    if (par) {
+      if (nptr_or_0 > 0)
+        launch_rs_apply(reinterpret_cast<const uint64_t *>(c->ptr_buf.p),
+                        nptr_or_0, (const int32_t *)par->in_idx.p, par->k,
+                        (const int32_t *)par->out_idx.p, par->nout,
+                        (const uint8_t *)par->tabs.p, shard_len, nstripes,
+                        c->stream);
+      else
+        launch_rs_apply_strided(strided_base, stripe_stride,
+                                (const int32_t *)par->in_idx.p, par->k,
+                                (const int32_t *)par->out_idx.p, par->nout,
+                                (const uint8_t *)par->tabs.p, shard_len,
+                                nstripes, c->stream);
+    }
+  }
+  return GFRS_OK;
+}
+
+/* Core reconstruct across global + local engines.  present covers all
+ * n+m+l global slots. */
+static int reconstruct_all(gfrs_ctx_impl *c, std::vector<uint8_t> &present,
+                           size_t shard_len, int nstripes, int data_only,
+                           uint64_t base, uint64_t stride, int nptr_or_0) {
+  const gfrs_tactic &t = c->t;
+  std::vector<int> gidx(t.n + t.m);
+  for (int i = 0; i < t.n + t.m; i++) gidx[i] = i;
+  std::vector<uint8_t> gpresent(present.begin(), present.begin() + t.n + t.m);
+  int rc = reconstruct_engine(c, t.n, t.m, c->enc_matrix, gidx, gpresent,
+                              shard_len, nstripes, data_only, base, stride,
+                              nptr_or_0);
+  if (rc != GFRS_OK) return rc;
+  if (t.l == 0 || data_only) return GFRS_OK;
+  /* regenerate bad local parities per AZ (lrcencoder.go:160-184); after
+   * the global pass everything below n+m is intact */
+  for (int az = 0; az < t.az_count; az++) {
+    auto idx = local_stripe(t, az);
+    bool need = false;
+    std::vector<uint8_t> lp(idx.size());
+    std::vector<int> li(idx.begin(), idx.end());
+    for (size_t i = 0; i < idx.size(); i++) {
+      lp[i] = idx[i] < t.n + t.m ? 1 : present[idx[i]];
+      if (!lp[i]) need = true;
+    }
+    if (!need) continue;
+    rc = reconstruct_engine(c, c->local_n, c->local_m, c->local_matrix, li,
+                            lp, shard_len, nstripes, /*data_only=*/0, base,
+                            stride, nptr_or_0);
+    if (rc != GFRS_OK) return rc;
+  }
+  return GFRS_OK;
+}
+
+}  // namespace gfrs
+
+/* ---------------- compute entry points ---------------- */
+
+extern "C" {
+
+int gfrs_encode(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
+                int nshards, int memloc) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  if (nshards != c->total) {
+    seterr("encode: want %d shards, got %d", c->total, nshards);
+    return GFRS_ERR_INVALID_SHARDS;
+  }
+  std::lock_guard<std::mutex> lk(c->mu);
+  StreamGuard g(c);
+  int rc;
+  if (memloc == GFRS_MEM_HOST) {
+    /* stage contiguous stripe, run strided, copy parity back */
+    size_t tot = size_t(c->total) * shard_len;
+    if ((rc = c->stage_dev.ensure(tot)) != GFRS_OK) return rc;
+    if ((rc = c->stage_pin.ensure(tot)) != GFRS_OK) return rc;
+    uint8_t *pin = (uint8_t *)c->stage_pin.p;
+    for (int i = 0; i < c->t.n; i++)
+      memcpy(pin + size_t(i) * shard_len, shards[i], shard_len);
+    uint8_t *dev = (uint8_t *)c->stage_dev.p;
+    HIP_TRY(hipMemcpyAsync(dev, pin, size_t(c->t.n) * shard_len,
+                           hipMemcpyHostToDevice, c->stream));
+    rc = gfrs_encode_batch(ctx, dev, shard_len, tot, 1);
+    if (rc != GFRS_OK) return rc;
+    HIP_TRY(hipMemcpyAsync(pin + size_t(c->t.n) * shard_len,
+                           dev + size_t(c->t.n) * shard_len,
+                           size_t(c->t.m + c->t.l) * shard_len,
+                           hipMemcpyDeviceToHost, c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    for (int i = c->t.n; i < c->total; i++)
+      memcpy(shards[i], pin + size_t(i) * shard_len, shard_len);
+    return GFRS_OK;
+  }
+  if ((rc = upload_ptrs(c, shards, nshards)) != GFRS_OK) return rc;
+  launch_rs_apply((const uint64_t *)c->ptr_buf.p, c->total,
+                  (const int32_t *)c->enc_plan.in_idx.p, c->enc_plan.k,
+                  (const int32_t *)c->enc_plan.out_idx.p, c->enc_plan.nout,
+                  (const uint8_t *)c->enc_plan.tabs.p, shard_len, 1,
+                  c->stream);
+  for (auto *lp : c->local_enc)
+    launch_rs_apply((const uint64_t *)c->ptr_buf.p, c->total,
+                    (const int32_t *)lp->in_idx.p, lp->k,
+                    (const int32_t *)lp->out_idx.p, lp->nout,
+                    (const uint8_t *)lp->tabs.p, shard_len, 1, c->stream);
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  return GFRS_OK;
+}
+
+int gfrs_encode_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
+                      size_t stripe_stride, int nstripes) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  StreamGuard g(c);
+  launch_rs_apply_strided((uint64_t)base, stripe_stride,
+                          (const int32_t *)c->enc_plan.in_idx.p,
+                          c->enc_plan.k, (const int32_t *)c->enc_plan.out_idx.p,
+                          c->enc_plan.nout, (const uint8_t *)c->enc_plan.tabs.p,
+                          shard_len, nstripes, c->stream);
+  for (auto *lp : c->local_enc)
+    launch_rs_apply_strided((uint64_t)base, stripe_stride,
+                            (const int32_t *)lp->in_idx.p, lp->k,
+                            (const int32_t *)lp->out_idx.p, lp->nout,
+                            (const uint8_t *)lp->tabs.p, shard_len, nstripes,
+                            c->stream);
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) return hip_fail("encode_batch launch", e);
+  return GFRS_OK;
+}
+
+int gfrs_verify(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
+                int nshards, int memloc, int *ok) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  if (nshards != c->total) return GFRS_ERR_INVALID_SHARDS;
+  std::lock_guard<std::mutex> lk(c->mu);
+  StreamGuard g(c);
+  int rc;
+  const uint64_t *pt;
+  if (memloc == GFRS_MEM_HOST) {
+    size_t tot = size_t(c->total) * shard_len;
+    if ((rc = c->stage_dev.ensure(tot + 64)) != GFRS_OK) return rc;
+    if ((rc = c->stage_pin.ensure(tot)) != GFRS_OK) return rc;
+    uint8_t *pin = (uint8_t *)c->stage_pin.p;
+    for (int i = 0; i < c->total; i++)
+      memcpy(pin + size_t(i) * shard_len, shards[i], shard_len);
+    HIP_TRY(hipMemcpyAsync(c->stage_dev.p, pin, tot, hipMemcpyHostToDevice,
+                           c->stream));
+    uint64_t fb = 0;
+    rc = gfrs_verify_batch(ctx, c->stage_dev.p, shard_len, tot, 1, &fb);
+    if (rc != GFRS_OK) return rc;
+    *ok = fb == 0;
+    return GFRS_OK;
+  }
+  if ((rc = upload_ptrs(c, shards, nshards)) != GFRS_OK) return rc;
+  if ((rc = c->fail_buf.ensure(4)) != GFRS_OK) return rc;
+  HIP_TRY(hipMemsetAsync(c->fail_buf.p, 0, 4, c->stream));
+  pt = (const uint64_t *)c->ptr_buf.p;
+  launch_rs_verify(pt, c->total, (const int32_t *)c->enc_plan.in_idx.p,
+                   c->enc_plan.k, (const int32_t *)c->enc_plan.out_idx.p,
+                   c->enc_plan.nout, (const uint8_t *)c->enc_plan.tabs.p,
+                   shard_len, 1, (uint32_t *)c->fail_buf.p, c->stream);
+  for (auto *lp : c->local_enc)
+    launch_rs_verify(pt, c->total, (const int32_t *)lp->in_idx.p, lp->k,
+                     (const int32_t *)lp->out_idx.p, lp->nout,
+                     (const uint8_t *)lp->tabs.p, shard_len, 1,
+                     (uint32_t *)c->fail_buf.p, c->stream);
+  uint32_t fail = 0;
+  HIP_TRY(hipMemcpyAsync(&fail, c->fail_buf.p, 4, hipMemcpyDeviceToHost,
+                         c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  *ok = fail == 0;
+  return GFRS_OK;
+}
+
+int gfrs_verify_batch(gfrs_ctx *ctx, const void *base, size_t shard_len,
+                      size_t stripe_stride, int nstripes,
+                      uint64_t *fail_bitmap) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  StreamGuard g(c);
+  int rc;
+  if ((rc = c->fail_buf.ensure(size_t(nstripes) * 4)) != GFRS_OK) return rc;
+  HIP_TRY(hipMemsetAsync(c->fail_buf.p, 0, size_t(nstripes) * 4, c->stream));
+  launch_rs_verify_strided((uint64_t)base, stripe_stride,
+                           (const int32_t *)c->enc_plan.in_idx.p,
+                           c->enc_plan.k,
+                           (const int32_t *)c->enc_plan.out_idx.p,
+                           c->enc_plan.nout,
+                           (const uint8_t *)c->enc_plan.tabs.p, shard_len,
+                           nstripes, (uint32_t *)c->fail_buf.p, c->stream);
+  for (auto *lp : c->local_enc)
+    launch_rs_verify_strided((uint64_t)base, stripe_stride,
+                             (const int32_t *)lp->in_idx.p, lp->k,
+                             (const int32_t *)lp->out_idx.p, lp->nout,
+                             (const uint8_t *)lp->tabs.p, shard_len, nstripes,
+                             (uint32_t *)c->fail_buf.p, c->stream);
+  std::vector<uint32_t> fails(nstripes);
+  HIP_TRY(hipMemcpyAsync(fails.data(), c->fail_buf.p, size_t(nstripes) * 4,
+                         hipMemcpyDeviceToHost, c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  if (fail_bitmap) {
+    memset(fail_bitmap, 0, ((nstripes + 63) / 64) * 8);
+    for (int s = 0; s < nstripes; s++)
+      if (fails[s]) fail_bitmap[s / 64] |= 1ull << (s % 64);
+  }
+  return GFRS_OK;
+}
+
+int gfrs_reconstruct(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
+                     int nshards, int memloc, const int32_t *bad_idx,
+                     int nbad, int data_only) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  const gfrs_tactic &t = c->t;
+  std::lock_guard<std::mutex> lk(c->mu);
+  StreamGuard g(c);
+  int rc;
+
+  /* full set or, for LRC, a single local stripe (lrcencoder.go:147-153) */
+  bool local_form = t.l > 0 && nshards == c->total / t.az_count;
+  if (nshards != c->total && !local_form) return GFRS_ERR_INVALID_SHARDS;
+
+  std::vector<uint8_t> present(nshards, 1);
+  for (int i = 0; i < nbad; i++) {
+    if (bad_idx[i] < 0 || bad_idx[i] >= nshards) return GFRS_ERR_INVALID_SHARDS;
+    present[bad_idx[i]] = 0;
+  }
+
+  if (memloc == GFRS_MEM_HOST) {
+    /* stage the whole stripe contiguously and run in strided mode */
+    size_t tot = size_t(nshards) * shard_len;
+    if ((rc = c->stage_dev.ensure(tot)) != GFRS_OK) return rc;
+    if ((rc = c->stage_pin.ensure(tot)) != GFRS_OK) return rc;
+    uint8_t *pin = (uint8_t *)c->stage_pin.p;
+    uint8_t *dev = (uint8_t *)c->stage_dev.p;
+    for (int i = 0; i < nshards; i++)
+      if (present[i]) memcpy(pin + size_t(i) * shard_len, shards[i], shard_len);
+    HIP_TRY(hipMemcpyAsync(dev, pin, tot, hipMemcpyHostToDevice, c->stream));
+    if (local_form) {
+      std::vector<int> li(nshards);
+      for (int i = 0; i < nshards; i++) li[i] = i;
+      rc = reconstruct_engine(c, c->local_n, c->local_m, c->local_matrix, li,
+                              present, shard_len, 1, data_only, (uint64_t)dev,
+                              tot, 0);
+    } else {
+      rc = reconstruct_all(c, present, shard_len, 1, data_only, (uint64_t)dev,
+                           tot, 0);
+    }
+    if (rc != GFRS_OK) return rc;
+    HIP_TRY(hipMemcpyAsync(pin, dev, tot, hipMemcpyDeviceToHost, c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    for (int i = 0; i < nshards; i++)
+      if (!present[i]) memcpy(shards[i], pin + size_t(i) * shard_len, shard_len);
+    return GFRS_OK;
+  }
+
+  if ((rc = upload_ptrs(c, shards, nshards)) != GFRS_OK) return rc;
+
+  if (local_form) {
+    std::vector<int> li(nshards);
+    for (int i = 0; i < nshards; i++) li[i] = i;
+    rc = reconstruct_engine(c, c->local_n, c->local_m, c->local_matrix, li,
+                            present, shard_len, 1, data_only, 0, 0, nshards);
+  } else {
+    rc = reconstruct_all(c, present, shard_len, 1, data_only, 0, 0, nshards);
+  }
+  if (rc != GFRS_OK) return rc;
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  return GFRS_OK;
+}
+
+int gfrs_reconstruct_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
+                           size_t stripe_stride, int nstripes,
+                           const int32_t *bad_idx, int nbad, int data_only) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  std::lock_guard<std::mutex> lk(c->mu);
+  StreamGuard g(c);
+  std::vector<uint8_t> present(c->total, 1);
+  for (int i = 0; i < nbad; i++) {
+    if (bad_idx[i] < 0 || bad_idx[i] >= c->total)
+      return GFRS_ERR_INVALID_SHARDS;
+    present[bad_idx[i]] = 0;
+  }
+  int rc = reconstruct_all(c, present, shard_len, nstripes, data_only,
+                           (uint64_t)base, stripe_stride, 0);
+  if (rc != GFRS_OK) return rc;
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) return hip_fail("reconstruct_batch launch", e);
+  return GFRS_OK;
+}
+
+/* ---------------- crc32block ---------------- */
+
+int64_t gfrs_crc32b_encode_size(int64_t size, int64_t block_len) {
+  if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
+  int64_t payload = block_len - 4;
+  return size + 4 * ((size + payload - 1) / payload);
+}
+
+int64_t gfrs_crc32b_decode_size(int64_t size, int64_t block_len) {
+  if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
+  return size - 4 * ((size + block_len - 1) / block_len);
+}
+
+int gfrs_crc32b_encode_batch(gfrs_ctx *ctx, void *dst, size_t dst_stride,
+                             const void *src, size_t src_stride, int64_t n,
+                             int64_t block_len, int nshards) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
+  if (n <= 0 || nshards <= 0) return GFRS_ERR_INVALID_SHARDS;
+  StreamGuard g(c);
+  launch_crc_encode((uint8_t *)dst, dst_stride, (const uint8_t *)src,
+                    src_stride, n, block_len, nshards, c->stream);
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) return hip_fail("crc encode launch", e);
+  return GFRS_OK;
+}
+
+int64_t gfrs_crc32b_encode(gfrs_ctx *ctx, void *dst, const void *src,
+                           int64_t n, int64_t block_len) {
+  int rc = gfrs_crc32b_encode_batch(ctx, dst, 0, src, 0, n, block_len, 1);
+  if (rc != GFRS_OK) return rc;
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  StreamGuard g(c);
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  return gfrs_crc32b_encode_size(n, block_len);
+}
+
+int gfrs_crc32b_verify_batch(gfrs_ctx *ctx, const void *framed, size_t stride,
+                             int64_t framed_len, int64_t block_len,
+                             int nshards, int64_t *bad_block_per_shard) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
+  if (framed_len <= 0 || nshards <= 0) return GFRS_ERR_INVALID_SHARDS;
+  std::lock_guard<std::mutex> lk(c->mu);
+  StreamGuard g(c);
+  int rc;
+  if ((rc = c->fail_buf.ensure(size_t(nshards) * 8)) != GFRS_OK) return rc;
+  std::vector<int64_t> bad(nshards, INT64_MAX);
+  HIP_TRY(hipMemcpyAsync(c->fail_buf.p, bad.data(), size_t(nshards) * 8,
+                         hipMemcpyHostToDevice, c->stream));
+  launch_crc_verify((const uint8_t *)framed, stride, framed_len, block_len,
+                    nshards, (int64_t *)c->fail_buf.p, c->stream);
+  HIP_TRY(hipMemcpyAsync(bad.data(), c->fail_buf.p, size_t(nshards) * 8,
+                         hipMemcpyDeviceToHost, c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  for (int i = 0; i < nshards; i++)
+    bad_block_per_shard[i] = bad[i] == INT64_MAX ? -1 : bad[i];
+  return GFRS_OK;
+}
+
+int gfrs_crc32b_verify(gfrs_ctx *ctx, const void *framed, int64_t framed_len,
+                       int64_t block_len, int64_t *bad_block) {
+  return gfrs_crc32b_verify_batch(ctx, framed, 0, framed_len, block_len, 1,
+                                  bad_block);
+}
+
+int64_t gfrs_crc32b_decode(gfrs_ctx *ctx, void *dst, const void *framed,
+                           int64_t framed_len, int64_t block_len) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
+  std::lock_guard<std::mutex> lk(c->mu);
+  StreamGuard g(c);
+  int rc;
+  if ((rc = c->fail_buf.ensure(8)) != GFRS_OK) return rc;
+  int64_t bad = INT64_MAX;
+  HIP_TRY(hipMemcpyAsync(c->fail_buf.p, &bad, 8, hipMemcpyHostToDevice,
+                         c->stream));
+  launch_crc_decode((uint8_t *)dst, 0, (const uint8_t *)framed, 0, framed_len,
+                    block_len, 1, (int64_t *)c->fail_buf.p, c->stream);
+  HIP_TRY(hipMemcpyAsync(&bad, c->fail_buf.p, 8, hipMemcpyDeviceToHost,
+                         c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  if (bad != INT64_MAX) return GFRS_ERR_MISMATCHED_CRC;
+  return gfrs_crc32b_decode_size(framed_len, block_len);
+}
+
+} /* extern "C" */

@@ -1,0 +1,95 @@
+/* cubefs_amd/csrc/gfrs_internal.h — internal declarations shared between
+ * the host runtime (gfrs_host.cpp) and kernel launchers (gfrs_kernels.hip).
+ * Product code: independent of oracle/.
+ */
+#ifndef GFRS_INTERNAL_H
+#define GFRS_INTERNAL_H
+
+#include <cstddef>
+#include <cstdint>
+#include <vector>
+
+#include <hip/hip_runtime.h>
+
+namespace gfrs {
+
+/* ---- GF(2^8) field, poly 0x11D (generator 29; galois.go:25) ---- */
+struct GfTables {
+  uint8_t log_t[256];
+  uint8_t exp_t[510];
+  uint8_t mul[256][256];
+  /* nibble tables: lo[c][x] = mul[c][x], hi[c][x] = mul[c][x<<4]
+   * (galois.go:340,596) */
+  uint8_t lo[256][16];
+  uint8_t hi[256][16];
+  GfTables();
+  uint8_t gexp(uint8_t a, int n) const; /* galois.go:892 */
+  uint8_t div(uint8_t a, uint8_t b) const;
+};
+const GfTables &gft();
+
+/* Row-major byte matrix helpers (matrix.go). */
+bool gf_invert(const uint8_t *in, int n, uint8_t *out); /* false = singular */
+/* klauspost default encode matrix: vandermonde(total,k) × inv(top k×k)
+ * (reedsolomon.go:220-244).  out is total×k. */
+bool gf_build_matrix(int k, int total, uint8_t *out);
+
+/* ---- kernel launch interface ---- */
+
+/* Matrix-apply over a stripe batch (serves encode and reconstruct).
+ * For each stripe s (0..nstripes), output r (0..nout), byte i:
+ *   out[r][i] = XOR_c coeff[r*k+c] * in[c][i]
+ * where in/out shard pointers come from ptrs[s*nptr + idx]:
+ *   in  c -> ptrs[s*nptr + in_idx[c]]
+ *   out r -> ptrs[s*nptr + out_idx[r]]
+ * tabs = device array [nout*k][32]: per-coefficient lo|hi nibble tables.
+ * All index/table arrays are device memory. */
+void launch_rs_apply(const uint64_t *ptrs, int nptr, const int32_t *in_idx,
+                     int k, const int32_t *out_idx, int nout,
+                     const uint8_t *tabs, size_t shard_len, int nstripes,
+                     hipStream_t s);
+
+/* Verify: recompute parity from tabs and OR a nonzero flag into
+ * fail[s] for any mismatching stripe. */
+void launch_rs_verify(const uint64_t *ptrs, int nptr, const int32_t *in_idx,
+                      int k, const int32_t *out_idx, int nout,
+                      const uint8_t *tabs, size_t shard_len, int nstripes,
+                      uint32_t *fail, hipStream_t s);
+
+/* crc32block kernels.  Shards framed independently; shard j raw bytes at
+ * src + j*src_stride (n bytes), framed at dst + j*dst_stride.
+ * suffix_ops: device array of fold operators (see gfrs_kernels.hip). */
+void launch_crc_encode(uint8_t *dst, size_t dst_stride, const uint8_t *src,
+                       size_t src_stride, int64_t n, int64_t block_len,
+                       int nshards, hipStream_t s);
+/* bad[j] = first failing block index in shard j, or -1 (preset by host). */
+void launch_crc_verify(const uint8_t *framed, size_t stride,
+                       int64_t framed_len, int64_t block_len, int nshards,
+                       int64_t *bad, hipStream_t s);
+void launch_crc_decode(uint8_t *dst, size_t dst_stride, const uint8_t *framed,
+                       size_t src_stride, int64_t framed_len,
+                       int64_t block_len, int nshards, int64_t *bad,
+                       hipStream_t s);
+
+/* v_perm semantics probe: returns 1 when `sel byte >= 8 -> 0x00` holds. */
+int probe_perm_device(void);
+
+/* One-time CRC table upload for the current device. */
+int crc_device_init_current(void);
+
+/* Strided-layout variants (batch APIs; stripe s shard i at
+ * base + s*stripe_stride + i*shard_len). */
+void launch_rs_apply_strided(uint64_t base, uint64_t stripe_stride,
+                             const int32_t *in_idx, int k,
+                             const int32_t *out_idx, int nout,
+                             const uint8_t *tabs, size_t shard_len,
+                             int nstripes, hipStream_t s);
+void launch_rs_verify_strided(uint64_t base, uint64_t stripe_stride,
+                              const int32_t *in_idx, int k,
+                              const int32_t *out_idx, int nout,
+                              const uint8_t *tabs, size_t shard_len,
+                              int nstripes, uint32_t *fail, hipStream_t s);
+
+}  // namespace gfrs
+
+#endif

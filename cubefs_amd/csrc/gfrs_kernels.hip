@@ -1,0 +1,517 @@
+/* cubefs_amd/csrc/gfrs_kernels.hip — CDNA4 (gfx950) kernels for the CubeFS
+ * blobstore EC/CRC hot path.
+ *
+ * Replaces (same arithmetic, MI355X-native data path):
+ *   galois_amd64.s / galois_gen_amd64.s  (GF(2^8) nibble-table multiply:
+ *     out[i] ^= lo[c][in[i]&0xF] ^ hi[c][in[i]>>4]; tables galois.go:340,596)
+ *   reedsolomon.go:807-985 codeSomeShards[P] (fused k-input × m-output
+ *     matrix apply; the goroutine byte-range split becomes the wave/grid
+ *     split)
+ *   Go hash/crc32 CLMUL path + crc32block/block.go:22-49 framing
+ *
+ * Design notes (HBM-bound byte work; no MFMA — GF(2^8) is not a dense
+ * contraction):
+ *  - rs_apply: each lane owns 16 B columns (uint4 loads, 1 KiB per wave
+ *    per instruction); all nout outputs accumulate in VGPRs so every input
+ *    byte is read from HBM exactly once per output-group of 4.
+ *  - GF multiply = two 16-entry nibble lookups done in-register with
+ *    v_perm_b32 over the 32 B coefficient table (staged in LDS, broadcast
+ *    reads) — no per-byte LDS gather.
+ *  - crc32block: one 256-thread workgroup per 64 KiB frame; each thread
+ *    CRCs a 256 B chunk (slice-by-4, tables in LDS), chunks are folded with
+ *    the GF(2) x^(8·len) shift operator and reduced in LDS.  This is the
+ *    parallel decomposition of the sequential Go loop; bit-identical.
+ */
+#include "gfrs_internal.h"
+
+namespace gfrs {
+
+#define GFRS_DEV __device__ __forceinline__
+
+/* ------------------------------------------------------------------ */
+/* GF(2^8) nibble multiply on packed u32, via v_perm_b32                */
+/* ------------------------------------------------------------------ */
+
+/* 16-entry byte table lookup for 4 packed nibbles (each byte of `nib` in
+ * 0..15).  Table held as two uint4 (t01 = bytes 0..7 in .x/.y, plus 8..15
+ * in .z/.w).  PERM0: rely on v_perm sel>=8 -> 0x00 (probed at runtime);
+ * otherwise mask-and-blend. */
+template <bool PERM0>
+GFRS_DEV uint32_t lut16(const uint4 t, uint32_t nib) {
+  if (PERM0) {
+    /* sel>=8 yields 0, so the two halves OR together by XOR. */
+    uint32_t a = __builtin_amdgcn_perm(t.y, t.x, nib);
+    uint32_t b = __builtin_amdgcn_perm(t.w, t.z, nib ^ 0x08080808u);
+    return a ^ b;
+  } else {
+    uint32_t s = nib & 0x07070707u;
+    uint32_t a = __builtin_amdgcn_perm(t.y, t.x, s);
+    uint32_t b = __builtin_amdgcn_perm(t.w, t.z, s);
+    uint32_t m = (nib & 0x08080808u) >> 3;
+    m *= 0xFFu; /* per-byte 0x00/0xFF, no carries */
+    return (a & ~m) | (b & m);
+  }
+}
+
+/* out ^= mul_c(v) for 4 packed bytes; tlo/thi are c's 16-B low/high nibble
+ * tables (galois_amd64.go:37-52 semantics). */
+template <bool PERM0>
+GFRS_DEV uint32_t gfmul4(uint32_t v, const uint4 tlo, const uint4 thi) {
+  return lut16<PERM0>(tlo, v & 0x0F0F0F0Fu) ^
+         lut16<PERM0>(thi, (v >> 4) & 0x0F0F0F0Fu);
+}
+
+template <bool PERM0>
+GFRS_DEV void gfmac16(uint4 &acc, const uint4 v, const uint4 tlo,
+                      const uint4 thi) {
+  acc.x ^= gfmul4<PERM0>(v.x, tlo, thi);
+  acc.y ^= gfmul4<PERM0>(v.y, tlo, thi);
+  acc.z ^= gfmul4<PERM0>(v.z, tlo, thi);
+  acc.w ^= gfmul4<PERM0>(v.w, tlo, thi);
+}
+
+/* ------------------------------------------------------------------ */
+/* rs_apply / rs_verify                                                 */
+/* ------------------------------------------------------------------ */
+
+constexpr int RS_BLOCK = 256;
+constexpr int RS_TILE = RS_BLOCK * 16; /* 4096 B of columns per tile */
+constexpr int MT = 4;                  /* output registers per pass */
+
+struct ShardAddr {
+  const uint64_t *ptrs; /* nstripes*nptr pointer table, or nullptr */
+  uint64_t base;        /* strided mode: base + s*stripe_stride + i*shard_len */
+  uint64_t stripe_stride;
+  int nptr;
+
+  GFRS_DEV const uint8_t *shard(size_t stripe, int idx, size_t shard_len) const {
+    if (ptrs) return reinterpret_cast<const uint8_t *>(ptrs[stripe * nptr + idx]);
+    return reinterpret_cast<const uint8_t *>(base + stripe * stripe_stride +
+                                             uint64_t(idx) * shard_len);
+  }
+};
+
+/* For each stripe and each output r: out[r] = XOR_c mul(coeff[r][c], in[c]).
+ * tabs: [nout*k][32] per-coefficient lo|hi tables, staged to LDS.
+ * Outputs processed in groups of MT so inputs stream from HBM once per
+ * group (arithmetic intensity k·m/(k+m) table-xors per byte; HBM-bound). */
+template <bool PERM0, bool VERIFY>
+__global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
+    ShardAddr addr, const int32_t *__restrict__ in_idx, int k,
+    const int32_t *__restrict__ out_idx, int nout,
+    const uint8_t *__restrict__ tabs, size_t shard_len, size_t nstripes,
+    uint32_t *fail) {
+  extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
+  uint4 *ltab = reinterpret_cast<uint4 *>(smem); /* [k*nout*2] */
+  const int ncoef = k * nout;
+  for (int i = threadIdx.x; i < ncoef * 2; i += RS_BLOCK)
+    ltab[i] = reinterpret_cast<const uint4 *>(tabs)[i];
+  __syncthreads();
+
+  const size_t tiles_per_shard = (shard_len + RS_TILE - 1) / RS_TILE;
+  const size_t total_tiles = tiles_per_shard * nstripes;
+
+  for (size_t tile = blockIdx.x; tile < total_tiles; tile += gridDim.x) {
+    const size_t stripe = tile / tiles_per_shard;
+    const size_t col0 = (tile - stripe * tiles_per_shard) * size_t(RS_TILE);
+    const size_t off = col0 + size_t(threadIdx.x) * 16;
+    bool mismatch = false;
+
+    if (off + 16 <= shard_len) {
+      for (int og = 0; og < nout; og += MT) {
+        const int gm = min(MT, nout - og);
+        uint4 acc[MT];
+#pragma unroll
+        for (int r = 0; r < MT; r++) acc[r] = uint4{0, 0, 0, 0};
+        for (int c = 0; c < k; c++) {
+          const uint8_t *in = addr.shard(stripe, in_idx[c], shard_len);
+          const uint4 v = *reinterpret_cast<const uint4 *>(in + off);
+#pragma unroll
+          for (int r = 0; r < MT; r++) {
+            if (r < gm) {
+              const int t2 = ((og + r) * k + c) * 2;
+              gfmac16<PERM0>(acc[r], v, ltab[t2], ltab[t2 + 1]);
+            }
+          }
+        }
+        for (int r = 0; r < gm; r++) {
+          uint8_t *out = const_cast<uint8_t *>(
+              addr.shard(stripe, out_idx[og + r], shard_len));
+          if (VERIFY) {
+            const uint4 e = *reinterpret_cast<const uint4 *>(out + off);
+            mismatch |= (e.x != acc[r].x) | (e.y != acc[r].y) |
+                        (e.z != acc[r].z) | (e.w != acc[r].w);
+          } else {
+            *reinterpret_cast<uint4 *>(out + off) = acc[r];
+          }
+        }
+      }
+    } else if (off < shard_len) {
+      /* ragged tail: per-byte path using the byte view of the LDS tables */
+      const uint8_t *bt = smem;
+      const size_t nb = shard_len - off;
+      for (int og = 0; og < nout; og++) {
+        const uint8_t *trow = bt + size_t(og * k) * 32;
+        uint8_t *out =
+            const_cast<uint8_t *>(addr.shard(stripe, out_idx[og], shard_len));
+        for (size_t i = 0; i < nb; i++) {
+          uint8_t v = 0;
+          for (int c = 0; c < k; c++) {
+            const uint8_t b = addr.shard(stripe, in_idx[c], shard_len)[off + i];
+            const uint8_t *t = trow + size_t(c) * 32;
+            v ^= t[b & 0x0F] ^ t[16 + (b >> 4)];
+          }
+          if (VERIFY)
+            mismatch |= (out[off + i] != v);
+          else
+            out[off + i] = v;
+        }
+      }
+    }
+    if (VERIFY) {
+      if (__ballot(mismatch) != 0) {
+        if ((threadIdx.x & 63) == 0) atomicOr(&fail[stripe], 1u);
+      }
+    }
+  }
+}
+
+static int rs_grid(size_t shard_len, size_t nstripes) {
+  size_t tiles = ((shard_len + RS_TILE - 1) / RS_TILE) * nstripes;
+  if (tiles == 0) tiles = 1;
+  /* memory-bound: cap and grid-stride (cdna_hip_programming.md G11) */
+  const size_t cap = 256 * 8;
+  return int(tiles < cap ? tiles : cap);
+}
+
+static bool perm0_ok(); /* below */
+
+void launch_rs_apply(const uint64_t *ptrs, int nptr, const int32_t *in_idx,
+                     int k, const int32_t *out_idx, int nout,
+                     const uint8_t *tabs, size_t shard_len, int nstripes,
+                     hipStream_t s) {
+  ShardAddr a{ptrs, 0, 0, nptr};
+  const int lds = k * nout * 32;
+  const int grid = rs_grid(shard_len, nstripes);
+  if (perm0_ok())
+    hipLaunchKernelGGL((rs_apply_k<true, false>), dim3(grid), dim3(RS_BLOCK),
+                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
+                       size_t(nstripes), nullptr);
+  else
+    hipLaunchKernelGGL((rs_apply_k<false, false>), dim3(grid), dim3(RS_BLOCK),
+                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
+                       size_t(nstripes), nullptr);
+}
+
+void launch_rs_apply_strided(uint64_t base, uint64_t stripe_stride,
+                             const int32_t *in_idx, int k,
+                             const int32_t *out_idx, int nout,
+                             const uint8_t *tabs, size_t shard_len,
+                             int nstripes, hipStream_t s) {
+  ShardAddr a{nullptr, base, stripe_stride, 0};
+  const int lds = k * nout * 32;
+  const int grid = rs_grid(shard_len, nstripes);
+  if (perm0_ok())
+    hipLaunchKernelGGL((rs_apply_k<true, false>), dim3(grid), dim3(RS_BLOCK),
+                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
+                       size_t(nstripes), nullptr);
+  else
+    hipLaunchKernelGGL((rs_apply_k<false, false>), dim3(grid), dim3(RS_BLOCK),
+                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
+                       size_t(nstripes), nullptr);
+}
+
+void launch_rs_verify(const uint64_t *ptrs, int nptr, const int32_t *in_idx,
+                      int k, const int32_t *out_idx, int nout,
+                      const uint8_t *tabs, size_t shard_len, int nstripes,
+                      uint32_t *fail, hipStream_t s) {
+  ShardAddr a{ptrs, 0, 0, nptr};
+  const int lds = k * nout * 32;
+  const int grid = rs_grid(shard_len, nstripes);
+  if (perm0_ok())
+    hipLaunchKernelGGL((rs_apply_k<true, true>), dim3(grid), dim3(RS_BLOCK),
+                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
+                       size_t(nstripes), fail);
+  else
+    hipLaunchKernelGGL((rs_apply_k<false, true>), dim3(grid), dim3(RS_BLOCK),
+                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
+                       size_t(nstripes), fail);
+}
+
+void launch_rs_verify_strided(uint64_t base, uint64_t stripe_stride,
+                              const int32_t *in_idx, int k,
+                              const int32_t *out_idx, int nout,
+                              const uint8_t *tabs, size_t shard_len,
+                              int nstripes, uint32_t *fail, hipStream_t s) {
+  ShardAddr a{nullptr, base, stripe_stride, 0};
+  const int lds = k * nout * 32;
+  const int grid = rs_grid(shard_len, nstripes);
+  if (perm0_ok())
+    hipLaunchKernelGGL((rs_apply_k<true, true>), dim3(grid), dim3(RS_BLOCK),
+                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
+                       size_t(nstripes), fail);
+  else
+    hipLaunchKernelGGL((rs_apply_k<false, true>), dim3(grid), dim3(RS_BLOCK),
+                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
+                       size_t(nstripes), fail);
+}
+
+/* ------------------------------------------------------------------ */
+/* v_perm semantics probe                                               */
+/* ------------------------------------------------------------------ */
+
+__global__ void probe_perm_k(uint32_t *out) {
+  if (threadIdx.x == 0) {
+    out[0] = __builtin_amdgcn_perm(0x44332211u, 0x88776655u, 0x03020100u);
+    out[1] = __builtin_amdgcn_perm(0x44332211u, 0x88776655u, 0x07060504u);
+    out[2] = __builtin_amdgcn_perm(0xAABBCCDDu, 0x11223344u, 0x0B0A0908u);
+  }
+}
+
+static int g_perm0 = -1; /* -1 unknown, 0 safe path, 1 fast path */
+
+int probe_perm_device(void) {
+  uint32_t *d = nullptr;
+  uint32_t h[3] = {1, 1, 1};
+  if (hipMalloc(&d, sizeof(h)) != hipSuccess) return -100;
+  hipLaunchKernelGGL(probe_perm_k, dim3(1), dim3(64), 0, nullptr, d);
+  if (hipMemcpy(h, d, sizeof(h), hipMemcpyDeviceToHost) != hipSuccess) {
+    hipFree(d);
+    return -100;
+  }
+  hipFree(d);
+  /* expected: sel 0-3 -> second operand's bytes, 4-7 -> first operand's,
+   * >=8 -> 0x00 */
+  const bool order_ok = (h[0] == 0x88776655u) && (h[1] == 0x44332211u);
+  if (!order_ok) return -103; /* would need a different lut16 — flag loudly */
+  g_perm0 = (h[2] == 0u) ? 1 : 0;
+  return g_perm0;
+}
+
+static bool perm0_ok() {
+  if (g_perm0 < 0) {
+    int r = probe_perm_device();
+    if (r < 0) g_perm0 = 0; /* safe path */
+  }
+  return g_perm0 == 1;
+}
+
+/* ------------------------------------------------------------------ */
+/* crc32block                                                           */
+/* ------------------------------------------------------------------ */
+
+#define CRC_POLY 0xEDB88320u
+#define CRC_LEN 4
+
+/* slice-by-4 tables, computed on host (gfrs_host.cpp) and copied to this
+ * symbol per device. */
+__device__ uint32_t g_crc_tab4[4][256];
+
+/* x^(8*2^j) mod P, reflected domain — host-filled alongside the tables. */
+__device__ uint32_t g_pow8[40];
+
+/* reflected-domain carry-less multiply mod P (see oracle/crc_ref.c for the
+ * derivation; independent implementation). */
+GFRS_DEV int64_t i64min(int64_t a, int64_t b) { return a < b ? a : b; }
+
+GFRS_DEV uint32_t gf2_mulmod_d(uint32_t a, uint32_t b) {
+  uint32_t prod = 0;
+#pragma unroll 8
+  for (int i = 31; i >= 0; i--) {
+    if ((a >> i) & 1) prod ^= b;
+    b = (b & 1) ? (b >> 1) ^ CRC_POLY : (b >> 1);
+  }
+  return prod;
+}
+
+/* x^(8*len) mod P via binary expansion of len. */
+GFRS_DEV uint32_t x8n_d(uint64_t len) {
+  uint32_t op = 0x80000000u; /* identity */
+  int j = 0;
+  while (len) {
+    if (len & 1) op = gf2_mulmod_d(op, g_pow8[j]);
+    len >>= 1;
+    j++;
+  }
+  return op;
+}
+
+constexpr int CRC_BLOCKT = 256;
+
+/* Raw (no init/final complement) CRC update of a chunk, 4 B at a time via
+ * LDS slice-by-4 tables, byte tail scalar. */
+GFRS_DEV uint32_t crc_chunk(const uint8_t *p, int len,
+                            const uint32_t (*tab)[256]) {
+  uint32_t c = 0;
+  int i = 0;
+  for (; i + 4 <= len; i += 4) {
+    c ^= *reinterpret_cast<const uint32_t *>(p + i);
+    c = tab[3][c & 0xFF] ^ tab[2][(c >> 8) & 0xFF] ^ tab[1][(c >> 16) & 0xFF] ^
+        tab[0][c >> 24];
+  }
+  for (; i < len; i++)
+    c = tab[0][(c ^ p[i]) & 0xFF] ^ (c >> 8);
+  return c;
+}
+
+/* One workgroup per frame.  MODE: 0 = encode (raw src -> framed dst),
+ * 1 = verify (framed src), 2 = decode (framed src -> raw dst). */
+template <int MODE>
+__global__ __launch_bounds__(CRC_BLOCKT) void crc32b_k(
+    uint8_t *__restrict__ dst, size_t dst_stride,
+    const uint8_t *__restrict__ src, size_t src_stride, int64_t n,
+    int64_t block_len, int64_t frames_per_shard, int64_t total_frames,
+    int64_t *__restrict__ bad) {
+  __shared__ uint32_t tab[4][256];
+  __shared__ uint32_t fold[CRC_BLOCKT];
+  for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
+    (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
+  __syncthreads();
+
+  const int64_t payload_full = block_len - CRC_LEN;
+  /* per-thread chunk: covers any block_len; multiple of 4 so every chunk
+   * start stays u32-aligned */
+  const int64_t chunk =
+      (((payload_full + CRC_BLOCKT - 1) / CRC_BLOCKT) + 3) & ~int64_t(3);
+
+  for (int64_t fr = blockIdx.x; fr < total_frames; fr += gridDim.x) {
+    const int64_t shard = fr / frames_per_shard;
+    const int64_t f = fr - shard * frames_per_shard;
+    /* n = raw bytes per shard (encode) or derived from framed_len; the
+     * launcher always passes raw payload total in n. */
+    const int64_t praw0 = f * payload_full;
+    const int64_t payload = i64min(payload_full, n - praw0);
+
+    const uint8_t *sbase = src + shard * src_stride;
+    uint8_t *dbase = dst ? dst + shard * dst_stride : nullptr;
+    /* chunk for this thread */
+    const int64_t c0 = int64_t(threadIdx.x) * chunk;
+    int clen = int(i64min(chunk, payload - c0));
+    if (clen < 0) clen = 0;
+    const uint8_t *payload_src =
+        (MODE == 0) ? sbase + praw0 : sbase + f * block_len + CRC_LEN;
+
+    uint32_t part = crc_chunk(payload_src + c0, clen, tab);
+    /* fold: contribution = part * x^(8*suffix) */
+    const int64_t suffix = clen > 0 ? payload - (c0 + clen) : 0;
+    part = clen > 0 ? gf2_mulmod_d(x8n_d(uint64_t(suffix)), part) : 0;
+    fold[threadIdx.x] = part;
+    __syncthreads();
+    for (int w = CRC_BLOCKT / 2; w > 0; w >>= 1) {
+      if (threadIdx.x < w) fold[threadIdx.x] ^= fold[threadIdx.x + w];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+      /* crc = ~( x^(8*payload)·(~0) ^ fold )  — see DESIGN.md */
+      const uint32_t raw =
+          gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu) ^ fold[0];
+      const uint32_t crc = ~raw;
+      if (MODE == 0) {
+        *reinterpret_cast<uint32_t *>(dbase + f * block_len) = crc;
+      } else {
+        uint32_t want;
+        __builtin_memcpy(&want, sbase + f * block_len, 4); /* LE load */
+        if (want != crc)
+          atomicMin(reinterpret_cast<unsigned long long *>(&bad[shard]),
+                    static_cast<unsigned long long>(f));
+      }
+    }
+    if (MODE == 0) {
+      /* coalesced payload copy (src likely L2-resident after the CRC pass) */
+      uint8_t *fdst = dbase + f * block_len;
+      const uint32_t *sw = reinterpret_cast<const uint32_t *>(sbase + praw0);
+      uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + CRC_LEN);
+      const int64_t words = payload / 4;
+      for (int64_t i = threadIdx.x; i < words; i += CRC_BLOCKT) dw[i] = sw[i];
+      if (threadIdx.x == 0)
+        for (int64_t i = words * 4; i < payload; i++)
+          fdst[CRC_LEN + i] = sbase[praw0 + i];
+    } else if (MODE == 2) {
+      const uint32_t *sw =
+          reinterpret_cast<const uint32_t *>(sbase + f * block_len + CRC_LEN);
+      uint32_t *dw = reinterpret_cast<uint32_t *>(dbase + praw0);
+      const int64_t words = payload / 4;
+      for (int64_t i = threadIdx.x; i < words; i += CRC_BLOCKT) dw[i] = sw[i];
+      if (threadIdx.x == 0)
+        for (int64_t i = words * 4; i < payload; i++)
+          dbase[praw0 + i] = sbase[f * block_len + CRC_LEN + i];
+    }
+    __syncthreads();
+  }
+}
+
+/* Host-side one-time init of the device CRC tables (g_crc_tab4, g_pow8)
+ * for the CURRENT device.  Called by gfrs_host.cpp under its device mutex. */
+int crc_device_init_current(void) {
+  uint32_t tab[4][256];
+  for (uint32_t i = 0; i < 256; i++) {
+    uint32_t c = i;
+    for (int j = 0; j < 8; j++) c = (c & 1) ? (c >> 1) ^ CRC_POLY : c >> 1;
+    tab[0][i] = c;
+  }
+  for (int t = 1; t < 4; t++)
+    for (int i = 0; i < 256; i++)
+      tab[t][i] = tab[0][tab[t - 1][i] & 0xFF] ^ (tab[t - 1][i] >> 8);
+  /* x^(8*2^j) mod P, reflected (identity = 0x80000000) */
+  auto mulmod = [](uint32_t a, uint32_t b) {
+    uint32_t prod = 0;
+    for (int i = 31; i >= 0; i--) {
+      if ((a >> i) & 1) prod ^= b;
+      b = (b & 1) ? (b >> 1) ^ CRC_POLY : (b >> 1);
+    }
+    return prod;
+  };
+  uint32_t pow8[40];
+  uint32_t p = 0x00800000u; /* x^8 reflected */
+  for (int j = 0; j < 40; j++) {
+    pow8[j] = p;
+    p = mulmod(p, p);
+  }
+  if (hipMemcpyToSymbol(HIP_SYMBOL(g_crc_tab4), tab, sizeof(tab)) != hipSuccess)
+    return -100;
+  if (hipMemcpyToSymbol(HIP_SYMBOL(g_pow8), pow8, sizeof(pow8)) != hipSuccess)
+    return -100;
+  return 0;
+}
+
+static int crc_grid(int64_t total_frames) {
+  if (total_frames <= 0) return 1;
+  const int64_t cap = 256 * 8;
+  return int(total_frames < cap ? total_frames : cap);
+}
+
+void launch_crc_encode(uint8_t *dst, size_t dst_stride, const uint8_t *src,
+                       size_t src_stride, int64_t n, int64_t block_len,
+                       int nshards, hipStream_t s) {
+  const int64_t payload = block_len - CRC_LEN;
+  const int64_t fps = (n + payload - 1) / payload;
+  const int64_t total = fps * nshards;
+  hipLaunchKernelGGL((crc32b_k<0>), dim3(crc_grid(total)), dim3(CRC_BLOCKT), 0,
+                     s, dst, dst_stride, src, src_stride, n, block_len, fps,
+                     total, nullptr);
+}
+
+void launch_crc_verify(const uint8_t *framed, size_t stride,
+                       int64_t framed_len, int64_t block_len, int nshards,
+                       int64_t *bad, hipStream_t s) {
+  const int64_t fps = (framed_len + block_len - 1) / block_len;
+  const int64_t n = framed_len - CRC_LEN * fps; /* raw payload total */
+  const int64_t total = fps * nshards;
+  hipLaunchKernelGGL((crc32b_k<1>), dim3(crc_grid(total)), dim3(CRC_BLOCKT), 0,
+                     s, nullptr, 0, framed, stride, n, block_len, fps, total,
+                     bad);
+}
+
+void launch_crc_decode(uint8_t *dst, size_t dst_stride, const uint8_t *framed,
+                       size_t src_stride, int64_t framed_len,
+                       int64_t block_len, int nshards, int64_t *bad,
+                       hipStream_t s) {
+  const int64_t fps = (framed_len + block_len - 1) / block_len;
+  const int64_t n = framed_len - CRC_LEN * fps;
+  const int64_t total = fps * nshards;
+  hipLaunchKernelGGL((crc32b_k<2>), dim3(crc_grid(total)), dim3(CRC_BLOCKT), 0,
+                     s, dst, dst_stride, framed, src_stride, n, block_len, fps,
+                     total, bad);
+}
+
+}  // namespace gfrs

@@ -1,0 +1,171 @@
+/* include/gfrs.h — C ABI of the MI355X-native erasure-coding + checksum
+ * engine for CubeFS blobstore ("gfrs" = GF(2^8) Reed-Solomon).
+ *
+ * This is the drop-in boundary replacing the compute underneath
+ * blobstore/common/ec.Encoder (encoder.go:41-62) and
+ * blobstore/common/crc32block.  The reference has no FFI of its own (the
+ * arithmetic is reached through Go interfaces), so each entry point below
+ * names the Go interface method it replaces; INTEGRATION.md shows the cgo
+ * shim a CubeFS maintainer would add so access/stream and blobnode/worker
+ * call this engine unchanged.
+ *
+ * Memory model: shard buffers are caller-owned, equal-length; data shards
+ * are read-only during Encode (reedsolomon.go:30-32).  Pointers may be HIP
+ * device pointers (GFRS_MEM_DEVICE) or plain host pointers
+ * (GFRS_MEM_HOST, staged through pinned buffers internally — the cgo path).
+ * A context is thread-safe and may be shared, like one ec.Encoder shared by
+ * many goroutines (encoder.go:114-151).
+ */
+#ifndef GFRS_H
+#define GFRS_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* Error codes: one per Go sentinel the cgo shim maps back.
+ * 0 = success; negative = failure. */
+enum {
+  GFRS_OK = 0,
+  GFRS_ERR_INVALID_CODEMODE = -1, /* ec.ErrInvalidCodeMode (encoder.go:35) */
+  GFRS_ERR_TOO_FEW_SHARDS = -2,   /* reedsolomon.ErrTooFewShards (:598) */
+  GFRS_ERR_SHARD_SIZE = -3,       /* reedsolomon.ErrShardSize */
+  GFRS_ERR_SHARD_NO_DATA = -4,    /* reedsolomon.ErrShardNoData */
+  GFRS_ERR_VERIFY = -5,           /* ec.ErrVerify (encoder.go:36) */
+  GFRS_ERR_INVALID_SHARDS = -6,   /* ec.ErrInvalidShards (encoder.go:37) */
+  GFRS_ERR_SHORT_DATA = -7,       /* ec.ErrShortData / rs.ErrShortData */
+  GFRS_ERR_SINGULAR = -8,         /* matrix.errSingular (matrix.go:186) */
+  GFRS_ERR_MISMATCHED_CRC = -9,   /* crc32block.ErrMismatchedCrc */
+  GFRS_ERR_INVALID_BLOCK = -10,   /* crc32block.ErrInvalidBlock */
+  GFRS_ERR_HIP = -100,            /* HIP runtime failure (see gfrs_last_error) */
+  GFRS_ERR_NO_GPU = -101,         /* no MI355X visible; the engine never
+                                     falls back to CPU — by design */
+  GFRS_ERR_NOMEM = -102,
+  GFRS_ERR_UNSUPPORTED = -103,
+};
+
+/* Where shard pointers live. */
+enum { GFRS_MEM_DEVICE = 0, GFRS_MEM_HOST = 1 };
+
+/* codemode.Tactic (codemode.go:156-190). */
+typedef struct gfrs_tactic {
+  int32_t n;              /* data shards */
+  int32_t m;              /* global parity shards */
+  int32_t l;              /* local (AZ) parity shards, 0 for plain RS */
+  int32_t az_count;       /* AZ count; n,m,l all divisible by it */
+  int32_t put_quorum;     /* carried for the shim; unused by compute */
+  int32_t get_quorum;
+  int32_t min_shard_size; /* ec.Buffer alignment floor (buf.go:79-83) */
+} gfrs_tactic;
+
+typedef struct gfrs_ctx gfrs_ctx;
+
+/* ---- lifecycle ---- */
+
+/* ec.NewEncoder(ec.Config{CodeMode: t}) (encoder.go:78-112).
+ * device < 0 selects the current HIP device.  Returns NULL on failure
+ * (gfrs_last_error() has the reason). */
+gfrs_ctx *gfrs_create(const gfrs_tactic *t, int device);
+void gfrs_destroy(gfrs_ctx *ctx);
+const char *gfrs_last_error(void);
+const char *gfrs_version(void);
+int gfrs_device_count(void);
+
+/* Optional: run on an existing HIP stream (e.g. torch's). stream==NULL
+ * reverts to the context's own stream. */
+int gfrs_set_stream(gfrs_ctx *ctx, void *hip_stream);
+int gfrs_synchronize(gfrs_ctx *ctx);
+
+/* ---- single stripe (the ec.Encoder surface) ----
+ * shards: n+m+l pointers, each shard_len bytes, all in `memloc` memory.
+ * All launches are async on the ctx stream when memloc==GFRS_MEM_DEVICE;
+ * host-memory calls block until the result is back. */
+
+/* Encoder.Encode (encoder.go:114 / lrcencoder.go:35): fills parity (and
+ * local parity when l > 0) from the data shards. */
+int gfrs_encode(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
+                int nshards, int memloc);
+
+/* Encoder.Verify (encoder.go:133): *ok=1 when parity matches. */
+int gfrs_verify(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
+                int nshards, int memloc, int *ok);
+
+/* Encoder.Reconstruct / ReconstructData (encoder.go:139-151,
+ * lrcencoder.go:133-207): bad_idx lists missing shard indices; their
+ * buffers must be allocated (shard_len bytes) and are overwritten. */
+int gfrs_reconstruct(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
+                     int nshards, int memloc, const int32_t *bad_idx,
+                     int nbad, int data_only);
+
+/* ---- stripe batches (blobnode repair/migrate bulk path,
+ * worker_slice_recover.go:804-888) ----
+ * Stripes laid out contiguously: stripe s shard i starts at
+ * base + s*stripe_stride + i*shard_len (the ec.Buffer layout, buf.go:24-35).
+ * base is always a device pointer. */
+
+int gfrs_encode_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
+                      size_t stripe_stride, int nstripes);
+/* fail_bitmap (host, optional): bit s set when stripe s fails verify. */
+int gfrs_verify_batch(gfrs_ctx *ctx, const void *base, size_t shard_len,
+                      size_t stripe_stride, int nstripes,
+                      uint64_t *fail_bitmap);
+/* One missing pattern shared by the whole batch (one repair tasklet). */
+int gfrs_reconstruct_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
+                           size_t stripe_stride, int nstripes,
+                           const int32_t *bad_idx, int nbad, int data_only);
+
+/* ---- crc32block (blobstore/common/crc32block) ----
+ * block_len must be a positive multiple of 4096 (util.go:40); frame =
+ * 4 B LE CRC32-IEEE ‖ payload (block.go:22-49).  dst/src are device
+ * pointers; *_host variants stage host memory. */
+
+int64_t gfrs_crc32b_encode_size(int64_t size, int64_t block_len);
+int64_t gfrs_crc32b_decode_size(int64_t size, int64_t block_len);
+
+/* Frame n raw bytes from src into dst (encode.go:86-106).
+ * dst capacity must be gfrs_crc32b_encode_size(n).  Returns bytes
+ * written or a negative error. */
+int64_t gfrs_crc32b_encode(gfrs_ctx *ctx, void *dst, const void *src,
+                           int64_t n, int64_t block_len);
+/* Check all frames (decode.go:84-107).  *bad_block = first failing block
+ * index or -1.  Returns GFRS_OK even when CRCs mismatch (inspect
+ * *bad_block); negative only on launch/arg errors. */
+int gfrs_crc32b_verify(gfrs_ctx *ctx, const void *framed, int64_t framed_len,
+                       int64_t block_len, int64_t *bad_block);
+/* Strip frames into dst, checking CRCs.  Returns payload bytes or
+ * GFRS_ERR_MISMATCHED_CRC. */
+int64_t gfrs_crc32b_decode(gfrs_ctx *ctx, void *dst, const void *framed,
+                           int64_t framed_len, int64_t block_len);
+
+/* Batched framing of many equal-length shards (datafile.go:342-445 write
+ * path): shard i raw at src + i*src_stride, framed at dst + i*dst_stride. */
+int gfrs_crc32b_encode_batch(gfrs_ctx *ctx, void *dst, size_t dst_stride,
+                             const void *src, size_t src_stride, int64_t n,
+                             int64_t block_len, int nshards);
+int gfrs_crc32b_verify_batch(gfrs_ctx *ctx, const void *framed,
+                             size_t stride, int64_t framed_len,
+                             int64_t block_len, int nshards,
+                             int64_t *bad_block_per_shard);
+
+/* ---- ec.Buffer size math (buf.go:67-133) ---- */
+int gfrs_buffer_sizes(const gfrs_tactic *t, int64_t data_size,
+                      int64_t *shard_size, int64_t *ec_data_size,
+                      int64_t *ec_size);
+
+/* ---- introspection for tests ---- */
+/* Copy the (n+m)×n encode matrix into out (row-major). */
+int gfrs_encode_matrix(gfrs_ctx *ctx, uint8_t *out);
+/* GPU-free: compute the klauspost default encode matrix (total×k) so CPU
+ * tests can pin the host math without a device. */
+int gfrs_compute_encode_matrix(int k, int total, uint8_t *out);
+/* Device probe used by tests: returns 1 when v_perm byte-select semantics
+ * match the kernel's assumption (sel>=8 yields 0), 0 otherwise, <0 error. */
+int gfrs_probe_perm(void);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* GFRS_H */

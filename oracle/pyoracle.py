@@ -66,6 +66,14 @@ def _shard_ptrs(shards):
     return arr
 
 
+def gf_mul(a, b):
+    return lib().orc_gf_mul(a, b)
+
+
+def gf_exp(a, n):
+    return lib().orc_gf_exp(a, n)
+
+
 def gf_tables():
     out = np.zeros(74494, dtype=np.uint8)
     lib().orc_gf_tables(_ptr(out), out.size)

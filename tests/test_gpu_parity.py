@@ -1,0 +1,256 @@
+"""GPU parity tests: the HIP engine vs the CPU oracle, bit-exact, through
+the C ABI (include/gfrs.h) — the way blobstore/common/ec's tests pin the
+reference engine (encoder_test.go:53-106), plus golden vectors.
+
+Inputs per SURVEY.md §8d: seeded uniform random bytes (GF LUT timing is
+data-independent but zero-skip fast paths must not exist — random data
+would expose them as mismatches anyway).
+"""
+import os
+
+import numpy as np
+import pytest
+
+torch = pytest.importorskip("torch")
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dev():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    return torch.device("cuda:0")
+
+
+def make_stripe(rng, k, extra, slen, dev):
+    """k random data shards + `extra` zeroed output shards on device."""
+    data = rng.integers(0, 256, (k, slen), dtype=np.uint8)
+    shards = [torch.from_numpy(data[i].copy()).to(dev) for i in range(k)]
+    shards += [torch.zeros(slen, dtype=torch.uint8, device=dev)
+               for _ in range(extra)]
+    return shards, data
+
+
+def cpu_copy(shards):
+    return [s.cpu().numpy().copy() for s in shards]
+
+
+def test_perm_probe():
+    """v_perm byte-select semantics must match the kernel's assumption."""
+    from cubefs_amd import runtime
+    r = runtime.lib().gfrs_probe_perm()
+    assert r >= 0, runtime.lib().gfrs_last_error()
+
+
+@pytest.mark.parametrize("name,slen", [
+    ("EC4P2", 1 << 20),       # BASELINE config 1 shape
+    ("EC6P3", 2048),
+    ("EC6P3", 4093),          # ragged tail (not multiple of 16/4096)
+    ("EC6P3", 1),             # minimum shard
+    ("EC12P4", 8192),
+    ("EC15P12", 4096),        # m > MT: multiple output groups
+    ("EC24P8", 2048),
+])
+def test_encode_matches_oracle(oracle, dev, name, slen):
+    from cubefs_amd import codemode, ec
+    if name == "EC4P2":
+        codemode.extend(241, "EC4P2", codemode.Tactic(4, 2, 0, 1, 5, 0, 2048))
+    t = codemode.get_tactic(name)
+    enc = ec.Encoder(t)
+    rng = np.random.default_rng(0xB10B5703 ^ hash(name) % 1000 ^ slen)
+    shards, _ = make_stripe(rng, t.N, t.M, slen, dev)
+    ref = cpu_copy(shards)
+    oracle.rs_encode(t.N, t.M, ref)
+    enc.encode(shards)
+    got = cpu_copy(shards)
+    for i in range(t.N + t.M):
+        assert np.array_equal(got[i], ref[i]), (name, slen, i)
+    assert enc.verify(shards)
+    # corruption must be detected
+    shards[t.N][slen // 2] ^= 0x5A
+    assert not enc.verify(shards)
+
+
+def test_encode_matches_golden(dev, golden_dir):
+    """Golden vectors frozen in-repo (tests/golden/rs_vectors.npz)."""
+    from cubefs_amd import codemode, ec
+    z = np.load(os.path.join(golden_dir, "rs_vectors.npz"))
+    codemode.extend(240, "LRC12P2L2", codemode.Tactic(12, 2, 2, 2, 14, 0, 2048))
+    for name, tac in [("EC6P3_2048", codemode.get_tactic("EC6P3")),
+                      ("EC12P4_2048", codemode.get_tactic("EC12P4")),
+                      ("LRC12P2L2_2048", codemode.get_tactic("LRC12P2L2")),
+                      ("EC16P20L2_1024", codemode.get_tactic("EC16P20L2")),
+                      ("EC6P10L2_1024", codemode.get_tactic("EC6P10L2"))]:
+        data = z[name + "/data"]
+        parity = z[name + "/parity"]
+        enc = ec.Encoder(tac)
+        shards = [torch.from_numpy(data[i].copy()).to(dev) for i in range(tac.N)]
+        shards += [torch.zeros(data.shape[1], dtype=torch.uint8, device=dev)
+                   for _ in range(tac.M + tac.L)]
+        enc.encode(shards)
+        got = np.stack(cpu_copy(shards)[tac.N:])
+        assert np.array_equal(got, parity), name
+
+
+@pytest.mark.parametrize("bad", [[0], [8], [3, 7], [0, 1, 2], [2, 6, 8]])
+def test_reconstruct_matches_oracle(oracle, dev, bad):
+    from cubefs_amd import codemode, ec
+    t = codemode.get_tactic("EC6P3")
+    enc = ec.Encoder(t)
+    slen = 8192 + 12  # ragged
+    rng = np.random.default_rng(42)
+    shards, _ = make_stripe(rng, t.N, t.M, slen, dev)
+    enc.encode(shards)
+    ref = cpu_copy(shards)
+    for i in bad:
+        shards[i].zero_()
+    enc.reconstruct(shards, bad)
+    got = cpu_copy(shards)
+    for i in range(t.N + t.M):
+        assert np.array_equal(got[i], ref[i]), i
+
+
+def test_reconstruct_data_only(dev):
+    from cubefs_amd import codemode, ec
+    t = codemode.get_tactic("EC6P3")
+    enc = ec.Encoder(t)
+    rng = np.random.default_rng(43)
+    shards, _ = make_stripe(rng, t.N, t.M, 4096, dev)
+    enc.encode(shards)
+    ref = cpu_copy(shards)
+    shards[1].zero_()
+    shards[7].zero_()
+    enc.reconstruct_data(shards, [1, 7])
+    got = cpu_copy(shards)
+    assert np.array_equal(got[1], ref[1])
+    assert not np.array_equal(got[7], ref[7])  # parity left missing
+
+
+def test_reconstruct_too_few(dev):
+    from cubefs_amd import codemode, ec
+    from cubefs_amd.runtime import GfrsError
+    t = codemode.get_tactic("EC6P3")
+    enc = ec.Encoder(t)
+    rng = np.random.default_rng(44)
+    shards, _ = make_stripe(rng, t.N, t.M, 2048, dev)
+    enc.encode(shards)
+    with pytest.raises(GfrsError) as ei:
+        enc.reconstruct(shards, [0, 1, 2, 3])  # 4 missing > m=3
+    assert ei.value.code == -2  # ErrTooFewShards
+
+
+def test_lrc_encode_reconstruct(oracle, dev):
+    """Azure-LRC(12,2,2) — BASELINE config 5 codemode."""
+    from cubefs_amd import codemode, ec
+    codemode.extend(240, "LRC12P2L2", codemode.Tactic(12, 2, 2, 2, 14, 0, 2048))
+    t = codemode.get_tactic("LRC12P2L2")
+    enc = ec.Encoder(t)
+    slen = 65536
+    rng = np.random.default_rng(45)
+    shards, _ = make_stripe(rng, t.N, t.M + t.L, slen, dev)
+    ref = cpu_copy(shards)
+    oracle.lrc_encode(t.N, t.M, t.L, t.AZCount, ref)
+    enc.encode(shards)
+    got = cpu_copy(shards)
+    for i in range(t.total):
+        assert np.array_equal(got[i], ref[i]), i
+    # lose data + global parity + local parity; full recovery incl. locals
+    for i in (3, 12, 14):
+        shards[i].zero_()
+    enc.reconstruct(shards, [3, 12, 14])
+    got = cpu_copy(shards)
+    for i in range(t.total):
+        assert np.array_equal(got[i], ref[i]), i
+
+
+def test_lrc_local_stripe_reconstruct(oracle, dev):
+    """Local-stripe form (lrcencoder.go:147-153): reconstruct inside one AZ
+    without the other AZ's shards."""
+    from cubefs_amd import codemode, ec
+    codemode.extend(240, "LRC12P2L2", codemode.Tactic(12, 2, 2, 2, 14, 0, 2048))
+    t = codemode.get_tactic("LRC12P2L2")
+    enc = ec.Encoder(t)
+    slen = 4096
+    rng = np.random.default_rng(46)
+    shards, _ = make_stripe(rng, t.N, t.M + t.L, slen, dev)
+    enc.encode(shards)
+    full = cpu_copy(shards)
+    # AZ 0 local stripe (global indices)
+    idx, ln_lm, ll = t.local_stripe_in_az(0)
+    local = [shards[i].clone() for i in idx]
+    local[2].zero_()  # lose one member, recover from the local stripe alone
+    enc.reconstruct(local, [2])
+    got = cpu_copy(local)
+    for j, gi in enumerate(idx):
+        assert np.array_equal(got[j], full[gi]), (j, gi)
+
+
+def test_host_memory_path(oracle, dev):
+    """GFRS_MEM_HOST staging (the cgo shim path): numpy in, numpy out."""
+    from cubefs_amd import codemode, ec
+    t = codemode.get_tactic("EC6P3")
+    enc = ec.Encoder(t)
+    rng = np.random.default_rng(47)
+    slen = 100000
+    shards = [rng.integers(0, 256, slen, dtype=np.uint8) for _ in range(t.N)]
+    shards += [np.zeros(slen, np.uint8) for _ in range(t.M)]
+    ref = [s.copy() for s in shards]
+    oracle.rs_encode(t.N, t.M, ref)
+    enc.encode(shards)
+    for i in range(t.N + t.M):
+        assert np.array_equal(shards[i], ref[i]), i
+    assert enc.verify(shards)
+    # host-mode reconstruct
+    shards[2][:] = 0
+    enc.reconstruct(shards, [2])
+    assert np.array_equal(shards[2], ref[2])
+
+
+def test_batch_apis(oracle, dev):
+    from cubefs_amd import codemode, ec
+    t = codemode.get_tactic("EC6P3")
+    enc = ec.Encoder(t)
+    ns, slen = 17, 8192
+    rng = np.random.default_rng(48)
+    batch = torch.from_numpy(
+        rng.integers(0, 256, (ns, t.total, slen), dtype=np.uint8)).to(dev)
+    refs = []
+    for s in range(ns):
+        st = [batch[s, i].cpu().numpy().copy() for i in range(t.total)]
+        oracle.rs_encode(t.N, t.M, st)
+        refs.append(st)
+    enc.encode_batch(batch)
+    enc.synchronize()
+    got = batch.cpu().numpy()
+    for s in range(ns):
+        for i in range(t.total):
+            assert np.array_equal(got[s, i], refs[s][i]), (s, i)
+    assert enc.verify_batch(batch) == [False] * ns
+    batch[3, t.N, 5] ^= 1
+    fails = enc.verify_batch(batch)
+    assert fails[3] and sum(fails) == 1
+    batch[3, t.N, 5] ^= 1
+    # batched reconstruct, uniform missing pattern
+    saved = batch[:, 1].clone()
+    batch[:, 1].zero_()
+    enc.reconstruct_batch(batch, [1])
+    enc.synchronize()
+    assert torch.equal(batch[:, 1], saved)
+
+
+def test_split_join(dev):
+    from cubefs_amd import codemode, ec
+    t = codemode.get_tactic("EC6P3")
+    enc = ec.Encoder(t)
+    rng = np.random.default_rng(49)
+    data = torch.from_numpy(rng.integers(0, 256, 1000, dtype=np.uint8)).to(dev)
+    shards = enc.split(data)
+    assert len(shards) == t.total
+    per = (1000 + t.N - 1) // t.N
+    assert all(int(s.shape[0]) == per for s in shards)
+    enc.encode(shards)
+    import io
+    out = io.BytesIO()
+    enc.join(out, shards, 1000)
+    assert out.getvalue() == data.cpu().numpy().tobytes()
