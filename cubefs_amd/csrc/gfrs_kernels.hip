@@ -123,26 +123,45 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
         uint4 acc[MT];
 #pragma unroll
         for (int r = 0; r < MT; r++) acc[r] = uint4{0, 0, 0, 0};
-        for (int c = 0; c < k; c++) {
-          const uint8_t *in = addr.shard(stripe, in_idx[c], shard_len);
-          const uint4 v = *reinterpret_cast<const uint4 *>(in + off);
+        /* issue up to KB=8 input loads before consuming any — 8 KiB of
+         * HBM reads in flight per wave instead of one dependent load per
+         * coefficient chain (memory-level parallelism, G7) */
+        constexpr int KB = 8;
+        for (int c0 = 0; c0 < k; c0 += KB) {
+          uint4 v[KB];
 #pragma unroll
-          for (int r = 0; r < MT; r++) {
-            if (r < gm) {
-              const int t2 = ((og + r) * k + c) * 2;
-              gfmac16<PERM0>(acc[r], v, ltab[t2], ltab[t2 + 1]);
+          for (int j = 0; j < KB; j++) {
+            if (c0 + j < k) {
+              const uint8_t *in =
+                  addr.shard(stripe, in_idx[c0 + j], shard_len);
+              v[j] = *reinterpret_cast<const uint4 *>(in + off);
+            }
+          }
+#pragma unroll
+          for (int j = 0; j < KB; j++) {
+            if (c0 + j < k) {
+#pragma unroll
+              for (int r = 0; r < MT; r++) {
+                if (r < gm) {
+                  const int t2 = ((og + r) * k + c0 + j) * 2;
+                  gfmac16<PERM0>(acc[r], v[j], ltab[t2], ltab[t2 + 1]);
+                }
+              }
             }
           }
         }
-        for (int r = 0; r < gm; r++) {
-          uint8_t *out = const_cast<uint8_t *>(
-              addr.shard(stripe, out_idx[og + r], shard_len));
-          if (VERIFY) {
-            const uint4 e = *reinterpret_cast<const uint4 *>(out + off);
-            mismatch |= (e.x != acc[r].x) | (e.y != acc[r].y) |
-                        (e.z != acc[r].z) | (e.w != acc[r].w);
-          } else {
-            *reinterpret_cast<uint4 *>(out + off) = acc[r];
+#pragma unroll
+        for (int r = 0; r < MT; r++) {
+          if (r < gm) {
+            uint8_t *out = const_cast<uint8_t *>(
+                addr.shard(stripe, out_idx[og + r], shard_len));
+            if (VERIFY) {
+              const uint4 e = *reinterpret_cast<const uint4 *>(out + off);
+              mismatch |= (e.x != acc[r].x) | (e.y != acc[r].y) |
+                          (e.z != acc[r].z) | (e.w != acc[r].w);
+            } else {
+              *reinterpret_cast<uint4 *>(out + off) = acc[r];
+            }
           }
         }
       }
@@ -265,6 +284,8 @@ __global__ void probe_perm_k(uint32_t *out) {
     out[0] = __builtin_amdgcn_perm(0x44332211u, 0x88776655u, 0x03020100u);
     out[1] = __builtin_amdgcn_perm(0x44332211u, 0x88776655u, 0x07060504u);
     out[2] = __builtin_amdgcn_perm(0xAABBCCDDu, 0x11223344u, 0x0B0A0908u);
+    out[3] = __builtin_amdgcn_perm(0xAABBCCDDu, 0x11223344u, 0x0F0E0D0Cu);
+    out[4] = __builtin_amdgcn_perm(0x80808080u, 0x7F7F7F7Fu, 0x0B0A0908u);
   }
 }
 
@@ -272,7 +293,7 @@ static int g_perm0 = -1; /* -1 unknown, 0 safe path, 1 fast path */
 
 int probe_perm_device(void) {
   uint32_t *d = nullptr;
-  uint32_t h[3] = {1, 1, 1};
+  uint32_t h[5] = {1, 1, 1, 1, 1};
   if (hipMalloc(&d, sizeof(h)) != hipSuccess) return -100;
   hipLaunchKernelGGL(probe_perm_k, dim3(1), dim3(64), 0, nullptr, d);
   if (hipMemcpy(h, d, sizeof(h), hipMemcpyDeviceToHost) != hipSuccess) {
@@ -284,6 +305,8 @@ int probe_perm_device(void) {
    * >=8 -> 0x00 */
   const bool order_ok = (h[0] == 0x88776655u) && (h[1] == 0x44332211u);
   if (!order_ok) return -103; /* would need a different lut16 — flag loudly */
+  printf("gfrs perm probe: sel8-11(lo=0x11223344,hi=0xAABBCCDD)=%08x "
+         "sel12-15=%08x sel8-11(sign)=%08x\n", h[2], h[3], h[4]);
   g_perm0 = (h[2] == 0u) ? 1 : 0;
   return g_perm0;
 }
@@ -339,18 +362,24 @@ GFRS_DEV uint32_t x8n_d(uint64_t len) {
 constexpr int CRC_BLOCKT = 256;
 
 /* Raw (no init/final complement) CRC update of a chunk, 4 B at a time via
- * LDS slice-by-4 tables, byte tail scalar. */
+ * LDS slice-by-4 tables, byte tail scalar.  When dst != nullptr the chunk
+ * is simultaneously copied there (the frame/strip move fused into the CRC
+ * pass so the payload crosses HBM exactly once each way). */
 GFRS_DEV uint32_t crc_chunk(const uint8_t *p, int len,
-                            const uint32_t (*tab)[256]) {
+                            const uint32_t (*tab)[256], uint8_t *dst) {
   uint32_t c = 0;
   int i = 0;
   for (; i + 4 <= len; i += 4) {
-    c ^= *reinterpret_cast<const uint32_t *>(p + i);
+    const uint32_t w = *reinterpret_cast<const uint32_t *>(p + i);
+    if (dst) *reinterpret_cast<uint32_t *>(dst + i) = w;
+    c ^= w;
     c = tab[3][c & 0xFF] ^ tab[2][(c >> 8) & 0xFF] ^ tab[1][(c >> 16) & 0xFF] ^
         tab[0][c >> 24];
   }
-  for (; i < len; i++)
+  for (; i < len; i++) {
+    if (dst) dst[i] = p[i];
     c = tab[0][(c ^ p[i]) & 0xFF] ^ (c >> 8);
+  }
   return c;
 }
 
@@ -390,8 +419,11 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_k(
     if (clen < 0) clen = 0;
     const uint8_t *payload_src =
         (MODE == 0) ? sbase + praw0 : sbase + f * block_len + CRC_LEN;
+    uint8_t *payload_dst = nullptr;
+    if (MODE == 0) payload_dst = dbase + f * block_len + CRC_LEN + c0;
+    if (MODE == 2) payload_dst = dbase + praw0 + c0;
 
-    uint32_t part = crc_chunk(payload_src + c0, clen, tab);
+    uint32_t part = crc_chunk(payload_src + c0, clen, tab, payload_dst);
     /* fold: contribution = part * x^(8*suffix) */
     const int64_t suffix = clen > 0 ? payload - (c0 + clen) : 0;
     part = clen > 0 ? gf2_mulmod_d(x8n_d(uint64_t(suffix)), part) : 0;
@@ -415,26 +447,6 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_k(
           atomicMin(reinterpret_cast<unsigned long long *>(&bad[shard]),
                     static_cast<unsigned long long>(f));
       }
-    }
-    if (MODE == 0) {
-      /* coalesced payload copy (src likely L2-resident after the CRC pass) */
-      uint8_t *fdst = dbase + f * block_len;
-      const uint32_t *sw = reinterpret_cast<const uint32_t *>(sbase + praw0);
-      uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + CRC_LEN);
-      const int64_t words = payload / 4;
-      for (int64_t i = threadIdx.x; i < words; i += CRC_BLOCKT) dw[i] = sw[i];
-      if (threadIdx.x == 0)
-        for (int64_t i = words * 4; i < payload; i++)
-          fdst[CRC_LEN + i] = sbase[praw0 + i];
-    } else if (MODE == 2) {
-      const uint32_t *sw =
-          reinterpret_cast<const uint32_t *>(sbase + f * block_len + CRC_LEN);
-      uint32_t *dw = reinterpret_cast<uint32_t *>(dbase + praw0);
-      const int64_t words = payload / 4;
-      for (int64_t i = threadIdx.x; i < words; i += CRC_BLOCKT) dw[i] = sw[i];
-      if (threadIdx.x == 0)
-        for (int64_t i = words * 4; i < payload; i++)
-          dbase[praw0 + i] = sbase[f * block_len + CRC_LEN + i];
     }
     __syncthreads();
   }
