@@ -36,7 +36,14 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--stripes", type=int, default=1024)
     p.add_argument("--shard-mib", type=int, default=8)
-    p.add_argument("--codemode", default="EC6P3")
+    p.add_argument("--codemode", default="EC6P3",
+                   help="EC6P3/EC12P4/... or LRC12P2L2 (registered via Extend)")
+    p.add_argument("--workload", default="encode",
+                   choices=["encode", "reconstruct"],
+                   help="encode = BASELINE configs[1]+CRC; reconstruct = "
+                        "configs[2]: 1-shard reconstruct + crc32block verify")
+    p.add_argument("--bad-idx", type=int, default=2,
+                   help="shard index reconstructed in --workload reconstruct")
     p.add_argument("--no-crc", action="store_true",
                    help="EC encode only (config label form)")
     p.add_argument("--cpu-sample-stripes", type=int, default=24)
@@ -123,6 +130,9 @@ def main():
     torch.cuda.set_device(local_rank)
     dev = torch.device("cuda", local_rank)
 
+    if args.codemode == "LRC12P2L2":
+        codemode.extend(240, "LRC12P2L2",
+                        codemode.Tactic(12, 2, 2, 2, 14, 0, 2048))
     t = codemode.get_tactic(args.codemode)
     S = args.shard_mib << 20
     ns = args.stripes
@@ -151,14 +161,32 @@ def main():
 
     flat = batch.view(ns * t.total, S)
 
+    if args.workload == "reconstruct":
+        # BASELINE configs[2]: parity must exist, frames pre-built; the
+        # step recomputes shard bad_idx for every stripe and CRC-verifies
+        # the framed shards (the repair read path)
+        enc.encode_batch(batch)
+        enc.synchronize()
+        if with_crc:
+            codec.encode_batch(framed, flat)
+            codec.synchronize()
+        bad = [args.bad_idx]
+
     def step(events=None):
         if events:
             events[0].record()
-        enc.encode_batch(batch)
-        if events:
-            events[1].record()
-        if with_crc:
-            codec.encode_batch(framed, flat)
+        if args.workload == "encode":
+            enc.encode_batch(batch)
+            if events:
+                events[1].record()
+            if with_crc:
+                codec.encode_batch(framed, flat)
+        else:
+            enc.reconstruct_batch(batch, bad)
+            if events:
+                events[1].record()
+            if with_crc:
+                codec.verify_batch(framed)
 
     # warmup
     for _ in range(args.warmup):
@@ -190,16 +218,23 @@ def main():
         total_src = sum(r["src_bytes"] for r in recs)
         value = total_src / GIB / elapsed
         total_shard = total_src / t.N * t.total / GIB / elapsed
-        workload = "RS(%d+%d)%s, %d MiB shards, %d stripes/GPU" % (
-            t.N, t.M, "+crc32block" if with_crc else "", args.shard_mib, ns)
+        kind = "encode" if args.workload == "encode" else             "reconstruct[%d]" % args.bad_idx
+        workload = "RS(%d+%d%s) %s%s, %d MiB shards, %d stripes/GPU" % (
+            t.N, t.M, "+L%d" % t.L if t.L else "", kind,
+            "+crc32block" if with_crc else "", args.shard_mib, ns)
         # roofline of the dominant kernel (rs_apply over the whole batch,
-        # one launch per step): algorithmic bytes = (k+m)·S·stripes
-        alg_bytes = float(t.total * S * ns)
+        # one launch per step): encode reads k·S writes (m+l)·S per stripe;
+        # 1-shard reconstruct reads k·S writes 1·S
+        if args.workload == "encode":
+            alg_bytes = float(t.total * S * ns)
+        else:
+            alg_bytes = float((t.N + len(bad)) * S * ns)
         avg_enc_s = (sum(enc_ms) / len(enc_ms)) / 1e3
         achieved = alg_bytes / avg_enc_s
         traffic = hbm_traffic_lookup(workload)
         cpu_base = None
-        if not args.skip_cpu_baseline and world == 1:
+        if (not args.skip_cpu_baseline and world == 1
+                and args.workload == "encode"):
             cpu_base = cpu_baseline_leg(t, S, args.cpu_sample_stripes, with_crc)
         out = {
             "metric": "GiB/s encode+CRC throughput, RS(k+m) per stripe, "

@@ -148,7 +148,9 @@ struct gfrs_ctx_impl {
   std::vector<uint8_t> enc_matrix;   /* (n+m)×n */
   std::vector<uint8_t> local_matrix; /* (local_n+local_m)×local_n */
   DevPlan enc_plan;                  /* global encode */
-  std::vector<DevPlan *> local_enc;  /* per-AZ local encode */
+  std::vector<DevPlan *> local_enc;  /* per-AZ local encode (global idx) */
+  DevPlan local_plan;                /* local engine, identity idx (one
+                                        local-stripe set, lrcencoder.go:94) */
   std::map<uint64_t, DevPlan *> dec_cache; /* missing-bitmask → data decode */
   std::map<uint64_t, DevPlan *> par_cache; /* missing-bitmask → parity rows */
 
@@ -287,6 +289,12 @@ gfrs_ctx *gfrs_create(const gfrs_tactic *t, int device) {
         auto *p = new DevPlan();
         ok = p->upload(in, out, lrows, c->stream) == GFRS_OK;
         c->local_enc.push_back(p);
+      }
+      if (ok) {
+        std::vector<int32_t> in(c->local_n), out(c->local_m);
+        for (int i = 0; i < c->local_n; i++) in[i] = i;
+        for (int i = 0; i < c->local_m; i++) out[i] = c->local_n + i;
+        ok = c->local_plan.upload(in, out, lrows, c->stream) == GFRS_OK;
       }
     }
   }
@@ -608,7 +616,51 @@ int gfrs_encode_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
 int gfrs_verify(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
                 int nshards, int memloc, int *ok) {
   auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
-  if (nshards != c->total) return GFRS_ERR_INVALID_SHARDS;
+  bool local_form = c->t.l > 0 && nshards == c->total / c->t.az_count;
+  if (nshards != c->total && !local_form) return GFRS_ERR_INVALID_SHARDS;
+  if (local_form) {
+    /* one local stripe set (lrcencoder.go:94-99) */
+    std::lock_guard<std::mutex> lk(c->mu);
+    StreamGuard g(c);
+    int rc;
+    if (memloc == GFRS_MEM_HOST) {
+      size_t tot = size_t(nshards) * shard_len;
+      if ((rc = c->stage_dev.ensure(tot)) != GFRS_OK) return rc;
+      if ((rc = c->stage_pin.ensure(tot)) != GFRS_OK) return rc;
+      uint8_t *pin = (uint8_t *)c->stage_pin.p;
+      for (int i = 0; i < nshards; i++)
+        memcpy(pin + size_t(i) * shard_len, shards[i], shard_len);
+      HIP_TRY(hipMemcpyAsync(c->stage_dev.p, pin, tot, hipMemcpyHostToDevice,
+                             c->stream));
+      if ((rc = c->fail_buf.ensure(4)) != GFRS_OK) return rc;
+      HIP_TRY(hipMemsetAsync(c->fail_buf.p, 0, 4, c->stream));
+      launch_rs_verify_strided((uint64_t)c->stage_dev.p, tot,
+                               (const int32_t *)c->local_plan.in_idx.p,
+                               c->local_plan.k,
+                               (const int32_t *)c->local_plan.out_idx.p,
+                               c->local_plan.nout,
+                               (const uint8_t *)c->local_plan.tabs.p,
+                               shard_len, 1, (uint32_t *)c->fail_buf.p,
+                               c->stream);
+    } else {
+      if ((rc = upload_ptrs(c, shards, nshards)) != GFRS_OK) return rc;
+      if ((rc = c->fail_buf.ensure(4)) != GFRS_OK) return rc;
+      HIP_TRY(hipMemsetAsync(c->fail_buf.p, 0, 4, c->stream));
+      launch_rs_verify((const uint64_t *)c->ptr_buf.p, nshards,
+                       (const int32_t *)c->local_plan.in_idx.p,
+                       c->local_plan.k,
+                       (const int32_t *)c->local_plan.out_idx.p,
+                       c->local_plan.nout,
+                       (const uint8_t *)c->local_plan.tabs.p, shard_len, 1,
+                       (uint32_t *)c->fail_buf.p, c->stream);
+    }
+    uint32_t fail = 0;
+    HIP_TRY(hipMemcpyAsync(&fail, c->fail_buf.p, 4, hipMemcpyDeviceToHost,
+                           c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    *ok = fail == 0;
+    return GFRS_OK;
+  }
   std::lock_guard<std::mutex> lk(c->mu);
   StreamGuard g(c);
   int rc;

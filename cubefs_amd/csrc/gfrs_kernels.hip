@@ -24,7 +24,16 @@
  */
 #include "gfrs_internal.h"
 
+#include <cstdlib>
+
 namespace gfrs {
+
+static int env_grid(const char *name, int dflt) {
+  const char *v = getenv(name);
+  if (!v) return dflt;
+  int x = atoi(v);
+  return x > 0 ? x : dflt;
+}
 
 #define GFRS_DEV __device__ __forceinline__
 
@@ -199,7 +208,7 @@ static int rs_grid(size_t shard_len, size_t nstripes) {
   size_t tiles = ((shard_len + RS_TILE - 1) / RS_TILE) * nstripes;
   if (tiles == 0) tiles = 1;
   /* memory-bound: cap and grid-stride (cdna_hip_programming.md G11) */
-  const size_t cap = 256 * 8;
+  const size_t cap = size_t(env_grid("GFRS_RS_GRID", 256 * 8));
   return int(tiles < cap ? tiles : cap);
 }
 
@@ -488,7 +497,7 @@ int crc_device_init_current(void) {
 
 static int crc_grid(int64_t total_frames) {
   if (total_frames <= 0) return 1;
-  const int64_t cap = 256 * 8;
+  const int64_t cap = env_grid("GFRS_CRC_GRID", 256 * 8);
   return int(total_frames < cap ? total_frames : cap);
 }
 

@@ -95,11 +95,9 @@ class Encoder:
         """Verify (encoder.go:133, lrcencoder.go:89).  Accepts a full set
         or, for LRC, one local stripe set."""
         t = self.tactic
-        if t.L > 0 and len(shards) == t.total // t.AZCount:
-            # local-stripe verify is a reconstruct-free parity check the
-            # C side does not expose directly; emulate via local encoder
-            raise NotImplementedError("local-stripe verify: use full set")
-        arr, ln, loc = self._ptrs(shards)
+        expect = len(shards) if (
+            t.L > 0 and len(shards) == t.total // t.AZCount) else t.total
+        arr, ln, loc = self._ptrs(shards, expect)
         ok = ctypes.c_int(0)
         check(lib().gfrs_verify(self._ctx, arr, ln, len(shards), loc,
                                 ctypes.byref(ok)), "verify")
