@@ -24,6 +24,7 @@
  */
 #include "gfrs_internal.h"
 
+#include <cstdio>
 #include <cstdlib>
 
 namespace gfrs {
@@ -33,6 +34,29 @@ static int env_grid(const char *name, int dflt) {
   if (!v) return dflt;
   int x = atoi(v);
   return x > 0 ? x : dflt;
+}
+
+typedef uint32_t u32x4 __attribute__((ext_vector_type(4)));
+
+static bool nt_enabled() {
+  static int v = -1;
+  if (v < 0) {
+    const char *e = getenv("GFRS_NT");
+    v = e ? (atoi(e) != 0) : 0;
+  }
+  return v == 1;
+}
+
+/* Streaming 16-B store; NT bypasses L2 (outputs are never re-read by the
+ * producing kernel). */
+template <bool NT>
+__device__ __forceinline__ void store16(uint8_t *p, const uint4 v) {
+  if (NT) {
+    u32x4 x = {v.x, v.y, v.z, v.w};
+    __builtin_nontemporal_store(x, reinterpret_cast<u32x4 *>(p));
+  } else {
+    *reinterpret_cast<uint4 *>(p) = v;
+  }
 }
 
 #define GFRS_DEV __device__ __forceinline__
@@ -104,7 +128,7 @@ struct ShardAddr {
  * tabs: [nout*k][32] per-coefficient lo|hi tables, staged to LDS.
  * Outputs processed in groups of MT so inputs stream from HBM once per
  * group (arithmetic intensity k·m/(k+m) table-xors per byte; HBM-bound). */
-template <bool PERM0, bool VERIFY>
+template <bool PERM0, bool VERIFY, bool NT>
 __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
     ShardAddr addr, const int32_t *__restrict__ in_idx, int k,
     const int32_t *__restrict__ out_idx, int nout,
@@ -169,7 +193,7 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
               mismatch |= (e.x != acc[r].x) | (e.y != acc[r].y) |
                           (e.z != acc[r].z) | (e.w != acc[r].w);
             } else {
-              *reinterpret_cast<uint4 *>(out + off) = acc[r];
+              store16<NT>(out + off, acc[r]);
             }
           }
         }
@@ -214,21 +238,36 @@ static int rs_grid(size_t shard_len, size_t nstripes) {
 
 static bool perm0_ok(); /* below */
 
+template <bool VERIFY>
+static void rs_dispatch(const ShardAddr &a, const int32_t *in_idx, int k,
+                        const int32_t *out_idx, int nout, const uint8_t *tabs,
+                        size_t shard_len, int nstripes, uint32_t *fail,
+                        hipStream_t s) {
+  const int lds = k * nout * 32;
+  const int grid = rs_grid(shard_len, nstripes);
+  const bool p0 = perm0_ok(), nt = nt_enabled() && !VERIFY;
+#define GFRS_GO(P, N)                                                       \
+  hipLaunchKernelGGL((rs_apply_k<P, VERIFY, N>), dim3(grid),                \
+                     dim3(RS_BLOCK), lds, s, a, in_idx, k, out_idx, nout,   \
+                     tabs, shard_len, size_t(nstripes), fail)
+  if (p0 && nt)
+    GFRS_GO(true, true);
+  else if (p0)
+    GFRS_GO(true, false);
+  else if (nt)
+    GFRS_GO(false, true);
+  else
+    GFRS_GO(false, false);
+#undef GFRS_GO
+}
+
 void launch_rs_apply(const uint64_t *ptrs, int nptr, const int32_t *in_idx,
                      int k, const int32_t *out_idx, int nout,
                      const uint8_t *tabs, size_t shard_len, int nstripes,
                      hipStream_t s) {
   ShardAddr a{ptrs, 0, 0, nptr};
-  const int lds = k * nout * 32;
-  const int grid = rs_grid(shard_len, nstripes);
-  if (perm0_ok())
-    hipLaunchKernelGGL((rs_apply_k<true, false>), dim3(grid), dim3(RS_BLOCK),
-                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
-                       size_t(nstripes), nullptr);
-  else
-    hipLaunchKernelGGL((rs_apply_k<false, false>), dim3(grid), dim3(RS_BLOCK),
-                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
-                       size_t(nstripes), nullptr);
+  rs_dispatch<false>(a, in_idx, k, out_idx, nout, tabs, shard_len, nstripes,
+                     nullptr, s);
 }
 
 void launch_rs_apply_strided(uint64_t base, uint64_t stripe_stride,
@@ -237,16 +276,8 @@ void launch_rs_apply_strided(uint64_t base, uint64_t stripe_stride,
                              const uint8_t *tabs, size_t shard_len,
                              int nstripes, hipStream_t s) {
   ShardAddr a{nullptr, base, stripe_stride, 0};
-  const int lds = k * nout * 32;
-  const int grid = rs_grid(shard_len, nstripes);
-  if (perm0_ok())
-    hipLaunchKernelGGL((rs_apply_k<true, false>), dim3(grid), dim3(RS_BLOCK),
-                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
-                       size_t(nstripes), nullptr);
-  else
-    hipLaunchKernelGGL((rs_apply_k<false, false>), dim3(grid), dim3(RS_BLOCK),
-                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
-                       size_t(nstripes), nullptr);
+  rs_dispatch<false>(a, in_idx, k, out_idx, nout, tabs, shard_len, nstripes,
+                     nullptr, s);
 }
 
 void launch_rs_verify(const uint64_t *ptrs, int nptr, const int32_t *in_idx,
@@ -254,16 +285,8 @@ void launch_rs_verify(const uint64_t *ptrs, int nptr, const int32_t *in_idx,
                       const uint8_t *tabs, size_t shard_len, int nstripes,
                       uint32_t *fail, hipStream_t s) {
   ShardAddr a{ptrs, 0, 0, nptr};
-  const int lds = k * nout * 32;
-  const int grid = rs_grid(shard_len, nstripes);
-  if (perm0_ok())
-    hipLaunchKernelGGL((rs_apply_k<true, true>), dim3(grid), dim3(RS_BLOCK),
-                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
-                       size_t(nstripes), fail);
-  else
-    hipLaunchKernelGGL((rs_apply_k<false, true>), dim3(grid), dim3(RS_BLOCK),
-                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
-                       size_t(nstripes), fail);
+  rs_dispatch<true>(a, in_idx, k, out_idx, nout, tabs, shard_len, nstripes,
+                    fail, s);
 }
 
 void launch_rs_verify_strided(uint64_t base, uint64_t stripe_stride,
@@ -272,16 +295,8 @@ void launch_rs_verify_strided(uint64_t base, uint64_t stripe_stride,
                               const uint8_t *tabs, size_t shard_len,
                               int nstripes, uint32_t *fail, hipStream_t s) {
   ShardAddr a{nullptr, base, stripe_stride, 0};
-  const int lds = k * nout * 32;
-  const int grid = rs_grid(shard_len, nstripes);
-  if (perm0_ok())
-    hipLaunchKernelGGL((rs_apply_k<true, true>), dim3(grid), dim3(RS_BLOCK),
-                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
-                       size_t(nstripes), fail);
-  else
-    hipLaunchKernelGGL((rs_apply_k<false, true>), dim3(grid), dim3(RS_BLOCK),
-                       lds, s, a, in_idx, k, out_idx, nout, tabs, shard_len,
-                       size_t(nstripes), fail);
+  rs_dispatch<true>(a, in_idx, k, out_idx, nout, tabs, shard_len, nstripes,
+                    fail, s);
 }
 
 /* ------------------------------------------------------------------ */
@@ -314,8 +329,9 @@ int probe_perm_device(void) {
    * >=8 -> 0x00 */
   const bool order_ok = (h[0] == 0x88776655u) && (h[1] == 0x44332211u);
   if (!order_ok) return -103; /* would need a different lut16 — flag loudly */
-  printf("gfrs perm probe: sel8-11(lo=0x11223344,hi=0xAABBCCDD)=%08x "
-         "sel12-15=%08x sel8-11(sign)=%08x\n", h[2], h[3], h[4]);
+  fprintf(stderr,
+          "gfrs perm probe: sel8-11(lo=0x11223344,hi=0xAABBCCDD)=%08x "
+          "sel12-15=%08x sel8-11(sign)=%08x\n", h[2], h[3], h[4]);
   g_perm0 = (h[2] == 0u) ? 1 : 0;
   return g_perm0;
 }
@@ -461,6 +477,129 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_k(
   }
 }
 
+static int crc_grid(int64_t total_frames) {
+  if (total_frames <= 0) return 1;
+  const int64_t cap = env_grid("GFRS_CRC_GRID", 256 * 8);
+  return int(total_frames < cap ? total_frames : cap);
+}
+
+/* LDS-staged crc32block kernel for the production 64 KiB block: the
+ * whole frame is staged through LDS so every global access is coalesced —
+ * read the payload once (fused copy to dst for encode/decode), CRC the
+ * 256 B chunks out of LDS (chunk stride 272 B keeps ds_read_b128
+ * conflict-free: lane t hits dword banks 68t+4i mod 64, distinct within
+ * each 16-lane group).  Traffic = S read + S write, the algorithmic
+ * minimum. */
+constexpr int STG_CHUNK = 256;
+constexpr int STG_STRIDE = STG_CHUNK + 16;
+
+template <int MODE, bool NT>
+__global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
+    uint8_t *__restrict__ dst, size_t dst_stride,
+    const uint8_t *__restrict__ src, size_t src_stride, int64_t n,
+    int64_t frames_per_shard, int64_t total_frames, int64_t *__restrict__ bad) {
+  constexpr int64_t block_len = 65536;
+  constexpr int64_t payload_full = block_len - CRC_LEN;
+  extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
+  uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
+  uint32_t *fold = reinterpret_cast<uint32_t *>(smem + 4096);
+  uint8_t *stage = smem + 4096 + 1024; /* 256 chunks x 272 B */
+  for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
+    (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
+  __syncthreads();
+
+  for (int64_t fr = blockIdx.x; fr < total_frames; fr += gridDim.x) {
+    const int64_t shard = fr / frames_per_shard;
+    const int64_t f = fr - shard * frames_per_shard;
+    const int64_t praw0 = f * payload_full;
+    const int64_t payload = i64min(payload_full, n - praw0);
+    const uint8_t *sbase = src + shard * src_stride;
+    uint8_t *dbase = dst ? dst + shard * dst_stride : nullptr;
+    const uint8_t *psrc =
+        (MODE == 0) ? sbase + praw0 : sbase + f * block_len + CRC_LEN;
+    uint8_t *pdst = nullptr;
+    if (MODE == 0) pdst = dbase + f * block_len + CRC_LEN;
+    if (MODE == 2) pdst = dbase + praw0;
+
+    /* stage in (coalesced) + fused copy out (coalesced) */
+    const int64_t words = payload >> 2;
+    for (int64_t w = threadIdx.x; w < words; w += CRC_BLOCKT) {
+      const uint32_t x = *reinterpret_cast<const uint32_t *>(psrc + 4 * w);
+      if (MODE != 1) {
+        if (NT)
+          __builtin_nontemporal_store(x,
+                                      reinterpret_cast<uint32_t *>(pdst + 4 * w));
+        else
+          *reinterpret_cast<uint32_t *>(pdst + 4 * w) = x;
+      }
+      const int64_t p = 4 * w;
+      *reinterpret_cast<uint32_t *>(
+          &stage[(p >> 8) * STG_STRIDE + (p & (STG_CHUNK - 1))]) = x;
+    }
+    if (threadIdx.x == 0)
+      for (int64_t p = words * 4; p < payload; p++) {
+        const uint8_t x = psrc[p];
+        if (MODE != 1) pdst[p] = x;
+        stage[(p >> 8) * STG_STRIDE + (p & (STG_CHUNK - 1))] = x;
+      }
+    __syncthreads();
+
+    /* per-chunk CRC out of LDS */
+    const int64_t c0 = int64_t(threadIdx.x) * STG_CHUNK;
+    int clen = int(i64min(int64_t(STG_CHUNK), payload - c0));
+    if (clen < 0) clen = 0;
+    uint32_t part =
+        crc_chunk(stage + threadIdx.x * STG_STRIDE, clen, tab, nullptr);
+    const int64_t suffix = clen > 0 ? payload - (c0 + clen) : 0;
+    part = clen > 0 ? gf2_mulmod_d(x8n_d(uint64_t(suffix)), part) : 0;
+    fold[threadIdx.x] = part;
+    __syncthreads();
+    for (int w = CRC_BLOCKT / 2; w > 0; w >>= 1) {
+      if (threadIdx.x < w) fold[threadIdx.x] ^= fold[threadIdx.x + w];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+      const uint32_t raw =
+          gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu) ^ fold[0];
+      const uint32_t crc = ~raw;
+      if (MODE == 0) {
+        *reinterpret_cast<uint32_t *>(dbase + f * block_len) = crc;
+      } else {
+        uint32_t want;
+        __builtin_memcpy(&want, sbase + f * block_len, 4);
+        if (want != crc)
+          atomicMin(reinterpret_cast<unsigned long long *>(&bad[shard]),
+                    static_cast<unsigned long long>(f));
+      }
+    }
+    __syncthreads();
+  }
+}
+
+constexpr int STG_LDS = 4096 + 1024 + 256 * STG_STRIDE;
+
+template <int MODE>
+static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
+                         size_t src_stride, int64_t n, int64_t block_len,
+                         int64_t fps, int64_t total, int64_t *bad,
+                         hipStream_t s) {
+  const int grid = crc_grid(total);
+  if (block_len == 65536) {
+    if (nt_enabled() && MODE != 1)
+      hipLaunchKernelGGL((crc32b_staged_k<MODE, true>), dim3(grid),
+                         dim3(CRC_BLOCKT), STG_LDS, s, dst, dst_stride, src,
+                         src_stride, n, fps, total, bad);
+    else
+      hipLaunchKernelGGL((crc32b_staged_k<MODE, false>), dim3(grid),
+                         dim3(CRC_BLOCKT), STG_LDS, s, dst, dst_stride, src,
+                         src_stride, n, fps, total, bad);
+  } else {
+    hipLaunchKernelGGL((crc32b_k<MODE>), dim3(grid), dim3(CRC_BLOCKT), 0, s,
+                       dst, dst_stride, src, src_stride, n, block_len, fps,
+                       total, bad);
+  }
+}
+
 /* Host-side one-time init of the device CRC tables (g_crc_tab4, g_pow8)
  * for the CURRENT device.  Called by gfrs_host.cpp under its device mutex. */
 int crc_device_init_current(void) {
@@ -495,21 +634,13 @@ int crc_device_init_current(void) {
   return 0;
 }
 
-static int crc_grid(int64_t total_frames) {
-  if (total_frames <= 0) return 1;
-  const int64_t cap = env_grid("GFRS_CRC_GRID", 256 * 8);
-  return int(total_frames < cap ? total_frames : cap);
-}
-
 void launch_crc_encode(uint8_t *dst, size_t dst_stride, const uint8_t *src,
                        size_t src_stride, int64_t n, int64_t block_len,
                        int nshards, hipStream_t s) {
   const int64_t payload = block_len - CRC_LEN;
   const int64_t fps = (n + payload - 1) / payload;
-  const int64_t total = fps * nshards;
-  hipLaunchKernelGGL((crc32b_k<0>), dim3(crc_grid(total)), dim3(CRC_BLOCKT), 0,
-                     s, dst, dst_stride, src, src_stride, n, block_len, fps,
-                     total, nullptr);
+  crc_dispatch<0>(dst, dst_stride, src, src_stride, n, block_len, fps,
+                  fps * nshards, nullptr, s);
 }
 
 void launch_crc_verify(const uint8_t *framed, size_t stride,
@@ -517,10 +648,8 @@ void launch_crc_verify(const uint8_t *framed, size_t stride,
                        int64_t *bad, hipStream_t s) {
   const int64_t fps = (framed_len + block_len - 1) / block_len;
   const int64_t n = framed_len - CRC_LEN * fps; /* raw payload total */
-  const int64_t total = fps * nshards;
-  hipLaunchKernelGGL((crc32b_k<1>), dim3(crc_grid(total)), dim3(CRC_BLOCKT), 0,
-                     s, nullptr, 0, framed, stride, n, block_len, fps, total,
-                     bad);
+  crc_dispatch<1>(nullptr, 0, framed, stride, n, block_len, fps,
+                  fps * nshards, bad, s);
 }
 
 void launch_crc_decode(uint8_t *dst, size_t dst_stride, const uint8_t *framed,
@@ -529,10 +658,8 @@ void launch_crc_decode(uint8_t *dst, size_t dst_stride, const uint8_t *framed,
                        hipStream_t s) {
   const int64_t fps = (framed_len + block_len - 1) / block_len;
   const int64_t n = framed_len - CRC_LEN * fps;
-  const int64_t total = fps * nshards;
-  hipLaunchKernelGGL((crc32b_k<2>), dim3(crc_grid(total)), dim3(CRC_BLOCKT), 0,
-                     s, dst, dst_stride, framed, src_stride, n, block_len, fps,
-                     total, bad);
+  crc_dispatch<2>(dst, dst_stride, framed, src_stride, n, block_len, fps,
+                  fps * nshards, bad, s);
 }
 
 }  // namespace gfrs
