@@ -816,6 +816,111 @@ int gfrs_reconstruct_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
   return GFRS_OK;
 }
 
+/* ---------------- blobnode shard images ---------------- */
+
+/* host-side CRC32-IEEE for the 28 header bytes (shard.go:258-260) */
+static uint32_t host_crc32(const uint8_t *p, size_t n) {
+  static uint32_t tab[256];
+  static bool init = false;
+  if (!init) {
+    for (uint32_t i = 0; i < 256; i++) {
+      uint32_t c = i;
+      for (int j = 0; j < 8; j++) c = (c & 1) ? (c >> 1) ^ 0xEDB88320u : c >> 1;
+      tab[i] = c;
+    }
+    init = true;
+  }
+  uint32_t c = 0xFFFFFFFFu;
+  for (size_t i = 0; i < n; i++) c = tab[(c ^ p[i]) & 0xFF] ^ (c >> 8);
+  return ~c;
+}
+
+static void put_be32(uint8_t *p, uint32_t v) {
+  p[0] = uint8_t(v >> 24); p[1] = uint8_t(v >> 16);
+  p[2] = uint8_t(v >> 8); p[3] = uint8_t(v);
+}
+static void put_be64(uint8_t *p, uint64_t v) {
+  put_be32(p, uint32_t(v >> 32));
+  put_be32(p + 4, uint32_t(v));
+}
+
+int64_t gfrs_shard_disk_size(int64_t size, int64_t block_len) {
+  int64_t body = gfrs_crc32b_encode_size(size, block_len);
+  if (body < 0) return body;
+  return 32 + body + 8; /* header + framed body + footer */
+}
+
+int gfrs_shard_write_batch(gfrs_ctx *ctx, void *dst, size_t dst_stride,
+                           const void *src, size_t src_stride, int64_t size,
+                           int64_t block_len, const uint64_t *bids,
+                           const uint64_t *vuids, int nshards) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
+  if (size <= 0 || nshards <= 0) return GFRS_ERR_INVALID_SHARDS;
+  std::lock_guard<std::mutex> lk(c->mu);
+  StreamGuard g(c);
+  int rc;
+  /* prebuild the 32 B headers on host (shard.go:241-261) */
+  std::vector<uint8_t> hdrs(size_t(nshards) * 32, 0);
+  for (int j = 0; j < nshards; j++) {
+    uint8_t *h = &hdrs[size_t(j) * 32];
+    h[4] = 0xab; h[5] = 0xcd; h[6] = 0xef; h[7] = 0xcc;
+    put_be64(h + 8, bids[j]);
+    put_be64(h + 16, vuids[j]);
+    put_be32(h + 24, uint32_t(size));
+    /* reserved h[28..32) = 0 */
+    put_be32(h, host_crc32(h + 4, 28));
+  }
+  if ((rc = c->stage_pin.ensure(hdrs.size())) != GFRS_OK) return rc;
+  memcpy(c->stage_pin.p, hdrs.data(), hdrs.size());
+  DevBuf &hbuf = c->ptr_buf; /* reuse scratch (stream-ordered) */
+  if ((rc = hbuf.ensure(hdrs.size())) != GFRS_OK) return rc;
+  HIP_TRY(hipMemcpyAsync(hbuf.p, c->stage_pin.p, hdrs.size(),
+                         hipMemcpyHostToDevice, c->stream));
+  /* framed body at +32 */
+  launch_crc_encode((uint8_t *)dst + 32, dst_stride, (const uint8_t *)src,
+                    src_stride, size, block_len, nshards, c->stream);
+  launch_shard_finalize((uint8_t *)dst, dst_stride, (const uint8_t *)hbuf.p,
+                        size, block_len, nshards, c->stream);
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) return hip_fail("shard_write launch", e);
+  return GFRS_OK;
+}
+
+int gfrs_shard_parse_batch(gfrs_ctx *ctx, const void *img, size_t stride,
+                           int64_t size, int64_t block_len,
+                           uint64_t *out_meta, int64_t *bad_block_per_shard,
+                           int nshards) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
+  if (size <= 0 || nshards <= 0) return GFRS_ERR_INVALID_SHARDS;
+  int64_t body = gfrs_crc32b_encode_size(size, block_len);
+  std::lock_guard<std::mutex> lk(c->mu);
+  StreamGuard g(c);
+  int rc;
+  if ((rc = c->fail_buf.ensure(size_t(nshards) * (8 + 32))) != GFRS_OK)
+    return rc;
+  int64_t *bad_dev = (int64_t *)c->fail_buf.p;
+  uint64_t *meta_dev = (uint64_t *)((uint8_t *)c->fail_buf.p + 8 * nshards);
+  std::vector<int64_t> bad(nshards, INT64_MAX);
+  HIP_TRY(hipMemcpyAsync(bad_dev, bad.data(), 8 * size_t(nshards),
+                         hipMemcpyHostToDevice, c->stream));
+  /* body block CRCs */
+  launch_crc_verify((const uint8_t *)img + 32, stride, body, block_len,
+                    nshards, bad_dev, c->stream);
+  /* header/footer */
+  launch_shard_parse((const uint8_t *)img, stride, size, block_len, nshards,
+                     meta_dev, c->stream);
+  HIP_TRY(hipMemcpyAsync(bad.data(), bad_dev, 8 * size_t(nshards),
+                         hipMemcpyDeviceToHost, c->stream));
+  HIP_TRY(hipMemcpyAsync(out_meta, meta_dev, 32 * size_t(nshards),
+                         hipMemcpyDeviceToHost, c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  for (int j = 0; j < nshards; j++)
+    bad_block_per_shard[j] = bad[j] == INT64_MAX ? -1 : bad[j];
+  return GFRS_OK;
+}
+
 /* ---------------- crc32block ---------------- */
 
 int64_t gfrs_crc32b_encode_size(int64_t size, int64_t block_len) {

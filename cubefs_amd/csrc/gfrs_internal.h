@@ -71,6 +71,14 @@ void launch_crc_decode(uint8_t *dst, size_t dst_stride, const uint8_t *framed,
                        int64_t block_len, int nshards, int64_t *bad,
                        hipStream_t s);
 
+/* blobnode on-disk shard codec (core/shard.go, datafile.go:342-445). */
+void launch_shard_finalize(uint8_t *dst, size_t dst_stride,
+                           const uint8_t *headers, int64_t raw_size,
+                           int64_t block_len, int nshards, hipStream_t s);
+void launch_shard_parse(const uint8_t *img, size_t stride, int64_t raw_size,
+                        int64_t block_len, int nshards, uint64_t *out,
+                        hipStream_t s);
+
 /* v_perm semantics probe: returns 1 when `sel byte >= 8 -> 0x00` holds. */
 int probe_perm_device(void);
 

@@ -232,7 +232,7 @@ static int rs_grid(size_t shard_len, size_t nstripes) {
   size_t tiles = ((shard_len + RS_TILE - 1) / RS_TILE) * nstripes;
   if (tiles == 0) tiles = 1;
   /* memory-bound: cap and grid-stride (cdna_hip_programming.md G11) */
-  const size_t cap = size_t(env_grid("GFRS_RS_GRID", 256 * 8));
+  const size_t cap = size_t(env_grid("GFRS_RS_GRID", 256 * 64));
   return int(tiles < cap ? tiles : cap);
 }
 
@@ -427,6 +427,13 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_k(
    * start stays u32-aligned */
   const int64_t chunk =
       (((payload_full + CRC_BLOCKT - 1) / CRC_BLOCKT) + 3) & ~int64_t(3);
+  /* full-frame fold operator depends only on the chunk position */
+  const int64_t cend_full =
+      i64min(i64min(int64_t(threadIdx.x) * chunk, payload_full) + chunk,
+             payload_full);
+  const uint32_t my_op_full = x8n_d(uint64_t(payload_full - cend_full));
+  const uint32_t init_full =
+      gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
 
   for (int64_t fr = blockIdx.x; fr < total_frames; fr += gridDim.x) {
     const int64_t shard = fr / frames_per_shard;
@@ -450,18 +457,21 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_k(
 
     uint32_t part = crc_chunk(payload_src + c0, clen, tab, payload_dst);
     /* fold: contribution = part * x^(8*suffix) */
-    const int64_t suffix = clen > 0 ? payload - (c0 + clen) : 0;
-    part = clen > 0 ? gf2_mulmod_d(x8n_d(uint64_t(suffix)), part) : 0;
-    fold[threadIdx.x] = part;
-    __syncthreads();
-    for (int w = CRC_BLOCKT / 2; w > 0; w >>= 1) {
-      if (threadIdx.x < w) fold[threadIdx.x] ^= fold[threadIdx.x + w];
-      __syncthreads();
+    uint32_t op = my_op_full, init_term = init_full;
+    if (payload != payload_full) { /* tail frame only */
+      const int64_t suffix = clen > 0 ? payload - (c0 + clen) : 0;
+      op = x8n_d(uint64_t(suffix));
+      init_term = gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
     }
+    part = clen > 0 ? gf2_mulmod_d(op, part) : 0;
+#pragma unroll
+    for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
+    if ((threadIdx.x & 63) == 0) fold[threadIdx.x >> 6] = part;
+    __syncthreads();
     if (threadIdx.x == 0) {
       /* crc = ~( x^(8*payload)·(~0) ^ fold )  — see DESIGN.md */
       const uint32_t raw =
-          gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu) ^ fold[0];
+          init_term ^ fold[0] ^ fold[1] ^ fold[2] ^ fold[3];
       const uint32_t crc = ~raw;
       if (MODE == 0) {
         *reinterpret_cast<uint32_t *>(dbase + f * block_len) = crc;
@@ -506,6 +516,15 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
   uint8_t *stage = smem + 4096 + 1024; /* 256 chunks x 272 B */
   for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
     (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
+  /* fold operators are a function of the thread's chunk position only for
+   * full frames — compute x^(8*suffix) once per block, not per frame */
+  const int64_t c0_full =
+      i64min(int64_t(threadIdx.x) * STG_CHUNK, payload_full);
+  const int64_t cend_full =
+      i64min(c0_full + STG_CHUNK, payload_full);
+  const uint32_t my_op_full = x8n_d(uint64_t(payload_full - cend_full));
+  const uint32_t init_full =
+      gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
   __syncthreads();
 
   for (int64_t fr = blockIdx.x; fr < total_frames; fr += gridDim.x) {
@@ -550,17 +569,21 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
     if (clen < 0) clen = 0;
     uint32_t part =
         crc_chunk(stage + threadIdx.x * STG_STRIDE, clen, tab, nullptr);
-    const int64_t suffix = clen > 0 ? payload - (c0 + clen) : 0;
-    part = clen > 0 ? gf2_mulmod_d(x8n_d(uint64_t(suffix)), part) : 0;
-    fold[threadIdx.x] = part;
-    __syncthreads();
-    for (int w = CRC_BLOCKT / 2; w > 0; w >>= 1) {
-      if (threadIdx.x < w) fold[threadIdx.x] ^= fold[threadIdx.x + w];
-      __syncthreads();
+    uint32_t op = my_op_full, init_term = init_full;
+    if (payload != payload_full) { /* tail frame only */
+      const int64_t suffix = clen > 0 ? payload - (c0 + clen) : 0;
+      op = x8n_d(uint64_t(suffix));
+      init_term = gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
     }
+    part = clen > 0 ? gf2_mulmod_d(op, part) : 0;
+    /* wave xor-reduce, then one LDS word per wave */
+#pragma unroll
+    for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
+    if ((threadIdx.x & 63) == 0) fold[threadIdx.x >> 6] = part;
+    __syncthreads();
     if (threadIdx.x == 0) {
       const uint32_t raw =
-          gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu) ^ fold[0];
+          init_term ^ fold[0] ^ fold[1] ^ fold[2] ^ fold[3];
       const uint32_t crc = ~raw;
       if (MODE == 0) {
         *reinterpret_cast<uint32_t *>(dbase + f * block_len) = crc;
@@ -598,6 +621,127 @@ static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
                        dst, dst_stride, src, src_stride, n, block_len, fps,
                        total, bad);
   }
+}
+
+/* ------------------------------------------------------------------ */
+/* blobnode on-disk shard codec (core/shard.go:42-111, datafile.go:342)  */
+/* ------------------------------------------------------------------ */
+
+/* disk image = header(32B) | crc32block body | footer(8B).
+ * header: crc32(BE, bytes 4..32) | magic ab cd ef cc | bid u64 BE |
+ *         vuid u64 BE | size u32 BE | reserved u32 0   (shard.go:42-58)
+ * footer: magic cc ef cd ab | crc32(raw data) u32 BE   (shard.go:66-73)
+ * The footer CRC is the finalized CRC of the UNFRAMED payload
+ * (datafile.go:338,381 io.TeeReader into crc32.NewIEEE).  Instead of a
+ * second pass over the data we GF(2)-combine the per-frame header CRCs the
+ * framing kernel just wrote: crc(A‖B) = x^(8|B|)·crc(A) ^ crc(B) on
+ * finalized values (proven against the oracle in tests). */
+
+/* One wave per shard: write the 32-B header (prebuilt on host), fold the
+ * frame CRCs into the footer. */
+__global__ void shard_finalize_k(uint8_t *__restrict__ dst, size_t dst_stride,
+                                 const uint8_t *__restrict__ headers,
+                                 int64_t raw_size, int64_t block_len,
+                                 int nshards) {
+  const int sh = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+  const int lane = threadIdx.x & 63;
+  if (sh >= nshards) return;
+  uint8_t *img = dst + size_t(sh) * dst_stride;
+  /* header copy, 32 B by the first 8 lanes */
+  if (lane < 8)
+    reinterpret_cast<uint32_t *>(img)[lane] =
+        reinterpret_cast<const uint32_t *>(headers + sh * 32)[lane];
+  if (lane != 0) return;
+  const int64_t payload_full = block_len - CRC_LEN;
+  const int64_t nframes = (raw_size + payload_full - 1) / payload_full;
+  uint32_t crc = 0;
+  int64_t remain = raw_size;
+  for (int64_t f = 0; f < nframes; f++) {
+    uint32_t fc;
+    __builtin_memcpy(&fc, img + 32 + f * block_len, 4); /* LE header */
+    const int64_t plen = remain < payload_full ? remain : payload_full;
+    crc = f == 0 ? fc : gf2_mulmod_d(x8n_d(uint64_t(plen)), crc) ^ fc;
+    remain -= plen;
+  }
+  uint8_t *ftr = img + 32 + (raw_size + CRC_LEN * nframes);
+  ftr[0] = 0xcc; ftr[1] = 0xef; ftr[2] = 0xcd; ftr[3] = 0xab;
+  ftr[4] = uint8_t(crc >> 24); ftr[5] = uint8_t(crc >> 16); /* BE */
+  ftr[6] = uint8_t(crc >> 8); ftr[7] = uint8_t(crc);
+}
+
+/* One wave per shard: parse+verify header and footer; body CRCs are
+ * checked separately by the crc verify kernel.  out layout per shard:
+ * {bid u64, vuid u64, size u64, err i64} (err: 0 ok, negative =
+ * GFRS_ERR_* from gfrs.h). */
+__global__ void shard_parse_k(const uint8_t *__restrict__ img0,
+                              size_t stride, int64_t raw_size,
+                              int64_t block_len, int nshards,
+                              uint64_t *__restrict__ out) {
+  const int sh = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+  const int lane = threadIdx.x & 63;
+  if (sh >= nshards || lane != 0) return;
+  const uint8_t *img = img0 + size_t(sh) * stride;
+  uint64_t *o = out + sh * 4;
+  int64_t err = 0;
+  /* header: magic + crc over bytes 4..32 (shard.go:278-305) */
+  if (!(img[4] == 0xab && img[5] == 0xcd && img[6] == 0xef && img[7] == 0xcc))
+    err = -9;
+  uint32_t hcrc = 0xFFFFFFFFu;
+  for (int i = 4; i < 32; i++) {
+    hcrc ^= img[i];
+    for (int b = 0; b < 8; b++)
+      hcrc = (hcrc & 1) ? (hcrc >> 1) ^ CRC_POLY : hcrc >> 1;
+  }
+  hcrc = ~hcrc;
+  const uint32_t want_h = (uint32_t(img[0]) << 24) | (uint32_t(img[1]) << 16) |
+                          (uint32_t(img[2]) << 8) | uint32_t(img[3]);
+  if (err == 0 && want_h != hcrc) err = -9;
+  uint64_t bid = 0, vuid = 0;
+  for (int i = 0; i < 8; i++) bid = (bid << 8) | img[8 + i];
+  for (int i = 0; i < 8; i++) vuid = (vuid << 8) | img[16 + i];
+  uint32_t size = (uint32_t(img[24]) << 24) | (uint32_t(img[25]) << 16) |
+                  (uint32_t(img[26]) << 8) | uint32_t(img[27]);
+  /* footer: magic + combined body CRC */
+  const int64_t payload_full = block_len - CRC_LEN;
+  const int64_t nframes = (raw_size + payload_full - 1) / payload_full;
+  const uint8_t *ftr = img + 32 + raw_size + CRC_LEN * nframes;
+  if (err == 0 && !(ftr[0] == 0xcc && ftr[1] == 0xef && ftr[2] == 0xcd &&
+                    ftr[3] == 0xab))
+    err = -9;
+  uint32_t crc = 0;
+  int64_t remain = raw_size;
+  for (int64_t f = 0; f < nframes; f++) {
+    uint32_t fc;
+    __builtin_memcpy(&fc, img + 32 + f * block_len, 4);
+    const int64_t plen = remain < payload_full ? remain : payload_full;
+    crc = f == 0 ? fc : gf2_mulmod_d(x8n_d(uint64_t(plen)), crc) ^ fc;
+    remain -= plen;
+  }
+  const uint32_t want_f = (uint32_t(ftr[4]) << 24) | (uint32_t(ftr[5]) << 16) |
+                          (uint32_t(ftr[6]) << 8) | uint32_t(ftr[7]);
+  if (err == 0 && want_f != crc) err = -9;
+  o[0] = bid;
+  o[1] = vuid;
+  o[2] = size;
+  o[3] = uint64_t(err);
+}
+
+void launch_shard_finalize(uint8_t *dst, size_t dst_stride,
+                           const uint8_t *headers, int64_t raw_size,
+                           int64_t block_len, int nshards, hipStream_t s) {
+  const int wps = 4; /* waves per block */
+  const int blocks = (nshards + wps - 1) / wps;
+  hipLaunchKernelGGL(shard_finalize_k, dim3(blocks), dim3(wps * 64), 0, s,
+                     dst, dst_stride, headers, raw_size, block_len, nshards);
+}
+
+void launch_shard_parse(const uint8_t *img, size_t stride, int64_t raw_size,
+                        int64_t block_len, int nshards, uint64_t *out,
+                        hipStream_t s) {
+  const int wps = 4;
+  const int blocks = (nshards + wps - 1) / wps;
+  hipLaunchKernelGGL(shard_parse_k, dim3(blocks), dim3(wps * 64), 0, s, img,
+                     stride, raw_size, block_len, nshards, out);
 }
 
 /* Host-side one-time init of the device CRC tables (g_crc_tab4, g_pow8)

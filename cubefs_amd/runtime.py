@@ -99,6 +99,13 @@ def lib():
         L.gfrs_crc32b_verify_batch.argtypes = [vp, vp, ctypes.c_size_t, i64, i64,
                                                ctypes.c_int, i64p]
         L.gfrs_buffer_sizes.argtypes = [ctypes.POINTER(Tactic), i64, i64p, i64p, i64p]
+        L.gfrs_shard_disk_size.restype = i64
+        L.gfrs_shard_disk_size.argtypes = [i64, i64]
+        L.gfrs_shard_write_batch.argtypes = [vp, vp, ctypes.c_size_t, vp,
+                                             ctypes.c_size_t, i64, i64, u64p,
+                                             u64p, ctypes.c_int]
+        L.gfrs_shard_parse_batch.argtypes = [vp, vp, ctypes.c_size_t, i64, i64,
+                                             u64p, i64p, ctypes.c_int]
         L.gfrs_encode_matrix.argtypes = [vp, ctypes.POINTER(ctypes.c_uint8)]
         _lib = L
     return _lib

@@ -150,6 +150,28 @@ int gfrs_crc32b_verify_batch(gfrs_ctx *ctx, const void *framed,
                              int64_t block_len, int nshards,
                              int64_t *bad_block_per_shard);
 
+/* ---- blobnode on-disk shard image (core/shard.go:42-111,
+ * datafile.go:342-445) ----
+ * image = 32 B header (crc|magic|bid|vuid|size|reserved, big-endian) ‖
+ * crc32block body ‖ 8 B footer (magic | crc32 of the raw data).  Writing
+ * a repaired shard through this produces bytes directly pwrite()-able by
+ * blobnode. */
+int64_t gfrs_shard_disk_size(int64_t size, int64_t block_len);
+/* Frame shard j (raw at src + j*src_stride, size bytes) into a full disk
+ * image at dst + j*dst_stride; bids/vuids are per-shard metadata. */
+int gfrs_shard_write_batch(gfrs_ctx *ctx, void *dst, size_t dst_stride,
+                           const void *src, size_t src_stride, int64_t size,
+                           int64_t block_len, const uint64_t *bids,
+                           const uint64_t *vuids, int nshards);
+/* Parse + fully verify images (header crc+magic, every body block crc,
+ * footer magic + whole-shard crc).  out_meta: per shard
+ * {bid, vuid, size, err} as 4 uint64 (err 0 or GFRS_ERR_MISMATCHED_CRC
+ * cast); bad_block_per_shard as in crc32b_verify_batch. */
+int gfrs_shard_parse_batch(gfrs_ctx *ctx, const void *img, size_t stride,
+                           int64_t size, int64_t block_len,
+                           uint64_t *out_meta, int64_t *bad_block_per_shard,
+                           int nshards);
+
 /* ---- ec.Buffer size math (buf.go:67-133) ---- */
 int gfrs_buffer_sizes(const gfrs_tactic *t, int64_t data_size,
                       int64_t *shard_size, int64_t *ec_data_size,

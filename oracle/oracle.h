@@ -118,6 +118,15 @@ int64_t orc_crc32b_verify(const uint8_t *framed, int64_t framed_len,
 int64_t orc_crc32b_decode(uint8_t *dst, const uint8_t *framed,
                           int64_t framed_len, int64_t block_len);
 
+/* ---- blobnode on-disk shard image (core/shard.go:42-111,
+ * datafile.go:342-407): 32 B header ‖ crc32block body ‖ 8 B footer. ---- */
+int64_t orc_shard_disk_size(int64_t size, int64_t block_len);
+int64_t orc_shard_write(uint8_t *dst, const uint8_t *src, int64_t size,
+                        int64_t block_len, uint64_t bid, uint64_t vuid);
+/* Returns 0 and fills bid/vuid/psize, or ORC_ERR_MISMATCHED_CRC. */
+int orc_shard_parse(const uint8_t *img, int64_t img_len, int64_t block_len,
+                    uint64_t *bid, uint64_t *vuid, uint32_t *psize);
+
 /* ---- multithreaded CPU baseline (bench.py cpu_baseline leg) ----
  * Identical nibble-table algorithm (galois_amd64.go:37-52 semantics),
  * OpenMP across stripes; AVX2 pshufb inner loop when compiled in.

@@ -50,6 +50,15 @@ def lib():
         L.orc_crc32b_verify.argtypes = [u8p, ctypes.c_int64, ctypes.c_int64]
         L.orc_crc32b_decode.restype = ctypes.c_int64
         L.orc_crc32b_decode.argtypes = [u8p, u8p, ctypes.c_int64, ctypes.c_int64]
+        L.orc_shard_disk_size.restype = ctypes.c_int64
+        L.orc_shard_disk_size.argtypes = [ctypes.c_int64, ctypes.c_int64]
+        L.orc_shard_write.restype = ctypes.c_int64
+        L.orc_shard_write.argtypes = [u8p, u8p, ctypes.c_int64, ctypes.c_int64,
+                                      ctypes.c_uint64, ctypes.c_uint64]
+        L.orc_shard_parse.argtypes = [u8p, ctypes.c_int64, ctypes.c_int64,
+                                      ctypes.POINTER(ctypes.c_uint64),
+                                      ctypes.POINTER(ctypes.c_uint64),
+                                      ctypes.POINTER(ctypes.c_uint32)]
         _lib = L
     return _lib
 
@@ -189,6 +198,30 @@ def crc32b_decode(framed, block_len=65536):
     if w < 0:
         raise ValueError("crc mismatch" if w == -9 else "err %d" % w)
     return out
+
+
+def shard_disk_size(size, block_len=65536):
+    return lib().orc_shard_disk_size(size, block_len)
+
+
+def shard_write(src, bid, vuid, block_len=65536):
+    n = src.size
+    out = np.zeros(shard_disk_size(n, block_len), dtype=np.uint8)
+    w = lib().orc_shard_write(_ptr(out), _ptr(src), n, block_len, bid, vuid)
+    assert w == out.size, (w, out.size)
+    return out
+
+
+def shard_parse(img, block_len=65536):
+    bid = ctypes.c_uint64()
+    vuid = ctypes.c_uint64()
+    size = ctypes.c_uint32()
+    rc = lib().orc_shard_parse(_ptr(img), img.size, block_len,
+                               ctypes.byref(bid), ctypes.byref(vuid),
+                               ctypes.byref(size))
+    if rc != 0:
+        raise ValueError("shard parse err %d" % rc)
+    return bid.value, vuid.value, size.value
 
 
 def rs_encode_mt(k, m, stripes, nthreads=0):
