@@ -921,6 +921,42 @@ int gfrs_shard_parse_batch(gfrs_ctx *ctx, const void *img, size_t stride,
   return GFRS_OK;
 }
 
+int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
+                      size_t stripe_stride, int nstripes,
+                      const int32_t *bad_idx, int nbad, void *disk_dst,
+                      size_t dst_stride, int64_t block_len,
+                      const uint64_t *bids, const uint64_t *vuids,
+                      uint64_t *fail_bitmap) {
+  if (nbad <= 0) return GFRS_ERR_INVALID_SHARDS;
+  int rc = gfrs_reconstruct_batch(ctx, base, shard_len, stripe_stride,
+                                  nstripes, bad_idx, nbad, 0);
+  if (rc != GFRS_OK) return rc;
+  rc = gfrs_verify_batch(ctx, base, shard_len, stripe_stride, nstripes,
+                         fail_bitmap);
+  if (rc != GFRS_OK) return rc;
+  /* frame each repaired shard; images laid out (stripe, bad) row-major.
+   * For bad shard b the raw source is strided across stripes.  The
+   * sub-calls take the ctx lock themselves; stream ordering serializes
+   * the kernels. */
+  std::vector<uint64_t> bb(nstripes), vv(nstripes);
+  for (int b = 0; b < nbad; b++) {
+    /* headers differ per stripe: gather this bad-shard column's ids */
+    for (int s2 = 0; s2 < nstripes; s2++) {
+      bb[s2] = bids[size_t(s2) * nbad + b];
+      vv[s2] = vuids[size_t(s2) * nbad + b];
+    }
+    /* write images for column b: dst row stride = nbad*dst_stride */
+    rc = gfrs_shard_write_batch(
+        ctx, (uint8_t *)disk_dst + size_t(b) * dst_stride,
+        size_t(nbad) * dst_stride,
+        (const uint8_t *)base + size_t(bad_idx[b]) * shard_len,
+        stripe_stride, int64_t(shard_len), block_len, bb.data(), vv.data(),
+        nstripes);
+    if (rc != GFRS_OK) return rc;
+  }
+  return GFRS_OK;
+}
+
 /* ---------------- crc32block ---------------- */
 
 int64_t gfrs_crc32b_encode_size(int64_t size, int64_t block_len) {

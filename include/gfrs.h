@@ -172,6 +172,25 @@ int gfrs_shard_parse_batch(gfrs_ctx *ctx, const void *img, size_t stride,
                            uint64_t *out_meta, int64_t *bad_block_per_shard,
                            int nshards);
 
+/* ---- fused repair pipeline (worker_slice_recover.go:804-888 +
+ * datafile.go:342-407) ----
+ * One stream-ordered call per repair tasklet: reconstruct the bad shards
+ * of every stripe, verify all parity (mandatory Verify, :865-874), then
+ * frame each repaired shard into a pwrite()-able disk image.  Intermediate
+ * data never leaves HBM.
+ *   disk_dst: nstripes*nbad images, image (s,b) at
+ *             disk_dst + (s*nbad + b)*dst_stride
+ *   bids/vuids: one per (stripe, bad shard), same order
+ *   fail_bitmap: bit s set when stripe s failed verify (those images are
+ *             still written; the caller drops them like the reference
+ *             drops failed bids). */
+int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
+                      size_t stripe_stride, int nstripes,
+                      const int32_t *bad_idx, int nbad, void *disk_dst,
+                      size_t dst_stride, int64_t block_len,
+                      const uint64_t *bids, const uint64_t *vuids,
+                      uint64_t *fail_bitmap);
+
 /* ---- ec.Buffer size math (buf.go:67-133) ---- */
 int gfrs_buffer_sizes(const gfrs_tactic *t, int64_t data_size,
                       int64_t *shard_size, int64_t *ec_data_size,
