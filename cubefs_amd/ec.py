@@ -212,6 +212,26 @@ class Encoder:
                                            len(bad_idx), int(data_only)),
               "reconstruct_batch")
 
+    def repair_batch(self, batch, bad_idx, disk_dst, bids, vuids,
+                     block_len=65536):
+        """Fused repair tasklet (worker_slice_recover.go:804-888 +
+        datafile.go:342): reconstruct bad shards, verify every stripe,
+        frame each repaired shard into a pwrite-able disk image.
+        disk_dst: [nstripes*len(bad_idx), disk_size] device tensor;
+        bids/vuids: flat per (stripe, bad) row-major.
+        Returns the per-stripe verify-fail list."""
+        p, ln, stride, ns = self._base(batch)
+        nb = len(bad_idx)
+        bad = (ctypes.c_int32 * nb)(*bad_idx)
+        ab = (ctypes.c_uint64 * (ns * nb))(*bids)
+        av = (ctypes.c_uint64 * (ns * nb))(*vuids)
+        nwords = (ns + 63) // 64
+        bm = (ctypes.c_uint64 * nwords)()
+        check(lib().gfrs_repair_batch(self._ctx, p, ln, stride, ns, bad, nb,
+                                      disk_dst.data_ptr(), disk_dst.stride(0),
+                                      block_len, ab, av, bm), "repair_batch")
+        return [bool(bm[s // 64] >> (s % 64) & 1) for s in range(ns)]
+
     def synchronize(self):
         check(lib().gfrs_synchronize(self._ctx), "synchronize")
 

@@ -75,3 +75,38 @@ def test_gpu_shard_write_parse(oracle):
     dst[5, 9] ^= 1
     metas = codec.parse_batch(dst, size)
     assert metas[5]["err"] == -9
+
+
+@pytest.mark.gpu
+def test_gpu_repair_pipeline(oracle):
+    """Fused repair: reconstruct+verify+disk images in one call, images
+    bit-identical to oracle shard_write of the oracle-reconstructed data."""
+    torch = pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from cubefs_amd import codemode, ec, shard
+    t = codemode.get_tactic("EC6P3")
+    enc = ec.Encoder(t)
+    ns, S = 5, 262144
+    rng = np.random.default_rng(13)
+    arr = rng.integers(0, 256, (ns, t.total, S), dtype=np.uint8)
+    batch = torch.from_numpy(arr).to("cuda:0")
+    enc.encode_batch(batch)
+    enc.synchronize()
+    ref = batch.cpu().numpy()
+    bad = [1, 7]
+    batch[:, 1].zero_()
+    batch[:, 7].zero_()
+    dsz = shard.disk_size(S)
+    imgs = torch.zeros((ns * len(bad), dsz), dtype=torch.uint8, device="cuda:0")
+    bids = [1000 * s + b for s in range(ns) for b in bad]
+    vuids = [77] * (ns * len(bad))
+    fails = enc.repair_batch(batch, bad, imgs, bids, vuids)
+    enc.synchronize()
+    assert fails == [False] * ns
+    got = imgs.cpu().numpy()
+    for s in range(ns):
+        for j, b in enumerate(bad):
+            want = oracle.shard_write(ref[s, b].copy(),
+                                      bid=1000 * s + b, vuid=77)
+            assert np.array_equal(got[s * len(bad) + j], want), (s, b)
