@@ -921,6 +921,50 @@ int gfrs_shard_parse_batch(gfrs_ctx *ctx, const void *img, size_t stride,
   return GFRS_OK;
 }
 
+int gfrs_encode_idx(gfrs_ctx *ctx, const void *data_shard, int idx,
+                    void *const *parity, size_t shard_len, int nparity) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  if (idx < 0 || idx >= c->t.n || nparity != c->t.m)
+    return GFRS_ERR_INVALID_SHARDS;
+  std::lock_guard<std::mutex> lk(c->mu);
+  StreamGuard g(c);
+  int rc;
+  /* pointer table: [data, parity...]; plan: k=1 input col idx */
+  std::vector<void *> ptrs(1 + nparity);
+  ptrs[0] = const_cast<void *>(data_shard);
+  for (int r = 0; r < nparity; r++) ptrs[1 + r] = parity[r];
+  if ((rc = upload_ptrs(c, ptrs.data(), int(ptrs.size()))) != GFRS_OK)
+    return rc;
+  /* per-idx plan cached in dec_cache keyed off a synthetic mask */
+  uint64_t key = 0xE0C0DE00ull + uint64_t(idx);
+  DevPlan *p;
+  auto it = c->dec_cache.find(key);
+  if (it != c->dec_cache.end()) {
+    p = it->second;
+  } else {
+    std::vector<int32_t> in{0};
+    std::vector<int32_t> out(nparity);
+    std::vector<uint8_t> rows(nparity);
+    for (int r = 0; r < nparity; r++) {
+      out[r] = 1 + r;
+      rows[r] = c->enc_matrix[size_t(c->t.n + r) * c->t.n + idx];
+    }
+    p = new DevPlan();
+    rc = p->upload(in, out, rows, c->stream);
+    if (rc != GFRS_OK) {
+      delete p;
+      return rc;
+    }
+    c->dec_cache[key] = p;
+  }
+  launch_rs_apply_xor((const uint64_t *)c->ptr_buf.p, 1 + nparity,
+                      (const int32_t *)p->in_idx.p, 1,
+                      (const int32_t *)p->out_idx.p, p->nout,
+                      (const uint8_t *)p->tabs.p, shard_len, 1, c->stream);
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  return GFRS_OK;
+}
+
 int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
                       size_t stripe_stride, int nstripes,
                       const int32_t *bad_idx, int nbad, void *disk_dst,

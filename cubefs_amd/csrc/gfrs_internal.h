@@ -71,6 +71,14 @@ void launch_crc_decode(uint8_t *dst, size_t dst_stride, const uint8_t *framed,
                        int64_t block_len, int nshards, int64_t *bad,
                        hipStream_t s);
 
+/* EncodeIdx-style accumulate apply (reedsolomon.go:631-668):
+ * out[r] ^= coeff[r]*in for one input shard. */
+void launch_rs_apply_xor(const uint64_t *ptrs, int nptr,
+                         const int32_t *in_idx, int k,
+                         const int32_t *out_idx, int nout,
+                         const uint8_t *tabs, size_t shard_len, int nstripes,
+                         hipStream_t s);
+
 /* blobnode on-disk shard codec (core/shard.go, datafile.go:342-445). */
 void launch_shard_finalize(uint8_t *dst, size_t dst_stride,
                            const uint8_t *headers, int64_t raw_size,

@@ -133,7 +133,7 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
     ShardAddr addr, const int32_t *__restrict__ in_idx, int k,
     const int32_t *__restrict__ out_idx, int nout,
     const uint8_t *__restrict__ tabs, size_t shard_len, size_t nstripes,
-    uint32_t *fail) {
+    uint32_t *fail, int xor_acc) {
   extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
   uint4 *ltab = reinterpret_cast<uint4 *>(smem); /* [k*nout*2] */
   const int ncoef = k * nout;
@@ -155,7 +155,14 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
        * groups so there is no runtime bound inside the unrolled loops */
       uint4 acc[GM];
 #pragma unroll
-      for (int r = 0; r < GM; r++) acc[r] = uint4{0, 0, 0, 0};
+      for (int r = 0; r < GM; r++) {
+        if (!VERIFY && xor_acc) {
+          const uint8_t *out = addr.shard(stripe, out_idx[r], shard_len);
+          acc[r] = *reinterpret_cast<const uint4 *>(out + off);
+        } else {
+          acc[r] = uint4{0, 0, 0, 0};
+        }
+      }
       /* issue up to KB=8 input loads before consuming any — 8 KiB of
        * HBM reads in flight per wave instead of one dependent load per
        * coefficient chain (memory-level parallelism, G7) */
@@ -201,7 +208,7 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
         uint8_t *out =
             const_cast<uint8_t *>(addr.shard(stripe, out_idx[og], shard_len));
         for (size_t i = 0; i < nb; i++) {
-          uint8_t v = 0;
+          uint8_t v = (!VERIFY && xor_acc) ? out[off + i] : uint8_t(0);
           for (int c = 0; c < k; c++) {
             const uint8_t b = addr.shard(stripe, in_idx[c], shard_len)[off + i];
             const uint8_t *t = trow + size_t(c) * 32;
@@ -236,17 +243,41 @@ template <bool VERIFY, int GM>
 static void rs_launch_one(const ShardAddr &a, const int32_t *in_idx, int k,
                           const int32_t *out_idx, const uint8_t *tabs,
                           size_t shard_len, int nstripes, uint32_t *fail,
-                          hipStream_t s) {
+                          hipStream_t s, int xor_acc = 0) {
   const int lds = k * GM * 32;
   const int grid = rs_grid(shard_len, nstripes);
-  if (nt_enabled() && !VERIFY)
+  if (nt_enabled() && !VERIFY && !xor_acc)
     hipLaunchKernelGGL((rs_apply_k<false, VERIFY, true, GM>), dim3(grid),
                        dim3(RS_BLOCK), lds, s, a, in_idx, k, out_idx, GM,
-                       tabs, shard_len, size_t(nstripes), fail);
+                       tabs, shard_len, size_t(nstripes), fail, xor_acc);
   else
     hipLaunchKernelGGL((rs_apply_k<false, VERIFY, false, GM>), dim3(grid),
                        dim3(RS_BLOCK), lds, s, a, in_idx, k, out_idx, GM,
-                       tabs, shard_len, size_t(nstripes), fail);
+                       tabs, shard_len, size_t(nstripes), fail, xor_acc);
+}
+
+/* EncodeIdx-style accumulate apply: out[r] ^= coeff[r]*in (one input). */
+void launch_rs_apply_xor(const uint64_t *ptrs, int nptr,
+                         const int32_t *in_idx, int k,
+                         const int32_t *out_idx, int nout,
+                         const uint8_t *tabs, size_t shard_len, int nstripes,
+                         hipStream_t s) {
+  ShardAddr a{ptrs, 0, 0, nptr};
+  for (int og = 0; og < nout; og += MT) {
+    const int gm = nout - og < MT ? nout - og : MT;
+    const int32_t *oi = out_idx + og;
+    const uint8_t *tb = tabs + size_t(og) * k * 32;
+    switch (gm) {
+      case 1: rs_launch_one<false, 1>(a, in_idx, k, oi, tb, shard_len,
+                                      nstripes, nullptr, s, 1); break;
+      case 2: rs_launch_one<false, 2>(a, in_idx, k, oi, tb, shard_len,
+                                      nstripes, nullptr, s, 1); break;
+      case 3: rs_launch_one<false, 3>(a, in_idx, k, oi, tb, shard_len,
+                                      nstripes, nullptr, s, 1); break;
+      default: rs_launch_one<false, 4>(a, in_idx, k, oi, tb, shard_len,
+                                       nstripes, nullptr, s, 1);
+    }
+  }
 }
 
 template <bool VERIFY>

@@ -212,6 +212,16 @@ class Encoder:
                                            len(bad_idx), int(data_only)),
               "reconstruct_batch")
 
+    def encode_idx(self, data_shard, idx, parity):
+        """EncodeIdx (reedsolomon.go:631): parity[r] ^= coeff[r][idx]*data.
+        Parity must start zeroed; call once per data shard."""
+        t = self.tactic
+        views = [_shard_view(s) for s in parity]
+        arr = (ctypes.c_void_p * len(parity))(*[v[0] for v in views])
+        ln = _shard_view(data_shard)[1]
+        check(lib().gfrs_encode_idx(self._ctx, data_shard.data_ptr(), idx,
+                                    arr, ln, len(parity)), "encode_idx")
+
     def repair_batch(self, batch, bad_idx, disk_dst, bids, vuids,
                      block_len=65536):
         """Fused repair tasklet (worker_slice_recover.go:804-888 +

@@ -172,6 +172,13 @@ int gfrs_shard_parse_batch(gfrs_ctx *ctx, const void *img, size_t stride,
                            uint64_t *out_meta, int64_t *bad_block_per_shard,
                            int nshards);
 
+/* ---- incremental parity (reedsolomon.go:631-668 EncodeIdx) ----
+ * Adds data shard idx's contribution into the m parity shards:
+ * parity[r] ^= coeff[r][idx]*data.  Parity must be zeroed before the
+ * first call; each data shard delivered exactly once.  Device pointers. */
+int gfrs_encode_idx(gfrs_ctx *ctx, const void *data_shard, int idx,
+                    void *const *parity, size_t shard_len, int nparity);
+
 /* ---- fused repair pipeline (worker_slice_recover.go:804-888 +
  * datafile.go:342-407) ----
  * One stream-ordered call per repair tasklet: reconstruct the bad shards

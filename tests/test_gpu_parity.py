@@ -254,3 +254,22 @@ def test_split_join(dev):
     out = io.BytesIO()
     enc.join(out, shards, 1000)
     assert out.getvalue() == data.cpu().numpy().tobytes()
+
+
+def test_encode_idx_matches_full_encode(oracle, dev):
+    """EncodeIdx accumulated over all data shards == Encode
+    (reedsolomon.go:627-631 contract)."""
+    from cubefs_amd import codemode, ec
+    t = codemode.get_tactic("EC6P3")
+    enc = ec.Encoder(t)
+    slen = 32768
+    rng = np.random.default_rng(50)
+    shards, _ = make_stripe(rng, t.N, t.M, slen, dev)
+    full = [s.clone() for s in shards]
+    enc.encode(full)
+    parity = [torch.zeros(slen, dtype=torch.uint8, device=dev)
+              for _ in range(t.M)]
+    for i in range(t.N):
+        enc.encode_idx(shards[i], i, parity)
+    for r in range(t.M):
+        assert torch.equal(parity[r], full[t.N + r]), r
