@@ -916,8 +916,14 @@ int gfrs_shard_parse_batch(gfrs_ctx *ctx, const void *img, size_t stride,
   HIP_TRY(hipMemcpyAsync(out_meta, meta_dev, 32 * size_t(nshards),
                          hipMemcpyDeviceToHost, c->stream));
   HIP_TRY(hipStreamSynchronize(c->stream));
-  for (int j = 0; j < nshards; j++)
+  for (int j = 0; j < nshards; j++) {
     bad_block_per_shard[j] = bad[j] == INT64_MAX ? -1 : bad[j];
+    /* body corruption (caught by the block-CRC recompute) must surface in
+     * err too: the frame-header fold in the footer check cannot see
+     * payload flips that keep the stored header */
+    if (bad_block_per_shard[j] >= 0 && int64_t(out_meta[4 * j + 3]) == 0)
+      out_meta[4 * j + 3] = uint64_t(int64_t(GFRS_ERR_MISMATCHED_CRC));
+  }
   return GFRS_OK;
 }
 

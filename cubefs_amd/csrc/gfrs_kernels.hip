@@ -576,8 +576,11 @@ static int crc_grid(int64_t total_frames) {
  * conflict-free: lane t hits dword banks 68t+4i mod 64, distinct within
  * each 16-lane group).  Traffic = S read + S write, the algorithmic
  * minimum. */
-constexpr int STG_CHUNK = 256;
+constexpr int STG_CHUNK = 128;  /* 512 chunks per 64 KiB frame, staged in
+                                   two 32 KiB passes; thread t owns chunks
+                                   t and t+256 */
 constexpr int STG_STRIDE = STG_CHUNK + 16;
+constexpr int STG_HALF = 256 * STG_CHUNK; /* payload bytes per pass */
 
 template <int MODE, bool NT>
 __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
@@ -589,16 +592,18 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
   extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
   uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
   uint32_t *fold = reinterpret_cast<uint32_t *>(smem + 4096);
-  uint8_t *stage = smem + 4096 + 1024; /* 256 chunks x 272 B */
+  uint8_t *stage = smem + 4096 + 1024; /* 256 chunk slots x (128+16) B */
   for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
     (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
-  /* fold operators are a function of the thread's chunk position only for
-   * full frames — compute x^(8*suffix) once per block, not per frame */
-  const int64_t c0_full =
-      i64min(int64_t(threadIdx.x) * STG_CHUNK, payload_full);
-  const int64_t cend_full =
-      i64min(c0_full + STG_CHUNK, payload_full);
-  const uint32_t my_op_full = x8n_d(uint64_t(payload_full - cend_full));
+  /* fold operators are a function of the thread's chunk positions only
+   * for full frames — compute x^(8*suffix) once per block, not per frame */
+  uint32_t op_full[2];
+#pragma unroll
+  for (int h = 0; h < 2; h++) {
+    const int64_t c0 = int64_t(h) * STG_HALF + int64_t(threadIdx.x) * STG_CHUNK;
+    const int64_t cend = i64min(c0 + STG_CHUNK, payload_full);
+    op_full[h] = x8n_d(uint64_t(payload_full - cend));
+  }
   const uint32_t init_full =
       gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
   __syncthreads();
@@ -616,42 +621,54 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
     if (MODE == 0) pdst = dbase + f * block_len + CRC_LEN;
     if (MODE == 2) pdst = dbase + praw0;
 
-    /* stage in (coalesced) + fused copy out (coalesced) */
-    const int64_t words = payload >> 2;
-    for (int64_t w = threadIdx.x; w < words; w += CRC_BLOCKT) {
-      const uint32_t x = *reinterpret_cast<const uint32_t *>(psrc + 4 * w);
-      if (MODE != 1) {
-        if (NT)
-          __builtin_nontemporal_store(x,
-                                      reinterpret_cast<uint32_t *>(pdst + 4 * w));
-        else
-          *reinterpret_cast<uint32_t *>(pdst + 4 * w) = x;
+    /* two half-frame passes: stage 32 KiB (coalesced, fused copy out),
+     * CRC the 128 B chunks out of LDS */
+    uint32_t acc = 0;
+    const uint32_t init_term =
+        payload == payload_full
+            ? init_full
+            : gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
+#pragma unroll
+    for (int h = 0; h < 2; h++) {
+      const int64_t h0 = int64_t(h) * STG_HALF;
+      const int64_t hbytes = i64min(int64_t(STG_HALF), payload - h0);
+      if (hbytes <= 0) break;
+      const int64_t words = hbytes >> 2;
+      for (int64_t w = threadIdx.x; w < words; w += CRC_BLOCKT) {
+        const uint32_t x =
+            *reinterpret_cast<const uint32_t *>(psrc + h0 + 4 * w);
+        if (MODE != 1) {
+          if (NT)
+            __builtin_nontemporal_store(
+                x, reinterpret_cast<uint32_t *>(pdst + h0 + 4 * w));
+          else
+            *reinterpret_cast<uint32_t *>(pdst + h0 + 4 * w) = x;
+        }
+        const int64_t p = 4 * w;
+        *reinterpret_cast<uint32_t *>(
+            &stage[(p >> 7) * STG_STRIDE + (p & (STG_CHUNK - 1))]) = x;
       }
-      const int64_t p = 4 * w;
-      *reinterpret_cast<uint32_t *>(
-          &stage[(p >> 8) * STG_STRIDE + (p & (STG_CHUNK - 1))]) = x;
-    }
-    if (threadIdx.x == 0)
-      for (int64_t p = words * 4; p < payload; p++) {
-        const uint8_t x = psrc[p];
-        if (MODE != 1) pdst[p] = x;
-        stage[(p >> 8) * STG_STRIDE + (p & (STG_CHUNK - 1))] = x;
+      if (threadIdx.x == 0)
+        for (int64_t p = words * 4; p < hbytes; p++) {
+          const uint8_t x = psrc[h0 + p];
+          if (MODE != 1) pdst[h0 + p] = x;
+          stage[(p >> 7) * STG_STRIDE + (p & (STG_CHUNK - 1))] = x;
+        }
+      __syncthreads();
+      const int64_t c0 = int64_t(threadIdx.x) * STG_CHUNK;
+      int clen = int(i64min(int64_t(STG_CHUNK), hbytes - c0));
+      if (clen < 0) clen = 0;
+      uint32_t part = crc_chunk16(stage + threadIdx.x * STG_STRIDE, clen, tab);
+      uint32_t op = op_full[h];
+      if (payload != payload_full) { /* tail frame only */
+        const int64_t suffix =
+            clen > 0 ? payload - (h0 + c0 + clen) : 0;
+        op = x8n_d(uint64_t(suffix));
       }
-    __syncthreads();
-
-    /* per-chunk CRC out of LDS */
-    const int64_t c0 = int64_t(threadIdx.x) * STG_CHUNK;
-    int clen = int(i64min(int64_t(STG_CHUNK), payload - c0));
-    if (clen < 0) clen = 0;
-    uint32_t part =
-        crc_chunk16(stage + threadIdx.x * STG_STRIDE, clen, tab);
-    uint32_t op = my_op_full, init_term = init_full;
-    if (payload != payload_full) { /* tail frame only */
-      const int64_t suffix = clen > 0 ? payload - (c0 + clen) : 0;
-      op = x8n_d(uint64_t(suffix));
-      init_term = gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
+      acc ^= clen > 0 ? gf2_mulmod_d(op, part) : 0;
+      __syncthreads(); /* stage reused by next half */
     }
-    part = clen > 0 ? gf2_mulmod_d(op, part) : 0;
+    uint32_t part = acc;
     /* wave xor-reduce, then one LDS word per wave */
 #pragma unroll
     for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
@@ -675,7 +692,7 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
   }
 }
 
-constexpr int STG_LDS = 4096 + 1024 + 256 * STG_STRIDE;
+constexpr int STG_LDS = 4096 + 1024 + 256 * STG_STRIDE; /* ~42 KB -> 3 blocks/CU */
 
 template <int MODE>
 static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
