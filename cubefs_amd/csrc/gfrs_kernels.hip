@@ -133,7 +133,7 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
     ShardAddr addr, const int32_t *__restrict__ in_idx, int k,
     const int32_t *__restrict__ out_idx, int nout,
     const uint8_t *__restrict__ tabs, size_t shard_len, size_t nstripes,
-    uint32_t *fail, int xor_acc) {
+    uint32_t *fail, int xor_acc, int seq_map) {
   extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
   uint4 *ltab = reinterpret_cast<uint4 *>(smem); /* [k*nout*2] */
   const int ncoef = k * nout;
@@ -143,8 +143,17 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
 
   const size_t tiles_per_shard = (shard_len + RS_TILE - 1) / RS_TILE;
   const size_t total_tiles = tiles_per_shard * nstripes;
+  /* seq_map: block walks consecutive tiles (long sequential bursts per
+   * stream, DRAM row locality) instead of grid-striding */
+  const size_t per_blk =
+      seq_map ? (total_tiles + gridDim.x - 1) / gridDim.x : 0;
+  const size_t t_lo = seq_map ? size_t(blockIdx.x) * per_blk : blockIdx.x;
+  const size_t t_hi =
+      seq_map ? (t_lo + per_blk < total_tiles ? t_lo + per_blk : total_tiles)
+              : total_tiles;
+  const size_t t_step = seq_map ? 1 : gridDim.x;
 
-  for (size_t tile = blockIdx.x; tile < total_tiles; tile += gridDim.x) {
+  for (size_t tile = t_lo; tile < t_hi; tile += t_step) {
     const size_t stripe = tile / tiles_per_shard;
     const size_t col0 = (tile - stripe * tiles_per_shard) * size_t(RS_TILE);
     const size_t off = col0 + size_t(threadIdx.x) * 16;
@@ -246,14 +255,18 @@ static void rs_launch_one(const ShardAddr &a, const int32_t *in_idx, int k,
                           hipStream_t s, int xor_acc = 0) {
   const int lds = k * GM * 32;
   const int grid = rs_grid(shard_len, nstripes);
+  static const int seq = []() {
+    const char *e = getenv("GFRS_RS_SEQ");
+    return e ? atoi(e) : 0;
+  }();
   if (nt_enabled() && !VERIFY && !xor_acc)
     hipLaunchKernelGGL((rs_apply_k<false, VERIFY, true, GM>), dim3(grid),
                        dim3(RS_BLOCK), lds, s, a, in_idx, k, out_idx, GM,
-                       tabs, shard_len, size_t(nstripes), fail, xor_acc);
+                       tabs, shard_len, size_t(nstripes), fail, xor_acc, seq);
   else
     hipLaunchKernelGGL((rs_apply_k<false, VERIFY, false, GM>), dim3(grid),
                        dim3(RS_BLOCK), lds, s, a, in_idx, k, out_idx, GM,
-                       tabs, shard_len, size_t(nstripes), fail, xor_acc);
+                       tabs, shard_len, size_t(nstripes), fail, xor_acc, seq);
 }
 
 /* EncodeIdx-style accumulate apply: out[r] ^= coeff[r]*in (one input). */
