@@ -22,6 +22,22 @@ def decode_size(size, block_len=DEFAULT_BLOCK):
     return check(lib().gfrs_crc32b_decode_size(size, block_len), "decode_size")
 
 
+def sized_encode_size(actual_size, block_len=DEFAULT_BLOCK):
+    """PartialEncodeSize with stableSize=0 (util.go:73-84): returns
+    (total incl. 512-B tail pad, tail)."""
+    size = ctypes.c_int64()
+    tail = ctypes.c_int64()
+    check(lib().gfrs_sized_encode_size(actual_size, block_len,
+                                       ctypes.byref(size), ctypes.byref(tail)),
+          "sized_encode_size")
+    return size.value, tail.value
+
+
+def sized_decode_size(total, tail, block_len=DEFAULT_BLOCK):
+    return check(lib().gfrs_sized_decode_size(total, tail, block_len),
+                 "sized_decode_size")
+
+
 class Codec:
     """Device-side encoder/decoder; one gfrs context (any valid tactic works,
     the CRC path ignores it)."""
@@ -76,6 +92,25 @@ class Codec:
                                              framed.stride(0), fl, block_len,
                                              ns, bad), "crc_verify_batch")
         return list(bad)
+
+    def sized_encode(self, dst, src, block_len=DEFAULT_BLOCK):
+        """rpc2 body framing (sized_coder.go ModeEncode): payload ‖ CRC(BE)
+        frames + 512-B zero tail.  Returns total bytes written."""
+        return check(lib().gfrs_sized_encode(self._ctx, dst.data_ptr(),
+                                             src.data_ptr(), src.numel(),
+                                             block_len), "sized_encode")
+
+    def sized_verify(self, framed, tail, block_len=DEFAULT_BLOCK):
+        bad = ctypes.c_int64(-1)
+        check(lib().gfrs_sized_verify(self._ctx, framed.data_ptr(),
+                                      framed.numel(), tail, block_len,
+                                      ctypes.byref(bad)), "sized_verify")
+        return bad.value
+
+    def sized_decode(self, dst, framed, tail, block_len=DEFAULT_BLOCK):
+        return check(lib().gfrs_sized_decode(self._ctx, dst.data_ptr(),
+                                             framed.data_ptr(), framed.numel(),
+                                             tail, block_len), "sized_decode")
 
     def synchronize(self):
         check(lib().gfrs_synchronize(self._ctx), "synchronize")

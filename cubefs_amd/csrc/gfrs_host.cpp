@@ -816,6 +816,86 @@ int gfrs_reconstruct_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
   return GFRS_OK;
 }
 
+/* ---------------- sized coder (rpc2 body framing) ---------------- */
+
+int gfrs_sized_encode_size(int64_t actual_size, int64_t block_len,
+                           int64_t *size, int64_t *tail) {
+  /* PartialEncodeSizeWith with stableSize == 0 (util.go:73-80),
+   * _alignment = 512 (rpc2/transport/allocator.go:12-15) */
+  int64_t enc = gfrs_crc32b_encode_size(actual_size, block_len);
+  if (enc < 0) return int(enc);
+  int64_t t = (512 - (enc & 511)) & 511;
+  *size = enc + t;
+  *tail = t;
+  return GFRS_OK;
+}
+
+int64_t gfrs_sized_decode_size(int64_t total, int64_t tail,
+                               int64_t block_len) {
+  /* PartialDecodeSizeWith, stableSize == 0 (util.go:86-94) */
+  return gfrs_crc32b_decode_size(total - tail, block_len);
+}
+
+int64_t gfrs_sized_encode(gfrs_ctx *ctx, void *dst, const void *src,
+                          int64_t n, int64_t block_len) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
+  if (n <= 0) return GFRS_ERR_INVALID_SHARDS;
+  int64_t size = 0, tail = 0;
+  int rc = gfrs_sized_encode_size(n, block_len, &size, &tail);
+  if (rc != GFRS_OK) return rc;
+  std::lock_guard<std::mutex> lk(c->mu);
+  StreamGuard g(c);
+  launch_sized_encode((uint8_t *)dst, 0, (const uint8_t *)src, 0, n,
+                      block_len, 1, c->stream);
+  if (tail)
+    HIP_TRY(hipMemsetAsync((uint8_t *)dst + size - tail, 0, size_t(tail),
+                           c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  return size;
+}
+
+int gfrs_sized_verify(gfrs_ctx *ctx, const void *framed, int64_t total,
+                      int64_t tail, int64_t block_len, int64_t *bad_block) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
+  std::lock_guard<std::mutex> lk(c->mu);
+  StreamGuard g(c);
+  int rc;
+  if ((rc = c->fail_buf.ensure(8)) != GFRS_OK) return rc;
+  int64_t bad = INT64_MAX;
+  HIP_TRY(hipMemcpyAsync(c->fail_buf.p, &bad, 8, hipMemcpyHostToDevice,
+                         c->stream));
+  launch_sized_verify((const uint8_t *)framed, 0, total - tail, block_len, 1,
+                      (int64_t *)c->fail_buf.p, c->stream);
+  HIP_TRY(hipMemcpyAsync(&bad, c->fail_buf.p, 8, hipMemcpyDeviceToHost,
+                         c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  *bad_block = bad == INT64_MAX ? -1 : bad;
+  return GFRS_OK;
+}
+
+int64_t gfrs_sized_decode(gfrs_ctx *ctx, void *dst, const void *framed,
+                          int64_t total, int64_t tail, int64_t block_len) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
+  std::lock_guard<std::mutex> lk(c->mu);
+  StreamGuard g(c);
+  int rc;
+  if ((rc = c->fail_buf.ensure(8)) != GFRS_OK) return rc;
+  int64_t bad = INT64_MAX;
+  HIP_TRY(hipMemcpyAsync(c->fail_buf.p, &bad, 8, hipMemcpyHostToDevice,
+                         c->stream));
+  launch_sized_decode((uint8_t *)dst, 0, (const uint8_t *)framed, 0,
+                      total - tail, block_len, 1, (int64_t *)c->fail_buf.p,
+                      c->stream);
+  HIP_TRY(hipMemcpyAsync(&bad, c->fail_buf.p, 8, hipMemcpyDeviceToHost,
+                         c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  if (bad != INT64_MAX) return GFRS_ERR_MISMATCHED_CRC;
+  return gfrs_sized_decode_size(total, tail, block_len);
+}
+
 /* ---------------- blobnode shard images ---------------- */
 
 /* host-side CRC32-IEEE for the 28 header bytes (shard.go:258-260) */

@@ -71,6 +71,19 @@ void launch_crc_decode(uint8_t *dst, size_t dst_stride, const uint8_t *framed,
                        int64_t block_len, int nshards, int64_t *bad,
                        hipStream_t s);
 
+/* sized coder (crc32block/sized_coder.go): payload ‖ CRC32(BE) frames,
+ * 512-B tail alignment handled by the host. */
+void launch_sized_encode(uint8_t *dst, size_t dst_stride, const uint8_t *src,
+                         size_t src_stride, int64_t n, int64_t block_len,
+                         int nshards, hipStream_t s);
+void launch_sized_verify(const uint8_t *framed, size_t stride,
+                         int64_t body_len, int64_t block_len, int nshards,
+                         int64_t *bad, hipStream_t s);
+void launch_sized_decode(uint8_t *dst, size_t dst_stride,
+                         const uint8_t *framed, size_t src_stride,
+                         int64_t body_len, int64_t block_len, int nshards,
+                         int64_t *bad, hipStream_t s);
+
 /* EncodeIdx-style accumulate apply (reedsolomon.go:631-668):
  * out[r] ^= coeff[r]*in for one input shard. */
 void launch_rs_apply_xor(const uint64_t *ptrs, int nptr,

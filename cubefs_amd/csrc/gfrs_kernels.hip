@@ -499,7 +499,7 @@ GFRS_DEV uint32_t crc_chunk16(const uint8_t *p, int len,
 
 /* One workgroup per frame.  MODE: 0 = encode (raw src -> framed dst),
  * 1 = verify (framed src), 2 = decode (framed src -> raw dst). */
-template <int MODE>
+template <int MODE, bool TAILCRC = false>
 __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_k(
     uint8_t *__restrict__ dst, size_t dst_stride,
     const uint8_t *__restrict__ src, size_t src_stride, int64_t n,
@@ -538,10 +538,11 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_k(
     const int64_t c0 = int64_t(threadIdx.x) * chunk;
     int clen = int(i64min(chunk, payload - c0));
     if (clen < 0) clen = 0;
+    const int64_t hdr_off = TAILCRC ? 0 : CRC_LEN;
     const uint8_t *payload_src =
-        (MODE == 0) ? sbase + praw0 : sbase + f * block_len + CRC_LEN;
+        (MODE == 0) ? sbase + praw0 : sbase + f * block_len + hdr_off;
     uint8_t *payload_dst = nullptr;
-    if (MODE == 0) payload_dst = dbase + f * block_len + CRC_LEN + c0;
+    if (MODE == 0) payload_dst = dbase + f * block_len + hdr_off + c0;
     if (MODE == 2) payload_dst = dbase + praw0 + c0;
 
     uint32_t part = crc_chunk(payload_src + c0, clen, tab, payload_dst);
@@ -562,11 +563,23 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_k(
       const uint32_t raw =
           init_term ^ fold[0] ^ fold[1] ^ fold[2] ^ fold[3];
       const uint32_t crc = ~raw;
+      uint8_t *hw = dbase ? dbase + f * block_len + (TAILCRC ? payload : 0)
+                          : nullptr;
+      const uint8_t *hr = sbase + f * block_len + (TAILCRC ? payload : 0);
       if (MODE == 0) {
-        *reinterpret_cast<uint32_t *>(dbase + f * block_len) = crc;
+        if (TAILCRC) { /* BE (sized_coder.go) */
+          hw[0] = uint8_t(crc >> 24); hw[1] = uint8_t(crc >> 16);
+          hw[2] = uint8_t(crc >> 8); hw[3] = uint8_t(crc);
+        } else {
+          *reinterpret_cast<uint32_t *>(hw) = crc;
+        }
       } else {
         uint32_t want;
-        __builtin_memcpy(&want, sbase + f * block_len, 4); /* LE load */
+        if (TAILCRC)
+          want = (uint32_t(hr[0]) << 24) | (uint32_t(hr[1]) << 16) |
+                 (uint32_t(hr[2]) << 8) | uint32_t(hr[3]);
+        else
+          __builtin_memcpy(&want, hr, 4); /* LE load */
         if (want != crc)
           atomicMin(reinterpret_cast<unsigned long long *>(&bad[shard]),
                     static_cast<unsigned long long>(f));
@@ -595,7 +608,7 @@ constexpr int STG_CHUNK = 128;  /* 512 chunks per 64 KiB frame, staged in
 constexpr int STG_STRIDE = STG_CHUNK + 16;
 constexpr int STG_HALF = 256 * STG_CHUNK; /* payload bytes per pass */
 
-template <int MODE, bool NT>
+template <int MODE, bool NT, bool TAILCRC>
 __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
     uint8_t *__restrict__ dst, size_t dst_stride,
     const uint8_t *__restrict__ src, size_t src_stride, int64_t n,
@@ -628,10 +641,13 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
     const int64_t payload = i64min(payload_full, n - praw0);
     const uint8_t *sbase = src + shard * src_stride;
     uint8_t *dbase = dst ? dst + shard * dst_stride : nullptr;
+    /* sized frames (sized_coder.go:256-279) put the CRC AFTER the
+     * payload, big-endian; block frames put it first, little-endian */
+    const int64_t hdr_off = TAILCRC ? 0 : CRC_LEN;
     const uint8_t *psrc =
-        (MODE == 0) ? sbase + praw0 : sbase + f * block_len + CRC_LEN;
+        (MODE == 0) ? sbase + praw0 : sbase + f * block_len + hdr_off;
     uint8_t *pdst = nullptr;
-    if (MODE == 0) pdst = dbase + f * block_len + CRC_LEN;
+    if (MODE == 0) pdst = dbase + f * block_len + hdr_off;
     if (MODE == 2) pdst = dbase + praw0;
 
     /* two half-frame passes: stage 32 KiB (coalesced, fused copy out),
@@ -691,11 +707,26 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
       const uint32_t raw =
           init_term ^ fold[0] ^ fold[1] ^ fold[2] ^ fold[3];
       const uint32_t crc = ~raw;
+      uint8_t *hpos_w = dbase ? dbase + f * block_len +
+                                    (TAILCRC ? payload : 0)
+                              : nullptr;
+      const uint8_t *hpos_r =
+          sbase + f * block_len + (TAILCRC ? payload : 0);
       if (MODE == 0) {
-        *reinterpret_cast<uint32_t *>(dbase + f * block_len) = crc;
+        if (TAILCRC) { /* big-endian (sized_coder.go be.Uint32) */
+          hpos_w[0] = uint8_t(crc >> 24); hpos_w[1] = uint8_t(crc >> 16);
+          hpos_w[2] = uint8_t(crc >> 8); hpos_w[3] = uint8_t(crc);
+        } else {
+          *reinterpret_cast<uint32_t *>(hpos_w) = crc;
+        }
       } else {
         uint32_t want;
-        __builtin_memcpy(&want, sbase + f * block_len, 4);
+        if (TAILCRC) {
+          want = (uint32_t(hpos_r[0]) << 24) | (uint32_t(hpos_r[1]) << 16) |
+                 (uint32_t(hpos_r[2]) << 8) | uint32_t(hpos_r[3]);
+        } else {
+          __builtin_memcpy(&want, hpos_r, 4);
+        }
         if (want != crc)
           atomicMin(reinterpret_cast<unsigned long long *>(&bad[shard]),
                     static_cast<unsigned long long>(f));
@@ -707,7 +738,7 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
 
 constexpr int STG_LDS = 4096 + 1024 + 256 * STG_STRIDE; /* ~42 KB -> 3 blocks/CU */
 
-template <int MODE>
+template <int MODE, bool TAILCRC = false>
 static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
                          size_t src_stride, int64_t n, int64_t block_len,
                          int64_t fps, int64_t total, int64_t *bad,
@@ -715,18 +746,46 @@ static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
   const int grid = crc_grid(total);
   if (block_len == 65536) {
     if (nt_enabled() && MODE != 1)
-      hipLaunchKernelGGL((crc32b_staged_k<MODE, true>), dim3(grid),
+      hipLaunchKernelGGL((crc32b_staged_k<MODE, true, TAILCRC>), dim3(grid),
                          dim3(CRC_BLOCKT), STG_LDS, s, dst, dst_stride, src,
                          src_stride, n, fps, total, bad);
     else
-      hipLaunchKernelGGL((crc32b_staged_k<MODE, false>), dim3(grid),
+      hipLaunchKernelGGL((crc32b_staged_k<MODE, false, TAILCRC>), dim3(grid),
                          dim3(CRC_BLOCKT), STG_LDS, s, dst, dst_stride, src,
                          src_stride, n, fps, total, bad);
   } else {
-    hipLaunchKernelGGL((crc32b_k<MODE>), dim3(grid), dim3(CRC_BLOCKT), 0, s,
-                       dst, dst_stride, src, src_stride, n, block_len, fps,
-                       total, bad);
+    hipLaunchKernelGGL((crc32b_k<MODE, TAILCRC>), dim3(grid),
+                       dim3(CRC_BLOCKT), 0, s, dst, dst_stride, src,
+                       src_stride, n, block_len, fps, total, bad);
   }
+}
+
+void launch_sized_encode(uint8_t *dst, size_t dst_stride, const uint8_t *src,
+                         size_t src_stride, int64_t n, int64_t block_len,
+                         int nshards, hipStream_t s) {
+  const int64_t payload = block_len - CRC_LEN;
+  const int64_t fps = (n + payload - 1) / payload;
+  crc_dispatch<0, true>(dst, dst_stride, src, src_stride, n, block_len, fps,
+                        fps * nshards, nullptr, s);
+}
+
+void launch_sized_verify(const uint8_t *framed, size_t stride,
+                         int64_t body_len, int64_t block_len, int nshards,
+                         int64_t *bad, hipStream_t s) {
+  const int64_t fps = (body_len + block_len - 1) / block_len;
+  const int64_t n = body_len - CRC_LEN * fps;
+  crc_dispatch<1, true>(nullptr, 0, framed, stride, n, block_len, fps,
+                        fps * nshards, bad, s);
+}
+
+void launch_sized_decode(uint8_t *dst, size_t dst_stride,
+                         const uint8_t *framed, size_t src_stride,
+                         int64_t body_len, int64_t block_len, int nshards,
+                         int64_t *bad, hipStream_t s) {
+  const int64_t fps = (body_len + block_len - 1) / block_len;
+  const int64_t n = body_len - CRC_LEN * fps;
+  crc_dispatch<2, true>(dst, dst_stride, framed, src_stride, n, block_len,
+                        fps, fps * nshards, bad, s);
 }
 
 /* ------------------------------------------------------------------ */

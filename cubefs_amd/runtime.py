@@ -99,6 +99,14 @@ def lib():
         L.gfrs_crc32b_verify_batch.argtypes = [vp, vp, ctypes.c_size_t, i64, i64,
                                                ctypes.c_int, i64p]
         L.gfrs_buffer_sizes.argtypes = [ctypes.POINTER(Tactic), i64, i64p, i64p, i64p]
+        L.gfrs_sized_encode_size.argtypes = [i64, i64, i64p, i64p]
+        L.gfrs_sized_decode_size.restype = i64
+        L.gfrs_sized_decode_size.argtypes = [i64, i64, i64]
+        L.gfrs_sized_encode.restype = i64
+        L.gfrs_sized_encode.argtypes = [vp, vp, vp, i64, i64]
+        L.gfrs_sized_verify.argtypes = [vp, vp, i64, i64, i64, i64p]
+        L.gfrs_sized_decode.restype = i64
+        L.gfrs_sized_decode.argtypes = [vp, vp, vp, i64, i64, i64]
         L.gfrs_shard_disk_size.restype = i64
         L.gfrs_shard_disk_size.argtypes = [i64, i64]
         L.gfrs_shard_write_batch.argtypes = [vp, vp, ctypes.c_size_t, vp,

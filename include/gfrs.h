@@ -150,6 +150,24 @@ int gfrs_crc32b_verify_batch(gfrs_ctx *ctx, const void *framed,
                              int64_t block_len, int nshards,
                              int64_t *bad_block_per_shard);
 
+/* ---- sized coder (crc32block/sized_coder.go, the rpc2 body framing) ----
+ * Frame = payload (block_len-4) ‖ CRC32-IEEE big-endian (ModeEncode,
+ * sized_coder.go:256-279); the encoded stream is zero-padded to the
+ * transport alignment of 512 (PartialEncodeSizeWith, util.go:73-80).
+ * *size returns total bytes incl. tail pad; *tail the pad length. */
+int gfrs_sized_encode_size(int64_t actual_size, int64_t block_len,
+                           int64_t *size, int64_t *tail);
+int64_t gfrs_sized_decode_size(int64_t total, int64_t tail, int64_t block_len);
+/* Encode n raw bytes into dst (device); writes the tail pad.  Returns
+ * total bytes written (incl. pad) or negative error. */
+int64_t gfrs_sized_encode(gfrs_ctx *ctx, void *dst, const void *src,
+                          int64_t n, int64_t block_len);
+/* Verify all frames of a sized body (total includes tail pad). */
+int gfrs_sized_verify(gfrs_ctx *ctx, const void *framed, int64_t total,
+                      int64_t tail, int64_t block_len, int64_t *bad_block);
+int64_t gfrs_sized_decode(gfrs_ctx *ctx, void *dst, const void *framed,
+                          int64_t total, int64_t tail, int64_t block_len);
+
 /* ---- blobnode on-disk shard image (core/shard.go:42-111,
  * datafile.go:342-445) ----
  * image = 32 B header (crc|magic|bid|vuid|size|reserved, big-endian) ‖

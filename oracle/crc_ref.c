@@ -145,3 +145,86 @@ int64_t orc_crc32b_decode(uint8_t *dst, const uint8_t *framed,
     }
     return w;
 }
+
+/* ---------------- sized coder (sized_coder.go, util.go:73-94) ----------
+ * ModeEncode with stableSize==0: frame = payload (<= block_len-4) ‖
+ * 4-byte BIG-endian CRC32-IEEE of the payload (sized_coder.go:256-279,
+ * be.Uint32 check :323-326); the whole encoded stream is zero-padded to
+ * the 512-byte transport alignment (PartialEncodeSizeWith, util.go:73-80,
+ * rpc2/transport/allocator.go:12-15). */
+
+#define SIZED_ALIGN 512
+
+int64_t orc_partial_encode_size(int64_t actual, int64_t stable,
+                                int64_t block_len, int64_t *tail) {
+    if (!valid_block_len(block_len)) return ORC_ERR_INVALID_BLOCK;
+    int64_t payload = block_len - CRC_LEN;
+    int64_t part = (stable % payload) & ~(int64_t)(SIZED_ALIGN - 1);
+    int64_t pad = (stable % payload) % SIZED_ALIGN;
+    int64_t size = orc_crc32b_encode_size(actual + part + pad, block_len) - part;
+    int64_t t = (SIZED_ALIGN - (size & (SIZED_ALIGN - 1))) % SIZED_ALIGN;
+    if (tail) *tail = t;
+    return size + t;
+}
+
+int64_t orc_partial_decode_size(int64_t total, int64_t tail, int64_t stable,
+                                int64_t block_len) {
+    if (!valid_block_len(block_len)) return ORC_ERR_INVALID_BLOCK;
+    int64_t payload = block_len - CRC_LEN;
+    int64_t part = (stable % payload) & ~(int64_t)(SIZED_ALIGN - 1);
+    int64_t pad = (stable % payload) % SIZED_ALIGN;
+    return orc_crc32b_decode_size(total - tail + part, block_len) - part - pad;
+}
+
+int64_t orc_sized_encode(uint8_t *dst, const uint8_t *src, int64_t n,
+                         int64_t block_len) {
+    if (!valid_block_len(block_len)) return ORC_ERR_INVALID_BLOCK;
+    int64_t payload = block_len - CRC_LEN;
+    int64_t w = 0;
+    for (int64_t off = 0; off < n; off += payload) {
+        int64_t take = n - off < payload ? n - off : payload;
+        memcpy(dst + w, src + off, (size_t)take);
+        uint32_t crc = orc_crc32(0, src + off, (size_t)take);
+        dst[w + take + 0] = (uint8_t)(crc >> 24); /* big-endian */
+        dst[w + take + 1] = (uint8_t)(crc >> 16);
+        dst[w + take + 2] = (uint8_t)(crc >> 8);
+        dst[w + take + 3] = (uint8_t)crc;
+        w += take + CRC_LEN;
+    }
+    int64_t t = (SIZED_ALIGN - (w & (SIZED_ALIGN - 1))) % SIZED_ALIGN;
+    memset(dst + w, 0, (size_t)t);
+    return w + t;
+}
+
+int64_t orc_sized_verify(const uint8_t *framed, int64_t total, int64_t tail,
+                         int64_t block_len) {
+    if (!valid_block_len(block_len)) return ORC_ERR_INVALID_BLOCK;
+    int64_t body = total - tail;
+    int64_t idx = 0;
+    for (int64_t off = 0; off < body; off += block_len, idx++) {
+        int64_t blk = body - off < block_len ? body - off : block_len;
+        if (blk <= CRC_LEN) return idx;
+        int64_t plen = blk - CRC_LEN;
+        uint32_t want = ((uint32_t)framed[off + plen] << 24) |
+                        ((uint32_t)framed[off + plen + 1] << 16) |
+                        ((uint32_t)framed[off + plen + 2] << 8) |
+                        (uint32_t)framed[off + plen + 3];
+        if (want != orc_crc32(0, framed + off, (size_t)plen)) return idx;
+    }
+    return -1;
+}
+
+int64_t orc_sized_decode(uint8_t *dst, const uint8_t *framed, int64_t total,
+                         int64_t tail, int64_t block_len) {
+    int64_t bad = orc_sized_verify(framed, total, tail, block_len);
+    if (bad == ORC_ERR_INVALID_BLOCK) return bad;
+    if (bad >= 0) return ORC_ERR_MISMATCHED_CRC;
+    int64_t body = total - tail;
+    int64_t w = 0;
+    for (int64_t off = 0; off < body; off += block_len) {
+        int64_t blk = body - off < block_len ? body - off : block_len;
+        memcpy(dst + w, framed + off, (size_t)(blk - CRC_LEN));
+        w += blk - CRC_LEN;
+    }
+    return w;
+}

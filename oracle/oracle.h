@@ -118,6 +118,20 @@ int64_t orc_crc32b_verify(const uint8_t *framed, int64_t framed_len,
 int64_t orc_crc32b_decode(uint8_t *dst, const uint8_t *framed,
                           int64_t framed_len, int64_t block_len);
 
+/* ---- sized coder (crc32block/sized_coder.go: payload ‖ CRC32 BE per
+ * block; stream zero-padded to the 512-B transport alignment;
+ * util.go:73-94 Partial sizes) ---- */
+int64_t orc_partial_encode_size(int64_t actual, int64_t stable,
+                                int64_t block_len, int64_t *tail);
+int64_t orc_partial_decode_size(int64_t total, int64_t tail, int64_t stable,
+                                int64_t block_len);
+int64_t orc_sized_encode(uint8_t *dst, const uint8_t *src, int64_t n,
+                         int64_t block_len);
+int64_t orc_sized_verify(const uint8_t *framed, int64_t total, int64_t tail,
+                         int64_t block_len);
+int64_t orc_sized_decode(uint8_t *dst, const uint8_t *framed, int64_t total,
+                         int64_t tail, int64_t block_len);
+
 /* ---- blobnode on-disk shard image (core/shard.go:42-111,
  * datafile.go:342-407): 32 B header ‖ crc32block body ‖ 8 B footer. ---- */
 int64_t orc_shard_disk_size(int64_t size, int64_t block_len);

@@ -50,6 +50,18 @@ def lib():
         L.orc_crc32b_verify.argtypes = [u8p, ctypes.c_int64, ctypes.c_int64]
         L.orc_crc32b_decode.restype = ctypes.c_int64
         L.orc_crc32b_decode.argtypes = [u8p, u8p, ctypes.c_int64, ctypes.c_int64]
+        L.orc_partial_encode_size.restype = ctypes.c_int64
+        L.orc_partial_encode_size.argtypes = [ctypes.c_int64, ctypes.c_int64,
+                                              ctypes.c_int64,
+                                              ctypes.POINTER(ctypes.c_int64)]
+        L.orc_partial_decode_size.restype = ctypes.c_int64
+        L.orc_partial_decode_size.argtypes = [ctypes.c_int64] * 4
+        L.orc_sized_encode.restype = ctypes.c_int64
+        L.orc_sized_encode.argtypes = [u8p, u8p, ctypes.c_int64, ctypes.c_int64]
+        L.orc_sized_verify.restype = ctypes.c_int64
+        L.orc_sized_verify.argtypes = [u8p, ctypes.c_int64, ctypes.c_int64, ctypes.c_int64]
+        L.orc_sized_decode.restype = ctypes.c_int64
+        L.orc_sized_decode.argtypes = [u8p, u8p, ctypes.c_int64, ctypes.c_int64, ctypes.c_int64]
         L.orc_shard_disk_size.restype = ctypes.c_int64
         L.orc_shard_disk_size.argtypes = [ctypes.c_int64, ctypes.c_int64]
         L.orc_shard_write.restype = ctypes.c_int64
@@ -197,6 +209,37 @@ def crc32b_decode(framed, block_len=65536):
     w = lib().orc_crc32b_decode(_ptr(out), _ptr(framed), framed.size, block_len)
     if w < 0:
         raise ValueError("crc mismatch" if w == -9 else "err %d" % w)
+    return out
+
+
+def partial_encode_size(actual, stable=0, block_len=65536):
+    t = ctypes.c_int64()
+    total = lib().orc_partial_encode_size(actual, stable, block_len, ctypes.byref(t))
+    return total, t.value
+
+
+def partial_decode_size(total, tail, stable=0, block_len=65536):
+    return lib().orc_partial_decode_size(total, tail, stable, block_len)
+
+
+def sized_encode(src, block_len=65536):
+    total, tail = partial_encode_size(src.size, 0, block_len)
+    out = np.zeros(total, dtype=np.uint8)
+    w = lib().orc_sized_encode(_ptr(out), _ptr(src), src.size, block_len)
+    assert w == total, (w, total)
+    return out, tail
+
+
+def sized_verify(framed, tail, block_len=65536):
+    return lib().orc_sized_verify(_ptr(framed), framed.size, tail, block_len)
+
+
+def sized_decode(framed, tail, block_len=65536):
+    n = partial_decode_size(framed.size, tail, 0, block_len)
+    out = np.zeros(n, dtype=np.uint8)
+    w = lib().orc_sized_decode(_ptr(out), _ptr(framed), framed.size, tail, block_len)
+    if w < 0:
+        raise ValueError("err %d" % w)
     return out
 
 
