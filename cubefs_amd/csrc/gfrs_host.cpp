@@ -1051,6 +1051,14 @@ int gfrs_encode_idx(gfrs_ctx *ctx, const void *data_shard, int idx,
   return GFRS_OK;
 }
 
+int gfrs_update_idx(gfrs_ctx *ctx, const void *old_shard,
+                    const void *new_shard, int idx, void *const *parity,
+                    size_t shard_len, int nparity) {
+  int rc = gfrs_encode_idx(ctx, old_shard, idx, parity, shard_len, nparity);
+  if (rc != GFRS_OK) return rc;
+  return gfrs_encode_idx(ctx, new_shard, idx, parity, shard_len, nparity);
+}
+
 int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
                       size_t stripe_stride, int nstripes,
                       const int32_t *bad_idx, int nbad, void *disk_dst,

@@ -222,6 +222,16 @@ class Encoder:
         check(lib().gfrs_encode_idx(self._ctx, data_shard.data_ptr(), idx,
                                     arr, ln, len(parity)), "encode_idx")
 
+    def update_idx(self, old_shard, new_shard, idx, parity):
+        """Update (reedsolomon.go:676): patch parity for a replaced data
+        shard: parity[r] ^= coeff[r][idx]*(old^new)."""
+        views = [_shard_view(s) for s in parity]
+        arr = (ctypes.c_void_p * len(parity))(*[v[0] for v in views])
+        ln = _shard_view(old_shard)[1]
+        check(lib().gfrs_update_idx(self._ctx, old_shard.data_ptr(),
+                                    new_shard.data_ptr(), idx, arr, ln,
+                                    len(parity)), "update_idx")
+
     def repair_batch(self, batch, bad_idx, disk_dst, bids, vuids,
                      block_len=65536):
         """Fused repair tasklet (worker_slice_recover.go:804-888 +

@@ -116,6 +116,8 @@ def lib():
                                              u64p, i64p, ctypes.c_int]
         L.gfrs_encode_idx.argtypes = [vp, vp, ctypes.c_int, vpp,
                                       ctypes.c_size_t, ctypes.c_int]
+        L.gfrs_update_idx.argtypes = [vp, vp, vp, ctypes.c_int, vpp,
+                                      ctypes.c_size_t, ctypes.c_int]
         L.gfrs_repair_batch.argtypes = [vp, vp, ctypes.c_size_t, ctypes.c_size_t,
                                         ctypes.c_int, i32p, ctypes.c_int, vp,
                                         ctypes.c_size_t, i64, u64p, u64p, u64p]

@@ -197,6 +197,14 @@ int gfrs_shard_parse_batch(gfrs_ctx *ctx, const void *img, size_t stride,
 int gfrs_encode_idx(gfrs_ctx *ctx, const void *data_shard, int idx,
                     void *const *parity, size_t shard_len, int nparity);
 
+/* Update (reedsolomon.go:676-766): replace data shard idx's content and
+ * patch the parity in place: parity[r] ^= coeff[r][idx]*(old ^ new) —
+ * applied as two accumulate passes by GF(2^8) linearity.  old_shard and
+ * new_shard are device pointers; the caller swaps its own data buffer. */
+int gfrs_update_idx(gfrs_ctx *ctx, const void *old_shard,
+                    const void *new_shard, int idx, void *const *parity,
+                    size_t shard_len, int nparity);
+
 /* ---- fused repair pipeline (worker_slice_recover.go:804-888 +
  * datafile.go:342-407) ----
  * One stream-ordered call per repair tasklet: reconstruct the bad shards

@@ -273,3 +273,20 @@ def test_encode_idx_matches_full_encode(oracle, dev):
         enc.encode_idx(shards[i], i, parity)
     for r in range(t.M):
         assert torch.equal(parity[r], full[t.N + r]), r
+
+
+def test_update_idx(oracle, dev):
+    """Update semantics: replacing a data shard and patching parity equals
+    a fresh encode of the new data (reedsolomon.go:676 contract)."""
+    from cubefs_amd import codemode, ec
+    t = codemode.get_tactic("EC6P3")
+    enc = ec.Encoder(t)
+    slen = 16384
+    rng = np.random.default_rng(51)
+    shards, _ = make_stripe(rng, t.N, t.M, slen, dev)
+    enc.encode(shards)
+    new2 = torch.from_numpy(
+        rng.integers(0, 256, slen, dtype=np.uint8)).to(dev)
+    enc.update_idx(shards[2], new2, 2, shards[t.N:])
+    shards[2] = new2
+    assert enc.verify(shards)
