@@ -602,30 +602,40 @@ static int crc_grid(int64_t total_frames) {
  * conflict-free: lane t hits dword banks 68t+4i mod 64, distinct within
  * each 16-lane group).  Traffic = S read + S write, the algorithmic
  * minimum. */
-constexpr int STG_CHUNK = 128;  /* 512 chunks per 64 KiB frame, staged in
-                                   two 32 KiB passes; thread t owns chunks
-                                   t and t+256 */
-constexpr int STG_STRIDE = STG_CHUNK + 16;
-constexpr int STG_HALF = 256 * STG_CHUNK; /* payload bytes per pass */
+/* Stage-pass chunk size trades LDS footprint (blocks/CU, latency overlap)
+ * against barrier count: 256 -> 1 pass/frame, 2 blocks/CU; 128 -> 2
+ * passes, 3 blocks/CU; 64 -> 4 passes, 6 blocks/CU.  Selected at launch
+ * via GFRS_CRC_CHUNK (default 128). */
+template <int CHUNK> struct StgGeom {
+  static constexpr int STRIDE = CHUNK + 16;
+  static constexpr int HALF = 256 * CHUNK; /* payload bytes per pass */
+  static constexpr int PASSES = (65532 + HALF - 1) / HALF;
+  static constexpr int LDS = 4096 + 1024 + 256 * STRIDE;
+};
 
-template <int MODE, bool NT, bool TAILCRC>
+template <int MODE, bool NT, bool TAILCRC, int CHUNK>
 __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
     uint8_t *__restrict__ dst, size_t dst_stride,
     const uint8_t *__restrict__ src, size_t src_stride, int64_t n,
     int64_t frames_per_shard, int64_t total_frames, int64_t *__restrict__ bad) {
   constexpr int64_t block_len = 65536;
   constexpr int64_t payload_full = block_len - CRC_LEN;
+  using G = StgGeom<CHUNK>;
+  constexpr int STG_CHUNK = CHUNK;
+  constexpr int STG_STRIDE = G::STRIDE;
+  constexpr int STG_HALF = G::HALF;
+  constexpr int STG_PASSES = G::PASSES;
   extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
   uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
   uint32_t *fold = reinterpret_cast<uint32_t *>(smem + 4096);
-  uint8_t *stage = smem + 4096 + 1024; /* 256 chunk slots x (128+16) B */
+  uint8_t *stage = smem + 4096 + 1024; /* 256 chunk slots x STRIDE B */
   for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
     (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
   /* fold operators are a function of the thread's chunk positions only
    * for full frames — compute x^(8*suffix) once per block, not per frame */
-  uint32_t op_full[2];
+  uint32_t op_full[STG_PASSES];
 #pragma unroll
-  for (int h = 0; h < 2; h++) {
+  for (int h = 0; h < STG_PASSES; h++) {
     const int64_t c0 = int64_t(h) * STG_HALF + int64_t(threadIdx.x) * STG_CHUNK;
     const int64_t cend = i64min(c0 + STG_CHUNK, payload_full);
     op_full[h] = x8n_d(uint64_t(payload_full - cend));
@@ -658,7 +668,7 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
             ? init_full
             : gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
 #pragma unroll
-    for (int h = 0; h < 2; h++) {
+    for (int h = 0; h < STG_PASSES; h++) {
       const int64_t h0 = int64_t(h) * STG_HALF;
       const int64_t hbytes = i64min(int64_t(STG_HALF), payload - h0);
       if (hbytes <= 0) break;
@@ -675,13 +685,13 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
         }
         const int64_t p = 4 * w;
         *reinterpret_cast<uint32_t *>(
-            &stage[(p >> 7) * STG_STRIDE + (p & (STG_CHUNK - 1))]) = x;
+            &stage[(p / STG_CHUNK) * STG_STRIDE + (p & (STG_CHUNK - 1))]) = x;
       }
       if (threadIdx.x == 0)
         for (int64_t p = words * 4; p < hbytes; p++) {
           const uint8_t x = psrc[h0 + p];
           if (MODE != 1) pdst[h0 + p] = x;
-          stage[(p >> 7) * STG_STRIDE + (p & (STG_CHUNK - 1))] = x;
+          stage[(p / STG_CHUNK) * STG_STRIDE + (p & (STG_CHUNK - 1))] = x;
         }
       __syncthreads();
       const int64_t c0 = int64_t(threadIdx.x) * STG_CHUNK;
@@ -736,23 +746,33 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
   }
 }
 
-constexpr int STG_LDS = 4096 + 1024 + 256 * STG_STRIDE; /* ~42 KB -> 3 blocks/CU */
-
 template <int MODE, bool TAILCRC = false>
 static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
                          size_t src_stride, int64_t n, int64_t block_len,
                          int64_t fps, int64_t total, int64_t *bad,
                          hipStream_t s) {
   const int grid = crc_grid(total);
+  static const int chunk_sel = []() {
+    const char *e = getenv("GFRS_CRC_CHUNK");
+    const int v = e ? atoi(e) : 128;
+    return (v == 64 || v == 128 || v == 256) ? v : 128;
+  }();
   if (block_len == 65536) {
-    if (nt_enabled() && MODE != 1)
-      hipLaunchKernelGGL((crc32b_staged_k<MODE, true, TAILCRC>), dim3(grid),
-                         dim3(CRC_BLOCKT), STG_LDS, s, dst, dst_stride, src,
-                         src_stride, n, fps, total, bad);
-    else
-      hipLaunchKernelGGL((crc32b_staged_k<MODE, false, TAILCRC>), dim3(grid),
-                         dim3(CRC_BLOCKT), STG_LDS, s, dst, dst_stride, src,
-                         src_stride, n, fps, total, bad);
+    const bool nt = nt_enabled() && MODE != 1;
+#define GFRS_CRC_GO(NTV, CH)                                                \
+  hipLaunchKernelGGL((crc32b_staged_k<MODE, NTV, TAILCRC, CH>), dim3(grid), \
+                     dim3(CRC_BLOCKT), StgGeom<CH>::LDS, s, dst,            \
+                     dst_stride, src, src_stride, n, fps, total, bad)
+    if (nt) {
+      if (chunk_sel == 64) GFRS_CRC_GO(true, 64);
+      else if (chunk_sel == 256) GFRS_CRC_GO(true, 256);
+      else GFRS_CRC_GO(true, 128);
+    } else {
+      if (chunk_sel == 64) GFRS_CRC_GO(false, 64);
+      else if (chunk_sel == 256) GFRS_CRC_GO(false, 256);
+      else GFRS_CRC_GO(false, 128);
+    }
+#undef GFRS_CRC_GO
   } else {
     hipLaunchKernelGGL((crc32b_k<MODE, TAILCRC>), dim3(grid),
                        dim3(CRC_BLOCKT), 0, s, dst, dst_stride, src,
