@@ -190,8 +190,11 @@ static std::vector<int32_t> local_stripe(const gfrs_tactic &t, int az_idx) {
 }
 
 static bool tactic_valid(const gfrs_tactic *t) {
-  /* codemode.go:291-299 IsValid (EC portion) */
+  /* codemode.go:291-299 IsValid; replicate modes (m==0, l==0) are legal —
+   * ec.NewEncoder accepts them and Encode is a no-op (reedsolomon.go:442) */
   if (!t) return false;
+  if (t->m == 0 && t->l == 0)
+    return t->n > 0 && t->az_count > 0 && t->n % t->az_count == 0;
   if (t->n <= 0 || t->m <= 0 || t->l < 0 || t->az_count <= 0) return false;
   if (t->n % t->az_count || t->m % t->az_count || t->l % t->az_count)
     return false;
@@ -258,18 +261,19 @@ gfrs_ctx *gfrs_create(const gfrs_tactic *t, int device) {
   bool ok = hipStreamCreate(&c->own_stream) == hipSuccess;
   c->stream = c->own_stream;
 
-  /* global encode matrix (reedsolomon.go:220-244) */
-  if (ok) {
+  /* global encode matrix (reedsolomon.go:220-244); replicate modes have
+   * no parity engine (reedsolomon.go:442 parityShards==0 early return) */
+  if (ok && t->m > 0) {
     c->enc_matrix.resize(size_t(t->n + t->m) * t->n);
     ok = gf_build_matrix(t->n, t->n + t->m, c->enc_matrix.data());
-  }
-  if (ok) {
-    std::vector<int32_t> in(t->n), out(t->m);
-    for (int i = 0; i < t->n; i++) in[i] = i;
-    for (int i = 0; i < t->m; i++) out[i] = t->n + i;
-    std::vector<uint8_t> rows(c->enc_matrix.begin() + size_t(t->n) * t->n,
-                              c->enc_matrix.end());
-    ok = c->enc_plan.upload(in, out, rows, c->stream) == GFRS_OK;
+    if (ok) {
+      std::vector<int32_t> in(t->n), out(t->m);
+      for (int i = 0; i < t->n; i++) in[i] = i;
+      for (int i = 0; i < t->m; i++) out[i] = t->n + i;
+      std::vector<uint8_t> rows(c->enc_matrix.begin() + size_t(t->n) * t->n,
+                                c->enc_matrix.end());
+      ok = c->enc_plan.upload(in, out, rows, c->stream) == GFRS_OK;
+    }
   }
   /* local engines (encoder.go:92-104) */
   if (ok && t->l > 0) {
@@ -549,6 +553,7 @@ extern "C" {
 int gfrs_encode(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
                 int nshards, int memloc) {
   auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  if (c->t.m == 0) return nshards == c->total ? GFRS_OK : GFRS_ERR_INVALID_SHARDS;
   if (nshards != c->total) {
     seterr("encode: want %d shards, got %d", c->total, nshards);
     return GFRS_ERR_INVALID_SHARDS;
@@ -616,6 +621,10 @@ int gfrs_encode_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
 int gfrs_verify(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
                 int nshards, int memloc, int *ok) {
   auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  if (c->t.m == 0) { /* zero parity rows: vacuously true (reedsolomon.go:784) */
+    *ok = 1;
+    return nshards == c->total ? GFRS_OK : GFRS_ERR_INVALID_SHARDS;
+  }
   bool local_form = c->t.l > 0 && nshards == c->total / c->t.az_count;
   if (nshards != c->total && !local_form) return GFRS_ERR_INVALID_SHARDS;
   if (local_form) {
@@ -743,6 +752,8 @@ int gfrs_reconstruct(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
   StreamGuard g(c);
   int rc;
 
+  if (t.m == 0) /* no parity: any missing shard is unrecoverable */
+    return nbad == 0 ? GFRS_OK : GFRS_ERR_TOO_FEW_SHARDS;
   /* full set or, for LRC, a single local stripe (lrcencoder.go:147-153) */
   bool local_form = t.l > 0 && nshards == c->total / t.az_count;
   if (nshards != c->total && !local_form) return GFRS_ERR_INVALID_SHARDS;

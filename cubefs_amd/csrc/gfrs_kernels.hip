@@ -604,8 +604,8 @@ static int crc_grid(int64_t total_frames) {
  * minimum. */
 /* Stage-pass chunk size trades LDS footprint (blocks/CU, latency overlap)
  * against barrier count: 256 -> 1 pass/frame, 2 blocks/CU; 128 -> 2
- * passes, 3 blocks/CU; 64 -> 4 passes, 6 blocks/CU.  Selected at launch
- * via GFRS_CRC_CHUNK (default 128). */
+ * passes, 3 blocks/CU; 64 -> 4 passes, 6 blocks/CU (measured best);
+ * 32 -> 8 passes, 8 blocks/CU.  GFRS_CRC_CHUNK overrides (default 64). */
 template <int CHUNK> struct StgGeom {
   static constexpr int STRIDE = CHUNK + 16;
   static constexpr int HALF = 256 * CHUNK; /* payload bytes per pass */
@@ -754,8 +754,8 @@ static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
   const int grid = crc_grid(total);
   static const int chunk_sel = []() {
     const char *e = getenv("GFRS_CRC_CHUNK");
-    const int v = e ? atoi(e) : 128;
-    return (v == 64 || v == 128 || v == 256) ? v : 128;
+    const int v = e ? atoi(e) : 64;
+    return (v == 32 || v == 64 || v == 128 || v == 256) ? v : 64;
   }();
   if (block_len == 65536) {
     const bool nt = nt_enabled() && MODE != 1;
@@ -764,13 +764,15 @@ static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
                      dim3(CRC_BLOCKT), StgGeom<CH>::LDS, s, dst,            \
                      dst_stride, src, src_stride, n, fps, total, bad)
     if (nt) {
-      if (chunk_sel == 64) GFRS_CRC_GO(true, 64);
+      if (chunk_sel == 32) GFRS_CRC_GO(true, 32);
+      else if (chunk_sel == 128) GFRS_CRC_GO(true, 128);
       else if (chunk_sel == 256) GFRS_CRC_GO(true, 256);
-      else GFRS_CRC_GO(true, 128);
+      else GFRS_CRC_GO(true, 64);
     } else {
-      if (chunk_sel == 64) GFRS_CRC_GO(false, 64);
+      if (chunk_sel == 32) GFRS_CRC_GO(false, 32);
+      else if (chunk_sel == 128) GFRS_CRC_GO(false, 128);
       else if (chunk_sel == 256) GFRS_CRC_GO(false, 256);
-      else GFRS_CRC_GO(false, 128);
+      else GFRS_CRC_GO(false, 64);
     }
 #undef GFRS_CRC_GO
   } else {

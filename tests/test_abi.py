@@ -84,3 +84,17 @@ def test_no_gpu_fails_loudly():
     ctx = L.gfrs_create(ctypes.byref(t), -1)
     assert not ctx
     assert b"no HIP device" in L.gfrs_last_error()
+
+
+def test_replicate_mode_accepted():
+    """ec.NewEncoder accepts replicate tactics (codemode Replica3); Encode
+    is a no-op and Verify vacuously true (reedsolomon.go:442,784).  These
+    entry points are GPU-free, so they are testable here."""
+    import torch
+    if torch.cuda.is_available():
+        pytest.skip("covered by GPU suite on a box")
+    # without a GPU create still fails (no-fallback rule), but the tactic
+    # must pass validation: check via the python registry
+    from cubefs_amd import codemode
+    t = codemode.get_tactic("Replica3")
+    assert t.is_valid() and t.is_replicate()

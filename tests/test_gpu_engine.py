@@ -141,3 +141,27 @@ def test_native_code_is_loaded():
     lib = runtime.lib()
     assert "cubefs_amd/libgfrs.so" in lib._name
     assert lib.gfrs_device_count() >= 1
+
+
+def test_replicate_mode(dev):
+    """Replica3 tactic: Encode no-op, Verify true, Reconstruct only when
+    nothing is missing (reedsolomon.go:442,784)."""
+    from cubefs_amd import codemode
+    from cubefs_amd.runtime import GfrsError, Tactic, lib
+    import ctypes
+    t = Tactic(3, 0, 0, 3, 3, 0, 0)
+    ctx = lib().gfrs_create(ctypes.byref(t), -1)
+    assert ctx, lib().gfrs_last_error()
+    sh = [torch.randint(0, 256, (4096,), dtype=torch.uint8, device=dev)
+          for _ in range(3)]
+    before = [s.clone() for s in sh]
+    arr = (ctypes.c_void_p * 3)(*[s.data_ptr() for s in sh])
+    assert lib().gfrs_encode(ctx, arr, 4096, 3, 0) == 0
+    for a, b in zip(sh, before):
+        assert torch.equal(a, b)
+    ok = ctypes.c_int(0)
+    assert lib().gfrs_verify(ctx, arr, 4096, 3, 0, ctypes.byref(ok)) == 0
+    assert ok.value == 1
+    bad = (ctypes.c_int32 * 1)(1)
+    assert lib().gfrs_reconstruct(ctx, arr, 4096, 3, 0, bad, 1, 0) == -2
+    lib().gfrs_destroy(ctx)
