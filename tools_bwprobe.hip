@@ -47,6 +47,18 @@ __global__ void k_copy(uint4 *__restrict__ d, const uint4 *__restrict__ s,
   }
 }
 
+/* encode-shaped mix: read 2 streams, write 1 (same ratio as RS(6+3)) */
+__global__ void k_mix21(uint4 *__restrict__ d, const uint4 *__restrict__ s0,
+                        const uint4 *__restrict__ s1, size_t n16) {
+  for (size_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n16;
+       i += size_t(gridDim.x) * blockDim.x) {
+    const uint4 a = s0[i];
+    const uint4 b = s1[i];
+    uint4 v{a.x ^ b.x, a.y ^ b.y, a.z ^ b.z, a.w ^ b.w};
+    d[i] = v;
+  }
+}
+
 int main() {
   const size_t bytes = size_t(8) << 30;
   const size_t n16 = bytes / 16;
@@ -109,6 +121,20 @@ int main() {
     hipEventRecord(e1); hipEventSynchronize(e1);
     float ms; hipEventElapsedTime(&ms, e0, e1);
     printf("%-14s %8.1f GB/s (moved)\n", "copy_nt", 2.0 * bytes / (ms / reps / 1e3) / 1e9);
+  }
+  {
+    uint4 *c;
+    hipMalloc(&c, bytes);
+    hipMemset(c, 3, bytes);
+    hipEvent_t e0, e1; hipEventCreate(&e0); hipEventCreate(&e1);
+    hipLaunchKernelGGL(k_mix21, grid, blk, 0, 0, c, a, b, n16);
+    hipDeviceSynchronize(); hipEventRecord(e0);
+    for (int r = 0; r < reps; r++)
+      hipLaunchKernelGGL(k_mix21, grid, blk, 0, 0, c, a, b, n16);
+    hipEventRecord(e1); hipEventSynchronize(e1);
+    float ms; hipEventElapsedTime(&ms, e0, e1);
+    printf("%-14s %8.1f GB/s (moved, 2r:1w)\n", "mix21",
+           3.0 * bytes / (ms / reps / 1e3) / 1e9);
   }
   return 0;
 }
