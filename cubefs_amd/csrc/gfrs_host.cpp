@@ -827,6 +827,36 @@ int gfrs_reconstruct_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
   return GFRS_OK;
 }
 
+int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
+                            size_t framed_stride, void *base,
+                            size_t shard_len, size_t stripe_stride,
+                            int nstripes, int64_t block_len) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  const gfrs_tactic &t = c->t;
+  if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
+  if (nstripes <= 0 || shard_len == 0) return GFRS_ERR_INVALID_SHARDS;
+  StreamGuard g(c);
+  const bool fused = block_len == 65536 && t.l == 0 && t.m >= 1 &&
+                     t.m <= 4 && t.n + t.m <= 16 && framed_stride % 4 == 0;
+  if (fused) {
+    launch_rs_encode_frame((uint8_t *)framed, framed_stride, (uint64_t)base,
+                           stripe_stride, shard_len, t.n, t.m,
+                           (const uint8_t *)c->enc_plan.tabs.p, nstripes,
+                           c->stream);
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess) return hip_fail("encode_frame launch", e);
+    return GFRS_OK;
+  }
+  /* fallback composition: needs the contiguous ec.Buffer batch layout */
+  if (stripe_stride != size_t(c->total) * shard_len)
+    return GFRS_ERR_UNSUPPORTED;
+  int rc = gfrs_encode_batch(ctx, base, shard_len, stripe_stride, nstripes);
+  if (rc != GFRS_OK) return rc;
+  return gfrs_crc32b_encode_batch(ctx, framed, framed_stride, base,
+                                  shard_len, int64_t(shard_len), block_len,
+                                  nstripes * c->total);
+}
+
 /* ---------------- sized coder (rpc2 body framing) ---------------- */
 
 int gfrs_sized_encode_size(int64_t actual_size, int64_t block_len,

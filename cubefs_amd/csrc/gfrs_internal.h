@@ -71,6 +71,13 @@ void launch_crc_decode(uint8_t *dst, size_t dst_stride, const uint8_t *framed,
                        int64_t block_len, int nshards, int64_t *bad,
                        hipStream_t s);
 
+/* fused encode+frame: parity + crc32block framed images in one pass
+ * (PUT/repair pipeline; data read once, framed written once). */
+void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
+                            uint64_t stripe_stride, size_t shard_len, int k,
+                            int m, const uint8_t *tabs, int nstripes,
+                            hipStream_t s);
+
 /* sized coder (crc32block/sized_coder.go): payload ‖ CRC32(BE) frames,
  * 512-B tail alignment handled by the host. */
 void launch_sized_encode(uint8_t *dst, size_t dst_stride, const uint8_t *src,

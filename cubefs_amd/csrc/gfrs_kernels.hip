@@ -811,6 +811,245 @@ void launch_sized_decode(uint8_t *dst, size_t dst_stride,
 }
 
 /* ------------------------------------------------------------------ */
+/* fused encode+frame: RS parity + crc32block framing in one pass        */
+/* ------------------------------------------------------------------ */
+
+/* The blobstore PUT/repair pipeline is encode (access/stream) followed by
+ * per-shard crc32block framing (blobnode datafile.go:342).  Run as two
+ * kernels that is 3 HBM passes over the data (encode: read k, write m;
+ * frame: read k+m, write k+m).  Fused, each data byte is read ONCE, the
+ * framed images are written ONCE, and the unframed parity never exists in
+ * HBM: read k·S + write (k+m)·(S+headers) — a 1.8x traffic cut at 6+3.
+ *
+ * Geometry: one workgroup per (stripe, 64 KiB frame); the frame's 65,532
+ * payload bytes are processed in 4 passes of 16,384 B.  Within a pass,
+ * lane w owns columns {i·4096 + w·16 : i<4} for the parity MACs
+ * (coalesced uint4 loads, identical to rs_apply) and chunk
+ * [w·64, w·64+64) of the pass for the CRC fold (read back from a 20 KB
+ * LDS stage with conflict-free b128 banking).  Each shard's range is
+ * staged, CRC'd and framed-written in turn; parity accumulates in VGPRs
+ * across the k data shards and then takes the same stage path. */
+
+constexpr int EF_PASS = 16384;  /* bytes per pass */
+constexpr int EF_CHUNK = 64;    /* CRC chunk per lane per pass */
+constexpr int EF_STRIDE = EF_CHUNK + 16;
+constexpr int EF_PASSES = 4;    /* ceil(65532 / 16384) */
+constexpr int EF_LDS = 4096 + 256 * EF_STRIDE; /* tables + stage = 24.5 KB */
+
+template <int GM>
+__global__ __launch_bounds__(CRC_BLOCKT) void rs_encode_frame_k(
+    uint8_t *__restrict__ dst, size_t dst_stride /* framed image stride */,
+    uint64_t base, uint64_t stripe_stride, size_t shard_len, int k,
+    const uint8_t *__restrict__ tabs /* [GM*k][32] */, int64_t total_frames,
+    int64_t frames_per_shard) {
+  constexpr int64_t block_len = 65536;
+  constexpr int64_t payload_full = block_len - CRC_LEN;
+  extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
+  uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
+  uint8_t *stage = smem + 4096;
+  uint4 *ltab_v = reinterpret_cast<uint4 *>(smem + 4096); /* transient */
+  /* coefficient tables live in registers: GM*k*32 B is too much for
+   * k>4, so re-read from global per MAC via __ldg-style loads would be
+   * slow; instead keep them in LDS *before* the stage area is used —
+   * they are consumed only during the MAC phase of each pass, while the
+   * stage is consumed in the CRC phase, so they can share space only if
+   * reloaded per pass.  Simpler: put them in the tail of the table area
+   * is impossible (4 KB exactly) — so carve GM*k*32 extra after stage. */
+  uint8_t *ctab = smem + 4096 + 256 * EF_STRIDE;
+  for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
+    (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
+  for (int i = threadIdx.x; i < GM * k * 2; i += CRC_BLOCKT)
+    reinterpret_cast<uint4 *>(ctab)[i] =
+        reinterpret_cast<const uint4 *>(tabs)[i];
+  (void)ltab_v;
+  /* per-(pass,lane) fold operator for full frames */
+  uint32_t op_full[EF_PASSES];
+#pragma unroll
+  for (int h = 0; h < EF_PASSES; h++) {
+    const int64_t c0 = int64_t(h) * EF_PASS + int64_t(threadIdx.x) * EF_CHUNK;
+    const int64_t cend = i64min(c0 + EF_CHUNK, payload_full);
+    op_full[h] = x8n_d(uint64_t(payload_full - (cend > c0 ? cend : c0)));
+  }
+  __syncthreads();
+
+  const uint4 *ltab = reinterpret_cast<const uint4 *>(ctab);
+
+  for (int64_t fr = blockIdx.x; fr < total_frames; fr += gridDim.x) {
+    const int64_t stripe = fr / frames_per_shard;
+    const int64_t f = fr - stripe * frames_per_shard;
+    const int64_t p0 = f * payload_full;
+    const int64_t payload = i64min(payload_full, int64_t(shard_len) - p0);
+    const uint8_t *sbase =
+        reinterpret_cast<const uint8_t *>(base + stripe * stripe_stride);
+
+    uint32_t crcacc[16]; /* fused fast path requires k+GM <= 16 */
+#pragma unroll
+    for (int j = 0; j < 16; j++) crcacc[j] = 0;
+    uint4 acc[GM][4];
+
+    for (int h = 0; h < EF_PASSES; h++) {
+      const int64_t r0 = int64_t(h) * EF_PASS;
+      const int64_t rbytes = i64min(int64_t(EF_PASS), payload - r0);
+      if (rbytes <= 0) break;
+      /* acc is per pass: each pass covers a fresh column range */
+#pragma unroll
+      for (int r = 0; r < GM; r++)
+#pragma unroll
+        for (int i = 0; i < 4; i++) acc[r][i] = uint4{0, 0, 0, 0};
+      /* ---- data shards: load, MAC, framed-write, stage, CRC ---- */
+      for (int c = 0; c < k; c++) {
+        const uint8_t *src = sbase + size_t(c) * shard_len + p0 + r0;
+        uint8_t *fdst =
+            dst + (stripe * (k + GM) + c) * dst_stride + f * block_len +
+            CRC_LEN + r0;
+        uint4 v[4];
+        const int64_t lane16 = int64_t(threadIdx.x) * 16;
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+          const int64_t off = int64_t(i) * 4096 + lane16;
+          if (off + 16 <= rbytes)
+            v[i] = *reinterpret_cast<const uint4 *>(src + off);
+          else
+            v[i] = uint4{0, 0, 0, 0};
+        }
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+          const int64_t off = int64_t(i) * 4096 + lane16;
+          if (off + 16 <= rbytes) {
+#pragma unroll
+            for (int r = 0; r < GM; r++) {
+              const int t2 = (r * k + c) * 2;
+              gfmac16<false>(acc[r][i], v[i], ltab[t2], ltab[t2 + 1]);
+            }
+            /* framed payload write (u32: +4 header breaks 16B align) */
+            uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+            dw[0] = v[i].x; dw[1] = v[i].y; dw[2] = v[i].z; dw[3] = v[i].w;
+            /* stage for the CRC phase */
+            const int64_t p = off;
+            *reinterpret_cast<uint4 *>(
+                &stage[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))]) =
+                v[i];
+          }
+        }
+        /* ragged tail of the last pass: bytes not covered by uint4 lanes */
+        if (rbytes < EF_PASS && threadIdx.x == 0) {
+          const int64_t start = (rbytes / 16) * 16;
+          for (int64_t p = start; p < rbytes; p++) {
+            const uint8_t x = src[p];
+            fdst[p] = x;
+            stage[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))] = x;
+          }
+          /* tail parity handled below after barrier by thread 0 via the
+           * byte table path -- accumulate into stage scratch is complex;
+           * instead recompute per-byte in the parity phase */
+        }
+        __syncthreads();
+        /* CRC this shard's chunks */
+        const int64_t c0b = int64_t(threadIdx.x) * EF_CHUNK;
+        int clen = int(i64min(int64_t(EF_CHUNK), rbytes - c0b));
+        if (clen < 0) clen = 0;
+        uint32_t part =
+            crc_chunk16(stage + threadIdx.x * EF_STRIDE, clen, tab);
+        uint32_t op = op_full[h];
+        if (payload != payload_full) {
+          const int64_t suffix = clen > 0 ? payload - (r0 + c0b + clen) : 0;
+          op = x8n_d(uint64_t(suffix));
+        }
+        crcacc[c] ^= clen > 0 ? gf2_mulmod_d(op, part) : 0;
+        __syncthreads();
+      }
+      /* ---- parity shards: ragged-tail fix, framed-write, stage, CRC -- */
+      const int64_t lane16 = int64_t(threadIdx.x) * 16;
+      for (int r = 0; r < GM; r++) {
+        uint8_t *fdst =
+            dst + (stripe * (k + GM) + k + r) * dst_stride + f * block_len +
+            CRC_LEN + r0;
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+          const int64_t off = int64_t(i) * 4096 + lane16;
+          if (off + 16 <= rbytes) {
+            uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+            dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
+            dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
+            const int64_t p = off;
+            *reinterpret_cast<uint4 *>(
+                &stage[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))]) =
+                acc[r][i];
+          }
+        }
+        if (rbytes < EF_PASS && threadIdx.x == 0) {
+          /* per-byte parity for the ragged tail */
+          const int64_t start = (rbytes / 16) * 16;
+          for (int64_t p = start; p < rbytes; p++) {
+            uint8_t pv = 0;
+            for (int c2 = 0; c2 < k; c2++) {
+              const uint8_t b =
+                  sbase[size_t(c2) * shard_len + p0 + r0 + p];
+              const uint8_t *t = ctab + size_t(r * k + c2) * 32;
+              pv ^= t[b & 0x0F] ^ t[16 + (b >> 4)];
+            }
+            fdst[p] = pv;
+            stage[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))] = pv;
+          }
+        }
+        __syncthreads();
+        const int64_t c0b = int64_t(threadIdx.x) * EF_CHUNK;
+        int clen = int(i64min(int64_t(EF_CHUNK), rbytes - c0b));
+        if (clen < 0) clen = 0;
+        uint32_t part =
+            crc_chunk16(stage + threadIdx.x * EF_STRIDE, clen, tab);
+        uint32_t op = op_full[h];
+        if (payload != payload_full) {
+          const int64_t suffix = clen > 0 ? payload - (r0 + c0b + clen) : 0;
+          op = x8n_d(uint64_t(suffix));
+        }
+        crcacc[k + r] ^= clen > 0 ? gf2_mulmod_d(op, part) : 0;
+        __syncthreads();
+      }
+    }
+    /* ---- reduce per-shard CRCs, write the 4 B LE headers ---- */
+    for (int j = 0; j < k + GM; j++) {
+      uint32_t part = crcacc[j];
+#pragma unroll
+      for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
+      if ((threadIdx.x & 63) == 0)
+        reinterpret_cast<uint32_t *>(stage)[threadIdx.x >> 6] = part;
+      __syncthreads();
+      if (threadIdx.x == 0) {
+        const uint32_t *fw = reinterpret_cast<uint32_t *>(stage);
+        const uint32_t it =
+            gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
+        const uint32_t crc = ~(it ^ fw[0] ^ fw[1] ^ fw[2] ^ fw[3]);
+        *reinterpret_cast<uint32_t *>(
+            dst + (stripe * (k + GM) + j) * dst_stride + f * block_len) = crc;
+      }
+      __syncthreads();
+    }
+  }
+}
+
+void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
+                            uint64_t stripe_stride, size_t shard_len, int k,
+                            int m, const uint8_t *tabs, int nstripes,
+                            hipStream_t s) {
+  const int64_t fps = (int64_t(shard_len) + 65531) / 65532;
+  const int64_t total = fps * nstripes;
+  const int grid = crc_grid(total);
+  const int lds = EF_LDS + m * k * 32;
+#define GFRS_EF_GO(G)                                                     \
+  hipLaunchKernelGGL((rs_encode_frame_k<G>), dim3(grid), dim3(CRC_BLOCKT),\
+                     lds, s, dst, dst_stride, base, stripe_stride,        \
+                     shard_len, k, tabs, total, fps)
+  switch (m) {
+    case 1: GFRS_EF_GO(1); break;
+    case 2: GFRS_EF_GO(2); break;
+    case 3: GFRS_EF_GO(3); break;
+    default: GFRS_EF_GO(4);
+  }
+#undef GFRS_EF_GO
+}
+
+/* ------------------------------------------------------------------ */
 /* blobnode on-disk shard codec (core/shard.go:42-111, datafile.go:342)  */
 /* ------------------------------------------------------------------ */
 

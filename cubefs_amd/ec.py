@@ -197,6 +197,16 @@ class Encoder:
         check(lib().gfrs_encode_batch(self._ctx, p, ln, stride, ns),
               "encode_batch")
 
+    def encode_frame_batch(self, framed, batch, block_len=65536):
+        """Fused PUT pipeline: parity + crc32block framed images in one
+        pass (stream_put.go:146 + datafile.go:342).  framed:
+        [nstripes*(n+m), encode_size(shard_len)] device tensor."""
+        p, ln, stride, ns = self._base(batch)
+        check(lib().gfrs_encode_frame_batch(self._ctx, framed.data_ptr(),
+                                            framed.stride(0), p, ln, stride,
+                                            ns, block_len),
+              "encode_frame_batch")
+
     def verify_batch(self, batch):
         p, ln, stride, ns = self._base(batch)
         nwords = (ns + 63) // 64

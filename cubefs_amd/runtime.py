@@ -116,6 +116,9 @@ def lib():
                                              u64p, i64p, ctypes.c_int]
         L.gfrs_encode_idx.argtypes = [vp, vp, ctypes.c_int, vpp,
                                       ctypes.c_size_t, ctypes.c_int]
+        L.gfrs_encode_frame_batch.argtypes = [vp, vp, ctypes.c_size_t, vp,
+                                              ctypes.c_size_t, ctypes.c_size_t,
+                                              ctypes.c_int, i64]
         L.gfrs_update_idx.argtypes = [vp, vp, vp, ctypes.c_int, vpp,
                                       ctypes.c_size_t, ctypes.c_int]
         L.gfrs_repair_batch.argtypes = [vp, vp, ctypes.c_size_t, ctypes.c_size_t,

@@ -150,6 +150,21 @@ int gfrs_crc32b_verify_batch(gfrs_ctx *ctx, const void *framed,
                              int64_t block_len, int nshards,
                              int64_t *bad_block_per_shard);
 
+/* ---- fused encode+frame (the PUT pipeline: stream_put.go:146 encode +
+ * datafile.go:342 framing in ONE pass) ----
+ * Reads the data shards once, computes parity in registers and writes
+ * ONLY the k+m crc32block-framed shard images (frame image j of stripe s
+ * at framed + (s*(n+m)+j)*framed_stride); the unframed parity never
+ * touches HBM.  Requires block_len 65536, 1<=m<=4, n+m<=16, l==0 and
+ * framed_stride % 4 == 0; other shapes fall back to
+ * encode_batch + crc32b_encode_batch (which needs
+ * stripe_stride == (n+m)*shard_len so the shard rows form one strided
+ * array).  Framed bytes are identical either way. */
+int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
+                            size_t framed_stride, void *base,
+                            size_t shard_len, size_t stripe_stride,
+                            int nstripes, int64_t block_len);
+
 /* ---- sized coder (crc32block/sized_coder.go, the rpc2 body framing) ----
  * Frame = payload (block_len-4) ‖ CRC32-IEEE big-endian (ModeEncode,
  * sized_coder.go:256-279); the encoded stream is zero-padded to the

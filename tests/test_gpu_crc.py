@@ -109,3 +109,32 @@ def test_other_block_lengths(oracle, dev, codec):
         codec.encode(dst, src, block_len=bl)
         assert np.array_equal(dst.cpu().numpy(), want), bl
         assert codec.verify(dst, block_len=bl) == -1
+
+
+def test_fused_encode_frame(oracle, dev):
+    """Fused encode+frame: framed images bit-identical to
+    oracle-encode followed by oracle-framing, across ragged sizes and
+    multi-frame shards; fallback path (EC15P12) agrees too."""
+    from cubefs_amd import codemode, crc32block, ec
+    for name, slen in [("EC6P3", 300000), ("EC6P3", 65532), ("EC6P3", 100),
+                       ("EC6P3", 1 << 20), ("EC12P4", 200000),
+                       ("EC15P12", 100000)]:  # EC15P12 -> fallback (m>4)
+        t = codemode.get_tactic(name)
+        ns = 3
+        rng = np.random.default_rng(slen ^ t.N)
+        arr = rng.integers(0, 256, (ns, t.total, slen), dtype=np.uint8)
+        batch = torch.from_numpy(arr.copy()).to(dev)
+        enc_sz = crc32block.encode_size(slen)
+        framed = torch.zeros((ns * t.total, enc_sz), dtype=torch.uint8,
+                             device=dev)
+        enc = ec.Encoder(t)
+        enc.encode_frame_batch(framed, batch)
+        enc.synchronize()
+        got = framed.cpu().numpy()
+        for s in range(ns):
+            sh = [arr[s, i].copy() for i in range(t.total)]
+            oracle.rs_encode(t.N, t.M, sh)
+            for j in range(t.total):
+                want = oracle.crc32b_encode(sh[j])
+                assert np.array_equal(got[s * t.total + j], want), \
+                    (name, slen, s, j)
