@@ -46,6 +46,9 @@ def parse_args():
                    help="shard index reconstructed in --workload reconstruct")
     p.add_argument("--no-crc", action="store_true",
                    help="EC encode only (config label form)")
+    p.add_argument("--no-fused", action="store_true",
+                   help="two-kernel encode-then-frame instead of the fused "
+                        "single-pass kernel")
     p.add_argument("--cpu-sample-stripes", type=int, default=24)
     p.add_argument("--skip-cpu-baseline", action="store_true")
     return p.parse_args()
@@ -172,10 +175,17 @@ def main():
             codec.synchronize()
         bad = [args.bad_idx]
 
+    fused = (args.workload == "encode" and with_crc and not args.no_fused)
+
     def step(events=None):
         if events:
             events[0].record()
-        if args.workload == "encode":
+        if fused:
+            # single-pass: parity + framed images, data read once
+            enc.encode_frame_batch(framed, batch)
+            if events:
+                events[1].record()
+        elif args.workload == "encode":
             enc.encode_batch(batch)
             if events:
                 events[1].record()
@@ -218,17 +228,27 @@ def main():
         total_src = sum(r["src_bytes"] for r in recs)
         value = total_src / GIB / elapsed
         total_shard = total_src / t.N * t.total / GIB / elapsed
-        kind = "encode" if args.workload == "encode" else             "reconstruct[%d]" % args.bad_idx
+        if args.workload != "encode":
+            kind = "reconstruct[%d]" % args.bad_idx
+        elif fused:
+            kind = "fused encode+frame"
+        else:
+            kind = "encode"
         workload = "RS(%d+%d%s) %s%s, %d MiB shards, %d stripes/GPU" % (
             t.N, t.M, "+L%d" % t.L if t.L else "", kind,
-            "+crc32block" if with_crc else "", args.shard_mib, ns)
-        # roofline of the dominant kernel (rs_apply over the whole batch,
-        # one launch per step): encode reads k·S writes (m+l)·S per stripe;
-        # 1-shard reconstruct reads k·S writes 1·S
-        if args.workload == "encode":
-            alg_bytes = float(t.total * S * ns)
-        else:
+            "+crc32block" if (with_crc and not fused) else "",
+            args.shard_mib, ns)
+        # roofline of the dominant kernel, one launch per step:
+        #   encode: read k·S, write (m+l)·S per stripe
+        #   fused encode+frame: read k·S, write (k+m)·(S+4·fps) per stripe
+        #   1-shard reconstruct: read k·S, write 1·S
+        fps = -(-S // 65532)
+        if args.workload != "encode":
             alg_bytes = float((t.N + len(bad)) * S * ns)
+        elif fused:
+            alg_bytes = float((t.N + t.total) * S + 4 * fps * t.total) * ns
+        else:
+            alg_bytes = float(t.total * S * ns)
         avg_enc_s = (sum(enc_ms) / len(enc_ms)) / 1e3
         achieved = alg_bytes / avg_enc_s
         traffic = hbm_traffic_lookup(workload)
