@@ -834,7 +834,8 @@ constexpr int EF_PASS = 16384;  /* bytes per pass */
 constexpr int EF_CHUNK = 64;    /* CRC chunk per lane per pass */
 constexpr int EF_STRIDE = EF_CHUNK + 16;
 constexpr int EF_PASSES = 4;    /* ceil(65532 / 16384) */
-constexpr int EF_LDS = 4096 + 256 * EF_STRIDE; /* tables + stage = 24.5 KB */
+constexpr int EF_STG_ONE = 256 * EF_STRIDE; /* one stage buffer (20 KB) */
+constexpr int EF_LDS = 4096 + 2 * EF_STG_ONE; /* tables + double stage */
 
 template <int GM>
 __global__ __launch_bounds__(CRC_BLOCKT) void rs_encode_frame_k(
@@ -887,144 +888,137 @@ __global__ __launch_bounds__(CRC_BLOCKT) void rs_encode_frame_k(
     for (int j = 0; j < 16; j++) crcacc[j] = 0;
     uint4 acc[GM][4];
 
+    /* Software pipeline over the k+GM shard units of each pass: unit u's
+     * range is written into stage[u&1] while unit u-1's chunks are CRC'd
+     * out of stage[(u-1)&1] — ONE barrier per unit, and the LDS/VALU CRC
+     * work overlaps the next unit's global traffic. */
     for (int h = 0; h < EF_PASSES; h++) {
       const int64_t r0 = int64_t(h) * EF_PASS;
       const int64_t rbytes = i64min(int64_t(EF_PASS), payload - r0);
       if (rbytes <= 0) break;
-      /* acc is per pass: each pass covers a fresh column range */
 #pragma unroll
       for (int r = 0; r < GM; r++)
 #pragma unroll
         for (int i = 0; i < 4; i++) acc[r][i] = uint4{0, 0, 0, 0};
-      /* ---- data shards: load, MAC, framed-write, stage, CRC ---- */
-      for (int c = 0; c < k; c++) {
-        const uint8_t *src = sbase + size_t(c) * shard_len + p0 + r0;
-        uint8_t *fdst =
-            dst + (stripe * (k + GM) + c) * dst_stride + f * block_len +
-            CRC_LEN + r0;
-        uint4 v[4];
-        const int64_t lane16 = int64_t(threadIdx.x) * 16;
-#pragma unroll
-        for (int i = 0; i < 4; i++) {
-          const int64_t off = int64_t(i) * 4096 + lane16;
-          if (off + 16 <= rbytes)
-            v[i] = *reinterpret_cast<const uint4 *>(src + off);
-          else
-            v[i] = uint4{0, 0, 0, 0};
-        }
-#pragma unroll
-        for (int i = 0; i < 4; i++) {
-          const int64_t off = int64_t(i) * 4096 + lane16;
-          if (off + 16 <= rbytes) {
-#pragma unroll
-            for (int r = 0; r < GM; r++) {
-              const int t2 = (r * k + c) * 2;
-              gfmac16<false>(acc[r][i], v[i], ltab[t2], ltab[t2 + 1]);
-            }
-            /* framed payload write (u32: +4 header breaks 16B align) */
-            uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
-            dw[0] = v[i].x; dw[1] = v[i].y; dw[2] = v[i].z; dw[3] = v[i].w;
-            /* stage for the CRC phase */
-            const int64_t p = off;
-            *reinterpret_cast<uint4 *>(
-                &stage[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))]) =
-                v[i];
-          }
-        }
-        /* ragged tail of the last pass: bytes not covered by uint4 lanes */
-        if (rbytes < EF_PASS && threadIdx.x == 0) {
-          const int64_t start = (rbytes / 16) * 16;
-          for (int64_t p = start; p < rbytes; p++) {
-            const uint8_t x = src[p];
-            fdst[p] = x;
-            stage[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))] = x;
-          }
-          /* tail parity handled below after barrier by thread 0 via the
-           * byte table path -- accumulate into stage scratch is complex;
-           * instead recompute per-byte in the parity phase */
-        }
-        __syncthreads();
-        /* CRC this shard's chunks */
-        const int64_t c0b = int64_t(threadIdx.x) * EF_CHUNK;
-        int clen = int(i64min(int64_t(EF_CHUNK), rbytes - c0b));
-        if (clen < 0) clen = 0;
-        uint32_t part =
-            crc_chunk16(stage + threadIdx.x * EF_STRIDE, clen, tab);
-        uint32_t op = op_full[h];
-        if (payload != payload_full) {
-          const int64_t suffix = clen > 0 ? payload - (r0 + c0b + clen) : 0;
-          op = x8n_d(uint64_t(suffix));
-        }
-        crcacc[c] ^= clen > 0 ? gf2_mulmod_d(op, part) : 0;
-        __syncthreads();
-      }
-      /* ---- parity shards: ragged-tail fix, framed-write, stage, CRC -- */
+
       const int64_t lane16 = int64_t(threadIdx.x) * 16;
-      for (int r = 0; r < GM; r++) {
-        uint8_t *fdst =
-            dst + (stripe * (k + GM) + k + r) * dst_stride + f * block_len +
-            CRC_LEN + r0;
+      const int64_t c0b = int64_t(threadIdx.x) * EF_CHUNK;
+      int clen = int(i64min(int64_t(EF_CHUNK), rbytes - c0b));
+      if (clen < 0) clen = 0;
+      uint32_t op = op_full[h];
+      if (payload != payload_full) {
+        const int64_t suffix = clen > 0 ? payload - (r0 + c0b + clen) : 0;
+        op = x8n_d(uint64_t(suffix));
+      }
+
+      for (int u = 0; u < k + GM; u++) {
+        uint8_t *stg = stage + (u & 1) * EF_STG_ONE;
+        if (u < k) {
+          /* data shard u: load, MAC, framed-write, stage */
+          const int c = u;
+          const uint8_t *src = sbase + size_t(c) * shard_len + p0 + r0;
+          uint8_t *fdst = dst + (stripe * (k + GM) + c) * dst_stride +
+                          f * block_len + CRC_LEN + r0;
 #pragma unroll
-        for (int i = 0; i < 4; i++) {
-          const int64_t off = int64_t(i) * 4096 + lane16;
-          if (off + 16 <= rbytes) {
-            uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
-            dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
-            dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
-            const int64_t p = off;
-            *reinterpret_cast<uint4 *>(
-                &stage[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))]) =
-                acc[r][i];
-          }
-        }
-        if (rbytes < EF_PASS && threadIdx.x == 0) {
-          /* per-byte parity for the ragged tail */
-          const int64_t start = (rbytes / 16) * 16;
-          for (int64_t p = start; p < rbytes; p++) {
-            uint8_t pv = 0;
-            for (int c2 = 0; c2 < k; c2++) {
-              const uint8_t b =
-                  sbase[size_t(c2) * shard_len + p0 + r0 + p];
-              const uint8_t *t = ctab + size_t(r * k + c2) * 32;
-              pv ^= t[b & 0x0F] ^ t[16 + (b >> 4)];
+          for (int i = 0; i < 4; i++) {
+            const int64_t off = int64_t(i) * 4096 + lane16;
+            if (off + 16 <= rbytes) {
+              const uint4 v = *reinterpret_cast<const uint4 *>(src + off);
+#pragma unroll
+              for (int r = 0; r < GM; r++) {
+                const int t2 = (r * k + c) * 2;
+                gfmac16<false>(acc[r][i], v, ltab[t2], ltab[t2 + 1]);
+              }
+              uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+              dw[0] = v.x; dw[1] = v.y; dw[2] = v.z; dw[3] = v.w;
+              *reinterpret_cast<uint4 *>(
+                  &stg[(off / EF_CHUNK) * EF_STRIDE +
+                       (off & (EF_CHUNK - 1))]) = v;
             }
-            fdst[p] = pv;
-            stage[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))] = pv;
+          }
+          if (rbytes < EF_PASS && threadIdx.x == 0) {
+            const int64_t t0 = (rbytes / 16) * 16;
+            for (int64_t p = t0; p < rbytes; p++) {
+              const uint8_t x = src[p];
+              fdst[p] = x;
+              stg[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))] = x;
+            }
+          }
+        } else {
+          /* parity shard u-k: framed-write the accumulator, stage it */
+          const int r = u - k;
+          uint8_t *fdst = dst + (stripe * (k + GM) + k + r) * dst_stride +
+                          f * block_len + CRC_LEN + r0;
+#pragma unroll
+          for (int i = 0; i < 4; i++) {
+            const int64_t off = int64_t(i) * 4096 + lane16;
+            if (off + 16 <= rbytes) {
+              uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+              dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
+              dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
+              *reinterpret_cast<uint4 *>(
+                  &stg[(off / EF_CHUNK) * EF_STRIDE +
+                       (off & (EF_CHUNK - 1))]) = acc[r][i];
+            }
+          }
+          if (rbytes < EF_PASS && threadIdx.x == 0) {
+            const int64_t t0 = (rbytes / 16) * 16;
+            for (int64_t p = t0; p < rbytes; p++) {
+              uint8_t pv = 0;
+              for (int c2 = 0; c2 < k; c2++) {
+                const uint8_t b = sbase[size_t(c2) * shard_len + p0 + r0 + p];
+                const uint8_t *t = ctab + size_t(r * k + c2) * 32;
+                pv ^= t[b & 0x0F] ^ t[16 + (b >> 4)];
+              }
+              fdst[p] = pv;
+              stg[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))] = pv;
+            }
           }
         }
         __syncthreads();
-        const int64_t c0b = int64_t(threadIdx.x) * EF_CHUNK;
-        int clen = int(i64min(int64_t(EF_CHUNK), rbytes - c0b));
-        if (clen < 0) clen = 0;
-        uint32_t part =
-            crc_chunk16(stage + threadIdx.x * EF_STRIDE, clen, tab);
-        uint32_t op = op_full[h];
-        if (payload != payload_full) {
-          const int64_t suffix = clen > 0 ? payload - (r0 + c0b + clen) : 0;
-          op = x8n_d(uint64_t(suffix));
+        /* CRC the PREVIOUS unit's staged range while unit u+1 streams */
+        if (u > 0 && clen > 0) {
+          const uint8_t *pstg = stage + ((u - 1) & 1) * EF_STG_ONE;
+          const uint32_t part =
+              crc_chunk16(pstg + threadIdx.x * EF_STRIDE, clen, tab);
+          crcacc[u - 1] ^= gf2_mulmod_d(op, part);
         }
-        crcacc[k + r] ^= clen > 0 ? gf2_mulmod_d(op, part) : 0;
-        __syncthreads();
+      }
+      __syncthreads();
+      if (clen > 0) {
+        const uint8_t *pstg = stage + ((k + GM - 1) & 1) * EF_STG_ONE;
+        const uint32_t part =
+            crc_chunk16(pstg + threadIdx.x * EF_STRIDE, clen, tab);
+        crcacc[k + GM - 1] ^= gf2_mulmod_d(op, part);
+      }
+      __syncthreads();
+    }
+
+    /* ---- reduce all shard CRCs at once, write the 4 B LE headers ---- */
+#pragma unroll
+    for (int j = 0; j < 16; j++) {
+      if (j < k + GM) {
+#pragma unroll
+        for (int sh = 32; sh > 0; sh >>= 1)
+          crcacc[j] ^= __shfl_xor(crcacc[j], sh, 64);
       }
     }
-    /* ---- reduce per-shard CRCs, write the 4 B LE headers ---- */
-    for (int j = 0; j < k + GM; j++) {
-      uint32_t part = crcacc[j];
-#pragma unroll
-      for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
-      if ((threadIdx.x & 63) == 0)
-        reinterpret_cast<uint32_t *>(stage)[threadIdx.x >> 6] = part;
-      __syncthreads();
-      if (threadIdx.x == 0) {
-        const uint32_t *fw = reinterpret_cast<uint32_t *>(stage);
-        const uint32_t it =
-            gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
-        const uint32_t crc = ~(it ^ fw[0] ^ fw[1] ^ fw[2] ^ fw[3]);
+    uint32_t *red = reinterpret_cast<uint32_t *>(stage);
+    if ((threadIdx.x & 63) == 0) {
+      const int w = threadIdx.x >> 6;
+      for (int j = 0; j < k + GM; j++) red[w * 16 + j] = crcacc[j];
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      const uint32_t it = gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
+      for (int j = 0; j < k + GM; j++) {
+        const uint32_t crc =
+            ~(it ^ red[j] ^ red[16 + j] ^ red[32 + j] ^ red[48 + j]);
         *reinterpret_cast<uint32_t *>(
             dst + (stripe * (k + GM) + j) * dst_stride + f * block_len) = crc;
       }
-      __syncthreads();
     }
+    __syncthreads();
   }
 }
 
