@@ -835,7 +835,9 @@ constexpr int EF_CHUNK = 64;    /* CRC chunk per lane per pass */
 constexpr int EF_STRIDE = EF_CHUNK + 16;
 constexpr int EF_PASSES = 4;    /* ceil(65532 / 16384) */
 constexpr int EF_STG_ONE = 256 * EF_STRIDE; /* one stage buffer (20 KB) */
-constexpr int EF_LDS = 4096 + 2 * EF_STG_ONE; /* tables + double stage */
+/* tables | per-frame CRC reduction slab (4 waves x 16 shards) | 2 stages */
+constexpr int EF_RED = 4 * 16 * 4;
+constexpr int EF_LDS = 4096 + EF_RED + 2 * EF_STG_ONE;
 
 template <int GM>
 __global__ __launch_bounds__(CRC_BLOCKT) void rs_encode_frame_k(
@@ -847,8 +849,8 @@ __global__ __launch_bounds__(CRC_BLOCKT) void rs_encode_frame_k(
   constexpr int64_t payload_full = block_len - CRC_LEN;
   extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
   uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
-  uint8_t *stage = smem + 4096;
-  uint4 *ltab_v = reinterpret_cast<uint4 *>(smem + 4096); /* transient */
+  uint32_t *red = reinterpret_cast<uint32_t *>(smem + 4096);
+  uint8_t *stage = smem + 4096 + EF_RED;
   /* coefficient tables live in registers: GM*k*32 B is too much for
    * k>4, so re-read from global per MAC via __ldg-style loads would be
    * slow; instead keep them in LDS *before* the stage area is used —
@@ -856,13 +858,13 @@ __global__ __launch_bounds__(CRC_BLOCKT) void rs_encode_frame_k(
    * stage is consumed in the CRC phase, so they can share space only if
    * reloaded per pass.  Simpler: put them in the tail of the table area
    * is impossible (4 KB exactly) — so carve GM*k*32 extra after stage. */
-  uint8_t *ctab = smem + 4096 + 256 * EF_STRIDE;
+  uint8_t *ctab = smem + EF_LDS - 0 + 0; /* carved after both stages */
+  ctab = smem + 4096 + EF_RED + 2 * EF_STG_ONE;
   for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
     (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
   for (int i = threadIdx.x; i < GM * k * 2; i += CRC_BLOCKT)
     reinterpret_cast<uint4 *>(ctab)[i] =
         reinterpret_cast<const uint4 *>(tabs)[i];
-  (void)ltab_v;
   /* per-(pass,lane) fold operator for full frames */
   uint32_t op_full[EF_PASSES];
 #pragma unroll
@@ -883,10 +885,11 @@ __global__ __launch_bounds__(CRC_BLOCKT) void rs_encode_frame_k(
     const uint8_t *sbase =
         reinterpret_cast<const uint8_t *>(base + stripe * stripe_stride);
 
-    uint32_t crcacc[16]; /* fused fast path requires k+GM <= 16 */
-#pragma unroll
-    for (int j = 0; j < 16; j++) crcacc[j] = 0;
+    /* per-frame CRC partials live in the LDS red slab (4 waves x 16
+     * shard slots): no runtime-indexed per-lane array -> no scratch */
+    for (int j = threadIdx.x; j < 64; j += CRC_BLOCKT) red[j] = 0;
     uint4 acc[GM][4];
+    __syncthreads();
 
     /* Software pipeline over the k+GM shard units of each pass: unit u's
      * range is written into stage[u&1] while unit u-1's chunks are CRC'd
@@ -911,104 +914,94 @@ __global__ __launch_bounds__(CRC_BLOCKT) void rs_encode_frame_k(
         op = x8n_d(uint64_t(suffix));
       }
 
-      for (int u = 0; u < k + GM; u++) {
-        uint8_t *stg = stage + (u & 1) * EF_STG_ONE;
-        if (u < k) {
-          /* data shard u: load, MAC, framed-write, stage */
-          const int c = u;
-          const uint8_t *src = sbase + size_t(c) * shard_len + p0 + r0;
-          uint8_t *fdst = dst + (stripe * (k + GM) + c) * dst_stride +
-                          f * block_len + CRC_LEN + r0;
+      /* helper: CRC the previous unit's staged buffer, fold, reduce */
+      auto crc_prev = [&](int unit) {
+        if (unit >= 0 && clen > 0) {
+          const uint8_t *pstg = stage + (unit & 1) * EF_STG_ONE;
+          uint32_t part =
+              gf2_mulmod_d(op,
+                           crc_chunk16(pstg + threadIdx.x * EF_STRIDE, clen,
+                                       tab));
 #pragma unroll
-          for (int i = 0; i < 4; i++) {
-            const int64_t off = int64_t(i) * 4096 + lane16;
-            if (off + 16 <= rbytes) {
-              const uint4 v = *reinterpret_cast<const uint4 *>(src + off);
+          for (int sh = 32; sh > 0; sh >>= 1)
+            part ^= __shfl_xor(part, sh, 64);
+          if ((threadIdx.x & 63) == 0)
+            red[(threadIdx.x >> 6) * 16 + unit] ^= part;
+        }
+      };
+
+      for (int c = 0; c < k; c++) {
+        uint8_t *stg = stage + (c & 1) * EF_STG_ONE;
+        const uint8_t *src = sbase + size_t(c) * shard_len + p0 + r0;
+        uint8_t *fdst = dst + (stripe * (k + GM) + c) * dst_stride +
+                        f * block_len + CRC_LEN + r0;
 #pragma unroll
-              for (int r = 0; r < GM; r++) {
-                const int t2 = (r * k + c) * 2;
-                gfmac16<false>(acc[r][i], v, ltab[t2], ltab[t2 + 1]);
-              }
-              uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
-              dw[0] = v.x; dw[1] = v.y; dw[2] = v.z; dw[3] = v.w;
-              *reinterpret_cast<uint4 *>(
-                  &stg[(off / EF_CHUNK) * EF_STRIDE +
-                       (off & (EF_CHUNK - 1))]) = v;
-            }
-          }
-          if (rbytes < EF_PASS && threadIdx.x == 0) {
-            const int64_t t0 = (rbytes / 16) * 16;
-            for (int64_t p = t0; p < rbytes; p++) {
-              const uint8_t x = src[p];
-              fdst[p] = x;
-              stg[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))] = x;
-            }
-          }
-        } else {
-          /* parity shard u-k: framed-write the accumulator, stage it */
-          const int r = u - k;
-          uint8_t *fdst = dst + (stripe * (k + GM) + k + r) * dst_stride +
-                          f * block_len + CRC_LEN + r0;
+        for (int i = 0; i < 4; i++) {
+          const int64_t off = int64_t(i) * 4096 + lane16;
+          if (off + 16 <= rbytes) {
+            const uint4 v = *reinterpret_cast<const uint4 *>(src + off);
 #pragma unroll
-          for (int i = 0; i < 4; i++) {
-            const int64_t off = int64_t(i) * 4096 + lane16;
-            if (off + 16 <= rbytes) {
-              uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
-              dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
-              dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
-              *reinterpret_cast<uint4 *>(
-                  &stg[(off / EF_CHUNK) * EF_STRIDE +
-                       (off & (EF_CHUNK - 1))]) = acc[r][i];
+            for (int r = 0; r < GM; r++) {
+              const int t2 = (r * k + c) * 2;
+              gfmac16<false>(acc[r][i], v, ltab[t2], ltab[t2 + 1]);
             }
-          }
-          if (rbytes < EF_PASS && threadIdx.x == 0) {
-            const int64_t t0 = (rbytes / 16) * 16;
-            for (int64_t p = t0; p < rbytes; p++) {
-              uint8_t pv = 0;
-              for (int c2 = 0; c2 < k; c2++) {
-                const uint8_t b = sbase[size_t(c2) * shard_len + p0 + r0 + p];
-                const uint8_t *t = ctab + size_t(r * k + c2) * 32;
-                pv ^= t[b & 0x0F] ^ t[16 + (b >> 4)];
-              }
-              fdst[p] = pv;
-              stg[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))] = pv;
-            }
+            uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+            dw[0] = v.x; dw[1] = v.y; dw[2] = v.z; dw[3] = v.w;
+            *reinterpret_cast<uint4 *>(
+                &stg[(off / EF_CHUNK) * EF_STRIDE +
+                     (off & (EF_CHUNK - 1))]) = v;
           }
         }
+        if (rbytes < EF_PASS && threadIdx.x == 0) {
+          const int64_t t0 = (rbytes / 16) * 16;
+          for (int64_t p = t0; p < rbytes; p++) {
+            const uint8_t x = src[p];
+            fdst[p] = x;
+            stg[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))] = x;
+          }
+        }
+        /* overlap: CRC unit c-1 (other buffer) before the reuse barrier */
+        crc_prev(c - 1);
         __syncthreads();
-        /* CRC the PREVIOUS unit's staged range while unit u+1 streams */
-        if (u > 0 && clen > 0) {
-          const uint8_t *pstg = stage + ((u - 1) & 1) * EF_STG_ONE;
-          const uint32_t part =
-              crc_chunk16(pstg + threadIdx.x * EF_STRIDE, clen, tab);
-          crcacc[u - 1] ^= gf2_mulmod_d(op, part);
+      }
+#pragma unroll
+      for (int r = 0; r < GM; r++) {
+        uint8_t *stg = stage + ((k + r) & 1) * EF_STG_ONE;
+        uint8_t *fdst = dst + (stripe * (k + GM) + k + r) * dst_stride +
+                        f * block_len + CRC_LEN + r0;
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+          const int64_t off = int64_t(i) * 4096 + lane16;
+          if (off + 16 <= rbytes) {
+            uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+            dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
+            dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
+            *reinterpret_cast<uint4 *>(
+                &stg[(off / EF_CHUNK) * EF_STRIDE +
+                     (off & (EF_CHUNK - 1))]) = acc[r][i];
+          }
         }
+        if (rbytes < EF_PASS && threadIdx.x == 0) {
+          const int64_t t0 = (rbytes / 16) * 16;
+          for (int64_t p = t0; p < rbytes; p++) {
+            uint8_t pv = 0;
+            for (int c2 = 0; c2 < k; c2++) {
+              const uint8_t b = sbase[size_t(c2) * shard_len + p0 + r0 + p];
+              const uint8_t *t = ctab + size_t(r * k + c2) * 32;
+              pv ^= t[b & 0x0F] ^ t[16 + (b >> 4)];
+            }
+            fdst[p] = pv;
+            stg[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))] = pv;
+          }
+        }
+        crc_prev(k + r - 1);
+        __syncthreads();
       }
-      __syncthreads();
-      if (clen > 0) {
-        const uint8_t *pstg = stage + ((k + GM - 1) & 1) * EF_STG_ONE;
-        const uint32_t part =
-            crc_chunk16(pstg + threadIdx.x * EF_STRIDE, clen, tab);
-        crcacc[k + GM - 1] ^= gf2_mulmod_d(op, part);
-      }
+      crc_prev(k + GM - 1);
       __syncthreads();
     }
 
-    /* ---- reduce all shard CRCs at once, write the 4 B LE headers ---- */
-#pragma unroll
-    for (int j = 0; j < 16; j++) {
-      if (j < k + GM) {
-#pragma unroll
-        for (int sh = 32; sh > 0; sh >>= 1)
-          crcacc[j] ^= __shfl_xor(crcacc[j], sh, 64);
-      }
-    }
-    uint32_t *red = reinterpret_cast<uint32_t *>(stage);
-    if ((threadIdx.x & 63) == 0) {
-      const int w = threadIdx.x >> 6;
-      for (int j = 0; j < k + GM; j++) red[w * 16 + j] = crcacc[j];
-    }
-    __syncthreads();
+    /* ---- combine wave partials, write the 4 B LE headers ---- */
     if (threadIdx.x == 0) {
       const uint32_t it = gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
       for (int j = 0; j < k + GM; j++) {
