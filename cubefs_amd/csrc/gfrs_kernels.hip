@@ -839,8 +839,8 @@ constexpr int EF_STG_ONE = 256 * EF_STRIDE; /* one stage buffer (20 KB) */
 constexpr int EF_RED = 4 * 16 * 4;
 constexpr int EF_LDS = 4096 + EF_RED + 2 * EF_STG_ONE;
 
-template <int GM>
-__global__ __launch_bounds__(CRC_BLOCKT) void rs_encode_frame_k(
+template <int GM, int NBUF, int WPS>
+__global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
     uint8_t *__restrict__ dst, size_t dst_stride /* framed image stride */,
     uint64_t base, uint64_t stripe_stride, size_t shard_len, int k,
     const uint8_t *__restrict__ tabs /* [GM*k][32] */, int64_t total_frames,
@@ -917,7 +917,8 @@ __global__ __launch_bounds__(CRC_BLOCKT) void rs_encode_frame_k(
       /* helper: CRC the previous unit's staged buffer, fold, reduce */
       auto crc_prev = [&](int unit) {
         if (unit >= 0 && clen > 0) {
-          const uint8_t *pstg = stage + (unit & 1) * EF_STG_ONE;
+          const uint8_t *pstg =
+              stage + (NBUF == 2 ? (unit & 1) : 0) * EF_STG_ONE;
           uint32_t part =
               gf2_mulmod_d(op,
                            crc_chunk16(pstg + threadIdx.x * EF_STRIDE, clen,
@@ -931,7 +932,7 @@ __global__ __launch_bounds__(CRC_BLOCKT) void rs_encode_frame_k(
       };
 
       for (int c = 0; c < k; c++) {
-        uint8_t *stg = stage + (c & 1) * EF_STG_ONE;
+        uint8_t *stg = stage + (NBUF == 2 ? (c & 1) : 0) * EF_STG_ONE;
         const uint8_t *src = sbase + size_t(c) * shard_len + p0 + r0;
         uint8_t *fdst = dst + (stripe * (k + GM) + c) * dst_stride +
                         f * block_len + CRC_LEN + r0;
@@ -960,13 +961,19 @@ __global__ __launch_bounds__(CRC_BLOCKT) void rs_encode_frame_k(
             stg[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))] = x;
           }
         }
-        /* overlap: CRC unit c-1 (other buffer) before the reuse barrier */
-        crc_prev(c - 1);
-        __syncthreads();
+        if (NBUF == 2) {
+          /* overlap: CRC unit c-1 (other buffer) before the reuse barrier */
+          crc_prev(c - 1);
+          __syncthreads();
+        } else {
+          __syncthreads();
+          crc_prev(c);
+          __syncthreads();
+        }
       }
 #pragma unroll
       for (int r = 0; r < GM; r++) {
-        uint8_t *stg = stage + ((k + r) & 1) * EF_STG_ONE;
+        uint8_t *stg = stage + (NBUF == 2 ? ((k + r) & 1) : 0) * EF_STG_ONE;
         uint8_t *fdst = dst + (stripe * (k + GM) + k + r) * dst_stride +
                         f * block_len + CRC_LEN + r0;
 #pragma unroll
@@ -994,11 +1001,19 @@ __global__ __launch_bounds__(CRC_BLOCKT) void rs_encode_frame_k(
             stg[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))] = pv;
           }
         }
-        crc_prev(k + r - 1);
+        if (NBUF == 2) {
+          crc_prev(k + r - 1);
+          __syncthreads();
+        } else {
+          __syncthreads();
+          crc_prev(k + r);
+          __syncthreads();
+        }
+      }
+      if (NBUF == 2) {
+        crc_prev(k + GM - 1);
         __syncthreads();
       }
-      crc_prev(k + GM - 1);
-      __syncthreads();
     }
 
     /* ---- combine wave partials, write the 4 B LE headers ---- */
@@ -1022,17 +1037,31 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
   const int64_t fps = (int64_t(shard_len) + 65531) / 65532;
   const int64_t total = fps * nstripes;
   const int grid = crc_grid(total);
-  const int lds = EF_LDS + m * k * 32;
-#define GFRS_EF_GO(G)                                                     \
-  hipLaunchKernelGGL((rs_encode_frame_k<G>), dim3(grid), dim3(CRC_BLOCKT),\
-                     lds, s, dst, dst_stride, base, stripe_stride,        \
-                     shard_len, k, tabs, total, fps)
-  switch (m) {
-    case 1: GFRS_EF_GO(1); break;
-    case 2: GFRS_EF_GO(2); break;
-    case 3: GFRS_EF_GO(3); break;
-    default: GFRS_EF_GO(4);
+  /* occupancy variant: GFRS_EF = NBUF*10 + waves-per-SIMD bound
+   * (23 = double stage, 3 w/SIMD — measured best so far) */
+  static const int var = []() {
+    const char *e = getenv("GFRS_EF");
+    const int v = e ? atoi(e) : 23;
+    return (v == 13 || v == 14 || v == 23 || v == 24) ? v : 23;
+  }();
+  const int nbuf = var / 10;
+  const int lds = 4096 + EF_RED + nbuf * EF_STG_ONE + m * k * 32;
+#define GFRS_EF_GO(G, NB, W)                                              \
+  hipLaunchKernelGGL((rs_encode_frame_k<G, NB, W>), dim3(grid),           \
+                     dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,     \
+                     stripe_stride, shard_len, k, tabs, total, fps)
+#define GFRS_EF_SW(NB, W)                                                 \
+  switch (m) {                                                            \
+    case 1: GFRS_EF_GO(1, NB, W); break;                                  \
+    case 2: GFRS_EF_GO(2, NB, W); break;                                  \
+    case 3: GFRS_EF_GO(3, NB, W); break;                                  \
+    default: GFRS_EF_GO(4, NB, W);                                        \
   }
+  if (var == 13) { GFRS_EF_SW(1, 3) }
+  else if (var == 14) { GFRS_EF_SW(1, 4) }
+  else if (var == 24) { GFRS_EF_SW(2, 4) }
+  else { GFRS_EF_SW(2, 3) }
+#undef GFRS_EF_SW
 #undef GFRS_EF_GO
 }
 
