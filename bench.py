@@ -46,9 +46,10 @@ def parse_args():
                    help="shard index reconstructed in --workload reconstruct")
     p.add_argument("--no-crc", action="store_true",
                    help="EC encode only (config label form)")
-    p.add_argument("--fused", action="store_true",
-                   help="single-pass fused encode+frame kernel (correct but "
-                        "still slower than the two-kernel path; being tuned)")
+    p.add_argument("--no-fused", action="store_true",
+                   help="two-kernel encode-then-frame instead of the fused "
+                        "single-pass kernel (the fused path is faster and "
+                        "is the default)")
     p.add_argument("--cpu-sample-stripes", type=int, default=24)
     p.add_argument("--skip-cpu-baseline", action="store_true")
     return p.parse_args()
@@ -175,7 +176,7 @@ def main():
             codec.synchronize()
         bad = [args.bad_idx]
 
-    fused = (args.workload == "encode" and with_crc and args.fused)
+    fused = (args.workload == "encode" and with_crc and not args.no_fused)
 
     def step(events=None):
         if events:

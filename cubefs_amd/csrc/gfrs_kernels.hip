@@ -1037,12 +1037,14 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
   const int64_t fps = (int64_t(shard_len) + 65531) / 65532;
   const int64_t total = fps * nstripes;
   const int grid = crc_grid(total);
-  /* occupancy variant: GFRS_EF = NBUF*10 + waves-per-SIMD bound
-   * (23 = double stage, 3 w/SIMD — measured best so far) */
+  /* occupancy variant: GFRS_EF = NBUF*10 + waves-per-SIMD bound.
+   * Measured @256 stripes RS(6+3): 14 -> 13.7 ms (single 20 KB stage,
+   * 4 blocks/CU even with a 40 B spill) beats 23 (15.3), 13 (15.4),
+   * 24 (16.5). */
   static const int var = []() {
     const char *e = getenv("GFRS_EF");
-    const int v = e ? atoi(e) : 23;
-    return (v == 13 || v == 14 || v == 23 || v == 24) ? v : 23;
+    const int v = e ? atoi(e) : 14;
+    return (v == 13 || v == 14 || v == 23 || v == 24) ? v : 14;
   }();
   const int nbuf = var / 10;
   const int lds = 4096 + EF_RED + nbuf * EF_STG_ONE + m * k * 32;
