@@ -15,7 +15,7 @@ import io
 
 import numpy as np
 
-from . import codemode, runtime
+from . import codemode
 from .runtime import GfrsError, MEM_DEVICE, MEM_HOST, Tactic, check, lib
 
 

@@ -147,7 +147,7 @@ def test_replicate_mode(dev):
     """Replica3 tactic: Encode no-op, Verify true, Reconstruct only when
     nothing is missing (reedsolomon.go:442,784)."""
     from cubefs_amd import codemode
-    from cubefs_amd.runtime import GfrsError, Tactic, lib
+    from cubefs_amd.runtime import Tactic, lib
     import ctypes
     t = Tactic(3, 0, 0, 3, 3, 0, 0)
     ctx = lib().gfrs_create(ctypes.byref(t), -1)
