@@ -858,8 +858,8 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
    * stage is consumed in the CRC phase, so they can share space only if
    * reloaded per pass.  Simpler: put them in the tail of the table area
    * is impossible (4 KB exactly) — so carve GM*k*32 extra after stage. */
-  uint8_t *ctab = smem + EF_LDS - 0 + 0; /* carved after both stages */
-  ctab = smem + 4096 + EF_RED + 2 * EF_STG_ONE;
+  /* coefficient tables, carved after the NBUF stage buffer(s) */
+  uint8_t *ctab = smem + 4096 + EF_RED + NBUF * EF_STG_ONE;
   for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
     (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
   for (int i = threadIdx.x; i < GM * k * 2; i += CRC_BLOCKT)
