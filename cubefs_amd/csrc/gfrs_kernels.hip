@@ -414,9 +414,9 @@ static bool perm0_ok() {
 #define CRC_POLY 0xEDB88320u
 #define CRC_LEN 4
 
-/* slice-by-4 tables, computed on host (gfrs_host.cpp) and copied to this
- * symbol per device. */
-__device__ uint32_t g_crc_tab4[4][256];
+/* slice-by-8 tables (zlib BYFOUR construction extended), host-computed,
+ * copied per device.  tab[0] is the base byte table. */
+__device__ uint32_t g_crc_tab4[8][256];
 
 /* x^(8*2^j) mod P, reflected domain — host-filled alongside the tables. */
 __device__ uint32_t g_pow8[40];
@@ -476,16 +476,20 @@ GFRS_DEV uint32_t crc_chunk(const uint8_t *p, int len,
  * LDS instructions than the byte-pointer path. */
 GFRS_DEV uint32_t crc_chunk16(const uint8_t *p, int len,
                               const uint32_t (*tab)[256]) {
+  /* slice-by-8: one chained step per 8 B (the 8 gathers within a step are
+   * independent, so the serial chain is half as deep as slice-by-4) */
   uint32_t c = 0;
   int i = 0;
   for (; i + 16 <= len; i += 16) {
     const uint4 q = *reinterpret_cast<const uint4 *>(p + i);
 #pragma unroll
-    for (int j = 0; j < 4; j++) {
-      const uint32_t w = j == 0 ? q.x : j == 1 ? q.y : j == 2 ? q.z : q.w;
-      c ^= w;
-      c = tab[3][c & 0xFF] ^ tab[2][(c >> 8) & 0xFF] ^
-          tab[1][(c >> 16) & 0xFF] ^ tab[0][c >> 24];
+    for (int j = 0; j < 2; j++) {
+      const uint32_t w0 = (j == 0 ? q.x : q.z) ^ c;
+      const uint32_t w1 = (j == 0 ? q.y : q.w);
+      c = tab[7][w0 & 0xFF] ^ tab[6][(w0 >> 8) & 0xFF] ^
+          tab[5][(w0 >> 16) & 0xFF] ^ tab[4][w0 >> 24] ^
+          tab[3][w1 & 0xFF] ^ tab[2][(w1 >> 8) & 0xFF] ^
+          tab[1][(w1 >> 16) & 0xFF] ^ tab[0][w1 >> 24];
     }
   }
   for (; i + 4 <= len; i += 4) {
@@ -505,9 +509,9 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_k(
     const uint8_t *__restrict__ src, size_t src_stride, int64_t n,
     int64_t block_len, int64_t frames_per_shard, int64_t total_frames,
     int64_t *__restrict__ bad) {
-  __shared__ uint32_t tab[4][256];
+  __shared__ uint32_t tab[8][256];
   __shared__ uint32_t fold[CRC_BLOCKT];
-  for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
+  for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
     (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
   __syncthreads();
 
@@ -610,7 +614,7 @@ template <int CHUNK> struct StgGeom {
   static constexpr int STRIDE = CHUNK + 16;
   static constexpr int HALF = 256 * CHUNK; /* payload bytes per pass */
   static constexpr int PASSES = (65532 + HALF - 1) / HALF;
-  static constexpr int LDS = 4096 + 1024 + 256 * STRIDE;
+  static constexpr int LDS = 8192 + 1024 + 256 * STRIDE;
 };
 
 template <int MODE, bool NT, bool TAILCRC, int CHUNK>
@@ -627,9 +631,9 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
   constexpr int STG_PASSES = G::PASSES;
   extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
   uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
-  uint32_t *fold = reinterpret_cast<uint32_t *>(smem + 4096);
-  uint8_t *stage = smem + 4096 + 1024; /* 256 chunk slots x STRIDE B */
-  for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
+  uint32_t *fold = reinterpret_cast<uint32_t *>(smem + 8192);
+  uint8_t *stage = smem + 8192 + 1024; /* 256 chunk slots x STRIDE B */
+  for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
     (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
   /* fold operators are a function of the thread's chunk positions only
    * for full frames — compute x^(8*suffix) once per block, not per frame */
@@ -837,7 +841,7 @@ constexpr int EF_PASSES = 4;    /* ceil(65532 / 16384) */
 constexpr int EF_STG_ONE = 256 * EF_STRIDE; /* one stage buffer (20 KB) */
 /* tables | per-frame CRC reduction slab (4 waves x 16 shards) | 2 stages */
 constexpr int EF_RED = 4 * 16 * 4;
-constexpr int EF_LDS = 4096 + EF_RED + 2 * EF_STG_ONE;
+constexpr int EF_LDS = 8192 + EF_RED + 2 * EF_STG_ONE;
 
 template <int GM, int NBUF, int WPS>
 __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
@@ -849,8 +853,8 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
   constexpr int64_t payload_full = block_len - CRC_LEN;
   extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
   uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
-  uint32_t *red = reinterpret_cast<uint32_t *>(smem + 4096);
-  uint8_t *stage = smem + 4096 + EF_RED;
+  uint32_t *red = reinterpret_cast<uint32_t *>(smem + 8192);
+  uint8_t *stage = smem + 8192 + EF_RED;
   /* coefficient tables live in registers: GM*k*32 B is too much for
    * k>4, so re-read from global per MAC via __ldg-style loads would be
    * slow; instead keep them in LDS *before* the stage area is used —
@@ -859,8 +863,8 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
    * reloaded per pass.  Simpler: put them in the tail of the table area
    * is impossible (4 KB exactly) — so carve GM*k*32 extra after stage. */
   /* coefficient tables, carved after the NBUF stage buffer(s) */
-  uint8_t *ctab = smem + 4096 + EF_RED + NBUF * EF_STG_ONE;
-  for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
+  uint8_t *ctab = smem + 8192 + EF_RED + NBUF * EF_STG_ONE;
+  for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
     (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
   for (int i = threadIdx.x; i < GM * k * 2; i += CRC_BLOCKT)
     reinterpret_cast<uint4 *>(ctab)[i] =
@@ -1047,7 +1051,7 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
     return (v == 13 || v == 14 || v == 23 || v == 24) ? v : 14;
   }();
   const int nbuf = var / 10;
-  const int lds = 4096 + EF_RED + nbuf * EF_STG_ONE + m * k * 32;
+  const int lds = 8192 + EF_RED + nbuf * EF_STG_ONE + m * k * 32;
 #define GFRS_EF_GO(G, NB, W)                                              \
   hipLaunchKernelGGL((rs_encode_frame_k<G, NB, W>), dim3(grid),           \
                      dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,     \
@@ -1191,13 +1195,13 @@ void launch_shard_parse(const uint8_t *img, size_t stride, int64_t raw_size,
 /* Host-side one-time init of the device CRC tables (g_crc_tab4, g_pow8)
  * for the CURRENT device.  Called by gfrs_host.cpp under its device mutex. */
 int crc_device_init_current(void) {
-  uint32_t tab[4][256];
+  uint32_t tab[8][256];
   for (uint32_t i = 0; i < 256; i++) {
     uint32_t c = i;
     for (int j = 0; j < 8; j++) c = (c & 1) ? (c >> 1) ^ CRC_POLY : c >> 1;
     tab[0][i] = c;
   }
-  for (int t = 1; t < 4; t++)
+  for (int t = 1; t < 8; t++)
     for (int i = 0; i < 256; i++)
       tab[t][i] = tab[0][tab[t - 1][i] & 0xFF] ^ (tab[t - 1][i] >> 8);
   /* x^(8*2^j) mod P, reflected (identity = 0x80000000) */
