@@ -843,7 +843,7 @@ constexpr int EF_STG_ONE = 256 * EF_STRIDE; /* one stage buffer (20 KB) */
 constexpr int EF_RED = 4 * 16 * 4;
 constexpr int EF_LDS = 8192 + EF_RED + 2 * EF_STG_ONE;
 
-template <int GM, int NBUF, int WPS>
+template <int GM, int NBUF, int WPS, int ABL = 0>
 __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
     uint8_t *__restrict__ dst, size_t dst_stride /* framed image stride */,
     uint64_t base, uint64_t stripe_stride, size_t shard_len, int k,
@@ -920,6 +920,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
 
       /* helper: CRC the previous unit's staged buffer, fold, reduce */
       auto crc_prev = [&](int unit) {
+        if (ABL == 1) return;
         if (unit >= 0 && clen > 0) {
           const uint8_t *pstg =
               stage + (NBUF == 2 ? (unit & 1) : 0) * EF_STG_ONE;
@@ -945,13 +946,17 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
           const int64_t off = int64_t(i) * 4096 + lane16;
           if (off + 16 <= rbytes) {
             const uint4 v = *reinterpret_cast<const uint4 *>(src + off);
+            if (ABL != 2) {
 #pragma unroll
-            for (int r = 0; r < GM; r++) {
-              const int t2 = (r * k + c) * 2;
-              gfmac16<false>(acc[r][i], v, ltab[t2], ltab[t2 + 1]);
+              for (int r = 0; r < GM; r++) {
+                const int t2 = (r * k + c) * 2;
+                gfmac16<false>(acc[r][i], v, ltab[t2], ltab[t2 + 1]);
+              }
             }
-            uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
-            dw[0] = v.x; dw[1] = v.y; dw[2] = v.z; dw[3] = v.w;
+            if (ABL != 3) {
+              uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+              dw[0] = v.x; dw[1] = v.y; dw[2] = v.z; dw[3] = v.w;
+            }
             *reinterpret_cast<uint4 *>(
                 &stg[(off / EF_CHUNK) * EF_STRIDE +
                      (off & (EF_CHUNK - 1))]) = v;
@@ -984,9 +989,11 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
         for (int i = 0; i < 4; i++) {
           const int64_t off = int64_t(i) * 4096 + lane16;
           if (off + 16 <= rbytes) {
-            uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
-            dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
-            dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
+            if (ABL != 3) {
+              uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+              dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
+              dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
+            }
             *reinterpret_cast<uint4 *>(
                 &stg[(off / EF_CHUNK) * EF_STRIDE +
                      (off & (EF_CHUNK - 1))]) = acc[r][i];
@@ -1062,6 +1069,26 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
     case 2: GFRS_EF_GO(2, NB, W); break;                                  \
     case 3: GFRS_EF_GO(3, NB, W); break;                                  \
     default: GFRS_EF_GO(4, NB, W);                                        \
+  }
+  static const int abl = []() {
+    const char *e = getenv("GFRS_EF_ABL");
+    return e ? atoi(e) : 0;
+  }();
+  if (abl == 1) {
+    switch (m) { case 3: hipLaunchKernelGGL((rs_encode_frame_k<3, 1, 4, 1>),
+        dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+        stripe_stride, shard_len, k, tabs, total, fps); return;
+      default: break; }
+  } else if (abl == 2) {
+    switch (m) { case 3: hipLaunchKernelGGL((rs_encode_frame_k<3, 1, 4, 2>),
+        dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+        stripe_stride, shard_len, k, tabs, total, fps); return;
+      default: break; }
+  } else if (abl == 3) {
+    switch (m) { case 3: hipLaunchKernelGGL((rs_encode_frame_k<3, 1, 4, 3>),
+        dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+        stripe_stride, shard_len, k, tabs, total, fps); return;
+      default: break; }
   }
   if (var == 13) { GFRS_EF_SW(1, 3) }
   else if (var == 14) { GFRS_EF_SW(1, 4) }
