@@ -501,25 +501,6 @@ GFRS_DEV uint32_t crc_chunk16(const uint8_t *p, int len,
   return c;
 }
 
-/* Full 64-B chunk CRC from four uint4s held in registers (slice-by-8,
- * fully unrolled, no addressable array -> no scratch). */
-GFRS_DEV uint32_t crc_regs_full(const uint4 a, const uint4 b, const uint4 e,
-                                const uint4 d, const uint32_t (*tab)[256]) {
-  uint32_t c = 0;
-#define GFRS_S8(w0v, w1v)                                                   \
-  {                                                                         \
-    const uint32_t w0 = (w0v) ^ c, w1 = (w1v);                              \
-    c = tab[7][w0 & 0xFF] ^ tab[6][(w0 >> 8) & 0xFF] ^                      \
-        tab[5][(w0 >> 16) & 0xFF] ^ tab[4][w0 >> 24] ^ tab[3][w1 & 0xFF] ^  \
-        tab[2][(w1 >> 8) & 0xFF] ^ tab[1][(w1 >> 16) & 0xFF] ^              \
-        tab[0][w1 >> 24];                                                   \
-  }
-  GFRS_S8(a.x, a.y) GFRS_S8(a.z, a.w) GFRS_S8(b.x, b.y) GFRS_S8(b.z, b.w)
-  GFRS_S8(e.x, e.y) GFRS_S8(e.z, e.w) GFRS_S8(d.x, d.y) GFRS_S8(d.z, d.w)
-#undef GFRS_S8
-  return c;
-}
-
 /* One workgroup per frame.  MODE: 0 = encode (raw src -> framed dst),
  * 1 = verify (framed src), 2 = decode (framed src -> raw dst). */
 template <int MODE, bool TAILCRC = false>
@@ -937,46 +918,21 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
         op = x8n_d(uint64_t(suffix));
       }
 
-      /* CRC split into a cheap LDS fetch (between the two stage barriers)
-       * and the expensive compute (gathers + fold), which runs from
-       * registers OVERLAPPED with the next unit's global phase. */
-      uint4 ch0, ch1, ch2, ch3; /* separate vars: no addressable array */
-      int pend_unit = -1;
-      uint32_t pend_part = 0;
-      auto crc_fetch = [&](int unit) {
-        if (ABL == 1 || clen <= 0) return;
-        const uint8_t *pstg =
-            stage + (NBUF == 2 ? (unit & 1) : 0) * EF_STG_ONE;
-        const uint8_t *cp = pstg + threadIdx.x * EF_STRIDE;
-        if (clen == 64) {
-          ch0 = *reinterpret_cast<const uint4 *>(cp);
-          ch1 = *reinterpret_cast<const uint4 *>(cp + 16);
-          ch2 = *reinterpret_cast<const uint4 *>(cp + 32);
-          ch3 = *reinterpret_cast<const uint4 *>(cp + 48);
-        } else {
-          /* ragged chunk (last lane / tail frame): CRC now, straight from
-           * LDS — must complete before the stage is overwritten anyway */
-          pend_part = crc_chunk16(cp, clen, tab);
-        }
-        pend_unit = unit;
-      };
-      auto crc_compute = [&]() {
-        if (ABL == 1 || pend_unit < 0) return;
-        const uint32_t raw =
-            clen == 64 ? crc_regs_full(ch0, ch1, ch2, ch3, tab) : pend_part;
-        uint32_t part = gf2_mulmod_d(op, raw);
-#pragma unroll
-        for (int sh = 32; sh > 0; sh >>= 1)
-          part ^= __shfl_xor(part, sh, 64);
-        if ((threadIdx.x & 63) == 0)
-          red[(threadIdx.x >> 6) * 16 + pend_unit] ^= part;
-        pend_unit = -1;
-      };
-      auto crc_prev = [&](int unit) { /* legacy combined form (NBUF==2) */
+      /* helper: CRC the previous unit's staged buffer, fold, reduce */
+      auto crc_prev = [&](int unit) {
         if (ABL == 1) return;
         if (unit >= 0 && clen > 0) {
-          crc_fetch(unit);
-          crc_compute();
+          const uint8_t *pstg =
+              stage + (NBUF == 2 ? (unit & 1) : 0) * EF_STG_ONE;
+          uint32_t part =
+              gf2_mulmod_d(op,
+                           crc_chunk16(pstg + threadIdx.x * EF_STRIDE, clen,
+                                       tab));
+#pragma unroll
+          for (int sh = 32; sh > 0; sh >>= 1)
+            part ^= __shfl_xor(part, sh, 64);
+          if ((threadIdx.x & 63) == 0)
+            red[(threadIdx.x >> 6) * 16 + unit] ^= part;
         }
       };
 
@@ -1019,9 +975,8 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
           crc_prev(c - 1);
           __syncthreads();
         } else {
-          crc_compute(); /* unit c-1, from registers, overlapped */
           __syncthreads();
-          crc_fetch(c);
+          crc_prev(c);
           __syncthreads();
         }
       }
@@ -1061,17 +1016,14 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
           crc_prev(k + r - 1);
           __syncthreads();
         } else {
-          crc_compute();
           __syncthreads();
-          crc_fetch(k + r);
+          crc_prev(k + r);
           __syncthreads();
         }
       }
       if (NBUF == 2) {
         crc_prev(k + GM - 1);
         __syncthreads();
-      } else {
-        crc_compute(); /* last unit of the pass */
       }
     }
 
