@@ -1092,6 +1092,117 @@ int gfrs_encode_idx(gfrs_ctx *ctx, const void *data_shard, int idx,
   return GFRS_OK;
 }
 
+int gfrs_reconstruct_verify_batch(gfrs_ctx *ctx, void *base,
+                                  size_t shard_len, size_t stripe_stride,
+                                  int nstripes, const int32_t *bad_idx,
+                                  int nbad, uint64_t *fail_bitmap) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  const gfrs_tactic &t = c->t;
+  if (nstripes <= 0) return GFRS_ERR_INVALID_SHARDS;
+  if (t.l != 0) { /* LRC fallback: two passes */
+    int rc = gfrs_reconstruct_batch(ctx, base, shard_len, stripe_stride,
+                                    nstripes, bad_idx, nbad, 0);
+    if (rc != GFRS_OK) return rc;
+    return gfrs_verify_batch(ctx, base, shard_len, stripe_stride, nstripes,
+                             fail_bitmap);
+  }
+  const int k = t.n, m = t.m;
+  std::vector<uint8_t> present(k + m, 1);
+  for (int i = 0; i < nbad; i++) {
+    if (bad_idx[i] < 0 || bad_idx[i] >= k + m) return GFRS_ERR_INVALID_SHARDS;
+    present[bad_idx[i]] = 0;
+  }
+  std::lock_guard<std::mutex> lk(c->mu);
+  StreamGuard g(c);
+
+  uint64_t key = 0;
+  for (int i = 0; i < k + m; i++)
+    if (!present[i]) key |= 1ull << i;
+  key = key * 777767ull + 0xFECBull; /* reconstruct+verify namespace */
+  DevPlan *plan = nullptr;
+  uint32_t cmp_mask = 0;
+  {
+    auto it = c->dec_cache.find(key);
+    if (it != c->dec_cache.end()) plan = it->second;
+  }
+  if (!plan) {
+    /* valid inputs: first k present in index order (reedsolomon.go:1453) */
+    std::vector<int> valid;
+    for (int i = 0; i < k + m && int(valid.size()) < k; i++)
+      if (present[i]) valid.push_back(i);
+    if (int(valid.size()) < k) return GFRS_ERR_TOO_FEW_SHARDS;
+    std::vector<uint8_t> sub(size_t(k) * k), dec(size_t(k) * k);
+    for (int r = 0; r < k; r++)
+      memcpy(&sub[size_t(r) * k], &c->enc_matrix[size_t(valid[r]) * k], k);
+    if (!gf_invert(sub.data(), k, dec.data())) return GFRS_ERR_SINGULAR;
+    /* input slot of each present data column */
+    std::vector<int> slot(k, -1);
+    for (int j = 0; j < k; j++)
+      if (valid[j] < k) slot[valid[j]] = j;
+
+    const GfTables &gt2 = gft();
+    std::vector<int32_t> in, out;
+    std::vector<uint8_t> rows;
+    for (int j = 0; j < k; j++) in.push_back(valid[j]);
+    /* 1) missing data: decode rows (write) */
+    for (int i = 0; i < k; i++)
+      if (!present[i]) {
+        out.push_back(i);
+        rows.insert(rows.end(), &dec[size_t(i) * k], &dec[size_t(i) * k + k]);
+      }
+    /* 2) every parity row composed over the valid inputs:
+     *    ver_p[j] = enc_p[d]·1[slot(d)==j] (+) Σ_missing enc_p[d]·dec[d][j] */
+    for (int p2 = k; p2 < k + m; p2++) {
+      std::vector<uint8_t> row(k, 0);
+      for (int d = 0; d < k; d++) {
+        const uint8_t coef = c->enc_matrix[size_t(p2) * k + d];
+        if (present[d]) {
+          row[slot[d]] ^= coef;
+        } else {
+          for (int j = 0; j < k; j++)
+            row[j] ^= gt2.mul[coef][dec[size_t(d) * k + j]];
+        }
+      }
+      out.push_back(p2);
+      rows.insert(rows.end(), row.begin(), row.end());
+    }
+    plan = new DevPlan();
+    int rc = plan->upload(in, out, rows, c->stream);
+    if (rc != GFRS_OK) {
+      delete plan;
+      return rc;
+    }
+    c->dec_cache[key] = plan;
+  }
+  /* compare bits: present parity rows; missing (data or parity) written */
+  {
+    int nmissdata = 0;
+    for (int i = 0; i < k; i++)
+      if (!present[i]) nmissdata++;
+    for (int p2 = k; p2 < k + m; p2++)
+      if (present[p2]) cmp_mask |= 1u << (nmissdata + (p2 - k));
+  }
+  int rc;
+  if ((rc = c->fail_buf.ensure(size_t(nstripes) * 4)) != GFRS_OK) return rc;
+  HIP_TRY(hipMemsetAsync(c->fail_buf.p, 0, size_t(nstripes) * 4, c->stream));
+  launch_rs_apply_mixed_strided((uint64_t)base, stripe_stride,
+                                (const int32_t *)plan->in_idx.p, plan->k,
+                                (const int32_t *)plan->out_idx.p, plan->nout,
+                                (const uint8_t *)plan->tabs.p, cmp_mask,
+                                shard_len, nstripes,
+                                (uint32_t *)c->fail_buf.p, c->stream);
+  std::vector<uint32_t> fails(nstripes);
+  HIP_TRY(hipMemcpyAsync(fails.data(), c->fail_buf.p, size_t(nstripes) * 4,
+                         hipMemcpyDeviceToHost, c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  if (fail_bitmap) {
+    memset(fail_bitmap, 0, ((nstripes + 63) / 64) * 8);
+    for (int s2 = 0; s2 < nstripes; s2++)
+      if (fails[s2]) fail_bitmap[s2 / 64] |= 1ull << (s2 % 64);
+  }
+  return GFRS_OK;
+}
+
 int gfrs_update_idx(gfrs_ctx *ctx, const void *old_shard,
                     const void *new_shard, int idx, void *const *parity,
                     size_t shard_len, int nparity) {
@@ -1107,11 +1218,10 @@ int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
                       const uint64_t *bids, const uint64_t *vuids,
                       uint64_t *fail_bitmap) {
   if (nbad <= 0) return GFRS_ERR_INVALID_SHARDS;
-  int rc = gfrs_reconstruct_batch(ctx, base, shard_len, stripe_stride,
-                                  nstripes, bad_idx, nbad, 0);
-  if (rc != GFRS_OK) return rc;
-  rc = gfrs_verify_batch(ctx, base, shard_len, stripe_stride, nstripes,
-                         fail_bitmap);
+  /* reconstruct + the mandatory verify in one data pass */
+  int rc = gfrs_reconstruct_verify_batch(ctx, base, shard_len, stripe_stride,
+                                         nstripes, bad_idx, nbad,
+                                         fail_bitmap);
   if (rc != GFRS_OK) return rc;
   /* frame each repaired shard; images laid out (stripe, bad) row-major.
    * For bad shard b the raw source is strided across stripes.  The

@@ -91,6 +91,14 @@ void launch_sized_decode(uint8_t *dst, size_t dst_stride,
                          int64_t body_len, int64_t block_len, int nshards,
                          int64_t *bad, hipStream_t s);
 
+/* Mixed write/compare strided apply (reconstruct+verify in one pass). */
+void launch_rs_apply_mixed_strided(uint64_t base, uint64_t stripe_stride,
+                                   const int32_t *in_idx, int k,
+                                   const int32_t *out_idx, int nout,
+                                   const uint8_t *tabs, uint32_t cmp_mask,
+                                   size_t shard_len, int nstripes,
+                                   uint32_t *fail, hipStream_t s);
+
 /* EncodeIdx-style accumulate apply (reedsolomon.go:631-668):
  * out[r] ^= coeff[r]*in for one input shard. */
 void launch_rs_apply_xor(const uint64_t *ptrs, int nptr,

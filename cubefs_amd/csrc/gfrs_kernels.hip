@@ -133,7 +133,7 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
     ShardAddr addr, const int32_t *__restrict__ in_idx, int k,
     const int32_t *__restrict__ out_idx, int nout,
     const uint8_t *__restrict__ tabs, size_t shard_len, size_t nstripes,
-    uint32_t *fail, int xor_acc, int seq_map) {
+    uint32_t *fail, int xor_acc, int seq_map, uint32_t cmp_mask) {
   extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
   uint4 *ltab = reinterpret_cast<uint4 *>(smem); /* [k*nout*2] */
   const int ncoef = k * nout;
@@ -200,7 +200,7 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
       for (int r = 0; r < GM; r++) {
         uint8_t *out = const_cast<uint8_t *>(
             addr.shard(stripe, out_idx[r], shard_len));
-        if (VERIFY) {
+        if (VERIFY || (cmp_mask >> r) & 1) {
           const uint4 e = *reinterpret_cast<const uint4 *>(out + off);
           mismatch |= (e.x != acc[r].x) | (e.y != acc[r].y) |
                       (e.z != acc[r].z) | (e.w != acc[r].w);
@@ -223,14 +223,14 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
             const uint8_t *t = trow + size_t(c) * 32;
             v ^= t[b & 0x0F] ^ t[16 + (b >> 4)];
           }
-          if (VERIFY)
+          if (VERIFY || (cmp_mask >> og) & 1)
             mismatch |= (out[off + i] != v);
           else
             out[off + i] = v;
         }
       }
     }
-    if (VERIFY) {
+    if (VERIFY || cmp_mask) {
       if (__ballot(mismatch) != 0) {
         if ((threadIdx.x & 63) == 0) atomicOr(&fail[stripe], 1u);
       }
@@ -252,21 +252,52 @@ template <bool VERIFY, int GM>
 static void rs_launch_one(const ShardAddr &a, const int32_t *in_idx, int k,
                           const int32_t *out_idx, const uint8_t *tabs,
                           size_t shard_len, int nstripes, uint32_t *fail,
-                          hipStream_t s, int xor_acc = 0) {
+                          hipStream_t s, int xor_acc = 0,
+                          uint32_t cmp_mask = 0) {
   const int lds = k * GM * 32;
   const int grid = rs_grid(shard_len, nstripes);
   static const int seq = []() {
     const char *e = getenv("GFRS_RS_SEQ");
     return e ? atoi(e) : 0;
   }();
-  if (nt_enabled() && !VERIFY && !xor_acc)
+  if (nt_enabled() && !VERIFY && !xor_acc && !cmp_mask)
     hipLaunchKernelGGL((rs_apply_k<false, VERIFY, true, GM>), dim3(grid),
                        dim3(RS_BLOCK), lds, s, a, in_idx, k, out_idx, GM,
-                       tabs, shard_len, size_t(nstripes), fail, xor_acc, seq);
+                       tabs, shard_len, size_t(nstripes), fail, xor_acc, seq,
+                       cmp_mask);
   else
     hipLaunchKernelGGL((rs_apply_k<false, VERIFY, false, GM>), dim3(grid),
                        dim3(RS_BLOCK), lds, s, a, in_idx, k, out_idx, GM,
-                       tabs, shard_len, size_t(nstripes), fail, xor_acc, seq);
+                       tabs, shard_len, size_t(nstripes), fail, xor_acc, seq,
+                       cmp_mask);
+}
+
+/* Mixed write/compare strided apply: rows with cmp_mask bit set are
+ * verified against the stored shard (fail[stripe] ORed on mismatch),
+ * others written — one data pass for reconstruct+verify. */
+void launch_rs_apply_mixed_strided(uint64_t base, uint64_t stripe_stride,
+                                   const int32_t *in_idx, int k,
+                                   const int32_t *out_idx, int nout,
+                                   const uint8_t *tabs, uint32_t cmp_mask,
+                                   size_t shard_len, int nstripes,
+                                   uint32_t *fail, hipStream_t s) {
+  ShardAddr a{nullptr, base, stripe_stride, 0};
+  for (int og = 0; og < nout; og += MT) {
+    const int gm = nout - og < MT ? nout - og : MT;
+    const int32_t *oi = out_idx + og;
+    const uint8_t *tb = tabs + size_t(og) * k * 32;
+    const uint32_t m = (cmp_mask >> og) & ((1u << gm) - 1);
+    switch (gm) {
+      case 1: rs_launch_one<false, 1>(a, in_idx, k, oi, tb, shard_len,
+                                      nstripes, fail, s, 0, m); break;
+      case 2: rs_launch_one<false, 2>(a, in_idx, k, oi, tb, shard_len,
+                                      nstripes, fail, s, 0, m); break;
+      case 3: rs_launch_one<false, 3>(a, in_idx, k, oi, tb, shard_len,
+                                      nstripes, fail, s, 0, m); break;
+      default: rs_launch_one<false, 4>(a, in_idx, k, oi, tb, shard_len,
+                                       nstripes, fail, s, 0, m);
+    }
+  }
 }
 
 /* EncodeIdx-style accumulate apply: out[r] ^= coeff[r]*in (one input). */

@@ -242,6 +242,19 @@ class Encoder:
                                     new_shard.data_ptr(), idx, arr, ln,
                                     len(parity)), "update_idx")
 
+    def reconstruct_verify_batch(self, batch, bad_idx):
+        """Fused Reconstruct + mandatory Verify (worker_slice_recover.go:
+        865-874) in one data pass.  Returns the per-stripe verify-fail
+        list."""
+        p, ln, stride, ns = self._base(batch)
+        bad = (ctypes.c_int32 * max(1, len(bad_idx)))(*bad_idx)
+        nwords = (ns + 63) // 64
+        bm = (ctypes.c_uint64 * nwords)()
+        check(lib().gfrs_reconstruct_verify_batch(self._ctx, p, ln, stride,
+                                                  ns, bad, len(bad_idx), bm),
+              "reconstruct_verify")
+        return [bool(bm[s // 64] >> (s % 64) & 1) for s in range(ns)]
+
     def repair_batch(self, batch, bad_idx, disk_dst, bids, vuids,
                      block_len=65536):
         """Fused repair tasklet (worker_slice_recover.go:804-888 +

@@ -119,6 +119,10 @@ def lib():
         L.gfrs_encode_frame_batch.argtypes = [vp, vp, ctypes.c_size_t, vp,
                                               ctypes.c_size_t, ctypes.c_size_t,
                                               ctypes.c_int, i64]
+        L.gfrs_reconstruct_verify_batch.argtypes = [vp, vp, ctypes.c_size_t,
+                                                    ctypes.c_size_t,
+                                                    ctypes.c_int, i32p,
+                                                    ctypes.c_int, u64p]
         L.gfrs_update_idx.argtypes = [vp, vp, vp, ctypes.c_int, vpp,
                                       ctypes.c_size_t, ctypes.c_int]
         L.gfrs_repair_batch.argtypes = [vp, vp, ctypes.c_size_t, ctypes.c_size_t,

@@ -205,6 +205,19 @@ int gfrs_shard_parse_batch(gfrs_ctx *ctx, const void *img, size_t stride,
                            uint64_t *out_meta, int64_t *bad_block_per_shard,
                            int nshards);
 
+/* Fused reconstruct+verify (worker_slice_recover.go:804-888: the
+ * mandatory Verify after Reconstruct, :865-874) in ONE data pass: the
+ * decode rows are substituted into the parity-check rows so every output
+ * — reconstructed shard or parity comparison — is a matrix apply over the
+ * same k valid inputs.  Stored parity is compared in-register; bad_idx
+ * shards are rewritten.  fail_bitmap bit s set when stripe s's parity
+ * does not match.  Plain RS only (l == 0); LRC falls back to
+ * reconstruct_batch + verify_batch. */
+int gfrs_reconstruct_verify_batch(gfrs_ctx *ctx, void *base,
+                                  size_t shard_len, size_t stripe_stride,
+                                  int nstripes, const int32_t *bad_idx,
+                                  int nbad, uint64_t *fail_bitmap);
+
 /* ---- incremental parity (reedsolomon.go:631-668 EncodeIdx) ----
  * Adds data shard idx's contribution into the m parity shards:
  * parity[r] ^= coeff[r][idx]*data.  Parity must be zeroed before the

@@ -290,3 +290,37 @@ def test_update_idx(oracle, dev):
     enc.update_idx(shards[2], new2, 2, shards[t.N:])
     shards[2] = new2
     assert enc.verify(shards)
+
+
+@pytest.mark.parametrize("bad", [[2], [1, 7], [0, 3, 8], [6], []])
+def test_reconstruct_verify_fused(oracle, dev, bad):
+    """One-pass reconstruct+verify == oracle reconstruct, and the fused
+    verify catches corruption in surviving parity."""
+    from cubefs_amd import codemode, ec
+    t = codemode.get_tactic("EC6P3")
+    enc = ec.Encoder(t)
+    ns, slen = 6, 65536
+    rng = np.random.default_rng(60 + len(bad))
+    arr = rng.integers(0, 256, (ns, t.total, slen), dtype=np.uint8)
+    batch = torch.from_numpy(arr).to(dev)
+    enc.encode_batch(batch)
+    enc.synchronize()
+    ref = batch.cpu().numpy()
+    for i in bad:
+        batch[:, i].zero_()
+    fails = enc.reconstruct_verify_batch(batch, bad)
+    enc.synchronize()
+    assert fails == [False] * ns, (bad, fails)
+    got = batch.cpu().numpy()
+    assert np.array_equal(got, ref), bad
+    # corrupt a SURVIVING parity shard byte in stripe 3: fused verify must
+    # flag that stripe (and reconstruct the bad set wrongly is fine — the
+    # reference drops failed stripes, worker_slice_recover.go:871-874)
+    surviving_parity = [p for p in range(t.N, t.total) if p not in bad]
+    if surviving_parity and bad:
+        batch[:, :, :] = torch.from_numpy(ref).to(dev)
+        batch[3, surviving_parity[0], 123] ^= 0x40
+        for i in bad:
+            batch[:, i].zero_()
+        fails = enc.reconstruct_verify_batch(batch, bad)
+        assert fails[3] and sum(fails) == 1, fails
