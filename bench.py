@@ -36,6 +36,8 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--stripes", type=int, default=1024)
     p.add_argument("--shard-mib", type=int, default=8)
+    p.add_argument("--shard-kib", type=int, default=0,
+                   help="overrides --shard-mib when > 0 (small-blob shapes)")
     p.add_argument("--codemode", default="EC6P3",
                    help="EC6P3/EC12P4/... or LRC12P2L2 (registered via Extend)")
     p.add_argument("--workload", default="encode",
@@ -138,7 +140,7 @@ def main():
         codemode.extend(240, "LRC12P2L2",
                         codemode.Tactic(12, 2, 2, 2, 14, 0, 2048))
     t = codemode.get_tactic(args.codemode)
-    S = args.shard_mib << 20
+    S = (args.shard_kib << 10) if args.shard_kib else (args.shard_mib << 20)
     ns = args.stripes
     with_crc = not args.no_crc
 
@@ -235,10 +237,11 @@ def main():
             kind = "fused encode+frame"
         else:
             kind = "encode"
-        workload = "RS(%d+%d%s) %s%s, %d MiB shards, %d stripes/GPU" % (
+        ssz = ("%d KiB" % args.shard_kib) if args.shard_kib else \
+            ("%d MiB" % args.shard_mib)
+        workload = "RS(%d+%d%s) %s%s, %s shards, %d stripes/GPU" % (
             t.N, t.M, "+L%d" % t.L if t.L else "", kind,
-            "+crc32block" if (with_crc and not fused) else "",
-            args.shard_mib, ns)
+            "+crc32block" if (with_crc and not fused) else "", ssz, ns)
         # roofline of the dominant kernel, one launch per step:
         #   encode: read k·S, write (m+l)·S per stripe
         #   fused encode+frame: read k·S, write (k+m)·(S+4·fps) per stripe

@@ -372,12 +372,15 @@ static int get_decode_plan(gfrs_ctx_impl *c, int k, int m,
                            const std::vector<uint8_t> &matrix,
                            const std::vector<int> &engine_idx /* global ids */,
                            const std::vector<uint8_t> &present,
-                           DevPlan **out_plan) {
+                           DevPlan **out_plan, int tag) {
+  /* key = missing bitmask | engine tag (global=1, az-local=2+az, ...):
+   * global and local engines can share k, first index AND mask, so the
+   * tag is load-bearing (a collision here once wrote a local parity row
+   * over a global one — caught by the randomized fuzz test) */
   uint64_t key = 0;
   for (int i = 0; i < k + m; i++)
     if (!present[i]) key |= 1ull << i;
-  /* separate cache namespaces for global vs local engines: tag with size */
-  key = key * 1000003ull + uint64_t(engine_idx[0]) * 31 + k;
+  key = (key << 6) | uint64_t(tag & 63);
   auto it = c->dec_cache.find(key);
   if (it != c->dec_cache.end()) {
     *out_plan = it->second;
@@ -417,11 +420,11 @@ static int get_parity_plan(gfrs_ctx_impl *c, int k, int m,
                            const std::vector<uint8_t> &matrix,
                            const std::vector<int> &engine_idx,
                            const std::vector<uint8_t> &present,
-                           DevPlan **out_plan) {
+                           DevPlan **out_plan, int tag) {
   uint64_t key = 0;
   for (int i = 0; i < k + m; i++)
     if (!present[i]) key |= 1ull << i;
-  key = key * 999983ull + uint64_t(engine_idx[0]) * 31 + k;
+  key = (key << 6) | uint64_t(tag & 63);
   auto it = c->par_cache.find(key);
   if (it != c->par_cache.end()) {
     *out_plan = it->second;
@@ -459,7 +462,7 @@ static int reconstruct_engine(gfrs_ctx_impl *c, int k, int m,
                               const std::vector<uint8_t> &present,
                               size_t shard_len, int nstripes, int data_only,
                               uint64_t strided_base, uint64_t stripe_stride,
-                              int nptr_or_0) {
+                              int nptr_or_0, int tag) {
   int npresent = 0, dpresent = 0;
   for (int i = 0; i < k + m; i++)
     if (present[i]) {
@@ -473,7 +476,7 @@ static int reconstruct_engine(gfrs_ctx_impl *c, int k, int m,
   int rc = GFRS_OK;
   bool have_missing_data = dpresent < k;
   if (have_missing_data) {
-    rc = get_decode_plan(c, k, m, matrix, engine_idx, present, &dec);
+    rc = get_decode_plan(c, k, m, matrix, engine_idx, present, &dec, tag);
     if (rc != GFRS_OK) return rc;
     if (nptr_or_0 > 0)
       launch_rs_apply(reinterpret_cast<const uint64_t *>(c->ptr_buf.p),
@@ -490,7 +493,7 @@ static int reconstruct_engine(gfrs_ctx_impl *c, int k, int m,
   }
   if (!data_only) {
     DevPlan *par = nullptr;
-    rc = get_parity_plan(c, k, m, matrix, engine_idx, present, &par);
+    rc = get_parity_plan(c, k, m, matrix, engine_idx, present, &par, tag);
     if (rc != GFRS_OK) return rc;
     if (par) {
       if (nptr_or_0 > 0)
@@ -521,7 +524,7 @@ static int reconstruct_all(gfrs_ctx_impl *c, std::vector<uint8_t> &present,
   std::vector<uint8_t> gpresent(present.begin(), present.begin() + t.n + t.m);
   int rc = reconstruct_engine(c, t.n, t.m, c->enc_matrix, gidx, gpresent,
                               shard_len, nstripes, data_only, base, stride,
-                              nptr_or_0);
+                              nptr_or_0, /*tag=*/1);
   if (rc != GFRS_OK) return rc;
   if (t.l == 0 || data_only) return GFRS_OK;
   /* regenerate bad local parities per AZ (lrcencoder.go:160-184); after
@@ -538,7 +541,7 @@ static int reconstruct_all(gfrs_ctx_impl *c, std::vector<uint8_t> &present,
     if (!need) continue;
     rc = reconstruct_engine(c, c->local_n, c->local_m, c->local_matrix, li,
                             lp, shard_len, nstripes, /*data_only=*/0, base,
-                            stride, nptr_or_0);
+                            stride, nptr_or_0, /*tag=*/2 + az);
     if (rc != GFRS_OK) return rc;
   }
   return GFRS_OK;
@@ -779,7 +782,7 @@ int gfrs_reconstruct(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
       for (int i = 0; i < nshards; i++) li[i] = i;
       rc = reconstruct_engine(c, c->local_n, c->local_m, c->local_matrix, li,
                               present, shard_len, 1, data_only, (uint64_t)dev,
-                              tot, 0);
+                              tot, 0, /*tag=*/50);
     } else {
       rc = reconstruct_all(c, present, shard_len, 1, data_only, (uint64_t)dev,
                            tot, 0);
@@ -798,7 +801,8 @@ int gfrs_reconstruct(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
     std::vector<int> li(nshards);
     for (int i = 0; i < nshards; i++) li[i] = i;
     rc = reconstruct_engine(c, c->local_n, c->local_m, c->local_matrix, li,
-                            present, shard_len, 1, data_only, 0, 0, nshards);
+                            present, shard_len, 1, data_only, 0, 0, nshards,
+                            /*tag=*/50);
   } else {
     rc = reconstruct_all(c, present, shard_len, 1, data_only, 0, 0, nshards);
   }
@@ -1063,7 +1067,7 @@ int gfrs_encode_idx(gfrs_ctx *ctx, const void *data_shard, int idx,
   if ((rc = upload_ptrs(c, ptrs.data(), int(ptrs.size()))) != GFRS_OK)
     return rc;
   /* per-idx plan cached in dec_cache keyed off a synthetic mask */
-  uint64_t key = 0xE0C0DE00ull + uint64_t(idx);
+  uint64_t key = (uint64_t(idx) << 6) | 62; /* EncodeIdx namespace */
   DevPlan *p;
   auto it = c->dec_cache.find(key);
   if (it != c->dec_cache.end()) {
@@ -1118,7 +1122,7 @@ int gfrs_reconstruct_verify_batch(gfrs_ctx *ctx, void *base,
   uint64_t key = 0;
   for (int i = 0; i < k + m; i++)
     if (!present[i]) key |= 1ull << i;
-  key = key * 777767ull + 0xFECBull; /* reconstruct+verify namespace */
+  key = (key << 6) | 63; /* reconstruct+verify namespace */
   DevPlan *plan = nullptr;
   uint32_t cmp_mask = 0;
   {

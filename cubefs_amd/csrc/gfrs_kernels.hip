@@ -142,7 +142,16 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
   __syncthreads();
 
   const size_t tiles_per_shard = (shard_len + RS_TILE - 1) / RS_TILE;
-  const size_t total_tiles = tiles_per_shard * nstripes;
+  /* small shards (< half a tile): pack several stripes per tile so all
+   * 256 lanes have columns — e.g. the 2 KiB MinShardSize foreground
+   * shapes would otherwise idle 7/8 of the block */
+  const size_t padded = (shard_len + 15) & ~size_t(15);
+  const size_t spb =
+      (tiles_per_shard == 1 && padded * 2 <= size_t(RS_TILE))
+          ? size_t(RS_TILE) / padded
+          : 1;
+  const size_t total_tiles =
+      spb > 1 ? (nstripes + spb - 1) / spb : tiles_per_shard * nstripes;
   /* seq_map: block walks consecutive tiles (long sequential bursts per
    * stream, DRAM row locality) instead of grid-striding */
   const size_t per_blk =
@@ -154,9 +163,18 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
   const size_t t_step = seq_map ? 1 : gridDim.x;
 
   for (size_t tile = t_lo; tile < t_hi; tile += t_step) {
-    const size_t stripe = tile / tiles_per_shard;
-    const size_t col0 = (tile - stripe * tiles_per_shard) * size_t(RS_TILE);
-    const size_t off = col0 + size_t(threadIdx.x) * 16;
+    size_t stripe, off;
+    if (spb > 1) {
+      const size_t lane_b = size_t(threadIdx.x) * 16;
+      const size_t sub = lane_b / padded;
+      stripe = tile * spb + sub;
+      off = lane_b - sub * padded;
+      if (stripe >= nstripes) continue;
+    } else {
+      stripe = tile / tiles_per_shard;
+      off = (tile - stripe * tiles_per_shard) * size_t(RS_TILE) +
+            size_t(threadIdx.x) * 16;
+    }
     bool mismatch = false;
 
     if (off + 16 <= shard_len) {
@@ -239,7 +257,14 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
 }
 
 static int rs_grid(size_t shard_len, size_t nstripes) {
-  size_t tiles = ((shard_len + RS_TILE - 1) / RS_TILE) * nstripes;
+  const size_t tps = (shard_len + RS_TILE - 1) / RS_TILE;
+  const size_t padded = (shard_len + 15) & ~size_t(15);
+  size_t tiles;
+  if (tps == 1 && padded * 2 <= size_t(RS_TILE))
+    tiles = (nstripes + size_t(RS_TILE) / padded - 1) /
+            (size_t(RS_TILE) / padded);
+  else
+    tiles = tps * nstripes;
   if (tiles == 0) tiles = 1;
   /* memory-bound: cap and grid-stride (cdna_hip_programming.md G11) */
   const size_t cap = size_t(env_grid("GFRS_RS_GRID", 256 * 64));

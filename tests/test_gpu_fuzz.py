@@ -16,7 +16,9 @@ def test_fuzz_encode_reconstruct(oracle):
     from cubefs_amd import codemode, ec, crc32block
     dev = torch.device("cuda:0")
     rng = np.random.default_rng(0xF0220001)
-    modes = [c for c in codemode.ec_code_modes() if c < 100 or c >= 200]
+    # fixed list: the registry may carry test-added Extend entries, and
+    # the draw sequence must not depend on test order
+    modes = [1, 2, 5, 8, 9, 10, 11, 12, 13, 14, 15, 200, 201]
     encoders = {}
     codec = crc32block.Codec()
     for it in range(CASES):
