@@ -840,8 +840,12 @@ int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
   if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
   if (nstripes <= 0 || shard_len == 0) return GFRS_ERR_INVALID_SHARDS;
   StreamGuard g(c);
+  /* the fused kernel gives a workgroup a whole 64 KiB frame; below a few
+   * frames per shard the two-kernel composition (whose rs_apply packs
+   * small stripes per tile) is the faster shape */
   const bool fused = block_len == 65536 && t.l == 0 && t.m >= 1 &&
-                     t.m <= 4 && t.n + t.m <= 16 && framed_stride % 4 == 0;
+                     t.m <= 4 && t.n + t.m <= 16 && framed_stride % 4 == 0 &&
+                     shard_len >= size_t(4) * 65532;
   if (fused) {
     launch_rs_encode_frame((uint8_t *)framed, framed_stride, (uint64_t)base,
                            stripe_stride, shard_len, t.n, t.m,
