@@ -662,6 +662,78 @@ static int crc_grid(int64_t total_frames) {
  * conflict-free: lane t hits dword banks 68t+4i mod 64, distinct within
  * each 16-lane group).  Traffic = S read + S write, the algorithmic
  * minimum. */
+/* Small-payload crc32block kernel: shards of at most 16 KiB produce one
+ * short frame each; a 256-lane workgroup per frame would idle most lanes
+ * (2 KiB MinShardSize shapes!).  Here each WAVE owns one frame: lane
+ * chunks are ceil(pay/64) rounded to 4 B, reduction is wave-local
+ * shfl_xor, there is no LDS staging and no barrier in the frame loop
+ * (short frames are line-local; L1/L2 absorb the per-lane strides). */
+template <int MODE, bool TAILCRC>
+__global__ __launch_bounds__(CRC_BLOCKT) void crc32b_small_k(
+    uint8_t *__restrict__ dst, size_t dst_stride,
+    const uint8_t *__restrict__ src, size_t src_stride, int64_t n,
+    int64_t block_len, int64_t total_frames, int64_t *__restrict__ bad) {
+  __shared__ uint32_t tab[8][256];
+  for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
+    (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
+  __syncthreads();
+  const int64_t payload = n; /* frames_per_shard == 1 */
+  const int64_t chunk = ((payload + 63) / 64 + 3) & ~int64_t(3);
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t c0 = int64_t(lane) * chunk;
+  int clen = int(i64min(chunk, payload - c0));
+  if (clen < 0) clen = 0;
+  const uint32_t my_op =
+      x8n_d(uint64_t(clen > 0 ? payload - (c0 + clen) : 0));
+  const uint32_t init_term =
+      gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
+
+  const int64_t waves_total = int64_t(gridDim.x) * 4;
+  for (int64_t fr = int64_t(blockIdx.x) * 4 + wave; fr < total_frames;
+       fr += waves_total) {
+    const int64_t shard = fr; /* one frame per shard */
+    const uint8_t *psrc =
+        (MODE == 0) ? src + shard * src_stride
+                    : src + shard * src_stride + (TAILCRC ? 0 : CRC_LEN);
+    uint8_t *pdst = nullptr;
+    if (MODE == 0)
+      pdst = dst + shard * dst_stride + (TAILCRC ? 0 : CRC_LEN);
+    if (MODE == 2) pdst = dst + shard * dst_stride;
+
+    uint32_t part = clen > 0
+                        ? crc_chunk(psrc + c0, clen, tab,
+                                    (MODE != 1 && pdst) ? pdst + c0 : nullptr)
+                        : 0;
+    part = clen > 0 ? gf2_mulmod_d(my_op, part) : 0;
+#pragma unroll
+    for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
+    if (lane == 0) {
+      const uint32_t crc = ~(init_term ^ part);
+      const int64_t hpos = TAILCRC ? payload : 0;
+      if (MODE == 0) {
+        uint8_t *h = dst + shard * dst_stride + hpos;
+        if (TAILCRC) {
+          h[0] = uint8_t(crc >> 24); h[1] = uint8_t(crc >> 16);
+          h[2] = uint8_t(crc >> 8); h[3] = uint8_t(crc);
+        } else {
+          *reinterpret_cast<uint32_t *>(h) = crc;
+        }
+      } else {
+        const uint8_t *h = src + shard * src_stride + hpos;
+        uint32_t want;
+        if (TAILCRC)
+          want = (uint32_t(h[0]) << 24) | (uint32_t(h[1]) << 16) |
+                 (uint32_t(h[2]) << 8) | uint32_t(h[3]);
+        else
+          __builtin_memcpy(&want, h, 4);
+        if (want != crc)
+          atomicMin(reinterpret_cast<unsigned long long *>(&bad[shard]), 0ull);
+      }
+    }
+  }
+}
+
 /* Stage-pass chunk size trades LDS footprint (blocks/CU, latency overlap)
  * against barrier count: 256 -> 1 pass/frame, 2 blocks/CU; 128 -> 2
  * passes, 3 blocks/CU; 64 -> 4 passes, 6 blocks/CU (measured best);
@@ -812,6 +884,14 @@ static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
                          int64_t fps, int64_t total, int64_t *bad,
                          hipStream_t s) {
   const int grid = crc_grid(total);
+  if (fps == 1 && n <= 16384) {
+    const int64_t g2l = (total + 3) / 4;
+    const int g2 = int(g2l < 2048 ? g2l : 2048);
+    hipLaunchKernelGGL((crc32b_small_k<MODE, TAILCRC>), dim3(g2 > 0 ? g2 : 1),
+                       dim3(CRC_BLOCKT), 0, s, dst, dst_stride, src,
+                       src_stride, n, block_len, total, bad);
+    return;
+  }
   static const int chunk_sel = []() {
     const char *e = getenv("GFRS_CRC_CHUNK");
     const int v = e ? atoi(e) : 64;
