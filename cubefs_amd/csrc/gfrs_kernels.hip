@@ -884,7 +884,7 @@ static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
                          int64_t fps, int64_t total, int64_t *bad,
                          hipStream_t s) {
   const int grid = crc_grid(total);
-  if (fps == 1 && n <= 16384) {
+  if (fps == 1 && n <= 8192) {
     const int64_t g2l = (total + 3) / 4;
     const int g2 = int(g2l < 2048 ? g2l : 2048);
     hipLaunchKernelGGL((crc32b_small_k<MODE, TAILCRC>), dim3(g2 > 0 ? g2 : 1),
