@@ -764,16 +764,22 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
   for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
     (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
   /* fold operators are a function of the thread's chunk positions only
-   * for full frames — compute x^(8*suffix) once per block, not per frame */
+   * for the block's UNIFORM frame payload (the full 65,532 B, or the
+   * single short frame when every shard has one frame) — computed once
+   * per block, not per frame */
+  const int64_t pf_uniform =
+      frames_per_shard == 1 ? i64min(payload_full, n) : payload_full;
   uint32_t op_full[STG_PASSES];
 #pragma unroll
   for (int h = 0; h < STG_PASSES; h++) {
-    const int64_t c0 = int64_t(h) * STG_HALF + int64_t(threadIdx.x) * STG_CHUNK;
-    const int64_t cend = i64min(c0 + STG_CHUNK, payload_full);
-    op_full[h] = x8n_d(uint64_t(payload_full - cend));
+    const int64_t c0 =
+        int64_t(h) * STG_HALF + int64_t(threadIdx.x) * STG_CHUNK;
+    const int64_t cend = i64min(i64min(c0 + STG_CHUNK, pf_uniform),
+                                c0 > pf_uniform ? c0 : pf_uniform);
+    op_full[h] = x8n_d(uint64_t(pf_uniform - (cend > c0 ? cend : c0)));
   }
   const uint32_t init_full =
-      gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
+      gf2_mulmod_d(x8n_d(uint64_t(pf_uniform)), 0xFFFFFFFFu);
   __syncthreads();
 
   for (int64_t fr = blockIdx.x; fr < total_frames; fr += gridDim.x) {
@@ -796,7 +802,7 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
      * CRC the 128 B chunks out of LDS */
     uint32_t acc = 0;
     const uint32_t init_term =
-        payload == payload_full
+        payload == pf_uniform
             ? init_full
             : gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
 #pragma unroll
@@ -831,7 +837,7 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
       if (clen < 0) clen = 0;
       uint32_t part = crc_chunk16(stage + threadIdx.x * STG_STRIDE, clen, tab);
       uint32_t op = op_full[h];
-      if (payload != payload_full) { /* tail frame only */
+      if (payload != pf_uniform) { /* true tail frame only */
         const int64_t suffix =
             clen > 0 ? payload - (h0 + c0 + clen) : 0;
         op = x8n_d(uint64_t(suffix));
