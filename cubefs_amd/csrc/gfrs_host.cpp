@@ -151,6 +151,10 @@ struct gfrs_ctx_impl {
   std::vector<DevPlan *> local_enc;  /* per-AZ local encode (global idx) */
   DevPlan local_plan;                /* local engine, identity idx (one
                                         local-stripe set, lrcencoder.go:94) */
+  DevPlan fused_lrc;                 /* m+l output rows composed over the
+                                        n data shards (locals are linear in
+                                        the data through the parity rows) */
+  bool fused_lrc_ok = false;
   std::map<uint64_t, DevPlan *> dec_cache; /* missing-bitmask → data decode */
   std::map<uint64_t, DevPlan *> par_cache; /* missing-bitmask → parity rows */
 
@@ -299,6 +303,43 @@ gfrs_ctx *gfrs_create(const gfrs_tactic *t, int device) {
         for (int i = 0; i < c->local_n; i++) in[i] = i;
         for (int i = 0; i < c->local_m; i++) out[i] = c->local_n + i;
         ok = c->local_plan.upload(in, out, lrows, c->stream) == GFRS_OK;
+      }
+      if (ok && t->m + t->l <= 4 && t->n + t->m + t->l <= 16) {
+        /* compose every output (global parity AND local parity) as a row
+         * over the n data shards: a local input that is a data shard
+         * contributes its coefficient directly; one that is a global
+         * parity contributes localcoef x its parity row (GF linearity) */
+        const GfTables &gt2 = gft();
+        std::vector<int32_t> in(t->n), out;
+        std::vector<uint8_t> rows;
+        for (int i = 0; i < t->n; i++) in[i] = i;
+        for (int p2 = 0; p2 < t->m; p2++) {
+          out.push_back(t->n + p2);
+          const uint8_t *er = &c->enc_matrix[size_t(t->n + p2) * t->n];
+          rows.insert(rows.end(), er, er + t->n);
+        }
+        for (int az = 0; az < t->az_count; az++) {
+          auto idx = local_stripe(*t, az);
+          for (int lp = 0; lp < c->local_m; lp++) {
+            std::vector<uint8_t> row(t->n, 0);
+            const uint8_t *lr =
+                &c->local_matrix[size_t(c->local_n + lp) * c->local_n];
+            for (int j = 0; j < c->local_n; j++) {
+              const int g2i = idx[j];
+              if (g2i < t->n) {
+                row[g2i] ^= lr[j];
+              } else { /* global parity input */
+                const uint8_t *er = &c->enc_matrix[size_t(g2i) * t->n];
+                for (int d = 0; d < t->n; d++)
+                  row[d] ^= gt2.mul[lr[j]][er[d]];
+              }
+            }
+            out.push_back(t->n + t->m + az * c->local_m + lp);
+            rows.insert(rows.end(), row.begin(), row.end());
+          }
+        }
+        ok = c->fused_lrc.upload(in, out, rows, c->stream) == GFRS_OK;
+        c->fused_lrc_ok = ok;
       }
     }
   }
@@ -842,15 +883,19 @@ int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
   StreamGuard g(c);
   /* the fused kernel gives a workgroup a whole 64 KiB frame; below a few
    * frames per shard the two-kernel composition (whose rs_apply packs
-   * small stripes per tile) is the faster shape */
-  const bool fused = block_len == 65536 && t.l == 0 && t.m >= 1 &&
-                     t.m <= 4 && t.n + t.m <= 16 && framed_stride % 4 == 0 &&
-                     shard_len >= size_t(4) * 65532;
+   * small stripes per tile) is the faster shape.  LRC runs fused through
+   * the composed plan: every global AND local parity is a row over the n
+   * data shards. */
+  const int gm_all = t.m + t.l;
+  const bool fused = block_len == 65536 && t.m >= 1 && gm_all <= 4 &&
+                     t.n + gm_all <= 16 && framed_stride % 4 == 0 &&
+                     shard_len >= size_t(4) * 65532 &&
+                     (t.l == 0 || c->fused_lrc_ok);
   if (fused) {
+    const DevPlan &pl = t.l == 0 ? c->enc_plan : c->fused_lrc;
     launch_rs_encode_frame((uint8_t *)framed, framed_stride, (uint64_t)base,
-                           stripe_stride, shard_len, t.n, t.m,
-                           (const uint8_t *)c->enc_plan.tabs.p, nstripes,
-                           c->stream);
+                           stripe_stride, shard_len, t.n, gm_all,
+                           (const uint8_t *)pl.tabs.p, nstripes, c->stream);
     hipError_t e = hipGetLastError();
     if (e != hipSuccess) return hip_fail("encode_frame launch", e);
     return GFRS_OK;

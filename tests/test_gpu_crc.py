@@ -116,9 +116,12 @@ def test_fused_encode_frame(oracle, dev):
     oracle-encode followed by oracle-framing, across ragged sizes and
     multi-frame shards; fallback path (EC15P12) agrees too."""
     from cubefs_amd import codemode, crc32block, ec
+    codemode.extend(240, "LRC12P2L2", codemode.Tactic(12, 2, 2, 2, 14, 0, 2048))
     for name, slen in [("EC6P3", 300000), ("EC6P3", 65532), ("EC6P3", 100),
                        ("EC6P3", 1 << 20), ("EC12P4", 200000),
-                       ("EC15P12", 100000)]:  # EC15P12 -> fallback (m>4)
+                       ("EC15P12", 100000),   # fallback (m > 4)
+                       ("LRC12P2L2", 300000),  # fused composed-LRC plan
+                       ("EC6P10L2", 300000)]:  # LRC fallback (m+l > 4)
         t = codemode.get_tactic(name)
         ns = 3
         rng = np.random.default_rng(slen ^ t.N)
@@ -133,7 +136,7 @@ def test_fused_encode_frame(oracle, dev):
         got = framed.cpu().numpy()
         for s in range(ns):
             sh = [arr[s, i].copy() for i in range(t.total)]
-            oracle.rs_encode(t.N, t.M, sh)
+            oracle.lrc_encode(t.N, t.M, t.L, t.AZCount, sh)
             for j in range(t.total):
                 want = oracle.crc32b_encode(sh[j])
                 assert np.array_equal(got[s * t.total + j], want), \
