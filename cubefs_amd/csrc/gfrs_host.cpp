@@ -627,6 +627,20 @@ int gfrs_encode(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
       memcpy(shards[i], pin + size_t(i) * shard_len, shard_len);
     return GFRS_OK;
   }
+  /* ec.Buffer lays shards out contiguously (buf.go:24-35): detect that
+   * and take the strided path — no pointer-table upload, one less
+   * dependency on the foreground latency path */
+  bool contig = true;
+  for (int i = 1; i < nshards && contig; i++)
+    contig = (const uint8_t *)shards[i] ==
+             (const uint8_t *)shards[0] + size_t(i) * shard_len;
+  if (contig) {
+    rc = gfrs_encode_batch(ctx, shards[0], shard_len,
+                           size_t(nshards) * shard_len, 1);
+    if (rc != GFRS_OK) return rc;
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    return GFRS_OK;
+  }
   if ((rc = upload_ptrs(c, shards, nshards)) != GFRS_OK) return rc;
   launch_rs_apply((const uint64_t *)c->ptr_buf.p, c->total,
                   (const int32_t *)c->enc_plan.in_idx.p, c->enc_plan.k,
