@@ -149,29 +149,38 @@ def run_host_streamed(args, t, S, ns, enc, rank, world, d):
               for _ in range(2)]
     dev_out = [torch.empty((B * t.total, enc_sz), dtype=torch.uint8,
                            device="cuda") for _ in range(2)]
-    copy_s = torch.cuda.Stream()
+    copy_s = torch.cuda.Stream()   # H2D
+    d2h_s = torch.cuda.Stream()    # D2H (own direction, own engine)
     comp_s = torch.cuda.Stream()
     ev_in = [torch.cuda.Event() for _ in range(2)]
     ev_comp = [torch.cuda.Event() for _ in range(2)]
     from cubefs_amd.runtime import lib
     lib().gfrs_set_stream(enc._ctx, comp_s.cuda_stream)
 
+    ev_d2h = [torch.cuda.Event() for _ in range(2)]
+
     def stream_queue():
         for i in range(nbatches):
             b = i % 2
+            with torch.cuda.stream(d2h_s):
+                if i >= 2:
+                    ev_comp[b].wait(d2h_s)
+                    host_out[b].copy_(dev_out[b], non_blocking=True)
+                    ev_d2h[b].record(d2h_s)
             with torch.cuda.stream(copy_s):
                 if i >= 2:
-                    ev_comp[b].wait(copy_s)
-                    host_out[b].copy_(dev_out[b], non_blocking=True)
-                dev_in[b].copy_(host_src, non_blocking=True)
+                    ev_d2h[b].wait(copy_s)
+                # only the data shards cross H2D (parity is computed)
+                dev_in[b][:, :t.N].copy_(host_src[:, :t.N],
+                                         non_blocking=True)
                 ev_in[b].record(copy_s)
             with torch.cuda.stream(comp_s):
                 ev_in[b].wait(comp_s)
                 enc.encode_frame_batch(dev_out[b], dev_in[b])
                 ev_comp[b].record(comp_s)
-        with torch.cuda.stream(copy_s):
+        with torch.cuda.stream(d2h_s):
             for b in range(2):
-                ev_comp[b].wait(copy_s)
+                ev_comp[b].wait(d2h_s)
                 host_out[b].copy_(dev_out[b], non_blocking=True)
         torch.cuda.synchronize()
 
