@@ -328,3 +328,24 @@ def test_reconstruct_verify_fused(oracle, dev, bad):
             batch[:, i].zero_()
         fails = enc.reconstruct_verify_batch(batch, bad)
         assert fails[3] and sum(fails) == 1, fails
+
+
+def test_lrc_reconstruct_data_only(oracle, dev):
+    """lrcEncoder.ReconstructData (lrcencoder.go:190-207): only the global
+    n+m prefix participates; local parities stay untouched."""
+    from cubefs_amd import codemode, ec
+    codemode.extend(240, "LRC12P2L2", codemode.Tactic(12, 2, 2, 2, 14, 0, 2048))
+    t = codemode.get_tactic("LRC12P2L2")
+    enc = ec.Encoder(t)
+    slen = 8192
+    rng = np.random.default_rng(70)
+    shards, _ = make_stripe(rng, t.N, t.M + t.L, slen, dev)
+    enc.encode(shards)
+    ref = cpu_copy(shards)
+    shards[5].zero_()
+    shards[12].zero_()  # a global parity, not required for data_only
+    shards[15].zero_()  # a local parity: must stay zero
+    enc.reconstruct_data(shards, [5, 12, 15])
+    got = cpu_copy(shards)
+    assert np.array_equal(got[5], ref[5])          # data restored
+    assert not got[12].any() and not got[15].any()  # parities untouched
