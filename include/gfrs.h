@@ -81,8 +81,11 @@ int gfrs_synchronize(gfrs_ctx *ctx);
 
 /* ---- single stripe (the ec.Encoder surface) ----
  * shards: n+m+l pointers, each shard_len bytes, all in `memloc` memory.
- * All launches are async on the ctx stream when memloc==GFRS_MEM_DEVICE;
- * host-memory calls block until the result is back. */
+ * Single-stripe calls BLOCK until the result is ready (ec.Encoder
+ * semantics: the caller owns the buffers on return).  The batch APIs
+ * below are async on the ctx stream; synchronize with
+ * gfrs_synchronize().  When the shards form one contiguous ec.Buffer
+ * layout (buf.go:24-35) the pointer-table upload is skipped. */
 
 /* Encoder.Encode (encoder.go:114 / lrcencoder.go:35): fills parity (and
  * local parity when l > 0) from the data shards. */
