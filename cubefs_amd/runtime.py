@@ -62,6 +62,20 @@ def lib():
                 "cubefs_amd/libgfrs.so not built — run `make -C cubefs_amd` "
                 "(or __graft_entry__.build()).  The gfrs engine has no CPU "
                 "fallback by design.")
+        # torch's wheel bundles its own ROCm runtime (unversioned
+        # libamdhip64.so / libhsa-runtime64.so sonames), while libgfrs
+        # links /opt/rocm's libamdhip64.so.7 — two HSA stacks coexist in
+        # the process and only the "torch's stack initializes first"
+        # order works (the second stack to init still enumerates the
+        # device; the other order leaves gfrs with hipGetDeviceCount==0).
+        # Enforce that order here so it doesn't depend on the caller's
+        # import sequence.
+        try:
+            import torch
+            if torch.cuda.is_available():
+                torch.cuda.init()
+        except ImportError:
+            pass  # pure-ctypes consumers (no torch) manage init order
         L = ctypes.CDLL(_LIB_PATH)
         vp, vpp = ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p)
         i32p = ctypes.POINTER(ctypes.c_int32)
