@@ -47,11 +47,18 @@ class Codec:
         self._ctx = lib().gfrs_create(ctypes.byref(t), device)
         if not self._ctx:
             raise GfrsError(-100, lib().gfrs_last_error().decode())
+        self._destroy = lib().gfrs_destroy
 
     def __del__(self):
+        # interpreter shutdown can clear module globals before __del__
+        # runs, so use the function handle captured at construction
         ctx = getattr(self, "_ctx", None)
-        if ctx:
-            lib().gfrs_destroy(ctx)
+        destroy = getattr(self, "_destroy", None)
+        if ctx and destroy is not None:
+            try:
+                destroy(ctx)
+            except TypeError:
+                pass
             self._ctx = None
 
     def encode(self, dst, src, block_len=DEFAULT_BLOCK):
