@@ -976,21 +976,25 @@ void launch_sized_decode(uint8_t *dst, size_t dst_stride,
  * staged, CRC'd and framed-written in turn; parity accumulates in VGPRs
  * across the k data shards and then takes the same stage path. */
 
-constexpr int EF_PASS = 16384;  /* bytes per pass */
-constexpr int EF_CHUNK = 64;    /* CRC chunk per lane per pass */
-constexpr int EF_STRIDE = EF_CHUNK + 16;
-constexpr int EF_PASSES = 4;    /* ceil(65532 / 16384) */
-constexpr int EF_STG_ONE = 256 * EF_STRIDE; /* one stage buffer (20 KB) */
-/* tables | per-frame CRC reduction slab (4 waves x 16 shards) | 2 stages */
+/* tables | per-frame CRC reduction slab (4 waves x 16 shards) | stage(s);
+ * pass/chunk/stage geometry is a template parameter of the kernel (NI). */
 constexpr int EF_RED = 4 * 16 * 4;
-constexpr int EF_LDS = 8192 + EF_RED + 2 * EF_STG_ONE;
 
-template <int GM, int NBUF, int WPS, int ABL = 0>
+template <int GM, int NBUF, int WPS, int ABL = 0, int NI = 4>
 __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
     uint8_t *__restrict__ dst, size_t dst_stride /* framed image stride */,
     uint64_t base, uint64_t stripe_stride, size_t shard_len, int k,
     const uint8_t *__restrict__ tabs /* [GM*k][32] */, int64_t total_frames,
     int64_t frames_per_shard) {
+  /* NI = 4096-B subtiles per pass.  NI=4 is the 16 KiB geometry described
+   * above; NI=2 halves the pass (8 KiB, 8 passes/frame): acc[] drops from
+   * GM*4 to GM*2 uint4s and the stage from 20 KB to 12 KB, trading more
+   * barriers + CRC folds for co-resident blocks (VGPR/LDS occupancy). */
+  constexpr int EF_PASS = NI * 4096;
+  constexpr int EF_CHUNK = NI * 16;          /* 256 lanes cover the pass */
+  constexpr int EF_STRIDE = EF_CHUNK + 16;   /* odd-16 stride: bank-clean */
+  constexpr int EF_PASSES = (65532 + EF_PASS - 1) / EF_PASS;
+  constexpr int EF_STG_ONE = 256 * EF_STRIDE;
   constexpr int64_t block_len = 65536;
   constexpr int64_t payload_full = block_len - CRC_LEN;
   extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
@@ -1034,7 +1038,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
     /* per-frame CRC partials live in the LDS red slab (4 waves x 16
      * shard slots): no runtime-indexed per-lane array -> no scratch */
     for (int j = threadIdx.x; j < 64; j += CRC_BLOCKT) red[j] = 0;
-    uint4 acc[GM][4];
+    uint4 acc[GM][NI];
     __syncthreads();
 
     /* Software pipeline over the k+GM shard units of each pass: unit u's
@@ -1048,7 +1052,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
 #pragma unroll
       for (int r = 0; r < GM; r++)
 #pragma unroll
-        for (int i = 0; i < 4; i++) acc[r][i] = uint4{0, 0, 0, 0};
+        for (int i = 0; i < NI; i++) acc[r][i] = uint4{0, 0, 0, 0};
 
       const int64_t lane16 = int64_t(threadIdx.x) * 16;
       const int64_t c0b = int64_t(threadIdx.x) * EF_CHUNK;
@@ -1084,7 +1088,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
         uint8_t *fdst = dst + (stripe * (k + GM) + c) * dst_stride +
                         f * block_len + CRC_LEN + r0;
 #pragma unroll
-        for (int i = 0; i < 4; i++) {
+        for (int i = 0; i < NI; i++) {
           const int64_t off = int64_t(i) * 4096 + lane16;
           if (off + 16 <= rbytes) {
             const uint4 v = *reinterpret_cast<const uint4 *>(src + off);
@@ -1128,7 +1132,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
         uint8_t *fdst = dst + (stripe * (k + GM) + k + r) * dst_stride +
                         f * block_len + CRC_LEN + r0;
 #pragma unroll
-        for (int i = 0; i < 4; i++) {
+        for (int i = 0; i < NI; i++) {
           const int64_t off = int64_t(i) * 4096 + lane16;
           if (off + 16 <= rbytes) {
             if (ABL != 3) {
@@ -1190,27 +1194,34 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
   const int64_t fps = (int64_t(shard_len) + 65531) / 65532;
   const int64_t total = fps * nstripes;
   const int grid = crc_grid(total);
-  /* occupancy variant: GFRS_EF = NBUF*10 + waves-per-SIMD bound.
-   * Measured @256 stripes RS(6+3): 14 -> 13.7 ms (single 20 KB stage,
-   * 4 blocks/CU even with a 40 B spill) beats 23 (15.3), 13 (15.4),
-   * 24 (16.5). */
+  /* occupancy variant: GFRS_EF = NBUF*10 + waves-per-SIMD bound (16 KiB
+   * pass), or a 3-digit NBUF*100 + WPS*10 + NI form for the 8 KiB-pass
+   * (NI=2) geometry.  Measured @256 stripes RS(6+3): 14 -> 13.7 ms
+   * (single 20 KB stage, 4 blocks/CU) beats 23 (15.3), 13 (15.4),
+   * 24 (16.5); NI=2 variants (142/152/162) target 5-6 blocks/CU. */
   static const int var = []() {
     const char *e = getenv("GFRS_EF");
     const int v = e ? atoi(e) : 14;
-    return (v == 13 || v == 14 || v == 23 || v == 24) ? v : 14;
+    switch (v) {
+      case 13: case 14: case 23: case 24:
+      case 142: case 152: case 162: return v;
+      default: return 14;
+    }
   }();
-  const int nbuf = var / 10;
-  const int lds = 8192 + EF_RED + nbuf * EF_STG_ONE + m * k * 32;
-#define GFRS_EF_GO(G, NB, W)                                              \
-  hipLaunchKernelGGL((rs_encode_frame_k<G, NB, W>), dim3(grid),           \
+  const int nbuf = (var >= 100 ? var / 100 : var / 10);
+  const int ni = (var >= 100 ? var % 10 : 4);
+  const int stg_one = 256 * (16 * ni + 16);
+  const int lds = 8192 + EF_RED + nbuf * stg_one + m * k * 32;
+#define GFRS_EF_GO(G, NB, W, I)                                           \
+  hipLaunchKernelGGL((rs_encode_frame_k<G, NB, W, 0, I>), dim3(grid),     \
                      dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,     \
                      stripe_stride, shard_len, k, tabs, total, fps)
-#define GFRS_EF_SW(NB, W)                                                 \
+#define GFRS_EF_SW(NB, W, I)                                              \
   switch (m) {                                                            \
-    case 1: GFRS_EF_GO(1, NB, W); break;                                  \
-    case 2: GFRS_EF_GO(2, NB, W); break;                                  \
-    case 3: GFRS_EF_GO(3, NB, W); break;                                  \
-    default: GFRS_EF_GO(4, NB, W);                                        \
+    case 1: GFRS_EF_GO(1, NB, W, I); break;                               \
+    case 2: GFRS_EF_GO(2, NB, W, I); break;                               \
+    case 3: GFRS_EF_GO(3, NB, W, I); break;                               \
+    default: GFRS_EF_GO(4, NB, W, I);                                     \
   }
   static const int abl = []() {
     const char *e = getenv("GFRS_EF_ABL");
@@ -1232,10 +1243,13 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
         stripe_stride, shard_len, k, tabs, total, fps); return;
       default: break; }
   }
-  if (var == 13) { GFRS_EF_SW(1, 3) }
-  else if (var == 14) { GFRS_EF_SW(1, 4) }
-  else if (var == 24) { GFRS_EF_SW(2, 4) }
-  else { GFRS_EF_SW(2, 3) }
+  if (var == 13) { GFRS_EF_SW(1, 3, 4) }
+  else if (var == 14) { GFRS_EF_SW(1, 4, 4) }
+  else if (var == 24) { GFRS_EF_SW(2, 4, 4) }
+  else if (var == 142) { GFRS_EF_SW(1, 4, 2) }
+  else if (var == 152) { GFRS_EF_SW(1, 5, 2) }
+  else if (var == 162) { GFRS_EF_SW(1, 6, 2) }
+  else { GFRS_EF_SW(2, 3, 4) }
 #undef GFRS_EF_SW
 #undef GFRS_EF_GO
 }
