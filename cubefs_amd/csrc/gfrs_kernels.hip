@@ -473,6 +473,10 @@ static bool perm0_ok() {
 /* slice-by-8 tables (zlib BYFOUR construction extended), host-computed,
  * copied per device.  tab[0] is the base byte table. */
 __device__ uint32_t g_crc_tab4[8][256];
+/* multiply a reflected CRC value by x^(8*4096): byte-sliced tables so the
+ * register-CRC fused kernel can Horner-chain a lane's four 4096-apart
+ * uint4 pieces with 4 LDS gathers per step instead of a 32-step mulmod */
+__device__ uint32_t g_shift4k[4][256];
 
 /* x^(8*2^j) mod P, reflected domain — host-filled alongside the tables. */
 __device__ uint32_t g_pow8[40];
@@ -1187,6 +1191,225 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
   }
 }
 
+/* raw CRC of 16 bytes held in a register uint4 (slice-by-8, 2 steps) */
+GFRS_DEV uint32_t crc16_reg(const uint4 q, const uint32_t (*tab)[256]) {
+  uint32_t c = 0;
+#pragma unroll
+  for (int j = 0; j < 2; j++) {
+    const uint32_t w0 = (j == 0 ? q.x : q.z) ^ c;
+    const uint32_t w1 = (j == 0 ? q.y : q.w);
+    c = tab[7][w0 & 0xFF] ^ tab[6][(w0 >> 8) & 0xFF] ^
+        tab[5][(w0 >> 16) & 0xFF] ^ tab[4][w0 >> 24] ^
+        tab[3][w1 & 0xFF] ^ tab[2][(w1 >> 8) & 0xFF] ^
+        tab[1][(w1 >> 16) & 0xFF] ^ tab[0][w1 >> 24];
+  }
+  return c;
+}
+
+GFRS_DEV uint32_t shift4k(uint32_t c, const uint32_t (*stab)[256]) {
+  return stab[0][c & 0xFF] ^ stab[1][(c >> 8) & 0xFF] ^
+         stab[2][(c >> 16) & 0xFF] ^ stab[3][c >> 24];
+}
+
+/* Register-CRC variant of the fused encode+frame kernel: no LDS stage at
+ * all.  Lane w's four uint4 pieces of a 16 KiB pass (subtile offsets
+ * i*4096 + 16w) are CRC'd straight out of the registers the MAC/store
+ * phase already holds, Horner-chained with the constant x^(8*4096) shift
+ * (4 LDS gathers), then folded once by the lane's position operator and
+ * wave-reduced — so the per-unit stage round-trip AND the two
+ * __syncthreads per unit disappear (only the per-frame reduction-slab
+ * barriers remain) and waves pipeline units/passes freely.  Same output
+ * bytes as rs_encode_frame_k (oracle parity-tested). */
+/* MAP: 0 = plain grid-stride over frames; 1 = each block owns a
+ * contiguous run of frames (stream locality across its iterations);
+ * 2 = run-per-block with the block ids permuted so the 8 XCDs (round-
+ * robin dispatch) each own a contiguous 1/8 of the frame space and an
+ * XCD's L2 sees only its own stripes' read/write streams. */
+/* SKEL=1 compiles out MAC+CRC, leaving loads + framed stores with the
+ * exact production addressing: the kernel's own empirical memory floor. */
+template <int GM, int WPS, int MAP = 0, int SKEL = 0>
+__global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
+    uint8_t *__restrict__ dst, size_t dst_stride, uint64_t base,
+    uint64_t stripe_stride, size_t shard_len, int k,
+    const uint8_t *__restrict__ tabs /* [GM*k][32] */, int64_t total_frames,
+    int64_t frames_per_shard) {
+  constexpr int EF_PASS = 16384;
+  constexpr int EF_PASSES = 4; /* ceil(65532 / 16384) */
+  constexpr int64_t block_len = 65536;
+  constexpr int64_t payload_full = block_len - CRC_LEN;
+  extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
+  /* crc tables 8 KB | shift4k 4 KB | red slab | coefficient tables */
+  uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
+  uint32_t(*stab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem + 8192);
+  uint32_t *red = reinterpret_cast<uint32_t *>(smem + 12288);
+  uint8_t *ctab = smem + 12288 + EF_RED;
+  for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
+    (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
+  for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
+    (&stab[0][0])[i] = (&g_shift4k[0][0])[i];
+  for (int i = threadIdx.x; i < GM * k * 2; i += CRC_BLOCKT)
+    reinterpret_cast<uint4 *>(ctab)[i] =
+        reinterpret_cast<const uint4 *>(tabs)[i];
+  const int64_t lane16 = int64_t(threadIdx.x) * 16;
+  /* fold operator for full frames: suffix after the lane's LAST present
+   * piece in pass h (pieces are a prefix in i since offsets grow) */
+  uint32_t op_full[EF_PASSES];
+#pragma unroll
+  for (int h = 0; h < EF_PASSES; h++) {
+    const int64_t r0 = int64_t(h) * EF_PASS;
+    const int64_t rb = i64min(int64_t(EF_PASS), payload_full - r0);
+    int np = 0;
+#pragma unroll
+    for (int i = 0; i < 4; i++)
+      if (int64_t(i) * 4096 + lane16 + 16 <= rb) np = i + 1;
+    const int64_t end = np ? r0 + int64_t(np - 1) * 4096 + lane16 + 16 : r0;
+    op_full[h] = x8n_d(uint64_t(payload_full - end));
+  }
+  __syncthreads();
+
+  const uint4 *ltab = reinterpret_cast<const uint4 *>(ctab);
+
+  int64_t fr0 = blockIdx.x, frN = total_frames, frS = gridDim.x;
+  if (MAP != 0) {
+    const int64_t nper = (total_frames + gridDim.x - 1) / gridDim.x;
+    const int64_t b = MAP == 2 ? int64_t(blockIdx.x & 7) * (gridDim.x >> 3) +
+                                     (blockIdx.x >> 3)
+                               : int64_t(blockIdx.x);
+    fr0 = b * nper;
+    frN = i64min(fr0 + nper, total_frames);
+    frS = 1;
+  }
+  for (int64_t fr = fr0; fr < frN; fr += frS) {
+    const int64_t stripe = fr / frames_per_shard;
+    const int64_t f = fr - stripe * frames_per_shard;
+    const int64_t p0 = f * payload_full;
+    const int64_t payload = i64min(payload_full, int64_t(shard_len) - p0);
+    const uint8_t *sbase =
+        reinterpret_cast<const uint8_t *>(base + stripe * stripe_stride);
+
+    for (int j = threadIdx.x; j < 64; j += CRC_BLOCKT) red[j] = 0;
+    uint4 acc[GM][4];
+    __syncthreads();
+
+    for (int h = 0; h < EF_PASSES; h++) {
+      const int64_t r0 = int64_t(h) * EF_PASS;
+      const int64_t rbytes = i64min(int64_t(EF_PASS), payload - r0);
+      if (rbytes <= 0) break;
+#pragma unroll
+      for (int r = 0; r < GM; r++)
+#pragma unroll
+        for (int i = 0; i < 4; i++) acc[r][i] = uint4{0, 0, 0, 0};
+
+      uint32_t op = op_full[h];
+      if (payload != payload_full) {
+        int np = 0;
+#pragma unroll
+        for (int i = 0; i < 4; i++)
+          if (int64_t(i) * 4096 + lane16 + 16 <= rbytes) np = i + 1;
+        const int64_t end =
+            np ? r0 + int64_t(np - 1) * 4096 + lane16 + 16 : r0;
+        op = x8n_d(uint64_t(payload - end));
+      }
+
+      for (int c = 0; c < k; c++) {
+        const uint8_t *src = sbase + size_t(c) * shard_len + p0 + r0;
+        uint8_t *fdst = dst + (stripe * (k + GM) + c) * dst_stride +
+                        f * block_len + CRC_LEN + r0;
+        uint32_t t = 0;
+        bool any = false;
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+          const int64_t off = int64_t(i) * 4096 + lane16;
+          if (off + 16 <= rbytes) {
+            const uint4 v = *reinterpret_cast<const uint4 *>(src + off);
+            if (SKEL == 0) {
+#pragma unroll
+              for (int r = 0; r < GM; r++) {
+                const int t2 = (r * k + c) * 2;
+                gfmac16<false>(acc[r][i], v, ltab[t2], ltab[t2 + 1]);
+              }
+            }
+            uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+            dw[0] = v.x; dw[1] = v.y; dw[2] = v.z; dw[3] = v.w;
+            if (SKEL == 0) {
+              t = any ? shift4k(t, stab) ^ crc16_reg(v, tab)
+                      : crc16_reg(v, tab);
+              any = true;
+            }
+          }
+        }
+        uint32_t part = any ? gf2_mulmod_d(op, t) : 0;
+        if (rbytes < EF_PASS && threadIdx.x == 0) {
+          /* pass tail ends exactly at the payload: identity fold */
+          const int64_t t0 = (rbytes / 16) * 16;
+          const uint32_t ct =
+              crc_chunk(src + t0, int(rbytes - t0), tab, fdst + t0);
+          if (SKEL == 0) part ^= ct;
+        }
+#pragma unroll
+        for (int sh = 32; sh > 0; sh >>= 1)
+          part ^= __shfl_xor(part, sh, 64);
+        if ((threadIdx.x & 63) == 0)
+          red[(threadIdx.x >> 6) * 16 + c] ^= part;
+      }
+#pragma unroll
+      for (int r = 0; r < GM; r++) {
+        uint8_t *fdst = dst + (stripe * (k + GM) + k + r) * dst_stride +
+                        f * block_len + CRC_LEN + r0;
+        uint32_t t = 0;
+        bool any = false;
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+          const int64_t off = int64_t(i) * 4096 + lane16;
+          if (off + 16 <= rbytes) {
+            uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+            dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
+            dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
+            if (SKEL == 0) {
+              t = any ? shift4k(t, stab) ^ crc16_reg(acc[r][i], tab)
+                      : crc16_reg(acc[r][i], tab);
+              any = true;
+            }
+          }
+        }
+        uint32_t part = any ? gf2_mulmod_d(op, t) : 0;
+        if (SKEL == 0 && rbytes < EF_PASS && threadIdx.x == 0) {
+          const int64_t t0 = (rbytes / 16) * 16;
+          uint32_t ct = 0;
+          for (int64_t p = t0; p < rbytes; p++) {
+            uint8_t pv = 0;
+            for (int c2 = 0; c2 < k; c2++) {
+              const uint8_t b = sbase[size_t(c2) * shard_len + p0 + r0 + p];
+              const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
+              pv ^= tt[b & 0x0F] ^ tt[16 + (b >> 4)];
+            }
+            fdst[p] = pv;
+            ct = tab[0][(ct ^ pv) & 0xFF] ^ (ct >> 8);
+          }
+          part ^= ct;
+        }
+#pragma unroll
+        for (int sh = 32; sh > 0; sh >>= 1)
+          part ^= __shfl_xor(part, sh, 64);
+        if ((threadIdx.x & 63) == 0)
+          red[(threadIdx.x >> 6) * 16 + k + r] ^= part;
+      }
+    }
+
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      const uint32_t it = gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
+      for (int j = 0; j < k + GM; j++) {
+        const uint32_t crc =
+            ~(it ^ red[j] ^ red[16 + j] ^ red[32 + j] ^ red[48 + j]);
+        *reinterpret_cast<uint32_t *>(
+            dst + (stripe * (k + GM) + j) * dst_stride + f * block_len) = crc;
+      }
+    }
+    __syncthreads();
+  }
+}
+
 void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
                             uint64_t stripe_stride, size_t shard_len, int k,
                             int m, const uint8_t *tabs, int nstripes,
@@ -1204,10 +1427,50 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
     const int v = e ? atoi(e) : 14;
     switch (v) {
       case 13: case 14: case 23: case 24:
+      case 74: case 75:
       case 142: case 152: case 162: return v;
       default: return 14;
     }
   }();
+  /* 7x = register-CRC kernel (no stage): crc tabs + shift tabs + red */
+  if (var == 74 || var == 75) {
+    const int lds = 12288 + EF_RED + m * k * 32;
+#define GFRS_EFR_GO(G, W, P) hipLaunchKernelGGL((rs_encode_frame_reg_k<G, W, P>), dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base, stripe_stride, shard_len, k, tabs, total, fps)
+#define GFRS_EFR_SW(W, P) switch (m) { case 1: GFRS_EFR_GO(1, W, P); break; case 2: GFRS_EFR_GO(2, W, P); break; case 3: GFRS_EFR_GO(3, W, P); break; default: GFRS_EFR_GO(4, W, P); }
+    static const int map = []() {
+      const char *e = getenv("GFRS_EF_MAP");
+      const int v = e ? atoi(e) : 0;
+      return (v >= 0 && v <= 2) ? v : 0;
+    }();
+    static const int rabl = []() {
+      const char *e = getenv("GFRS_EF_ABL");
+      return e && atoi(e) == 4 ? 4 : 0;
+    }();
+    if (rabl == 4) { /* memory-floor skeleton (diagnostic only) */
+      switch (m) {
+        case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 4, 1, 1>),
+            dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+            stripe_stride, shard_len, k, tabs, total, fps); break;
+        case 3: hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 4, 1, 1>),
+            dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+            stripe_stride, shard_len, k, tabs, total, fps); break;
+        default: hipLaunchKernelGGL((rs_encode_frame_reg_k<4, 4, 1, 1>),
+            dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+            stripe_stride, shard_len, k, tabs, total, fps);
+      }
+      return;
+    }
+    if (map == 0) {
+      if (var == 74) { GFRS_EFR_SW(4, 0) } else { GFRS_EFR_SW(5, 0) }
+    } else if (map == 1 || (grid & 7) != 0) { /* XCD split needs grid%8==0 */
+      if (var == 74) { GFRS_EFR_SW(4, 1) } else { GFRS_EFR_SW(5, 1) }
+    } else {
+      if (var == 74) { GFRS_EFR_SW(4, 2) } else { GFRS_EFR_SW(5, 2) }
+    }
+#undef GFRS_EFR_SW
+#undef GFRS_EFR_GO
+    return;
+  }
   const int nbuf = (var >= 100 ? var / 100 : var / 10);
   const int ni = (var >= 100 ? var % 10 : 4);
   const int stg_one = 256 * (16 * ni + 16);
@@ -1402,6 +1665,15 @@ int crc_device_init_current(void) {
     pow8[j] = p;
     p = mulmod(p, p);
   }
+  /* g_shift4k[j][b] = x^(8*4096) * (b << 8j): mulmod is GF(2)-linear in
+   * its second argument, so XORing the four byte-slice lookups of a value
+   * multiplies the whole value by x^(8*4096).  pow8[12] = x^(8*2^12). */
+  uint32_t s4k[4][256];
+  for (int j = 0; j < 4; j++)
+    for (uint32_t b = 0; b < 256; b++)
+      s4k[j][b] = mulmod(pow8[12], b << (8 * j));
+  if (hipMemcpyToSymbol(HIP_SYMBOL(g_shift4k), s4k, sizeof(s4k)) != hipSuccess)
+    return -100;
   if (hipMemcpyToSymbol(HIP_SYMBOL(g_crc_tab4), tab, sizeof(tab)) != hipSuccess)
     return -100;
   if (hipMemcpyToSymbol(HIP_SYMBOL(g_pow8), pow8, sizeof(pow8)) != hipSuccess)
