@@ -1227,7 +1227,11 @@ GFRS_DEV uint32_t shift4k(uint32_t c, const uint32_t (*stab)[256]) {
  * XCD's L2 sees only its own stripes' read/write streams. */
 /* SKEL=1 compiles out MAC+CRC, leaving loads + framed stores with the
  * exact production addressing: the kernel's own empirical memory floor. */
-template <int GM, int WPS, int MAP = 0, int SKEL = 0>
+/* PIPE=1: explicit one-unit lookahead - unit c+1's four piece loads are
+ * issued before unit c's MAC/CRC/store consume their values, so the
+ * compute of each unit overlaps the memory latency of the next.  Costs
+ * ~32 VGPRs of double-buffer; pair with WPS=3. */
+template <int GM, int WPS, int MAP = 0, int SKEL = 0, int PIPE = 0>
 __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     uint8_t *__restrict__ dst, size_t dst_stride, uint64_t base,
     uint64_t stripe_stride, size_t shard_len, int k,
@@ -1311,17 +1315,48 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
         op = x8n_d(uint64_t(payload - end));
       }
 
+      uint4 vnext[PIPE ? 4 : 1];
+      if (PIPE) {
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+          const int64_t off = int64_t(i) * 4096 + lane16;
+          vnext[PIPE ? i : 0] =
+              off + 16 <= rbytes
+                  ? *reinterpret_cast<const uint4 *>(sbase + p0 + r0 + off)
+                  : uint4{0, 0, 0, 0};
+        }
+      }
       for (int c = 0; c < k; c++) {
         const uint8_t *src = sbase + size_t(c) * shard_len + p0 + r0;
         uint8_t *fdst = dst + (stripe * (k + GM) + c) * dst_stride +
                         f * block_len + CRC_LEN + r0;
+        uint4 vcur[PIPE ? 4 : 1];
+        if (PIPE) {
+#pragma unroll
+          for (int i = 0; i < 4; i++) vcur[PIPE ? i : 0] = vnext[PIPE ? i : 0];
+          if (c + 1 < k) {
+            const uint8_t *nsrc = sbase + size_t(c + 1) * shard_len + p0 + r0;
+#pragma unroll
+            for (int i = 0; i < 4; i++) {
+              const int64_t off = int64_t(i) * 4096 + lane16;
+              vnext[PIPE ? i : 0] =
+                  off + 16 <= rbytes
+                      ? *reinterpret_cast<const uint4 *>(nsrc + off)
+                      : uint4{0, 0, 0, 0};
+            }
+          }
+        }
+        /* shift4k(0) == 0, so the Horner chain can start from t = 0 and
+         * run unconditionally: no 'any' tracking, less register state,
+         * and the 4 piece loads stay in flight together. */
         uint32_t t = 0;
-        bool any = false;
 #pragma unroll
         for (int i = 0; i < 4; i++) {
           const int64_t off = int64_t(i) * 4096 + lane16;
           if (off + 16 <= rbytes) {
-            const uint4 v = *reinterpret_cast<const uint4 *>(src + off);
+            const uint4 v =
+                PIPE ? vcur[PIPE ? i : 0]
+                     : *reinterpret_cast<const uint4 *>(src + off);
             if (SKEL == 0) {
 #pragma unroll
               for (int r = 0; r < GM; r++) {
@@ -1331,14 +1366,10 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
             }
             uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
             dw[0] = v.x; dw[1] = v.y; dw[2] = v.z; dw[3] = v.w;
-            if (SKEL == 0) {
-              t = any ? shift4k(t, stab) ^ crc16_reg(v, tab)
-                      : crc16_reg(v, tab);
-              any = true;
-            }
+            if (SKEL == 0) t = shift4k(t, stab) ^ crc16_reg(v, tab);
           }
         }
-        uint32_t part = any ? gf2_mulmod_d(op, t) : 0;
+        uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
         if (rbytes < EF_PASS && threadIdx.x == 0) {
           /* pass tail ends exactly at the payload: identity fold */
           const int64_t t0 = (rbytes / 16) * 16;
@@ -1357,7 +1388,6 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
         uint8_t *fdst = dst + (stripe * (k + GM) + k + r) * dst_stride +
                         f * block_len + CRC_LEN + r0;
         uint32_t t = 0;
-        bool any = false;
 #pragma unroll
         for (int i = 0; i < 4; i++) {
           const int64_t off = int64_t(i) * 4096 + lane16;
@@ -1365,14 +1395,11 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
             uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
             dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
             dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
-            if (SKEL == 0) {
-              t = any ? shift4k(t, stab) ^ crc16_reg(acc[r][i], tab)
-                      : crc16_reg(acc[r][i], tab);
-              any = true;
-            }
+            if (SKEL == 0)
+              t = shift4k(t, stab) ^ crc16_reg(acc[r][i], tab);
           }
         }
-        uint32_t part = any ? gf2_mulmod_d(op, t) : 0;
+        uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
         if (SKEL == 0 && rbytes < EF_PASS && threadIdx.x == 0) {
           const int64_t t0 = (rbytes / 16) * 16;
           uint32_t ct = 0;
@@ -1427,12 +1454,30 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
     const int v = e ? atoi(e) : 14;
     switch (v) {
       case 13: case 14: case 23: case 24:
-      case 74: case 75:
+      case 74: case 75: case 76:
       case 142: case 152: case 162: return v;
       default: return 14;
     }
   }();
   /* 7x = register-CRC kernel (no stage): crc tabs + shift tabs + red */
+  if (var == 76) { /* one-unit-lookahead pipeline at 3 waves/SIMD */
+    const int lds = 12288 + EF_RED + m * k * 32;
+    switch (m) {
+      case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 3, 0, 0, 1>),
+          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps); break;
+      case 2: hipLaunchKernelGGL((rs_encode_frame_reg_k<2, 3, 0, 0, 1>),
+          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps); break;
+      case 3: hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 0, 1>),
+          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps); break;
+      default: hipLaunchKernelGGL((rs_encode_frame_reg_k<4, 3, 0, 0, 1>),
+          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps);
+    }
+    return;
+  }
   if (var == 74 || var == 75) {
     const int lds = 12288 + EF_RED + m * k * 32;
 #define GFRS_EFR_GO(G, W, P) hipLaunchKernelGGL((rs_encode_frame_reg_k<G, W, P>), dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base, stripe_stride, shard_len, k, tabs, total, fps)
