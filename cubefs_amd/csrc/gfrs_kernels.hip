@@ -1291,8 +1291,21 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     const uint8_t *sbase =
         reinterpret_cast<const uint8_t *>(base + stripe * stripe_stride);
 
-    for (int j = threadIdx.x; j < 64; j += CRC_BLOCKT) red[j] = 0;
     uint4 acc[GM][4];
+    uint4 vnext[PIPE ? 4 : 1];
+    if (PIPE && fr == fr0) { /* later frames are prefetched by the
+                                previous frame's last pass */
+      const int64_t rb0 = i64min(int64_t(EF_PASS), payload);
+#pragma unroll
+      for (int i = 0; i < 4; i++) {
+        const int64_t off = int64_t(i) * 4096 + lane16;
+        vnext[PIPE ? i : 0] =
+            off + 16 <= rb0
+                ? *reinterpret_cast<const uint4 *>(sbase + p0 + off)
+                : uint4{0, 0, 0, 0};
+      }
+    }
+    for (int j = threadIdx.x; j < 64; j += CRC_BLOCKT) red[j] = 0;
     __syncthreads();
 
     for (int h = 0; h < EF_PASSES; h++) {
@@ -1315,17 +1328,6 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
         op = x8n_d(uint64_t(payload - end));
       }
 
-      uint4 vnext[PIPE ? 4 : 1];
-      if (PIPE) {
-#pragma unroll
-        for (int i = 0; i < 4; i++) {
-          const int64_t off = int64_t(i) * 4096 + lane16;
-          vnext[PIPE ? i : 0] =
-              off + 16 <= rbytes
-                  ? *reinterpret_cast<const uint4 *>(sbase + p0 + r0 + off)
-                  : uint4{0, 0, 0, 0};
-        }
-      }
       for (int c = 0; c < k; c++) {
         const uint8_t *src = sbase + size_t(c) * shard_len + p0 + r0;
         uint8_t *fdst = dst + (stripe * (k + GM) + c) * dst_stride +
@@ -1382,6 +1384,31 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
           part ^= __shfl_xor(part, sh, 64);
         if ((threadIdx.x & 63) == 0)
           red[(threadIdx.x >> 6) * 16 + c] ^= part;
+      }
+      if (PIPE) { /* next pass's (or next frame's) unit-0 loads fly
+                     during the parity rows and the frame epilogue */
+        int64_t r0n = r0 + EF_PASS;
+        int64_t rbn = i64min(int64_t(EF_PASS), payload - r0n);
+        const uint8_t *nbase = sbase + p0;
+        if (rbn <= 0 && fr + frS < frN) {
+          const int64_t fr2 = fr + frS;
+          const int64_t st2 = fr2 / frames_per_shard;
+          const int64_t p02 = (fr2 - st2 * frames_per_shard) * payload_full;
+          rbn = i64min(int64_t(EF_PASS), int64_t(shard_len) - p02);
+          nbase = reinterpret_cast<const uint8_t *>(
+                      base + st2 * stripe_stride) + p02;
+          r0n = 0;
+        }
+        if (rbn > 0) {
+#pragma unroll
+          for (int i = 0; i < 4; i++) {
+            const int64_t off = int64_t(i) * 4096 + lane16;
+            vnext[PIPE ? i : 0] =
+                off + 16 <= rbn
+                    ? *reinterpret_cast<const uint4 *>(nbase + r0n + off)
+                    : uint4{0, 0, 0, 0};
+          }
+        }
       }
 #pragma unroll
       for (int r = 0; r < GM; r++) {
