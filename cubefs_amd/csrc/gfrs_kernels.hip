@@ -61,6 +61,15 @@ __device__ __forceinline__ void store16(uint8_t *p, const uint4 v) {
 
 #define GFRS_DEV __device__ __forceinline__
 
+/* Pointers rebuilt from integers (stripe math, pointer tables) lose the
+ * global address space and compile to flat_load/flat_store with per-access
+ * 64-bit vaddr arithmetic.  A round-trip through an explicit AS(1) cast
+ * lets InferAddressSpaces prove the access is global (global_load_* with
+ * scalar base + 32-bit voffset). */
+GFRS_DEV const uint8_t *as_global(uint64_t p) {
+  return (const uint8_t *)(const __attribute__((address_space(1))) uint8_t *)p;
+}
+
 /* ------------------------------------------------------------------ */
 /* GF(2^8) nibble multiply on packed u32, via v_perm_b32                */
 /* ------------------------------------------------------------------ */
@@ -118,9 +127,8 @@ struct ShardAddr {
   int nptr;
 
   GFRS_DEV const uint8_t *shard(size_t stripe, int idx, size_t shard_len) const {
-    if (ptrs) return reinterpret_cast<const uint8_t *>(ptrs[stripe * nptr + idx]);
-    return reinterpret_cast<const uint8_t *>(base + stripe * stripe_stride +
-                                             uint64_t(idx) * shard_len);
+    if (ptrs) return as_global(ptrs[stripe * nptr + idx]);
+    return as_global(base + stripe * stripe_stride + uint64_t(idx) * shard_len);
   }
 };
 
@@ -1036,8 +1044,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
     const int64_t f = fr - stripe * frames_per_shard;
     const int64_t p0 = f * payload_full;
     const int64_t payload = i64min(payload_full, int64_t(shard_len) - p0);
-    const uint8_t *sbase =
-        reinterpret_cast<const uint8_t *>(base + stripe * stripe_stride);
+    const uint8_t *sbase = as_global(base + stripe * stripe_stride);
 
     /* per-frame CRC partials live in the LDS red slab (4 waves x 16
      * shard slots): no runtime-indexed per-lane array -> no scratch */
@@ -1288,8 +1295,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     const int64_t f = fr - stripe * frames_per_shard;
     const int64_t p0 = f * payload_full;
     const int64_t payload = i64min(payload_full, int64_t(shard_len) - p0);
-    const uint8_t *sbase =
-        reinterpret_cast<const uint8_t *>(base + stripe * stripe_stride);
+    const uint8_t *sbase = as_global(base + stripe * stripe_stride);
 
     uint4 acc[GM][4];
     uint4 vnext[PIPE ? 4 : 1];
@@ -1359,7 +1365,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
             const uint4 v =
                 PIPE ? vcur[PIPE ? i : 0]
                      : *reinterpret_cast<const uint4 *>(src + off);
-            if (SKEL == 0) {
+            if (SKEL != 1 && SKEL != 4) {
 #pragma unroll
               for (int r = 0; r < GM; r++) {
                 const int t2 = (r * k + c) * 2;
@@ -1368,16 +1374,16 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
             }
             uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
             dw[0] = v.x; dw[1] = v.y; dw[2] = v.z; dw[3] = v.w;
-            if (SKEL == 0) t = shift4k(t, stab) ^ crc16_reg(v, tab);
+            if (SKEL != 1 && SKEL != 3) t = shift4k(t, stab) ^ crc16_reg(v, tab);
           }
         }
-        uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
+        uint32_t part = SKEL == 2 ? t : (t ? gf2_mulmod_d(op, t) : 0);
         if (rbytes < EF_PASS && threadIdx.x == 0) {
           /* pass tail ends exactly at the payload: identity fold */
           const int64_t t0 = (rbytes / 16) * 16;
           const uint32_t ct =
               crc_chunk(src + t0, int(rbytes - t0), tab, fdst + t0);
-          if (SKEL == 0) part ^= ct;
+          if (SKEL != 1) part ^= ct;
         }
 #pragma unroll
         for (int sh = 32; sh > 0; sh >>= 1)
@@ -1395,8 +1401,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
           const int64_t st2 = fr2 / frames_per_shard;
           const int64_t p02 = (fr2 - st2 * frames_per_shard) * payload_full;
           rbn = i64min(int64_t(EF_PASS), int64_t(shard_len) - p02);
-          nbase = reinterpret_cast<const uint8_t *>(
-                      base + st2 * stripe_stride) + p02;
+          nbase = as_global(base + st2 * stripe_stride) + p02;
           r0n = 0;
         }
         if (rbn > 0) {
@@ -1422,12 +1427,12 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
             uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
             dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
             dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
-            if (SKEL == 0)
+            if (SKEL != 1 && SKEL != 3)
               t = shift4k(t, stab) ^ crc16_reg(acc[r][i], tab);
           }
         }
-        uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
-        if (SKEL == 0 && rbytes < EF_PASS && threadIdx.x == 0) {
+        uint32_t part = SKEL == 2 ? t : (t ? gf2_mulmod_d(op, t) : 0);
+        if (SKEL != 1 && rbytes < EF_PASS && threadIdx.x == 0) {
           const int64_t t0 = (rbytes / 16) * 16;
           uint32_t ct = 0;
           for (int64_t p = t0; p < rbytes; p++) {
@@ -1478,7 +1483,7 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
    * 24 (16.5); NI=2 variants (142/152/162) target 5-6 blocks/CU. */
   static const int var = []() {
     const char *e = getenv("GFRS_EF");
-    const int v = e ? atoi(e) : 14;
+    const int v = e ? atoi(e) : 76; /* register-CRC + load lookahead */
     switch (v) {
       case 13: case 14: case 23: case 24:
       case 74: case 75: case 76:
@@ -1516,8 +1521,21 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
     }();
     static const int rabl = []() {
       const char *e = getenv("GFRS_EF_ABL");
-      return e && atoi(e) == 4 ? 4 : 0;
+      const int v = e ? atoi(e) : 0;
+      return (v >= 4 && v <= 7) ? v : 0;
     }();
+    if (rabl >= 5 && var == 76 && m == 3) { /* phase diagnostics */
+      const int lds5 = 12288 + EF_RED + m * k * 32;
+      if (rabl == 5) { hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 2, 1>),
+          dim3(grid), dim3(CRC_BLOCKT), lds5, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps); return; }
+      if (rabl == 6) { hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 3, 1>),
+          dim3(grid), dim3(CRC_BLOCKT), lds5, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps); return; }
+      if (rabl == 7) { hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 4, 1>),
+          dim3(grid), dim3(CRC_BLOCKT), lds5, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps); return; }
+    }
     if (rabl == 4) { /* memory-floor skeleton (diagnostic only) */
       switch (m) {
         case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 4, 1, 1>),
