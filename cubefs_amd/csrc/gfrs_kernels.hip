@@ -1262,6 +1262,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     reinterpret_cast<uint4 *>(ctab)[i] =
         reinterpret_cast<const uint4 *>(tabs)[i];
   const int64_t lane16 = int64_t(threadIdx.x) * 16;
+  const int lane16i = int(threadIdx.x) * 16;
   /* fold operator for full frames: suffix after the lane's LAST present
    * piece in pass h (pieces are a prefix in i since offsets grow) */
   uint32_t op_full[EF_PASSES];
@@ -1290,9 +1291,16 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     frN = i64min(fr0 + nper, total_frames);
     frS = 1;
   }
-  for (int64_t fr = fr0; fr < frN; fr += frS) {
-    const int64_t stripe = fr / frames_per_shard;
-    const int64_t f = fr - stripe * frames_per_shard;
+  /* track (stripe, f) incrementally: one i64 division per block instead
+   * of a ~100-instruction software divide per frame iteration */
+  int64_t stripe = fr0 / frames_per_shard;
+  int64_t f = fr0 - stripe * frames_per_shard;
+  const int64_t dstripe = frS / frames_per_shard;
+  const int64_t drem = frS - dstripe * frames_per_shard;
+  for (int64_t fr = fr0; fr < frN; fr += frS,
+               stripe += dstripe, f += drem,
+               (f >= frames_per_shard ? (f -= frames_per_shard, ++stripe)
+                                      : int64_t(0))) {
     const int64_t p0 = f * payload_full;
     const int64_t payload = i64min(payload_full, int64_t(shard_len) - p0);
     const uint8_t *sbase = as_global(base + stripe * stripe_stride);
@@ -1301,10 +1309,10 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     uint4 vnext[PIPE ? 4 : 1];
     if (PIPE && fr == fr0) { /* later frames are prefetched by the
                                 previous frame's last pass */
-      const int64_t rb0 = i64min(int64_t(EF_PASS), payload);
+      const int rb0 = int(i64min(int64_t(EF_PASS), payload));
 #pragma unroll
       for (int i = 0; i < 4; i++) {
-        const int64_t off = int64_t(i) * 4096 + lane16;
+        const int off = i * 4096 + lane16i;
         vnext[PIPE ? i : 0] =
             off + 16 <= rb0
                 ? *reinterpret_cast<const uint4 *>(sbase + p0 + off)
@@ -1334,6 +1342,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
         op = x8n_d(uint64_t(payload - end));
       }
 
+      const int rbi = int(rbytes);
       for (int c = 0; c < k; c++) {
         const uint8_t *src = sbase + size_t(c) * shard_len + p0 + r0;
         uint8_t *fdst = dst + (stripe * (k + GM) + c) * dst_stride +
@@ -1346,9 +1355,9 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
             const uint8_t *nsrc = sbase + size_t(c + 1) * shard_len + p0 + r0;
 #pragma unroll
             for (int i = 0; i < 4; i++) {
-              const int64_t off = int64_t(i) * 4096 + lane16;
+              const int off = i * 4096 + lane16i;
               vnext[PIPE ? i : 0] =
-                  off + 16 <= rbytes
+                  off + 16 <= rbi
                       ? *reinterpret_cast<const uint4 *>(nsrc + off)
                       : uint4{0, 0, 0, 0};
             }
@@ -1360,8 +1369,8 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
         uint32_t t = 0;
 #pragma unroll
         for (int i = 0; i < 4; i++) {
-          const int64_t off = int64_t(i) * 4096 + lane16;
-          if (off + 16 <= rbytes) {
+          const int off = i * 4096 + lane16i;
+          if (off + 16 <= rbi) {
             const uint4 v =
                 PIPE ? vcur[PIPE ? i : 0]
                      : *reinterpret_cast<const uint4 *>(src + off);
@@ -1405,11 +1414,12 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
           r0n = 0;
         }
         if (rbn > 0) {
+          const int rbni = int(rbn);
 #pragma unroll
           for (int i = 0; i < 4; i++) {
-            const int64_t off = int64_t(i) * 4096 + lane16;
+            const int off = i * 4096 + lane16i;
             vnext[PIPE ? i : 0] =
-                off + 16 <= rbn
+                off + 16 <= rbni
                     ? *reinterpret_cast<const uint4 *>(nbase + r0n + off)
                     : uint4{0, 0, 0, 0};
           }
@@ -1422,8 +1432,8 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
         uint32_t t = 0;
 #pragma unroll
         for (int i = 0; i < 4; i++) {
-          const int64_t off = int64_t(i) * 4096 + lane16;
-          if (off + 16 <= rbytes) {
+          const int off = i * 4096 + lane16i;
+          if (off + 16 <= rbi) {
             uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
             dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
             dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
