@@ -91,7 +91,9 @@ GFRS_DEV uint32_t lut16(const uint4 t, uint32_t nib) {
     uint32_t b = __builtin_amdgcn_perm(t.w, t.z, s);
     uint32_t m = (nib & 0x08080808u) >> 3;
     m *= 0xFFu; /* per-byte 0x00/0xFF, no carries */
-    return (a & ~m) | (b & m);
+    /* one v_bitop3_b32: imm bit index = src0*4 + src1*2 + src2, so
+     * select(m ? b : a) = 0xD8 (bits 3,4,6,7) */
+    return __builtin_amdgcn_bitop3_b32(a, b, m, 0xD8);
   }
 }
 
@@ -517,6 +519,10 @@ GFRS_DEV uint32_t x8n_d(uint64_t len) {
 
 constexpr int CRC_BLOCKT = 256;
 
+GFRS_DEV uint32_t xor3(uint32_t a, uint32_t b, uint32_t c) {
+  return __builtin_amdgcn_bitop3_b32(a, b, c, 0x96); /* a ^ b ^ c */
+}
+
 /* Raw (no init/final complement) CRC update of a chunk, 4 B at a time via
  * LDS slice-by-4 tables, byte tail scalar.  When dst != nullptr the chunk
  * is simultaneously copied there (the frame/strip move fused into the CRC
@@ -554,10 +560,11 @@ GFRS_DEV uint32_t crc_chunk16(const uint8_t *p, int len,
     for (int j = 0; j < 2; j++) {
       const uint32_t w0 = (j == 0 ? q.x : q.z) ^ c;
       const uint32_t w1 = (j == 0 ? q.y : q.w);
-      c = tab[7][w0 & 0xFF] ^ tab[6][(w0 >> 8) & 0xFF] ^
-          tab[5][(w0 >> 16) & 0xFF] ^ tab[4][w0 >> 24] ^
-          tab[3][w1 & 0xFF] ^ tab[2][(w1 >> 8) & 0xFF] ^
-          tab[1][(w1 >> 16) & 0xFF] ^ tab[0][w1 >> 24];
+      c = xor3(xor3(tab[7][w0 & 0xFF], tab[6][(w0 >> 8) & 0xFF],
+                    tab[5][(w0 >> 16) & 0xFF]),
+               xor3(tab[4][w0 >> 24], tab[3][w1 & 0xFF],
+                    tab[2][(w1 >> 8) & 0xFF]),
+               tab[1][(w1 >> 16) & 0xFF] ^ tab[0][w1 >> 24]);
     }
   }
   for (; i + 4 <= len; i += 4) {
@@ -1205,17 +1212,19 @@ GFRS_DEV uint32_t crc16_reg(const uint4 q, const uint32_t (*tab)[256]) {
   for (int j = 0; j < 2; j++) {
     const uint32_t w0 = (j == 0 ? q.x : q.z) ^ c;
     const uint32_t w1 = (j == 0 ? q.y : q.w);
-    c = tab[7][w0 & 0xFF] ^ tab[6][(w0 >> 8) & 0xFF] ^
-        tab[5][(w0 >> 16) & 0xFF] ^ tab[4][w0 >> 24] ^
-        tab[3][w1 & 0xFF] ^ tab[2][(w1 >> 8) & 0xFF] ^
-        tab[1][(w1 >> 16) & 0xFF] ^ tab[0][w1 >> 24];
+    c = xor3(xor3(tab[7][w0 & 0xFF], tab[6][(w0 >> 8) & 0xFF],
+                  tab[5][(w0 >> 16) & 0xFF]),
+             xor3(tab[4][w0 >> 24], tab[3][w1 & 0xFF],
+                  tab[2][(w1 >> 8) & 0xFF]),
+             tab[1][(w1 >> 16) & 0xFF] ^ tab[0][w1 >> 24]);
   }
   return c;
 }
 
 GFRS_DEV uint32_t shift4k(uint32_t c, const uint32_t (*stab)[256]) {
-  return stab[0][c & 0xFF] ^ stab[1][(c >> 8) & 0xFF] ^
-         stab[2][(c >> 16) & 0xFF] ^ stab[3][c >> 24];
+  return xor3(stab[0][c & 0xFF], stab[1][(c >> 8) & 0xFF],
+              stab[2][(c >> 16) & 0xFF]) ^
+         stab[3][c >> 24];
 }
 
 /* Register-CRC variant of the fused encode+frame kernel: no LDS stage at
