@@ -1272,20 +1272,14 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
         reinterpret_cast<const uint4 *>(tabs)[i];
   const int64_t lane16 = int64_t(threadIdx.x) * 16;
   const int lane16i = int(threadIdx.x) * 16;
-  /* fold operator for full frames: suffix after the lane's LAST present
-   * piece in pass h (pieces are a prefix in i since offsets grow) */
-  uint32_t op_full[EF_PASSES];
-#pragma unroll
-  for (int h = 0; h < EF_PASSES; h++) {
-    const int64_t r0 = int64_t(h) * EF_PASS;
-    const int64_t rb = i64min(int64_t(EF_PASS), payload_full - r0);
-    int np = 0;
-#pragma unroll
-    for (int i = 0; i < 4; i++)
-      if (int64_t(i) * 4096 + lane16 + 16 <= rb) np = i + 1;
-    const int64_t end = np ? r0 + int64_t(np - 1) * 4096 + lane16 + 16 : r0;
-    op_full[h] = x8n_d(uint64_t(payload_full - end));
-  }
+  /* Full-frame fold operator, chained: op for pass h+1 = op for pass h
+   * times x^(-8*16384) (constant 0x479933FC, precomputed inverse).  Only
+   * lane 255 in the last pass has 3 pieces instead of 4; its operator is
+   * the chained value times x^(8*4096), i.e. one shift4k.  One register
+   * instead of an op_full[4] array. */
+  constexpr uint32_t INV16K = 0x479933FCu;
+  const uint32_t op_first =
+      x8n_d(uint64_t(payload_full - (3 * 4096 + lane16 + 16)));
   __syncthreads();
 
   const uint4 *ltab = reinterpret_cast<const uint4 *>(ctab);
@@ -1331,6 +1325,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     for (int j = threadIdx.x; j < 64; j += CRC_BLOCKT) red[j] = 0;
     __syncthreads();
 
+    uint32_t op_chain = op_first;
     for (int h = 0; h < EF_PASSES; h++) {
       const int64_t r0 = int64_t(h) * EF_PASS;
       const int64_t rbytes = i64min(int64_t(EF_PASS), payload - r0);
@@ -1340,7 +1335,10 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
 #pragma unroll
         for (int i = 0; i < 4; i++) acc[r][i] = uint4{0, 0, 0, 0};
 
-      uint32_t op = op_full[h];
+      uint32_t op = op_chain;
+      if (h == EF_PASSES - 1 && threadIdx.x == 255)
+        op = shift4k(op, stab); /* lane 255's last pass has 3 pieces */
+      op_chain = gf2_mulmod_d(op_chain, INV16K);
       if (payload != payload_full) {
         int np = 0;
 #pragma unroll
@@ -1502,15 +1500,33 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
    * 24 (16.5); NI=2 variants (142/152/162) target 5-6 blocks/CU. */
   static const int var = []() {
     const char *e = getenv("GFRS_EF");
-    const int v = e ? atoi(e) : 76; /* register-CRC + load lookahead */
+    const int v = e ? atoi(e) : 77; /* register-CRC + load lookahead */
     switch (v) {
       case 13: case 14: case 23: case 24:
-      case 74: case 75: case 76:
+      case 74: case 75: case 76: case 77:
       case 142: case 152: case 162: return v;
       default: return 14;
     }
   }();
   /* 7x = register-CRC kernel (no stage): crc tabs + shift tabs + red */
+  if (var == 77) { /* lookahead pipeline squeezed to 4 waves/SIMD */
+    const int lds = 12288 + EF_RED + m * k * 32;
+    switch (m) {
+      case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 4, 0, 0, 1>),
+          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps); break;
+      case 2: hipLaunchKernelGGL((rs_encode_frame_reg_k<2, 4, 0, 0, 1>),
+          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps); break;
+      case 3: hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 4, 0, 0, 1>),
+          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps); break;
+      default: hipLaunchKernelGGL((rs_encode_frame_reg_k<4, 4, 0, 0, 1>),
+          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps);
+    }
+    return;
+  }
   if (var == 76) { /* one-unit-lookahead pipeline at 3 waves/SIMD */
     const int lds = 12288 + EF_RED + m * k * 32;
     switch (m) {
