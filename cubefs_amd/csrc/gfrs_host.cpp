@@ -901,10 +901,19 @@ int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
    * the composed plan: every global AND local parity is a row over the n
    * data shards. */
   const int gm_all = t.m + t.l;
+  /* Measured crossover (pipelined fused kernel, RS(6+3), fixed 4 GiB
+   * source): fused wins from 8 KiB shards up (837 vs 306 GiB/s at 8 KiB,
+   * ~1180 vs ~780 at 16 KiB-8 MiB); the two-kernel path with rs_apply
+   * stripe packing wins at 2-4 KiB (684 vs 320 at 2 KiB).
+   * GFRS_FUSED_MIN overrides the threshold (bytes). */
+  static const size_t fused_min = []() {
+    const char *e = getenv("GFRS_FUSED_MIN");
+    const long v = e ? atol(e) : 0;
+    return v > 0 ? size_t(v) : size_t(6144);
+  }();
   const bool fused = block_len == 65536 && t.m >= 1 && gm_all <= 4 &&
                      t.n + gm_all <= 16 && framed_stride % 4 == 0 &&
-                     shard_len >= size_t(4) * 65532 &&
-                     (t.l == 0 || c->fused_lrc_ok);
+                     shard_len >= fused_min && (t.l == 0 || c->fused_lrc_ok);
   if (fused) {
     const DevPlan &pl = t.l == 0 ? c->enc_plan : c->fused_lrc;
     launch_rs_encode_frame((uint8_t *)framed, framed_stride, (uint64_t)base,
