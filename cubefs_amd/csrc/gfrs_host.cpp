@@ -25,6 +25,7 @@
 #include <mutex>
 #include <set>
 #include <string>
+#include <algorithm>
 #include <vector>
 
 namespace gfrs {
@@ -109,9 +110,10 @@ struct DevPlan {
   int k = 0, nout = 0;
   int upload(const std::vector<int32_t> &in, const std::vector<int32_t> &out,
              const std::vector<uint8_t> &rows /* nout*k coefficients */,
-             hipStream_t s) {
+             hipStream_t s, int rowk = -1 /* row width; in may carry extra
+             (cmp) indices after the first rowk inputs */) {
     const GfTables &t = gft();
-    k = int(in.size());
+    k = rowk >= 0 ? rowk : int(in.size());
     nout = int(out.size());
     std::vector<uint8_t> tb(size_t(nout) * k * 32);
     for (int r = 0; r < nout; r++)
@@ -1294,7 +1296,154 @@ int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
                       const uint64_t *bids, const uint64_t *vuids,
                       uint64_t *fail_bitmap) {
   if (nbad <= 0) return GFRS_ERR_INVALID_SHARDS;
-  /* reconstruct + the mandatory verify in one data pass */
+  auto *cc = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  {
+    const gfrs_tactic &t = cc->t;
+    /* Fused repair: one kernel reads the k inputs + surviving-parity
+     * check shards and writes ONLY the repaired shards' framed bodies -
+     * the raw reconstruction never touches HBM and `base` is NOT
+     * mutated (the legacy path below reconstructs in place). */
+    static const size_t fmin = []() {
+      const char *e = getenv("GFRS_FUSED_MIN");
+      const long v = e ? atol(e) : 0;
+      return v > 0 ? size_t(v) : size_t(6144);
+    }();
+    if (t.l == 0 && block_len == 65536 && t.m >= 1 && t.m <= 4 &&
+        t.n + t.m <= 16 && nbad <= t.m && shard_len >= fmin &&
+        dst_stride % 4 == 0 && nstripes > 0) {
+      const int k = t.n, m = t.m;
+      std::vector<uint8_t> present(k + m, 1);
+      std::vector<int> badv(bad_idx, bad_idx + nbad);
+      for (int i = 0; i < nbad; i++) {
+        if (badv[i] < 0 || badv[i] >= k + m) return GFRS_ERR_INVALID_SHARDS;
+        present[badv[i]] = 0;
+      }
+      std::vector<int> sbad = badv;
+      std::sort(sbad.begin(), sbad.end());
+      uint32_t colpack = 0;
+      for (int r = 0; r < nbad; r++) {
+        int col = int(std::find(badv.begin(), badv.end(), sbad[r]) -
+                      badv.begin());
+        colpack |= uint32_t(col) << (4 * r);
+      }
+      std::lock_guard<std::mutex> lk(cc->mu);
+      StreamGuard g(cc);
+      uint64_t key = 0;
+      for (int i = 0; i < k + m; i++)
+        if (!present[i]) key |= 1ull << i;
+      key = (key << 6) | 61; /* fused-repair namespace */
+      DevPlan *plan = nullptr;
+      {
+        auto it = cc->dec_cache.find(key);
+        if (it != cc->dec_cache.end()) plan = it->second;
+      }
+      if (!plan) {
+        std::vector<int> valid;
+        for (int i = 0; i < k + m && int(valid.size()) < k; i++)
+          if (present[i]) valid.push_back(i);
+        if (int(valid.size()) < k) return GFRS_ERR_TOO_FEW_SHARDS;
+        std::vector<uint8_t> sub(size_t(k) * k), dec(size_t(k) * k);
+        for (int r = 0; r < k; r++)
+          memcpy(&sub[size_t(r) * k], &cc->enc_matrix[size_t(valid[r]) * k],
+                 k);
+        if (!gf_invert(sub.data(), k, dec.data()))
+          return GFRS_ERR_SINGULAR;
+        std::vector<int> slot(k, -1);
+        for (int j = 0; j < k; j++)
+          if (valid[j] < k) slot[valid[j]] = j;
+        const GfTables &gt2 = gft();
+        auto parity_row = [&](int p2, std::vector<uint8_t> &row) {
+          row.assign(k, 0);
+          for (int d = 0; d < k; d++) {
+            const uint8_t coef = cc->enc_matrix[size_t(p2) * k + d];
+            if (present[d]) {
+              row[slot[d]] ^= coef;
+            } else {
+              for (int j = 0; j < k; j++)
+                row[j] ^= gt2.mul[coef][dec[size_t(d) * k + j]];
+            }
+          }
+        };
+        std::vector<int32_t> in, out;
+        std::vector<uint8_t> rows, row;
+        for (int j = 0; j < k; j++) in.push_back(valid[j]);
+        /* rebuild rows in sorted-bad order */
+        for (int b : sbad) {
+          out.push_back(b);
+          if (b < k) {
+            rows.insert(rows.end(), &dec[size_t(b) * k],
+                        &dec[size_t(b) * k + k]);
+          } else {
+            parity_row(b, row);
+            rows.insert(rows.end(), row.begin(), row.end());
+          }
+        }
+        /* check rows: surviving parity NOT already an input */
+        for (int p2 = k; p2 < k + m; p2++) {
+          if (!present[p2]) continue;
+          if (std::find(valid.begin(), valid.end(), p2) != valid.end())
+            continue; /* input parity: its check is the identity */
+          out.push_back(p2);
+          parity_row(p2, row);
+          rows.insert(rows.end(), row.begin(), row.end());
+          in.push_back(p2); /* cmp shard index rides after the k inputs */
+        }
+        plan = new DevPlan();
+        int rc2 = plan->upload(in, out, rows, cc->stream, k);
+        if (rc2 != GFRS_OK) {
+          delete plan;
+          return rc2;
+        }
+        cc->dec_cache[key] = plan;
+      }
+      const int gm = plan->nout;
+      int rc2;
+      if ((rc2 = cc->fail_buf.ensure(size_t(nstripes) * 4)) != GFRS_OK)
+        return rc2;
+      HIP_TRY(hipMemsetAsync(cc->fail_buf.p, 0, size_t(nstripes) * 4,
+                             cc->stream));
+      /* headers for every (stripe, bad) image, caller's column order */
+      std::vector<uint8_t> hdrs(size_t(nstripes) * nbad * 32, 0);
+      for (size_t j = 0; j < size_t(nstripes) * nbad; j++) {
+        uint8_t *h = &hdrs[j * 32];
+        h[4] = 0xab; h[5] = 0xcd; h[6] = 0xef; h[7] = 0xcc;
+        put_be64(h + 8, bids[j]);
+        put_be64(h + 16, vuids[j]);
+        put_be32(h + 24, uint32_t(shard_len));
+        put_be32(h, host_crc32(h + 4, 28));
+      }
+      if ((rc2 = cc->stage_pin.ensure(hdrs.size())) != GFRS_OK) return rc2;
+      memcpy(cc->stage_pin.p, hdrs.data(), hdrs.size());
+      DevBuf &hbuf = cc->ptr_buf;
+      if ((rc2 = hbuf.ensure(hdrs.size())) != GFRS_OK) return rc2;
+      HIP_TRY(hipMemcpyAsync(hbuf.p, cc->stage_pin.p, hdrs.size(),
+                             hipMemcpyHostToDevice, cc->stream));
+      launch_rs_repair_frame((uint8_t *)disk_dst + 32, dst_stride,
+                             (uint64_t)base, stripe_stride, shard_len,
+                             plan->k, gm, nbad,
+                             (const int32_t *)plan->in_idx.p,
+                             (const uint8_t *)plan->tabs.p, colpack,
+                             (uint32_t *)cc->fail_buf.p, nstripes,
+                             cc->stream);
+      launch_shard_finalize((uint8_t *)disk_dst, dst_stride,
+                            (const uint8_t *)hbuf.p, int64_t(shard_len),
+                            block_len, nstripes * nbad, cc->stream);
+      hipError_t e = hipGetLastError();
+      if (e != hipSuccess) return hip_fail("repair_frame launch", e);
+      std::vector<uint32_t> fails(nstripes);
+      HIP_TRY(hipMemcpyAsync(fails.data(), cc->fail_buf.p,
+                             size_t(nstripes) * 4, hipMemcpyDeviceToHost,
+                             cc->stream));
+      HIP_TRY(hipStreamSynchronize(cc->stream));
+      if (fail_bitmap) {
+        memset(fail_bitmap, 0, ((nstripes + 63) / 64) * 8);
+        for (int s2 = 0; s2 < nstripes; s2++)
+          if (fails[s2]) fail_bitmap[s2 / 64] |= 1ull << (s2 % 64);
+      }
+      return GFRS_OK;
+    }
+  }
+  /* legacy path: reconstruct + the mandatory verify in one data pass */
   int rc = gfrs_reconstruct_verify_batch(ctx, base, shard_len, stripe_stride,
                                          nstripes, bad_idx, nbad,
                                          fail_bitmap);

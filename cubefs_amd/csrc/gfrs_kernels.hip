@@ -1486,6 +1486,270 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
   }
 }
 
+/* ------------------------------------------------------------------ */
+/* fused repair: reconstruct + verify + crc32block-framed images        */
+/* ------------------------------------------------------------------ */
+
+/* The blobnode repair tasklet (worker_slice_recover.go) reconstructs the
+ * lost shards from k survivors, verifies parity consistency, and writes
+ * the repaired shards as framed disk images.  As three passes that is
+ * read k+ncmp, write nbad raw, read nbad, write nbad framed.  Fused:
+ * the k inputs and ncmp check shards are read ONCE and only the nbad
+ * framed bodies are written - the raw reconstruction never exists in
+ * HBM.  Rows 0..nw-1 rebuild lost shards (framed output; image column
+ * from colpack, 4 bits per row); rows nw.. are surviving-parity checks
+ * (compared against the shard itself, mismatch -> fail[stripe]).
+ * Same op-chain CRC machinery as rs_encode_frame_reg_k. */
+template <int GM>
+__global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
+    uint8_t *__restrict__ dst /* first image body (+32 into the image) */,
+    size_t dst_stride /* whole-image stride */, uint64_t base,
+    uint64_t stripe_stride, size_t shard_len, int k,
+    const int32_t *__restrict__ imap /* k inputs then GM-nw cmp shards */,
+    const uint8_t *__restrict__ tabs /* [GM*k][32] */, int nw,
+    uint32_t colpack, uint32_t *__restrict__ fail, int64_t total_frames,
+    int64_t frames_per_shard) {
+  constexpr int EF_PASS = 16384;
+  constexpr int EF_PASSES = 4;
+  constexpr int64_t block_len = 65536;
+  constexpr int64_t payload_full = block_len - CRC_LEN;
+  extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
+  uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
+  uint32_t(*stab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem + 8192);
+  uint32_t *red = reinterpret_cast<uint32_t *>(smem + 12288);
+  uint8_t *ctab = smem + 12288 + EF_RED;
+  for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
+    (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
+  for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
+    (&stab[0][0])[i] = (&g_shift4k[0][0])[i];
+  for (int i = threadIdx.x; i < GM * k * 2; i += CRC_BLOCKT)
+    reinterpret_cast<uint4 *>(ctab)[i] =
+        reinterpret_cast<const uint4 *>(tabs)[i];
+  const int64_t lane16 = int64_t(threadIdx.x) * 16;
+  const int lane16i = int(threadIdx.x) * 16;
+  constexpr uint32_t INV16K = 0x479933FCu;
+  const uint32_t op_first =
+      x8n_d(uint64_t(payload_full - (3 * 4096 + lane16 + 16)));
+  __syncthreads();
+
+  const uint4 *ltab = reinterpret_cast<const uint4 *>(ctab);
+
+  int64_t stripe = int64_t(blockIdx.x) / frames_per_shard;
+  int64_t f = int64_t(blockIdx.x) - stripe * frames_per_shard;
+  const int64_t dstripe = int64_t(gridDim.x) / frames_per_shard;
+  const int64_t drem = int64_t(gridDim.x) - dstripe * frames_per_shard;
+  for (int64_t fr = blockIdx.x; fr < total_frames; fr += gridDim.x,
+               stripe += dstripe, f += drem,
+               (f >= frames_per_shard ? (f -= frames_per_shard, ++stripe)
+                                      : int64_t(0))) {
+    const int64_t p0 = f * payload_full;
+    const int64_t payload = i64min(payload_full, int64_t(shard_len) - p0);
+    const uint8_t *sbase = as_global(base + stripe * stripe_stride);
+
+    uint4 acc[GM][4];
+    uint4 vnext[4];
+    { /* prefetch pass 0, unit 0 */
+      const int rb0 = int(i64min(int64_t(EF_PASS), payload));
+      const uint8_t *src0 = sbase + size_t(imap[0]) * shard_len + p0;
+#pragma unroll
+      for (int i = 0; i < 4; i++) {
+        const int off = i * 4096 + lane16i;
+        vnext[i] = off + 16 <= rb0
+                       ? *reinterpret_cast<const uint4 *>(src0 + off)
+                       : uint4{0, 0, 0, 0};
+      }
+    }
+    for (int j = threadIdx.x; j < 64; j += CRC_BLOCKT) red[j] = 0;
+    __syncthreads();
+
+    uint32_t mismatch = 0;
+    uint32_t op_chain = op_first;
+    for (int h = 0; h < EF_PASSES; h++) {
+      const int64_t r0 = int64_t(h) * EF_PASS;
+      const int64_t rbytes = i64min(int64_t(EF_PASS), payload - r0);
+      if (rbytes <= 0) break;
+#pragma unroll
+      for (int r = 0; r < GM; r++)
+#pragma unroll
+        for (int i = 0; i < 4; i++) acc[r][i] = uint4{0, 0, 0, 0};
+
+      uint32_t op = op_chain;
+      if (h == EF_PASSES - 1 && threadIdx.x == 255)
+        op = shift4k(op, stab);
+      op_chain = gf2_mulmod_d(op_chain, INV16K);
+      if (payload != payload_full) {
+        int np = 0;
+#pragma unroll
+        for (int i = 0; i < 4; i++)
+          if (int64_t(i) * 4096 + lane16 + 16 <= rbytes) np = i + 1;
+        const int64_t end =
+            np ? r0 + int64_t(np - 1) * 4096 + lane16 + 16 : r0;
+        op = x8n_d(uint64_t(payload - end));
+      }
+
+      const int rbi = int(rbytes);
+      for (int c = 0; c < k; c++) {
+        uint4 vcur[4];
+#pragma unroll
+        for (int i = 0; i < 4; i++) vcur[i] = vnext[i];
+        if (c + 1 < k) { /* next input's loads fly over this MAC */
+          const uint8_t *nsrc =
+              sbase + size_t(imap[c + 1]) * shard_len + p0 + r0;
+#pragma unroll
+          for (int i = 0; i < 4; i++) {
+            const int off = i * 4096 + lane16i;
+            vnext[i] = off + 16 <= rbi
+                           ? *reinterpret_cast<const uint4 *>(nsrc + off)
+                           : uint4{0, 0, 0, 0};
+          }
+        }
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+          const int off = i * 4096 + lane16i;
+          if (off + 16 <= rbi) {
+#pragma unroll
+            for (int r = 0; r < GM; r++) {
+              const int t2 = (r * k + c) * 2;
+              gfmac16<false>(acc[r][i], vcur[i], ltab[t2], ltab[t2 + 1]);
+            }
+          }
+        }
+      }
+      { /* prefetch next pass's (or frame's) unit-0 loads */
+        int64_t r0n = r0 + EF_PASS;
+        int64_t rbn = i64min(int64_t(EF_PASS), payload - r0n);
+        const uint8_t *nbase = sbase + size_t(imap[0]) * shard_len + p0;
+        if (rbn <= 0 && fr + gridDim.x < total_frames) {
+          const int64_t fr2 = fr + gridDim.x;
+          const int64_t st2 = fr2 / frames_per_shard;
+          const int64_t p02 = (fr2 - st2 * frames_per_shard) * payload_full;
+          rbn = i64min(int64_t(EF_PASS), int64_t(shard_len) - p02);
+          nbase = as_global(base + st2 * stripe_stride) +
+                  size_t(imap[0]) * shard_len + p02;
+          r0n = 0;
+        }
+        if (rbn > 0) {
+          const int rbni = int(rbn);
+#pragma unroll
+          for (int i = 0; i < 4; i++) {
+            const int off = i * 4096 + lane16i;
+            vnext[i] = off + 16 <= rbni
+                           ? *reinterpret_cast<const uint4 *>(nbase + r0n + off)
+                           : uint4{0, 0, 0, 0};
+          }
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < GM; r++) {
+        if (r < nw) { /* rebuild row: framed image output */
+          const int col = int((colpack >> (4 * r)) & 0xF);
+          uint8_t *fdst = dst + (stripe * nw + col) * dst_stride +
+                          f * block_len + CRC_LEN + r0;
+          uint32_t t = 0;
+#pragma unroll
+          for (int i = 0; i < 4; i++) {
+            const int off = i * 4096 + lane16i;
+            if (off + 16 <= rbi) {
+              uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+              dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
+              dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
+              t = shift4k(t, stab) ^ crc16_reg(acc[r][i], tab);
+            }
+          }
+          uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
+          if (rbytes < EF_PASS && threadIdx.x == 0) {
+            const int64_t t0 = (rbytes / 16) * 16;
+            uint32_t ct = 0;
+            for (int64_t p = t0; p < rbytes; p++) {
+              uint8_t pv = 0;
+              for (int c2 = 0; c2 < k; c2++) {
+                const uint8_t b2 =
+                    sbase[size_t(imap[c2]) * shard_len + p0 + r0 + p];
+                const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
+                pv ^= tt[b2 & 0x0F] ^ tt[16 + (b2 >> 4)];
+              }
+              fdst[p] = pv;
+              ct = tab[0][(ct ^ pv) & 0xFF] ^ (ct >> 8);
+            }
+            part ^= ct;
+          }
+#pragma unroll
+          for (int sh = 32; sh > 0; sh >>= 1)
+            part ^= __shfl_xor(part, sh, 64);
+          if ((threadIdx.x & 63) == 0)
+            red[(threadIdx.x >> 6) * 16 + r] ^= part;
+        } else { /* check row: compare against the surviving parity */
+          const uint8_t *cshard =
+              sbase + size_t(imap[k + (r - nw)]) * shard_len + p0 + r0;
+          uint32_t d = 0;
+#pragma unroll
+          for (int i = 0; i < 4; i++) {
+            const int off = i * 4096 + lane16i;
+            if (off + 16 <= rbi) {
+              const uint4 w =
+                  *reinterpret_cast<const uint4 *>(cshard + off);
+              d |= (w.x ^ acc[r][i].x) | (w.y ^ acc[r][i].y) |
+                   (w.z ^ acc[r][i].z) | (w.w ^ acc[r][i].w);
+            }
+          }
+          if (rbytes < EF_PASS && threadIdx.x == 0) {
+            const int64_t t0 = (rbytes / 16) * 16;
+            for (int64_t p = t0; p < rbytes; p++) {
+              uint8_t pv = 0;
+              for (int c2 = 0; c2 < k; c2++) {
+                const uint8_t b2 =
+                    sbase[size_t(imap[c2]) * shard_len + p0 + r0 + p];
+                const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
+                pv ^= tt[b2 & 0x0F] ^ tt[16 + (b2 >> 4)];
+              }
+              d |= uint32_t(pv ^ cshard[p]);
+            }
+          }
+          mismatch |= d;
+        }
+      }
+    }
+
+    /* fail flag + the 4 B LE frame headers of the rebuilt images */
+#pragma unroll
+    for (int sh = 32; sh > 0; sh >>= 1)
+      mismatch |= __shfl_xor(mismatch, sh, 64);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      const uint32_t it = gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
+      for (int r = 0; r < nw; r++) {
+        const int col = int((colpack >> (4 * r)) & 0xF);
+        const uint32_t crc =
+            ~(it ^ red[r] ^ red[16 + r] ^ red[32 + r] ^ red[48 + r]);
+        *reinterpret_cast<uint32_t *>(
+            dst + (stripe * nw + col) * dst_stride + f * block_len) = crc;
+      }
+    }
+    if ((threadIdx.x & 63) == 0 && mismatch)
+      atomicOr(&fail[stripe], 1u);
+    __syncthreads();
+  }
+}
+
+void launch_rs_repair_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
+                            uint64_t stripe_stride, size_t shard_len, int k,
+                            int gm, int nw, const int32_t *imap,
+                            const uint8_t *tabs, uint32_t colpack,
+                            uint32_t *fail, int nstripes, hipStream_t s) {
+  const int64_t fps = (int64_t(shard_len) + 65531) / 65532;
+  const int64_t total = fps * nstripes;
+  const int grid = crc_grid(total);
+  const int lds = 12288 + EF_RED + gm * k * 32;
+#define GFRS_RP_GO(G)                                                       hipLaunchKernelGGL((rs_repair_frame_k<G>), dim3(grid),                                       dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,                          stripe_stride, shard_len, k, imap, tabs, nw,                              colpack, fail, total, fps)
+  switch (gm) {
+    case 1: GFRS_RP_GO(1); break;
+    case 2: GFRS_RP_GO(2); break;
+    case 3: GFRS_RP_GO(3); break;
+    default: GFRS_RP_GO(4);
+  }
+#undef GFRS_RP_GO
+}
+
 void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
                             uint64_t stripe_stride, size_t shard_len, int k,
                             int m, const uint8_t *tabs, int nstripes,

@@ -349,3 +349,62 @@ def test_lrc_reconstruct_data_only(oracle, dev):
     got = cpu_copy(shards)
     assert np.array_equal(got[5], ref[5])          # data restored
     assert not got[12].any() and not got[15].any()  # parities untouched
+
+
+@pytest.mark.parametrize("bad", [[7, 2], [8], [1, 6, 7], [6, 7, 8]])
+def test_repair_batch_images(oracle, dev, bad):
+    """Fused repair tasklet: framed disk images of the lost shards are
+    byte-identical to oracle-built images of the ORIGINAL shards, in the
+    caller's (unsorted) bad order (worker_slice_recover.go:804-888)."""
+    import torch
+    from cubefs_amd import codemode, ec, shard
+    t = codemode.get_tactic("EC6P3")
+    enc = ec.Encoder(t)
+    ns, slen = 4, 200_000
+    rng = np.random.default_rng(700 + len(bad))
+    arr = rng.integers(0, 256, (ns, t.total, slen), dtype=np.uint8)
+    batch = torch.from_numpy(arr).to(dev)
+    enc.encode_batch(batch)
+    enc.synchronize()
+    ref = batch.cpu().numpy()
+    for i in bad:
+        batch[:, i].zero_()
+    nb = len(bad)
+    dsz = shard.disk_size(slen)
+    imgs = torch.zeros((ns * nb, dsz), dtype=torch.uint8, device=dev)
+    bids = [9000 + s * nb + b for s in range(ns) for b in range(nb)]
+    vuids = [77] * (ns * nb)
+    fails = enc.repair_batch(batch, bad, imgs, bids, vuids)
+    enc.synchronize()
+    assert fails == [False] * ns, (bad, fails)
+    got = imgs.cpu().numpy()
+    for s in range(ns):
+        for b, shard_idx in enumerate(bad):
+            want = oracle.shard_write(ref[s, shard_idx].copy(),
+                                      bid=9000 + s * nb + b, vuid=77)
+            assert np.array_equal(got[s * nb + b], want), (bad, s, b)
+
+
+def test_repair_batch_detects_corruption(oracle, dev):
+    """With spare parity (nbad < m) a corrupted surviving shard flips the
+    per-stripe fail bit; clean stripes stay clean."""
+    import torch
+    from cubefs_amd import codemode, ec, shard
+    t = codemode.get_tactic("EC6P3")
+    enc = ec.Encoder(t)
+    ns, slen = 5, 150_000
+    rng = np.random.default_rng(314)
+    arr = rng.integers(0, 256, (ns, t.total, slen), dtype=np.uint8)
+    batch = torch.from_numpy(arr).to(dev)
+    enc.encode_batch(batch)
+    enc.synchronize()
+    bad = [3]
+    batch[2, 8, 140_001] ^= 0x10   # surviving parity, stripe 2
+    batch[4, 0, 7] ^= 0x01         # surviving data, stripe 4
+    for i in bad:
+        batch[:, i].zero_()
+    dsz = shard.disk_size(slen)
+    imgs = torch.zeros((ns, dsz), dtype=torch.uint8, device=dev)
+    fails = enc.repair_batch(batch, bad, imgs, list(range(ns)), [1] * ns)
+    enc.synchronize()
+    assert fails == [False, False, True, False, True], fails
