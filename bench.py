@@ -41,9 +41,11 @@ def parse_args():
     p.add_argument("--codemode", default="EC6P3",
                    help="EC6P3/EC12P4/... or LRC12P2L2 (registered via Extend)")
     p.add_argument("--workload", default="encode",
-                   choices=["encode", "reconstruct"],
+                   choices=["encode", "reconstruct", "repair"],
                    help="encode = BASELINE configs[1]+CRC; reconstruct = "
-                        "configs[2]: 1-shard reconstruct + crc32block verify")
+                        "configs[2]: 1-shard reconstruct + crc32block "
+                        "verify; repair = fused repair tasklet (2 lost "
+                        "shards -> verified framed disk images)")
     p.add_argument("--bad-idx", type=int, default=2,
                    help="shard index reconstructed in --workload reconstruct")
     p.add_argument("--no-crc", action="store_true",
@@ -285,6 +287,21 @@ def main():
             codec.synchronize()
         bad = [args.bad_idx]
 
+    if args.workload == "repair":
+        # blobnode repair tasklet: lose a data and a parity shard, rebuild
+        # them as verified, framed, pwrite-able disk images in one call
+        from cubefs_amd import shard as shardmod
+        enc.encode_batch(batch)
+        enc.synchronize()
+        bad = [args.bad_idx, t.N] if t.M >= 2 else [args.bad_idx]
+        for i in bad:
+            batch[:, i].zero_()
+        dsz = shardmod.disk_size(S)
+        imgs = torch.empty((ns * len(bad), dsz), dtype=torch.uint8,
+                           device=dev)
+        rbids = list(range(ns * len(bad)))
+        rvuids = [1] * (ns * len(bad))
+
     fused = (args.workload == "encode" and with_crc and not args.no_fused)
 
     if args.host_streamed:
@@ -305,6 +322,10 @@ def main():
                 events[1].record()
             if with_crc:
                 codec.encode_batch(framed, flat)
+        elif args.workload == "repair":
+            enc.repair_batch(batch, bad, imgs, rbids, rvuids)
+            if events:
+                events[1].record()
         else:
             enc.reconstruct_batch(batch, bad)
             if events:
@@ -342,7 +363,9 @@ def main():
         total_src = sum(r["src_bytes"] for r in recs)
         value = total_src / GIB / elapsed
         total_shard = total_src / t.N * t.total / GIB / elapsed
-        if args.workload != "encode":
+        if args.workload == "repair":
+            kind = "repair[%s]" % ",".join(str(b) for b in bad)
+        elif args.workload != "encode":
             kind = "reconstruct[%d]" % args.bad_idx
         elif fused:
             kind = "fused encode+frame"
@@ -352,13 +375,19 @@ def main():
             ("%d MiB" % args.shard_mib)
         workload = "RS(%d+%d%s) %s%s, %s shards, %d stripes/GPU" % (
             t.N, t.M, "+L%d" % t.L if t.L else "", kind,
-            "+crc32block" if (with_crc and not fused) else "", ssz, ns)
+            "+crc32block" if (with_crc and not fused
+                              and args.workload != "repair") else "",
+            ssz, ns)
         # roofline of the dominant kernel, one launch per step:
         #   encode: read k·S, write (m+l)·S per stripe
         #   fused encode+frame: read k·S, write (k+m)·(S+4·fps) per stripe
         #   1-shard reconstruct: read k·S, write 1·S
         fps = -(-S // 65532)
-        if args.workload != "encode":
+        if args.workload == "repair":
+            # read k inputs + (m-nbad) check shards, write nbad framed
+            alg_bytes = float((t.N + t.M - len(bad)) * S +
+                              len(bad) * (S + 4 * fps)) * ns
+        elif args.workload != "encode":
             alg_bytes = float((t.N + len(bad)) * S * ns)
         elif fused:
             alg_bytes = float((t.N + t.total) * S + 4 * fps * t.total) * ns
