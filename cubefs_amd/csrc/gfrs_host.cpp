@@ -1308,16 +1308,24 @@ int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
       const long v = e ? atol(e) : 0;
       return v > 0 ? size_t(v) : size_t(6144);
     }();
-    if (t.l == 0 && block_len == 65536 && t.m >= 1 && t.m <= 4 &&
-        t.n + t.m <= 16 && nbad <= t.m && shard_len >= fmin &&
-        dst_stride % 4 == 0 && nstripes > 0) {
+    int nglobad = 0;
+    for (int i = 0; i < nbad; i++)
+      if (bad_idx[i] < t.n + t.m) nglobad++;
+    if ((t.l == 0 || cc->fused_lrc_ok) && block_len == 65536 && t.m >= 1 &&
+        t.n + t.m + t.l <= 16 && nglobad <= t.m && nbad <= 4 &&
+        shard_len >= fmin && dst_stride % 4 == 0 && nstripes > 0) {
       const int k = t.n, m = t.m;
+      const int total_sh = k + m + t.l;
       std::vector<uint8_t> present(k + m, 1);
       std::vector<int> badv(bad_idx, bad_idx + nbad);
+      bool badrange = false;
       for (int i = 0; i < nbad; i++) {
-        if (badv[i] < 0 || badv[i] >= k + m) return GFRS_ERR_INVALID_SHARDS;
-        present[badv[i]] = 0;
+        if (badv[i] < 0 || badv[i] >= total_sh) return GFRS_ERR_INVALID_SHARDS;
+        if (badv[i] < k + m) present[badv[i]] = 0;
+        for (int j = 0; j < i; j++)
+          if (badv[j] == badv[i]) badrange = true;
       }
+      if (badrange) return GFRS_ERR_INVALID_SHARDS;
       std::vector<int> sbad = badv;
       std::sort(sbad.begin(), sbad.end());
       uint32_t colpack = 0;
@@ -1329,8 +1337,7 @@ int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
       std::lock_guard<std::mutex> lk(cc->mu);
       StreamGuard g(cc);
       uint64_t key = 0;
-      for (int i = 0; i < k + m; i++)
-        if (!present[i]) key |= 1ull << i;
+      for (int b : badv) key |= 1ull << b; /* incl. local-parity bads */
       key = (key << 6) | 61; /* fused-repair namespace */
       DevPlan *plan = nullptr;
       {
@@ -1352,10 +1359,39 @@ int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
         for (int j = 0; j < k; j++)
           if (valid[j] < k) slot[valid[j]] = j;
         const GfTables &gt2 = gft();
-        auto parity_row = [&](int p2, std::vector<uint8_t> &row) {
+        /* shard idx -> its row over the n data shards (locals compose
+         * through the global-parity rows, as in the fused_lrc plan) */
+        auto dspace_row = [&](int sh, std::vector<uint8_t> &drow) {
+          drow.assign(k, 0);
+          if (sh < k) {
+            drow[sh] = 1;
+          } else if (sh < k + m) {
+            memcpy(drow.data(), &cc->enc_matrix[size_t(sh) * k], k);
+          } else {
+            const int az = (sh - k - m) / cc->local_m;
+            const int lp = (sh - k - m) % cc->local_m;
+            auto idx = local_stripe(t, az);
+            const uint8_t *lr =
+                &cc->local_matrix[size_t(cc->local_n + lp) * cc->local_n];
+            for (int j = 0; j < cc->local_n; j++) {
+              const int g2i = idx[j];
+              if (g2i < k) {
+                drow[g2i] ^= lr[j];
+              } else {
+                const uint8_t *er = &cc->enc_matrix[size_t(g2i) * k];
+                for (int d = 0; d < k; d++)
+                  drow[d] ^= gt2.mul[lr[j]][er[d]];
+              }
+            }
+          }
+        };
+        /* data-space row -> row over the k chosen inputs */
+        auto xform = [&](const std::vector<uint8_t> &drow,
+                         std::vector<uint8_t> &row) {
           row.assign(k, 0);
           for (int d = 0; d < k; d++) {
-            const uint8_t coef = cc->enc_matrix[size_t(p2) * k + d];
+            const uint8_t coef = drow[d];
+            if (!coef) continue;
             if (present[d]) {
               row[slot[d]] ^= coef;
             } else {
@@ -1365,29 +1401,35 @@ int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
           }
         };
         std::vector<int32_t> in, out;
-        std::vector<uint8_t> rows, row;
+        std::vector<uint8_t> rows, row, drow;
         for (int j = 0; j < k; j++) in.push_back(valid[j]);
-        /* rebuild rows in sorted-bad order */
+        /* rebuild rows in sorted-bad order (data, global or local) */
         for (int b : sbad) {
           out.push_back(b);
-          if (b < k) {
-            rows.insert(rows.end(), &dec[size_t(b) * k],
-                        &dec[size_t(b) * k + k]);
-          } else {
-            parity_row(b, row);
-            rows.insert(rows.end(), row.begin(), row.end());
-          }
+          dspace_row(b, drow);
+          xform(drow, row);
+          rows.insert(rows.end(), row.begin(), row.end());
         }
-        /* check rows: surviving parity NOT already an input */
+        /* check rows while they fit in the 4-row kernel budget:
+         * surviving globals not already inputs, then surviving locals
+         * (the locals restore detection when npresent == k globally) */
+        auto add_check = [&](int sh) {
+          if (int(out.size()) >= 4) return;
+          out.push_back(sh);
+          dspace_row(sh, drow);
+          xform(drow, row);
+          rows.insert(rows.end(), row.begin(), row.end());
+          in.push_back(sh); /* cmp shard index rides after the k inputs */
+        };
         for (int p2 = k; p2 < k + m; p2++) {
           if (!present[p2]) continue;
           if (std::find(valid.begin(), valid.end(), p2) != valid.end())
             continue; /* input parity: its check is the identity */
-          out.push_back(p2);
-          parity_row(p2, row);
-          rows.insert(rows.end(), row.begin(), row.end());
-          in.push_back(p2); /* cmp shard index rides after the k inputs */
+          add_check(p2);
         }
+        for (int q = k + m; q < total_sh; q++)
+          if (std::find(badv.begin(), badv.end(), q) == badv.end())
+            add_check(q);
         plan = new DevPlan();
         int rc2 = plan->upload(in, out, rows, cc->stream, k);
         if (rc2 != GFRS_OK) {

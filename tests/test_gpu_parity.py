@@ -408,3 +408,64 @@ def test_repair_batch_detects_corruption(oracle, dev):
     fails = enc.repair_batch(batch, bad, imgs, list(range(ns)), [1] * ns)
     enc.synchronize()
     assert fails == [False, False, True, False, True], fails
+
+
+@pytest.mark.parametrize("bad", [[14, 5], [13, 2], [15], [3, 9]])
+def test_repair_batch_lrc_images(oracle, dev, bad):
+    """LRC fused repair: lost data / global-parity / local-parity shards
+    come back as framed disk images bit-identical to oracle-built images
+    of the original shards."""
+    import torch
+    from cubefs_amd import codemode, ec, shard
+    t = codemode.get_tactic("LRC12P2L2")
+    enc = ec.Encoder(t)
+    ns, slen = 3, 100_000
+    rng = np.random.default_rng(900 + len(bad) + bad[0])
+    arr = rng.integers(0, 256, (ns, t.total, slen), dtype=np.uint8)
+    ref = [[arr[s, i].copy() for i in range(t.total)] for s in range(ns)]
+    for s in range(ns):
+        oracle.lrc_encode(t.N, t.M, t.L, t.AZCount, ref[s])
+    batch = torch.from_numpy(
+        np.stack([np.stack(r) for r in ref])).to(dev)
+    for i in bad:
+        batch[:, i].zero_()
+    nb = len(bad)
+    dsz = shard.disk_size(slen)
+    imgs = torch.zeros((ns * nb, dsz), dtype=torch.uint8, device=dev)
+    bids = [4000 + s * nb + b for s in range(ns) for b in range(nb)]
+    fails = enc.repair_batch(batch, bad, imgs, bids, [9] * (ns * nb))
+    enc.synchronize()
+    assert fails == [False] * ns, (bad, fails)
+    got = imgs.cpu().numpy()
+    for s in range(ns):
+        for b, shard_idx in enumerate(bad):
+            want = oracle.shard_write(ref[s][shard_idx].copy(),
+                                      bid=4000 + s * nb + b, vuid=9)
+            assert np.array_equal(got[s * nb + b], want), (bad, s, b)
+
+
+def test_repair_batch_lrc_local_check_detects(oracle, dev):
+    """At nbad == m the global stripe has no spare check equation, but
+    the surviving LOCAL parities do: corruption is still detected."""
+    import torch
+    from cubefs_amd import codemode, ec, shard
+    t = codemode.get_tactic("LRC12P2L2")
+    enc = ec.Encoder(t)
+    ns, slen = 4, 80_000
+    rng = np.random.default_rng(4242)
+    arr = rng.integers(0, 256, (ns, t.total, slen), dtype=np.uint8)
+    ref = [[arr[s, i].copy() for i in range(t.total)] for s in range(ns)]
+    for s in range(ns):
+        oracle.lrc_encode(t.N, t.M, t.L, t.AZCount, ref[s])
+    batch = torch.from_numpy(
+        np.stack([np.stack(r) for r in ref])).to(dev)
+    bad = [1, 7]                       # nbad == m: 12 survivors == k
+    batch[2, 4, 50_000] ^= 0x80        # corrupt surviving data, stripe 2
+    for i in bad:
+        batch[:, i].zero_()
+    dsz = shard.disk_size(slen)
+    imgs = torch.zeros((ns * 2, dsz), dtype=torch.uint8, device=dev)
+    fails = enc.repair_batch(batch, bad, imgs,
+                             list(range(ns * 2)), [1] * (ns * 2))
+    enc.synchronize()
+    assert fails == [False, False, True, False], fails
