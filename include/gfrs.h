@@ -241,7 +241,12 @@ int gfrs_update_idx(gfrs_ctx *ctx, const void *old_shard,
  * One stream-ordered call per repair tasklet: reconstruct the bad shards
  * of every stripe, verify all parity (mandatory Verify, :865-874), then
  * frame each repaired shard into a pwrite()-able disk image.  Intermediate
- * data never leaves HBM.
+ * data never leaves HBM; on the fused path the raw reconstruction is
+ * never materialized, so the bad shards' regions of `base` are left
+ * UNSPECIFIED after the call (the reference worker also only consumes
+ * the repaired bytes through the written bids).  For LRC the surviving
+ * local parities serve as extra verify equations, so corruption is
+ * detected even when the global stripe has no spare parity.
  *   disk_dst: nstripes*nbad images, image (s,b) at
  *             disk_dst + (s*nbad + b)*dst_stride
  *   bids/vuids: one per (stripe, bad shard), same order
