@@ -74,7 +74,9 @@ def hbm_traffic_lookup(workload):
     try:
         with open(path) as f:
             d = json.load(f)
-        return d.get(workload, {}).get("bytes_per_launch")
+        entry = d.get(workload)
+        return entry.get("bytes_per_launch") if isinstance(entry, dict) \
+            else None
     except (OSError, ValueError):
         return None
 
