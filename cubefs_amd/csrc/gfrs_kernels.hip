@@ -1310,6 +1310,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
 
     uint4 acc[GM][4];
     uint4 vnext[PIPE ? 4 : 1];
+    uint4 vnext2[PIPE == 2 ? 4 : 1];
     if (PIPE && fr == fr0) { /* later frames are prefetched by the
                                 previous frame's last pass */
       const int rb0 = int(i64min(int64_t(EF_PASS), payload));
@@ -1320,6 +1321,16 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
             off + 16 <= rb0
                 ? *reinterpret_cast<const uint4 *>(sbase + p0 + off)
                 : uint4{0, 0, 0, 0};
+      }
+      if (PIPE == 2) {
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+          const int off = i * 4096 + lane16i;
+          vnext2[PIPE == 2 ? i : 0] =
+              off + 16 <= rb0 ? *reinterpret_cast<const uint4 *>(
+                                    sbase + shard_len + p0 + off)
+                              : uint4{0, 0, 0, 0};
+        }
       }
     }
     for (int j = threadIdx.x; j < 64; j += CRC_BLOCKT) red[j] = 0;
@@ -1355,7 +1366,24 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
         uint8_t *fdst = dst + (stripe * (k + GM) + c) * dst_stride +
                         f * block_len + CRC_LEN + r0;
         uint4 vcur[PIPE ? 4 : 1];
-        if (PIPE) {
+        if (PIPE == 2) {
+#pragma unroll
+          for (int i = 0; i < 4; i++) {
+            vcur[PIPE ? i : 0] = vnext[PIPE ? i : 0];
+            vnext[PIPE ? i : 0] = vnext2[PIPE == 2 ? i : 0];
+          }
+          if (c + 2 < k) {
+            const uint8_t *nsrc = sbase + size_t(c + 2) * shard_len + p0 + r0;
+#pragma unroll
+            for (int i = 0; i < 4; i++) {
+              const int off = i * 4096 + lane16i;
+              vnext2[PIPE == 2 ? i : 0] =
+                  off + 16 <= rbi
+                      ? *reinterpret_cast<const uint4 *>(nsrc + off)
+                      : uint4{0, 0, 0, 0};
+            }
+          }
+        } else if (PIPE) {
 #pragma unroll
           for (int i = 0; i < 4; i++) vcur[PIPE ? i : 0] = vnext[PIPE ? i : 0];
           if (c + 1 < k) {
@@ -1429,6 +1457,16 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
                 off + 16 <= rbni
                     ? *reinterpret_cast<const uint4 *>(nbase + r0n + off)
                     : uint4{0, 0, 0, 0};
+          }
+          if (PIPE == 2) {
+#pragma unroll
+            for (int i = 0; i < 4; i++) {
+              const int off = i * 4096 + lane16i;
+              vnext2[PIPE == 2 ? i : 0] =
+                  off + 16 <= rbni ? *reinterpret_cast<const uint4 *>(
+                                         nbase + shard_len + r0n + off)
+                                   : uint4{0, 0, 0, 0};
+            }
           }
         }
       }
@@ -1767,12 +1805,30 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
     const int v = e ? atoi(e) : 77; /* register-CRC + load lookahead */
     switch (v) {
       case 13: case 14: case 23: case 24:
-      case 74: case 75: case 76: case 77:
+      case 74: case 75: case 76: case 77: case 78:
       case 142: case 152: case 162: return v;
       default: return 14;
     }
   }();
   /* 7x = register-CRC kernel (no stage): crc tabs + shift tabs + red */
+  if (var == 78) { /* two-unit-deep lookahead at 3 waves/SIMD */
+    const int lds = 12288 + EF_RED + m * k * 32;
+    switch (m) {
+      case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 3, 0, 0, 2>),
+          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps); break;
+      case 2: hipLaunchKernelGGL((rs_encode_frame_reg_k<2, 3, 0, 0, 2>),
+          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps); break;
+      case 3: hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 0, 2>),
+          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps); break;
+      default: hipLaunchKernelGGL((rs_encode_frame_reg_k<4, 3, 0, 0, 2>),
+          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, tabs, total, fps);
+    }
+    return;
+  }
   if (var == 77) { /* lookahead pipeline squeezed to 4 waves/SIMD */
     const int lds = 12288 + EF_RED + m * k * 32;
     switch (m) {
