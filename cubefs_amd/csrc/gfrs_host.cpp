@@ -1170,6 +1170,146 @@ int gfrs_encode_idx(gfrs_ctx *ctx, const void *data_shard, int idx,
   return GFRS_OK;
 }
 
+/* LRC reconstruct+verify in ONE rs_apply_mixed pass: write rows for the
+ * bad shards (data, global or local parity, composed over the k global
+ * decode inputs), compare rows for every surviving global parity not
+ * already an input and every surviving local parity.  Requires the bad
+ * set to be globally decodable; returns GFRS_ERR_UNSUPPORTED otherwise
+ * so the caller can fall back to local-stripe decode + full verify. */
+static int lrc_mixed_reconstruct_verify(gfrs_ctx_impl *cc, void *base,
+                                        size_t shard_len,
+                                        size_t stripe_stride, int nstripes,
+                                        const int32_t *bad_idx, int nbad,
+                                        uint64_t *fail_bitmap) {
+  const gfrs_tactic &t = cc->t;
+  const int k = t.n, m = t.m;
+  const int total_sh = k + m + t.l;
+  std::vector<uint8_t> present(k + m, 1);
+  std::vector<int> badv(bad_idx, bad_idx + nbad);
+  for (int i = 0; i < nbad; i++) {
+    if (badv[i] < k + m) present[badv[i]] = 0;
+    for (int j = 0; j < i; j++)
+      if (badv[j] == badv[i]) return GFRS_ERR_INVALID_SHARDS;
+  }
+  std::lock_guard<std::mutex> lk(cc->mu);
+  StreamGuard g(cc);
+  uint64_t key = 0;
+  for (int b : badv) key |= 1ull << b;
+  key = (key << 6) | 60; /* LRC mixed reconstruct+verify namespace */
+  DevPlan *plan = nullptr;
+  uint32_t cmp_mask = 0;
+  {
+    auto it = cc->dec_cache.find(key);
+    if (it != cc->dec_cache.end()) plan = it->second;
+  }
+  /* cmp_mask is derivable from the (cached) row order: rows after nbad
+   * are compares */
+  if (!plan) {
+    std::vector<int> valid;
+    for (int i = 0; i < k + m && int(valid.size()) < k; i++)
+      if (present[i]) valid.push_back(i);
+    if (int(valid.size()) < k) return GFRS_ERR_UNSUPPORTED;
+    std::vector<uint8_t> sub(size_t(k) * k), dec(size_t(k) * k);
+    for (int r = 0; r < k; r++)
+      memcpy(&sub[size_t(r) * k], &cc->enc_matrix[size_t(valid[r]) * k], k);
+    if (!gf_invert(sub.data(), k, dec.data())) return GFRS_ERR_SINGULAR;
+    std::vector<int> slot(k, -1);
+    for (int j = 0; j < k; j++)
+      if (valid[j] < k) slot[valid[j]] = j;
+    const GfTables &gt2 = gft();
+    auto dspace_row = [&](int sh, std::vector<uint8_t> &drow) {
+      drow.assign(k, 0);
+      if (sh < k) {
+        drow[sh] = 1;
+      } else if (sh < k + m) {
+        memcpy(drow.data(), &cc->enc_matrix[size_t(sh) * k], k);
+      } else {
+        const int az = (sh - k - m) / cc->local_m;
+        const int lp = (sh - k - m) % cc->local_m;
+        auto idx = local_stripe(t, az);
+        const uint8_t *lr =
+            &cc->local_matrix[size_t(cc->local_n + lp) * cc->local_n];
+        for (int j = 0; j < cc->local_n; j++) {
+          const int g2i = idx[j];
+          if (g2i < k) {
+            drow[g2i] ^= lr[j];
+          } else {
+            const uint8_t *er = &cc->enc_matrix[size_t(g2i) * k];
+            for (int d = 0; d < k; d++)
+              drow[d] ^= gt2.mul[lr[j]][er[d]];
+          }
+        }
+      }
+    };
+    auto xform = [&](const std::vector<uint8_t> &drow,
+                     std::vector<uint8_t> &row) {
+      row.assign(k, 0);
+      for (int d = 0; d < k; d++) {
+        const uint8_t coef = drow[d];
+        if (!coef) continue;
+        if (present[d]) {
+          row[slot[d]] ^= coef;
+        } else {
+          for (int j = 0; j < k; j++)
+            row[j] ^= gt2.mul[coef][dec[size_t(d) * k + j]];
+        }
+      }
+    };
+    std::vector<int32_t> in, out;
+    std::vector<uint8_t> rows, row, drow;
+    for (int j = 0; j < k; j++) in.push_back(valid[j]);
+    for (int b : badv) { /* write rows */
+      out.push_back(b);
+      dspace_row(b, drow);
+      xform(drow, row);
+      rows.insert(rows.end(), row.begin(), row.end());
+    }
+    auto add_check = [&](int sh) {
+      out.push_back(sh);
+      dspace_row(sh, drow);
+      xform(drow, row);
+      rows.insert(rows.end(), row.begin(), row.end());
+    };
+    for (int p2 = k; p2 < k + m; p2++) {
+      if (!present[p2]) continue;
+      if (std::find(valid.begin(), valid.end(), p2) != valid.end()) continue;
+      add_check(p2);
+    }
+    for (int q = k + m; q < total_sh; q++)
+      if (std::find(badv.begin(), badv.end(), q) == badv.end())
+        add_check(q);
+    if (out.size() > 32) return GFRS_ERR_UNSUPPORTED; /* cmp_mask width */
+    plan = new DevPlan();
+    int rc = plan->upload(in, out, rows, cc->stream, k);
+    if (rc != GFRS_OK) {
+      delete plan;
+      return rc;
+    }
+    cc->dec_cache[key] = plan;
+  }
+  for (int r = nbad; r < plan->nout; r++) cmp_mask |= 1u << r;
+  int rc;
+  if ((rc = cc->fail_buf.ensure(size_t(nstripes) * 4)) != GFRS_OK) return rc;
+  HIP_TRY(hipMemsetAsync(cc->fail_buf.p, 0, size_t(nstripes) * 4,
+                         cc->stream));
+  launch_rs_apply_mixed_strided((uint64_t)base, stripe_stride,
+                                (const int32_t *)plan->in_idx.p, plan->k,
+                                (const int32_t *)plan->out_idx.p, plan->nout,
+                                (const uint8_t *)plan->tabs.p, cmp_mask,
+                                shard_len, nstripes,
+                                (uint32_t *)cc->fail_buf.p, cc->stream);
+  std::vector<uint32_t> fails(nstripes);
+  HIP_TRY(hipMemcpyAsync(fails.data(), cc->fail_buf.p, size_t(nstripes) * 4,
+                         hipMemcpyDeviceToHost, cc->stream));
+  HIP_TRY(hipStreamSynchronize(cc->stream));
+  if (fail_bitmap) {
+    memset(fail_bitmap, 0, ((nstripes + 63) / 64) * 8);
+    for (int s2 = 0; s2 < nstripes; s2++)
+      if (fails[s2]) fail_bitmap[s2 / 64] |= 1ull << (s2 % 64);
+  }
+  return GFRS_OK;
+}
+
 int gfrs_reconstruct_verify_batch(gfrs_ctx *ctx, void *base,
                                   size_t shard_len, size_t stripe_stride,
                                   int nstripes, const int32_t *bad_idx,
@@ -1177,7 +1317,26 @@ int gfrs_reconstruct_verify_batch(gfrs_ctx *ctx, void *base,
   auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
   const gfrs_tactic &t = c->t;
   if (nstripes <= 0) return GFRS_ERR_INVALID_SHARDS;
-  if (t.l != 0) { /* LRC fallback: two passes */
+  if (t.l != 0) {
+    /* LRC single pass: every bad shard (data, global or local parity)
+     * is a write row composed over the k global-decode inputs, and
+     * every surviving global/local parity is a compare row - ONE
+     * rs_apply_mixed pass instead of reconstruct + full re-read verify.
+     * Falls back to two passes only when the bad set needs local-stripe
+     * decode (globally undecodable). */
+    int nglobad = 0;
+    bool ok_range = true;
+    for (int i = 0; i < nbad; i++) {
+      if (bad_idx[i] < 0 || bad_idx[i] >= t.n + t.m + t.l) ok_range = false;
+      else if (bad_idx[i] < t.n + t.m) nglobad++;
+    }
+    if (!ok_range) return GFRS_ERR_INVALID_SHARDS;
+    if (nglobad <= t.m && c->fused_lrc_ok) {
+      int rc = lrc_mixed_reconstruct_verify(c, base, shard_len,
+                                            stripe_stride, nstripes,
+                                            bad_idx, nbad, fail_bitmap);
+      if (rc != GFRS_ERR_UNSUPPORTED) return rc;
+    }
     int rc = gfrs_reconstruct_batch(ctx, base, shard_len, stripe_stride,
                                     nstripes, bad_idx, nbad, 0);
     if (rc != GFRS_OK) return rc;

@@ -469,3 +469,39 @@ def test_repair_batch_lrc_local_check_detects(oracle, dev):
                              list(range(ns * 2)), [1] * (ns * 2))
     enc.synchronize()
     assert fails == [False, False, True, False], fails
+
+
+@pytest.mark.parametrize("bad", [[2], [0, 13], [15], [1, 7]])
+def test_lrc_reconstruct_verify(oracle, dev, bad):
+    """LRC reconstruct+verify in one mixed pass: reconstruction bit-exact
+    and corruption detected through global AND local check equations."""
+    import torch
+    from cubefs_amd import codemode, ec
+    t = codemode.get_tactic("LRC12P2L2")
+    enc = ec.Encoder(t)
+    ns, slen = 4, 120_000
+    rng = np.random.default_rng(1100 + bad[0] + len(bad))
+    arr = rng.integers(0, 256, (ns, t.total, slen), dtype=np.uint8)
+    ref = [[arr[s, i].copy() for i in range(t.total)] for s in range(ns)]
+    for s in range(ns):
+        oracle.lrc_encode(t.N, t.M, t.L, t.AZCount, ref[s])
+    batch = torch.from_numpy(np.stack([np.stack(r) for r in ref])).to(dev)
+    for i in bad:
+        batch[:, i].zero_()
+    fails = enc.reconstruct_verify_batch(batch, bad)
+    enc.synchronize()
+    assert fails == [False] * ns, (bad, fails)
+    got = batch.cpu().numpy()
+    for s in range(ns):
+        for i in range(t.total):
+            assert np.array_equal(got[s, i], ref[s][i]), (bad, s, i)
+    # corrupt a surviving data shard in stripe 1: with nbad == m == 2 the
+    # global stripe is check-less, but the local equations still catch it
+    surv = next(i for i in range(t.N) if i not in bad)
+    batch[:] = torch.from_numpy(np.stack([np.stack(r) for r in ref])).to(dev)
+    batch[1, surv, 100_001] ^= 0x04
+    for i in bad:
+        batch[:, i].zero_()
+    fails = enc.reconstruct_verify_batch(batch, bad)
+    enc.synchronize()
+    assert fails[1] and sum(fails) == 1, (bad, fails)
