@@ -1272,7 +1272,12 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
   uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
   uint32_t(*stab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem + 8192);
   uint32_t *red = reinterpret_cast<uint32_t *>(smem + 12288);
-  uint8_t *ctab = smem + 12288 + EF_RED;
+  /* x^(8*j), j<16: position operators for the <16-byte pass tails */
+  uint32_t *x8tab = reinterpret_cast<uint32_t *>(smem + 12288 + EF_RED);
+  /* staged tail bytes of the (up to 16) input shards: the parity tails
+   * read these from LDS instead of k dependent global loads per lane */
+  uint8_t *tailb = smem + 12288 + EF_RED + 64;
+  uint8_t *ctab = smem + 12288 + EF_RED + 64 + 256;
   for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
     (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
   for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
@@ -1280,6 +1285,13 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
   for (int i = threadIdx.x; i < GM * k * 2; i += CRC_BLOCKT)
     reinterpret_cast<uint4 *>(ctab)[i] =
         reinterpret_cast<const uint4 *>(tabs)[i];
+  if (threadIdx.x == 0) {
+    uint32_t v = 0x80000000u;
+    for (int j = 0; j < 16; j++) {
+      x8tab[j] = v;
+      v = gf2_mulmod_d(v, g_pow8[0]);
+    }
+  }
   const int64_t lane16 = int64_t(threadIdx.x) * 16;
   const int lane16i = int(threadIdx.x) * 16;
   /* Full-frame fold operator, chained: op for pass h+1 = op for pass h
@@ -1432,12 +1444,19 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
           }
         }
         uint32_t part = SKEL == 2 ? t : (t ? gf2_mulmod_d(op, t) : 0);
-        if (rbytes < EF_PASS && threadIdx.x == 0) {
-          /* pass tail ends exactly at the payload: identity fold */
-          const int64_t t0 = (rbytes / 16) * 16;
-          const uint32_t ct =
-              crc_chunk(src + t0, int(rbytes - t0), tab, fdst + t0);
-          if (SKEL != 1) part ^= ct;
+        if (rbytes < EF_PASS) {
+          /* lane-parallel tail: one byte per lane, folded by its own
+           * position operator (the serial thread-0 loop dominated tiny
+           * last frames: 9 units of dependent global loads) */
+          const int t0 = (rbi / 16) * 16;
+          const int p = t0 + int(threadIdx.x);
+          if (p < rbi) {
+            const uint8_t x = src[p];
+            fdst[p] = x;
+            tailb[c * 16 + (p - t0)] = x;
+            if (SKEL != 1)
+              part ^= gf2_mulmod_d(x8tab[rbi - 1 - p], tab[0][x]);
+          }
         }
 #pragma unroll
         for (int sh = 32; sh > 0; sh >>= 1)
@@ -1445,6 +1464,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
         if ((threadIdx.x & 63) == 0)
           red[(threadIdx.x >> 6) * 16 + c] ^= part;
       }
+      if (rbytes < EF_PASS) __syncthreads(); /* tailb visible to all */
       if (PIPE) { /* next pass's (or next frame's) unit-0 loads fly
                      during the parity rows and the frame epilogue */
         int64_t r0n = r0 + EF_PASS;
@@ -1497,20 +1517,19 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
           }
         }
         uint32_t part = SKEL == 2 ? t : (t ? gf2_mulmod_d(op, t) : 0);
-        if (SKEL != 1 && rbytes < EF_PASS && threadIdx.x == 0) {
-          const int64_t t0 = (rbytes / 16) * 16;
-          uint32_t ct = 0;
-          for (int64_t p = t0; p < rbytes; p++) {
+        if (SKEL != 1 && rbytes < EF_PASS) {
+          const int t0 = (rbi / 16) * 16;
+          const int p = t0 + int(threadIdx.x);
+          if (p < rbi) {
             uint8_t pv = 0;
             for (int c2 = 0; c2 < k; c2++) {
-              const uint8_t b = sbase[size_t(c2) * shard_len + p0 + r0 + p];
+              const uint8_t b = tailb[c2 * 16 + (p - t0)];
               const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
               pv ^= tt[b & 0x0F] ^ tt[16 + (b >> 4)];
             }
             fdst[p] = pv;
-            ct = tab[0][(ct ^ pv) & 0xFF] ^ (ct >> 8);
+            part ^= gf2_mulmod_d(x8tab[rbi - 1 - p], tab[0][pv]);
           }
-          part ^= ct;
         }
 #pragma unroll
         for (int sh = 32; sh > 0; sh >>= 1)
@@ -1565,7 +1584,12 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
   uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
   uint32_t(*stab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem + 8192);
   uint32_t *red = reinterpret_cast<uint32_t *>(smem + 12288);
-  uint8_t *ctab = smem + 12288 + EF_RED;
+  /* x^(8*j), j<16: position operators for the <16-byte pass tails */
+  uint32_t *x8tab = reinterpret_cast<uint32_t *>(smem + 12288 + EF_RED);
+  /* staged tail bytes of the (up to 16) input shards: the parity tails
+   * read these from LDS instead of k dependent global loads per lane */
+  uint8_t *tailb = smem + 12288 + EF_RED + 64;
+  uint8_t *ctab = smem + 12288 + EF_RED + 64 + 256;
   for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
     (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
   for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
@@ -1573,6 +1597,13 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
   for (int i = threadIdx.x; i < GM * k * 2; i += CRC_BLOCKT)
     reinterpret_cast<uint4 *>(ctab)[i] =
         reinterpret_cast<const uint4 *>(tabs)[i];
+  if (threadIdx.x == 0) {
+    uint32_t v = 0x80000000u;
+    for (int j = 0; j < 16; j++) {
+      x8tab[j] = v;
+      v = gf2_mulmod_d(v, g_pow8[0]);
+    }
+  }
   const int64_t lane16 = int64_t(threadIdx.x) * 16;
   const int lane16i = int(threadIdx.x) * 16;
   constexpr uint32_t INV16K = 0x479933FCu;
@@ -1687,6 +1718,14 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
           }
         }
       }
+      if (rbytes < EF_PASS) { /* stage input tail bytes: lane (c,j) */
+        const int tt0 = (rbi / 16) * 16;
+        const int c2 = int(threadIdx.x) >> 4, j = int(threadIdx.x) & 15;
+        if (c2 < k && tt0 + j < rbi)
+          tailb[c2 * 16 + j] =
+              sbase[size_t(imap[c2]) * shard_len + p0 + r0 + tt0 + j];
+        __syncthreads();
+      }
 #pragma unroll
       for (int r = 0; r < GM; r++) {
         if (r < nw) { /* rebuild row: framed image output */
@@ -1705,21 +1744,19 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
             }
           }
           uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
-          if (rbytes < EF_PASS && threadIdx.x == 0) {
-            const int64_t t0 = (rbytes / 16) * 16;
-            uint32_t ct = 0;
-            for (int64_t p = t0; p < rbytes; p++) {
+          if (rbytes < EF_PASS) { /* lane-parallel tail (see encode) */
+            const int t0 = (rbi / 16) * 16;
+            const int p = t0 + int(threadIdx.x);
+            if (p < rbi) {
               uint8_t pv = 0;
               for (int c2 = 0; c2 < k; c2++) {
-                const uint8_t b2 =
-                    sbase[size_t(imap[c2]) * shard_len + p0 + r0 + p];
+                const uint8_t b2 = tailb[c2 * 16 + (p - t0)];
                 const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
                 pv ^= tt[b2 & 0x0F] ^ tt[16 + (b2 >> 4)];
               }
               fdst[p] = pv;
-              ct = tab[0][(ct ^ pv) & 0xFF] ^ (ct >> 8);
+              part ^= gf2_mulmod_d(x8tab[rbi - 1 - p], tab[0][pv]);
             }
-            part ^= ct;
           }
 #pragma unroll
           for (int sh = 32; sh > 0; sh >>= 1)
@@ -1740,13 +1777,13 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
                    (w.z ^ acc[r][i].z) | (w.w ^ acc[r][i].w);
             }
           }
-          if (rbytes < EF_PASS && threadIdx.x == 0) {
-            const int64_t t0 = (rbytes / 16) * 16;
-            for (int64_t p = t0; p < rbytes; p++) {
+          if (rbytes < EF_PASS) {
+            const int t0 = (rbi / 16) * 16;
+            const int p = t0 + int(threadIdx.x);
+            if (p < rbi) {
               uint8_t pv = 0;
               for (int c2 = 0; c2 < k; c2++) {
-                const uint8_t b2 =
-                    sbase[size_t(imap[c2]) * shard_len + p0 + r0 + p];
+                const uint8_t b2 = tailb[c2 * 16 + (p - t0)];
                 const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
                 pv ^= tt[b2 & 0x0F] ^ tt[16 + (b2 >> 4)];
               }
@@ -1787,7 +1824,7 @@ void launch_rs_repair_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
   const int64_t fps = (int64_t(shard_len) + 65531) / 65532;
   const int64_t total = fps * nstripes;
   const int grid = fused_grid(total);
-  const int lds = 12288 + EF_RED + gm * k * 32;
+  const int lds = 12288 + EF_RED + 64 + 256 + gm * k * 32;
 #define GFRS_RP_GO(G)                                                       hipLaunchKernelGGL((rs_repair_frame_k<G>), dim3(grid),                                       dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,                          stripe_stride, shard_len, k, imap, tabs, nw,                              colpack, fail, total, fps)
   switch (gm) {
     case 1: GFRS_RP_GO(1); break;
@@ -1822,7 +1859,7 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
   }();
   /* 7x = register-CRC kernel (no stage): crc tabs + shift tabs + red */
   if (var == 78) { /* two-unit-deep lookahead at 3 waves/SIMD */
-    const int lds = 12288 + EF_RED + m * k * 32;
+    const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
     switch (m) {
       case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 3, 0, 0, 2>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
@@ -1840,7 +1877,7 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
     return;
   }
   if (var == 77) { /* lookahead pipeline squeezed to 4 waves/SIMD */
-    const int lds = 12288 + EF_RED + m * k * 32;
+    const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
     switch (m) {
       case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 4, 0, 0, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
@@ -1858,7 +1895,7 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
     return;
   }
   if (var == 76) { /* one-unit-lookahead pipeline at 3 waves/SIMD */
-    const int lds = 12288 + EF_RED + m * k * 32;
+    const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
     switch (m) {
       case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 3, 0, 0, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
@@ -1876,7 +1913,7 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
     return;
   }
   if (var == 74 || var == 75) {
-    const int lds = 12288 + EF_RED + m * k * 32;
+    const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
 #define GFRS_EFR_GO(G, W, P) hipLaunchKernelGGL((rs_encode_frame_reg_k<G, W, P>), dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base, stripe_stride, shard_len, k, tabs, total, fps)
 #define GFRS_EFR_SW(W, P) switch (m) { case 1: GFRS_EFR_GO(1, W, P); break; case 2: GFRS_EFR_GO(2, W, P); break; case 3: GFRS_EFR_GO(3, W, P); break; default: GFRS_EFR_GO(4, W, P); }
     static const int map = []() {
@@ -1890,7 +1927,7 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
       return (v >= 4 && v <= 7) ? v : 0;
     }();
     if (rabl >= 5 && var == 76 && m == 3) { /* phase diagnostics */
-      const int lds5 = 12288 + EF_RED + m * k * 32;
+      const int lds5 = 12288 + EF_RED + 64 + 256 + m * k * 32;
       if (rabl == 5) { hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 2, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds5, s, dst, dst_stride, base,
           stripe_stride, shard_len, k, tabs, total, fps); return; }
