@@ -1302,6 +1302,8 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
   constexpr uint32_t INV16K = 0x479933FCu;
   const uint32_t op_first =
       x8n_d(uint64_t(payload_full - (3 * 4096 + lane16 + 16)));
+  const uint32_t it_full =
+      gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
   __syncthreads();
 
   const uint4 *ltab = reinterpret_cast<const uint4 *>(ctab);
@@ -1540,9 +1542,13 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     }
 
     __syncthreads();
-    if (threadIdx.x == 0) {
-      const uint32_t it = gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
-      for (int j = 0; j < k + GM; j++) {
+    { /* one frame-header lane per shard (j < k+GM <= 16, all in wave 0) */
+      const int j = int(threadIdx.x);
+      if (j < k + GM) {
+        const uint32_t it =
+            payload == payload_full
+                ? it_full
+                : gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
         const uint32_t crc =
             ~(it ^ red[j] ^ red[16 + j] ^ red[32 + j] ^ red[48 + j]);
         *reinterpret_cast<uint32_t *>(
@@ -1609,6 +1615,8 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
   constexpr uint32_t INV16K = 0x479933FCu;
   const uint32_t op_first =
       x8n_d(uint64_t(payload_full - (3 * 4096 + lane16 + 16)));
+  const uint32_t it_full =
+      gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
   __syncthreads();
 
   const uint4 *ltab = reinterpret_cast<const uint4 *>(ctab);
@@ -1800,9 +1808,13 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
     for (int sh = 32; sh > 0; sh >>= 1)
       mismatch |= __shfl_xor(mismatch, sh, 64);
     __syncthreads();
-    if (threadIdx.x == 0) {
-      const uint32_t it = gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
-      for (int r = 0; r < nw; r++) {
+    {
+      const int r = int(threadIdx.x);
+      if (r < nw) {
+        const uint32_t it =
+            payload == payload_full
+                ? it_full
+                : gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
         const int col = int((colpack >> (4 * r)) & 0xF);
         const uint32_t crc =
             ~(it ^ red[r] ^ red[16 + r] ^ red[32 + r] ^ red[48 + r]);
