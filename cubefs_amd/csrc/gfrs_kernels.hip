@@ -668,10 +668,20 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_k(
   }
 }
 
-static int crc_grid(int64_t total_frames) {
+static int crc_grid(int64_t total_frames, int64_t frames_per_shard) {
   if (total_frames <= 0) return 1;
   const int64_t cap = env_grid("GFRS_CRC_GRID", 256 * 8);
-  return int(total_frames < cap ? total_frames : cap);
+  int64_t g = total_frames < cap ? total_frames : cap;
+  /* keep the stride coprime with fps so blocks see a mix of frame
+   * positions (see fused_grid) */
+  if (g < total_frames && frames_per_shard > 1)
+    while (g > 1) {
+      int64_t a = g, b = frames_per_shard;
+      while (b) { int64_t t2 = a % b; a = b; b = t2; }
+      if (a == 1) break;
+      g--;
+    }
+  return int(g);
 }
 
 /* the pipelined fused kernels prefer many more blocks than the staged
@@ -929,7 +939,7 @@ static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
                          size_t src_stride, int64_t n, int64_t block_len,
                          int64_t fps, int64_t total, int64_t *bad,
                          hipStream_t s) {
-  const int grid = crc_grid(total);
+  const int grid = crc_grid(total, fps);
   if (fps == 1 && n <= 8192) {
     const int64_t g2l = (total + 3) / 4;
     const int g2 = int(g2l < 2048 ? g2l : 2048);
