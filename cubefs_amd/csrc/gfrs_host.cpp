@@ -913,9 +913,21 @@ int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
     const long v = e ? atol(e) : 0;
     return v > 0 ? size_t(v) : size_t(6144);
   }();
-  const bool fused = block_len == 65536 && t.m >= 1 && gm_all <= 4 &&
-                     t.n + gm_all <= 16 && framed_stride % 4 == 0 &&
-                     shard_len >= fused_min && (t.l == 0 || c->fused_lrc_ok);
+  const bool shapes_ok = block_len == 65536 && t.m >= 1 && gm_all <= 4 &&
+                         t.n + gm_all <= 16 && framed_stride % 4 == 0 &&
+                         (t.l == 0 || c->fused_lrc_ok);
+  if (shapes_ok && shard_len <= 4096) {
+    /* MinShardSize-class shapes: wave-per-stripe fused kernel */
+    const DevPlan &pl = t.l == 0 ? c->enc_plan : c->fused_lrc;
+    launch_rs_encode_frame_small((uint8_t *)framed, framed_stride,
+                                 (uint64_t)base, stripe_stride, shard_len,
+                                 t.n, gm_all, (const uint8_t *)pl.tabs.p,
+                                 nstripes, c->stream);
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess) return hip_fail("encode_frame_small launch", e);
+    return GFRS_OK;
+  }
+  const bool fused = shapes_ok && shard_len >= fused_min;
   if (fused) {
     const DevPlan &pl = t.l == 0 ? c->enc_plan : c->fused_lrc;
     launch_rs_encode_frame((uint8_t *)framed, framed_stride, (uint64_t)base,

@@ -73,6 +73,12 @@ void launch_crc_decode(uint8_t *dst, size_t dst_stride, const uint8_t *framed,
 
 /* fused encode+frame: parity + crc32block framed images in one pass
  * (PUT/repair pipeline; data read once, framed written once). */
+void launch_rs_encode_frame_small(uint8_t *dst, size_t dst_stride,
+                                  uint64_t base, uint64_t stripe_stride,
+                                  size_t shard_len, int k, int m,
+                                  const uint8_t *tabs, int nstripes,
+                                  hipStream_t s);
+
 void launch_rs_repair_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
                             uint64_t stripe_stride, size_t shard_len, int k,
                             int gm, int nw, const int32_t *imap,

@@ -487,6 +487,8 @@ __device__ uint32_t g_crc_tab4[8][256];
  * register-CRC fused kernel can Horner-chain a lane's four 4096-apart
  * uint4 pieces with 4 LDS gathers per step instead of a 32-step mulmod */
 __device__ uint32_t g_shift4k[4][256];
+/* same, for x^(8*1024): the small-shard fused kernel's piece stride */
+__device__ uint32_t g_shift1k[4][256];
 
 /* x^(8*2^j) mod P, reflected domain — host-filled alongside the tables. */
 __device__ uint32_t g_pow8[40];
@@ -1868,6 +1870,172 @@ void launch_rs_repair_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
 #undef GFRS_RP_GO
 }
 
+/* Small-shard fused encode+frame: one 64-lane WAVE per stripe, no LDS
+ * staging, no barriers (everything wave-local).  Shard <= 4096 B is one
+ * short crc32block frame; lane w owns pieces {i*1024 + 16w : i < NI}
+ * (64 lanes x 16 B = 1 KiB per subtile) Horner-chained by the constant
+ * x^(8*1024) byte-sliced tables.  Four independent waves per block.
+ * Covers the reference's 2 KiB MinShardSize PUT shapes in one launch. */
+template <int GM, int NI>
+__global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_encode_frame_small_k(
+    uint8_t *__restrict__ dst, size_t dst_stride, uint64_t base,
+    uint64_t stripe_stride, size_t shard_len, int k,
+    const uint8_t *__restrict__ tabs /* [GM*k][32] */, int64_t nstripes) {
+  extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
+  uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
+  uint32_t(*stab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem + 8192);
+  uint32_t *x8tab = reinterpret_cast<uint32_t *>(smem + 12288);
+  uint8_t *tailb = smem + 12288 + 64; /* per-wave 16x16 tail bytes */
+  uint8_t *ctab = smem + 12288 + 64 + 4 * 256;
+  for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
+    (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
+  for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
+    (&stab[0][0])[i] = (&g_shift1k[0][0])[i];
+  for (int i = threadIdx.x; i < GM * k * 2; i += CRC_BLOCKT)
+    reinterpret_cast<uint4 *>(ctab)[i] =
+        reinterpret_cast<const uint4 *>(tabs)[i];
+  if (threadIdx.x == 0) {
+    uint32_t v = 0x80000000u;
+    for (int j = 0; j < 16; j++) {
+      x8tab[j] = v;
+      v = gf2_mulmod_d(v, g_pow8[0]);
+    }
+  }
+  __syncthreads();
+
+  const int wv = int(threadIdx.x) >> 6, lane = int(threadIdx.x) & 63;
+  const int lane16i = lane * 16;
+  uint8_t *wtail = tailb + wv * 256;
+  const uint4 *ltab = reinterpret_cast<const uint4 *>(ctab);
+  const int64_t payload = int64_t(shard_len);
+  const int pli = int(payload);
+  /* lane's fold operator: suffix after its LAST present piece */
+  int np = 0;
+#pragma unroll
+  for (int i = 0; i < NI; i++)
+    if (i * 1024 + lane16i + 16 <= pli) np = i + 1;
+  const uint32_t op =
+      np ? x8n_d(uint64_t(pli - ((np - 1) * 1024 + lane16i + 16))) : 0;
+  const uint32_t it = gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
+  const int t0 = (pli / 16) * 16;
+
+  for (int64_t stripe = int64_t(blockIdx.x) * 4 + wv; stripe < nstripes;
+       stripe += int64_t(gridDim.x) * 4) {
+    const uint8_t *sbase = as_global(base + stripe * stripe_stride);
+    uint4 acc[GM][NI];
+#pragma unroll
+    for (int r = 0; r < GM; r++)
+#pragma unroll
+      for (int i = 0; i < NI; i++) acc[r][i] = uint4{0, 0, 0, 0};
+
+    for (int c = 0; c < k; c++) {
+      const uint8_t *src = sbase + size_t(c) * shard_len;
+      uint8_t *fdst =
+          dst + (stripe * (k + GM) + c) * dst_stride + CRC_LEN;
+      uint32_t t = 0;
+#pragma unroll
+      for (int i = 0; i < NI; i++) {
+        const int off = i * 1024 + lane16i;
+        if (off + 16 <= pli) {
+          const uint4 v = *reinterpret_cast<const uint4 *>(src + off);
+#pragma unroll
+          for (int r = 0; r < GM; r++) {
+            const int t2 = (r * k + c) * 2;
+            gfmac16<false>(acc[r][i], v, ltab[t2], ltab[t2 + 1]);
+          }
+          uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+          dw[0] = v.x; dw[1] = v.y; dw[2] = v.z; dw[3] = v.w;
+          t = shift4k(t, stab) ^ crc16_reg(v, tab); /* stab = x^(8*1024) */
+        }
+      }
+      uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
+      { /* tail bytes, one per lane; stage for the parity tails */
+        const int p = t0 + lane;
+        if (p < pli) {
+          const uint8_t x = src[p];
+          fdst[p] = x;
+          wtail[c * 16 + (p - t0)] = x;
+          part ^= gf2_mulmod_d(x8tab[pli - 1 - p], tab[0][x]);
+        }
+      }
+#pragma unroll
+      for (int sh = 32; sh > 0; sh >>= 1)
+        part ^= __shfl_xor(part, sh, 64);
+      if (lane == 0)
+        *reinterpret_cast<uint32_t *>(
+            dst + (stripe * (k + GM) + c) * dst_stride) = ~(it ^ part);
+    }
+#pragma unroll
+    for (int r = 0; r < GM; r++) {
+      uint8_t *fdst =
+          dst + (stripe * (k + GM) + k + r) * dst_stride + CRC_LEN;
+      uint32_t t = 0;
+#pragma unroll
+      for (int i = 0; i < NI; i++) {
+        const int off = i * 1024 + lane16i;
+        if (off + 16 <= pli) {
+          uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+          dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
+          dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
+          t = shift4k(t, stab) ^ crc16_reg(acc[r][i], tab);
+        }
+      }
+      uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
+      {
+        const int p = t0 + lane;
+        if (p < pli) { /* same-wave LDS visibility: no barrier needed */
+          uint8_t pv = 0;
+          for (int c2 = 0; c2 < k; c2++) {
+            const uint8_t b = wtail[c2 * 16 + (p - t0)];
+            const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
+            pv ^= tt[b & 0x0F] ^ tt[16 + (b >> 4)];
+          }
+          fdst[p] = pv;
+          part ^= gf2_mulmod_d(x8tab[pli - 1 - p], tab[0][pv]);
+        }
+      }
+#pragma unroll
+      for (int sh = 32; sh > 0; sh >>= 1)
+        part ^= __shfl_xor(part, sh, 64);
+      if (lane == 0)
+        *reinterpret_cast<uint32_t *>(
+            dst + (stripe * (k + GM) + k + r) * dst_stride) = ~(it ^ part);
+    }
+  }
+}
+
+void launch_rs_encode_frame_small(uint8_t *dst, size_t dst_stride,
+                                  uint64_t base, uint64_t stripe_stride,
+                                  size_t shard_len, int k, int m,
+                                  const uint8_t *tabs, int nstripes,
+                                  hipStream_t s) {
+  const int64_t groups = (int64_t(nstripes) + 3) / 4;
+  const int64_t cap = env_grid("GFRS_CRC_GRID", 16384);
+  int grid = int(groups < cap ? groups : cap);
+  if (grid < 1) grid = 1;
+  const int ni = int((shard_len + 1023) / 1024);
+  const int lds = 12288 + 64 + 1024 + m * k * 32;
+#define GFRS_SM_GO(G, I)                                                  \
+  hipLaunchKernelGGL((rs_encode_frame_small_k<G, I>), dim3(grid),         \
+                     dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,     \
+                     stripe_stride, shard_len, k, tabs, nstripes)
+#define GFRS_SM_NI(G)                                                     \
+  switch (ni) {                                                           \
+    case 1: GFRS_SM_GO(G, 1); break;                                      \
+    case 2: GFRS_SM_GO(G, 2); break;                                      \
+    case 3: GFRS_SM_GO(G, 3); break;                                      \
+    default: GFRS_SM_GO(G, 4);                                            \
+  }
+  switch (m) {
+    case 1: GFRS_SM_NI(1); break;
+    case 2: GFRS_SM_NI(2); break;
+    case 3: GFRS_SM_NI(3); break;
+    default: GFRS_SM_NI(4);
+  }
+#undef GFRS_SM_NI
+#undef GFRS_SM_GO
+}
+
 void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
                             uint64_t stripe_stride, size_t shard_len, int k,
                             int m, const uint8_t *tabs, int nstripes,
@@ -2198,6 +2366,11 @@ int crc_device_init_current(void) {
     for (uint32_t b = 0; b < 256; b++)
       s4k[j][b] = mulmod(pow8[12], b << (8 * j));
   if (hipMemcpyToSymbol(HIP_SYMBOL(g_shift4k), s4k, sizeof(s4k)) != hipSuccess)
+    return -100;
+  for (int j = 0; j < 4; j++)
+    for (uint32_t b = 0; b < 256; b++)
+      s4k[j][b] = mulmod(pow8[10], b << (8 * j)); /* x^(8*2^10) */
+  if (hipMemcpyToSymbol(HIP_SYMBOL(g_shift1k), s4k, sizeof(s4k)) != hipSuccess)
     return -100;
   if (hipMemcpyToSymbol(HIP_SYMBOL(g_crc_tab4), tab, sizeof(tab)) != hipSuccess)
     return -100;

@@ -121,7 +121,14 @@ def test_fused_encode_frame(oracle, dev):
                        ("EC6P3", 1 << 20), ("EC12P4", 200000),
                        ("EC15P12", 100000),   # fallback (m > 4)
                        ("LRC12P2L2", 300000),  # fused composed-LRC plan
-                       ("EC6P10L2", 300000)]:  # LRC fallback (m+l > 4)
+                       ("EC6P10L2", 300000),   # LRC fallback (m+l > 4)
+                       # wave-per-stripe small kernel (<= 4096) across
+                       # NI boundaries, odd sizes, and the composed-LRC
+                       # plan; 5000 lands in the two-kernel gap
+                       ("EC6P3", 2048), ("EC6P3", 1024), ("EC6P3", 1040),
+                       ("EC6P3", 17), ("EC6P3", 2049), ("EC6P3", 4095),
+                       ("EC6P3", 4096), ("EC12P4", 3000),
+                       ("LRC12P2L2", 2048), ("EC6P3", 5000)]:
         t = codemode.get_tactic(name)
         ns = 3
         rng = np.random.default_rng(slen ^ t.N)
