@@ -907,11 +907,13 @@ int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
    * source): fused wins from 8 KiB shards up (837 vs 306 GiB/s at 8 KiB,
    * ~1180 vs ~780 at 16 KiB-8 MiB); the two-kernel path with rs_apply
    * stripe packing wins at 2-4 KiB (684 vs 320 at 2 KiB).
-   * GFRS_FUSED_MIN overrides the threshold (bytes). */
+   * GFRS_FUSED_MIN overrides the threshold (bytes); shards
+   * <= 4096 take the wave-per-stripe small kernel, and at
+   * 5 KiB the big fused kernel beats two-kernel 622 vs 392. */
   static const size_t fused_min = []() {
     const char *e = getenv("GFRS_FUSED_MIN");
     const long v = e ? atol(e) : 0;
-    return v > 0 ? size_t(v) : size_t(6144);
+    return v > 0 ? size_t(v) : size_t(4097);
   }();
   const bool shapes_ok = block_len == 65536 && t.m >= 1 && gm_all <= 4 &&
                          t.n + gm_all <= 16 && framed_stride % 4 == 0 &&
@@ -1477,7 +1479,7 @@ int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
     static const size_t fmin = []() {
       const char *e = getenv("GFRS_FUSED_MIN");
       const long v = e ? atol(e) : 0;
-      return v > 0 ? size_t(v) : size_t(6144);
+      return v > 0 ? size_t(v) : size_t(4097);
     }();
     int nglobad = 0;
     for (int i = 0; i < nbad; i++)
