@@ -505,3 +505,34 @@ def test_lrc_reconstruct_verify(oracle, dev, bad):
     fails = enc.reconstruct_verify_batch(batch, bad)
     enc.synchronize()
     assert fails[1] and sum(fails) == 1, (bad, fails)
+
+
+def test_repair_batch_small_legacy_path(oracle, dev):
+    """Shards below the fused-repair gate take the legacy
+    reconstruct_verify + shard_write path; images stay bit-exact."""
+    import torch
+    from cubefs_amd import codemode, ec, shard
+    t = codemode.get_tactic("EC6P3")
+    enc = ec.Encoder(t)
+    ns, slen = 6, 3000
+    rng = np.random.default_rng(505)
+    arr = rng.integers(0, 256, (ns, t.total, slen), dtype=np.uint8)
+    batch = torch.from_numpy(arr).to(dev)
+    enc.encode_batch(batch)
+    enc.synchronize()
+    ref = batch.cpu().numpy()
+    bad = [4, 8]
+    for i in bad:
+        batch[:, i].zero_()
+    dsz = shard.disk_size(slen)
+    imgs = torch.zeros((ns * 2, dsz), dtype=torch.uint8, device=dev)
+    fails = enc.repair_batch(batch, bad, imgs,
+                             list(range(ns * 2)), [3] * (ns * 2))
+    enc.synchronize()
+    assert fails == [False] * ns, fails
+    got = imgs.cpu().numpy()
+    for s in range(ns):
+        for b, shard_idx in enumerate(bad):
+            want = oracle.shard_write(ref[s, shard_idx].copy(),
+                                      bid=s * 2 + b, vuid=3)
+            assert np.array_equal(got[s * 2 + b], want), (s, b)
