@@ -301,8 +301,8 @@ def main():
         dsz = shardmod.disk_size(S)
         imgs = torch.empty((ns * len(bad), dsz), dtype=torch.uint8,
                            device=dev)
-        rbids = list(range(ns * len(bad)))
-        rvuids = [1] * (ns * len(bad))
+        rbids = np.arange(ns * len(bad), dtype=np.uint64)
+        rvuids = np.ones(ns * len(bad), dtype=np.uint64)
 
     fused = (args.workload == "encode" and with_crc and not args.no_fused)
 

@@ -1073,28 +1073,24 @@ int gfrs_shard_write_batch(gfrs_ctx *ctx, void *dst, size_t dst_stride,
   std::lock_guard<std::mutex> lk(c->mu);
   StreamGuard g(c);
   int rc;
-  /* prebuild the 32 B headers on host (shard.go:241-261) */
-  std::vector<uint8_t> hdrs(size_t(nshards) * 32, 0);
-  for (int j = 0; j < nshards; j++) {
-    uint8_t *h = &hdrs[size_t(j) * 32];
-    h[4] = 0xab; h[5] = 0xcd; h[6] = 0xef; h[7] = 0xcc;
-    put_be64(h + 8, bids[j]);
-    put_be64(h + 16, vuids[j]);
-    put_be32(h + 24, uint32_t(size));
-    /* reserved h[28..32) = 0 */
-    put_be32(h, host_crc32(h + 4, 28));
-  }
-  if ((rc = c->stage_pin.ensure(hdrs.size())) != GFRS_OK) return rc;
-  memcpy(c->stage_pin.p, hdrs.data(), hdrs.size());
+  /* ship the raw id arrays; the finalize kernel builds the 32 B headers
+   * (shard.go:241-261) on device - host-side construction was the
+   * bottleneck at millions of shards per call */
+  const size_t idbytes = size_t(nshards) * 8;
+  if ((rc = c->stage_pin.ensure(idbytes * 2)) != GFRS_OK) return rc;
+  memcpy(c->stage_pin.p, bids, idbytes);
+  memcpy((uint8_t *)c->stage_pin.p + idbytes, vuids, idbytes);
   DevBuf &hbuf = c->ptr_buf; /* reuse scratch (stream-ordered) */
-  if ((rc = hbuf.ensure(hdrs.size())) != GFRS_OK) return rc;
-  HIP_TRY(hipMemcpyAsync(hbuf.p, c->stage_pin.p, hdrs.size(),
+  if ((rc = hbuf.ensure(idbytes * 2)) != GFRS_OK) return rc;
+  HIP_TRY(hipMemcpyAsync(hbuf.p, c->stage_pin.p, idbytes * 2,
                          hipMemcpyHostToDevice, c->stream));
   /* framed body at +32 */
   launch_crc_encode((uint8_t *)dst + 32, dst_stride, (const uint8_t *)src,
                     src_stride, size, block_len, nshards, c->stream);
-  launch_shard_finalize((uint8_t *)dst, dst_stride, (const uint8_t *)hbuf.p,
-                        size, block_len, nshards, c->stream);
+  launch_shard_finalize((uint8_t *)dst, dst_stride,
+                        (const uint64_t *)hbuf.p,
+                        (const uint64_t *)hbuf.p + nshards, size, block_len,
+                        nshards, c->stream);
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) return hip_fail("shard_write launch", e);
   return GFRS_OK;
@@ -1486,7 +1482,8 @@ int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
       if (bad_idx[i] < t.n + t.m) nglobad++;
     if ((t.l == 0 || cc->fused_lrc_ok) && block_len == 65536 && t.m >= 1 &&
         t.n + t.m + t.l <= 16 && nglobad <= t.m && nbad <= 4 &&
-        shard_len >= fmin && dst_stride % 4 == 0 && nstripes > 0) {
+        (shard_len >= fmin || shard_len <= 4096) && dst_stride % 4 == 0 &&
+        nstripes > 0) {
       const int k = t.n, m = t.m;
       const int total_sh = k + m + t.l;
       std::vector<uint8_t> present(k + m, 1);
@@ -1617,32 +1614,37 @@ int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
         return rc2;
       HIP_TRY(hipMemsetAsync(cc->fail_buf.p, 0, size_t(nstripes) * 4,
                              cc->stream));
-      /* headers for every (stripe, bad) image, caller's column order */
-      std::vector<uint8_t> hdrs(size_t(nstripes) * nbad * 32, 0);
-      for (size_t j = 0; j < size_t(nstripes) * nbad; j++) {
-        uint8_t *h = &hdrs[j * 32];
-        h[4] = 0xab; h[5] = 0xcd; h[6] = 0xef; h[7] = 0xcc;
-        put_be64(h + 8, bids[j]);
-        put_be64(h + 16, vuids[j]);
-        put_be32(h + 24, uint32_t(shard_len));
-        put_be32(h, host_crc32(h + 4, 28));
-      }
-      if ((rc2 = cc->stage_pin.ensure(hdrs.size())) != GFRS_OK) return rc2;
-      memcpy(cc->stage_pin.p, hdrs.data(), hdrs.size());
+      /* id arrays for every (stripe, bad) image, caller's column order;
+       * headers themselves are built by the finalize kernel */
+      const size_t nimg = size_t(nstripes) * nbad, idbytes = nimg * 8;
+      if ((rc2 = cc->stage_pin.ensure(idbytes * 2)) != GFRS_OK) return rc2;
+      memcpy(cc->stage_pin.p, bids, idbytes);
+      memcpy((uint8_t *)cc->stage_pin.p + idbytes, vuids, idbytes);
       DevBuf &hbuf = cc->ptr_buf;
-      if ((rc2 = hbuf.ensure(hdrs.size())) != GFRS_OK) return rc2;
-      HIP_TRY(hipMemcpyAsync(hbuf.p, cc->stage_pin.p, hdrs.size(),
+      if ((rc2 = hbuf.ensure(idbytes * 2)) != GFRS_OK) return rc2;
+      HIP_TRY(hipMemcpyAsync(hbuf.p, cc->stage_pin.p, idbytes * 2,
                              hipMemcpyHostToDevice, cc->stream));
-      launch_rs_repair_frame((uint8_t *)disk_dst + 32, dst_stride,
-                             (uint64_t)base, stripe_stride, shard_len,
-                             plan->k, gm, nbad,
-                             (const int32_t *)plan->in_idx.p,
-                             (const uint8_t *)plan->tabs.p, colpack,
-                             (uint32_t *)cc->fail_buf.p, nstripes,
-                             cc->stream);
+      if (shard_len <= 4096)
+        launch_rs_repair_frame_small((uint8_t *)disk_dst + 32, dst_stride,
+                                     (uint64_t)base, stripe_stride,
+                                     shard_len, plan->k, gm, nbad,
+                                     (const int32_t *)plan->in_idx.p,
+                                     (const uint8_t *)plan->tabs.p, colpack,
+                                     (uint32_t *)cc->fail_buf.p, nstripes,
+                                     cc->stream);
+      else
+        launch_rs_repair_frame((uint8_t *)disk_dst + 32, dst_stride,
+                               (uint64_t)base, stripe_stride, shard_len,
+                               plan->k, gm, nbad,
+                               (const int32_t *)plan->in_idx.p,
+                               (const uint8_t *)plan->tabs.p, colpack,
+                               (uint32_t *)cc->fail_buf.p, nstripes,
+                               cc->stream);
       launch_shard_finalize((uint8_t *)disk_dst, dst_stride,
-                            (const uint8_t *)hbuf.p, int64_t(shard_len),
-                            block_len, nstripes * nbad, cc->stream);
+                            (const uint64_t *)hbuf.p,
+                            (const uint64_t *)hbuf.p + nimg,
+                            int64_t(shard_len), block_len, nstripes * nbad,
+                            cc->stream);
       hipError_t e = hipGetLastError();
       if (e != hipSuccess) return hip_fail("repair_frame launch", e);
       std::vector<uint32_t> fails(nstripes);

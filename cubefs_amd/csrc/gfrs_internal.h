@@ -79,6 +79,13 @@ void launch_rs_encode_frame_small(uint8_t *dst, size_t dst_stride,
                                   const uint8_t *tabs, int nstripes,
                                   hipStream_t s);
 
+void launch_rs_repair_frame_small(uint8_t *dst, size_t dst_stride,
+                                  uint64_t base, uint64_t stripe_stride,
+                                  size_t shard_len, int k, int gm, int nw,
+                                  const int32_t *imap, const uint8_t *tabs,
+                                  uint32_t colpack, uint32_t *fail,
+                                  int nstripes, hipStream_t s);
+
 void launch_rs_repair_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
                             uint64_t stripe_stride, size_t shard_len, int k,
                             int gm, int nw, const int32_t *imap,
@@ -121,7 +128,8 @@ void launch_rs_apply_xor(const uint64_t *ptrs, int nptr,
 
 /* blobnode on-disk shard codec (core/shard.go, datafile.go:342-445). */
 void launch_shard_finalize(uint8_t *dst, size_t dst_stride,
-                           const uint8_t *headers, int64_t raw_size,
+                           const uint64_t *bids, const uint64_t *vuids,
+                           int64_t raw_size,
                            int64_t block_len, int nshards, hipStream_t s);
 void launch_shard_parse(const uint8_t *img, size_t stride, int64_t raw_size,
                         int64_t block_len, int nshards, uint64_t *out,

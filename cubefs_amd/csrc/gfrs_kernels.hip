@@ -1851,6 +1851,194 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
   }
 }
 
+/* Small-shard fused repair: wave-per-stripe variant of rs_repair_frame_k
+ * for shards <= 4096 B (one short frame).  Same plan/imap/colpack
+ * contract; inputs MAC-only, rebuild rows become framed image bodies,
+ * check rows compare against the surviving shard; input tail bytes are
+ * staged in a wave-local LDS slab (same-wave visibility, no barriers). */
+template <int GM, int NI>
+__global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_small_k(
+    uint8_t *__restrict__ dst, size_t dst_stride, uint64_t base,
+    uint64_t stripe_stride, size_t shard_len, int k,
+    const int32_t *__restrict__ imap, const uint8_t *__restrict__ tabs,
+    int nw, uint32_t colpack, uint32_t *__restrict__ fail,
+    int64_t nstripes) {
+  extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
+  uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
+  uint32_t(*stab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem + 8192);
+  uint32_t *x8tab = reinterpret_cast<uint32_t *>(smem + 12288);
+  uint8_t *tailb = smem + 12288 + 64;
+  uint8_t *ctab = smem + 12288 + 64 + 4 * 256;
+  for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
+    (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
+  for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
+    (&stab[0][0])[i] = (&g_shift1k[0][0])[i];
+  for (int i = threadIdx.x; i < GM * k * 2; i += CRC_BLOCKT)
+    reinterpret_cast<uint4 *>(ctab)[i] =
+        reinterpret_cast<const uint4 *>(tabs)[i];
+  if (threadIdx.x == 0) {
+    uint32_t v = 0x80000000u;
+    for (int j = 0; j < 16; j++) {
+      x8tab[j] = v;
+      v = gf2_mulmod_d(v, g_pow8[0]);
+    }
+  }
+  __syncthreads();
+
+  const int wv = int(threadIdx.x) >> 6, lane = int(threadIdx.x) & 63;
+  const int lane16i = lane * 16;
+  uint8_t *wtail = tailb + wv * 256;
+  const uint4 *ltab = reinterpret_cast<const uint4 *>(ctab);
+  const int pli = int(shard_len);
+  int np = 0;
+#pragma unroll
+  for (int i = 0; i < NI; i++)
+    if (i * 1024 + lane16i + 16 <= pli) np = i + 1;
+  const uint32_t op =
+      np ? x8n_d(uint64_t(pli - ((np - 1) * 1024 + lane16i + 16))) : 0;
+  const uint32_t it = gf2_mulmod_d(x8n_d(uint64_t(pli)), 0xFFFFFFFFu);
+  const int t0 = (pli / 16) * 16;
+
+  for (int64_t stripe = int64_t(blockIdx.x) * 4 + wv; stripe < nstripes;
+       stripe += int64_t(gridDim.x) * 4) {
+    const uint8_t *sbase = as_global(base + stripe * stripe_stride);
+    uint4 acc[GM][NI];
+#pragma unroll
+    for (int r = 0; r < GM; r++)
+#pragma unroll
+      for (int i = 0; i < NI; i++) acc[r][i] = uint4{0, 0, 0, 0};
+
+    for (int c = 0; c < k; c++) {
+      const uint8_t *src = sbase + size_t(imap[c]) * shard_len;
+#pragma unroll
+      for (int i = 0; i < NI; i++) {
+        const int off = i * 1024 + lane16i;
+        if (off + 16 <= pli) {
+          const uint4 v = *reinterpret_cast<const uint4 *>(src + off);
+#pragma unroll
+          for (int r = 0; r < GM; r++) {
+            const int t2 = (r * k + c) * 2;
+            gfmac16<false>(acc[r][i], v, ltab[t2], ltab[t2 + 1]);
+          }
+        }
+      }
+    }
+    if (t0 < pli) { /* stage input tail bytes, 4 shards per sweep */
+      for (int cb = 0; cb < k; cb += 4) {
+        const int c2 = cb + (lane >> 4), j = lane & 15;
+        if (c2 < k && t0 + j < pli)
+          wtail[c2 * 16 + j] =
+              sbase[size_t(imap[c2]) * shard_len + t0 + j];
+      }
+    }
+
+    uint32_t mismatch = 0;
+#pragma unroll
+    for (int r = 0; r < GM; r++) {
+      if (r < nw) {
+        const int col = int((colpack >> (4 * r)) & 0xF);
+        uint8_t *fdst =
+            dst + (stripe * nw + col) * dst_stride + CRC_LEN;
+        uint32_t t = 0;
+#pragma unroll
+        for (int i = 0; i < NI; i++) {
+          const int off = i * 1024 + lane16i;
+          if (off + 16 <= pli) {
+            uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
+            dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
+            dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
+            t = shift4k(t, stab) ^ crc16_reg(acc[r][i], tab);
+          }
+        }
+        uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
+        {
+          const int p = t0 + lane;
+          if (p < pli) {
+            uint8_t pv = 0;
+            for (int c2 = 0; c2 < k; c2++) {
+              const uint8_t b = wtail[c2 * 16 + (p - t0)];
+              const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
+              pv ^= tt[b & 0x0F] ^ tt[16 + (b >> 4)];
+            }
+            fdst[p] = pv;
+            part ^= gf2_mulmod_d(x8tab[pli - 1 - p], tab[0][pv]);
+          }
+        }
+#pragma unroll
+        for (int sh = 32; sh > 0; sh >>= 1)
+          part ^= __shfl_xor(part, sh, 64);
+        if (lane == 0)
+          *reinterpret_cast<uint32_t *>(
+              dst + (stripe * nw + col) * dst_stride) = ~(it ^ part);
+      } else {
+        const uint8_t *cshard =
+            sbase + size_t(imap[k + (r - nw)]) * shard_len;
+        uint32_t d2 = 0;
+#pragma unroll
+        for (int i = 0; i < NI; i++) {
+          const int off = i * 1024 + lane16i;
+          if (off + 16 <= pli) {
+            const uint4 w = *reinterpret_cast<const uint4 *>(cshard + off);
+            d2 |= (w.x ^ acc[r][i].x) | (w.y ^ acc[r][i].y) |
+                  (w.z ^ acc[r][i].z) | (w.w ^ acc[r][i].w);
+          }
+        }
+        {
+          const int p = t0 + lane;
+          if (p < pli) {
+            uint8_t pv = 0;
+            for (int c2 = 0; c2 < k; c2++) {
+              const uint8_t b = wtail[c2 * 16 + (p - t0)];
+              const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
+              pv ^= tt[b & 0x0F] ^ tt[16 + (b >> 4)];
+            }
+            d2 |= uint32_t(pv ^ cshard[p]);
+          }
+        }
+        mismatch |= d2;
+      }
+    }
+#pragma unroll
+    for (int sh = 32; sh > 0; sh >>= 1)
+      mismatch |= __shfl_xor(mismatch, sh, 64);
+    if (lane == 0 && mismatch) atomicOr(&fail[stripe], 1u);
+  }
+}
+
+void launch_rs_repair_frame_small(uint8_t *dst, size_t dst_stride,
+                                  uint64_t base, uint64_t stripe_stride,
+                                  size_t shard_len, int k, int gm, int nw,
+                                  const int32_t *imap, const uint8_t *tabs,
+                                  uint32_t colpack, uint32_t *fail,
+                                  int nstripes, hipStream_t s) {
+  const int64_t groups = (int64_t(nstripes) + 3) / 4;
+  const int64_t cap = env_grid("GFRS_CRC_GRID", 16384);
+  int grid = int(groups < cap ? groups : cap);
+  if (grid < 1) grid = 1;
+  const int ni = int((shard_len + 1023) / 1024);
+  const int lds = 12288 + 64 + 1024 + gm * k * 32;
+#define GFRS_RPS_GO(G, I)                                                 \
+  hipLaunchKernelGGL((rs_repair_frame_small_k<G, I>), dim3(grid),         \
+                     dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,     \
+                     stripe_stride, shard_len, k, imap, tabs, nw,         \
+                     colpack, fail, nstripes)
+#define GFRS_RPS_NI(G)                                                    \
+  switch (ni) {                                                           \
+    case 1: GFRS_RPS_GO(G, 1); break;                                     \
+    case 2: GFRS_RPS_GO(G, 2); break;                                     \
+    case 3: GFRS_RPS_GO(G, 3); break;                                     \
+    default: GFRS_RPS_GO(G, 4);                                           \
+  }
+  switch (gm) {
+    case 1: GFRS_RPS_NI(1); break;
+    case 2: GFRS_RPS_NI(2); break;
+    case 3: GFRS_RPS_NI(3); break;
+    default: GFRS_RPS_NI(4);
+  }
+#undef GFRS_RPS_NI
+#undef GFRS_RPS_GO
+}
+
 void launch_rs_repair_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
                             uint64_t stripe_stride, size_t shard_len, int k,
                             int gm, int nw, const int32_t *imap,
@@ -2229,18 +2417,41 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
 /* One wave per shard: write the 32-B header (prebuilt on host), fold the
  * frame CRCs into the footer. */
 __global__ void shard_finalize_k(uint8_t *__restrict__ dst, size_t dst_stride,
-                                 const uint8_t *__restrict__ headers,
+                                 const uint64_t *__restrict__ bids,
+                                 const uint64_t *__restrict__ vuids,
                                  int64_t raw_size, int64_t block_len,
                                  int nshards) {
   const int sh = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
   const int lane = threadIdx.x & 63;
   if (sh >= nshards) return;
   uint8_t *img = dst + size_t(sh) * dst_stride;
-  /* header copy, 32 B by the first 8 lanes */
-  if (lane < 8)
-    reinterpret_cast<uint32_t *>(img)[lane] =
-        reinterpret_cast<const uint32_t *>(headers + sh * 32)[lane];
   if (lane != 0) return;
+  /* build the 32 B header on device (shard.go:241-261): host-side
+   * construction was the bottleneck at millions of shards per call.
+   * Memory is LE; the format stores bid/vuid/size/crc big-endian. */
+  {
+    const uint64_t bid = bids[sh], vuid = vuids[sh];
+    uint32_t w[8];
+    w[1] = 0xccefcdabu; /* ab cd ef cc */
+    w[2] = __builtin_bswap32(uint32_t(bid >> 32));
+    w[3] = __builtin_bswap32(uint32_t(bid));
+    w[4] = __builtin_bswap32(uint32_t(vuid >> 32));
+    w[5] = __builtin_bswap32(uint32_t(vuid));
+    w[6] = __builtin_bswap32(uint32_t(raw_size));
+    w[7] = 0;
+    uint32_t c = 0xFFFFFFFFu;
+#pragma unroll
+    for (int i = 1; i < 8; i++) {
+      const uint32_t x = w[i];
+#pragma unroll
+      for (int b = 0; b < 4; b++)
+        c = g_crc_tab4[0][(c ^ (x >> (8 * b))) & 0xFF] ^ (c >> 8);
+    }
+    w[0] = __builtin_bswap32(~c);
+    uint32_t *d32 = reinterpret_cast<uint32_t *>(img);
+#pragma unroll
+    for (int i = 0; i < 8; i++) d32[i] = w[i];
+  }
   const int64_t payload_full = block_len - CRC_LEN;
   const int64_t nframes = (raw_size + payload_full - 1) / payload_full;
   uint32_t crc = 0;
@@ -2316,12 +2527,14 @@ __global__ void shard_parse_k(const uint8_t *__restrict__ img0,
 }
 
 void launch_shard_finalize(uint8_t *dst, size_t dst_stride,
-                           const uint8_t *headers, int64_t raw_size,
-                           int64_t block_len, int nshards, hipStream_t s) {
+                           const uint64_t *bids, const uint64_t *vuids,
+                           int64_t raw_size, int64_t block_len, int nshards,
+                           hipStream_t s) {
   const int wps = 4; /* waves per block */
   const int blocks = (nshards + wps - 1) / wps;
   hipLaunchKernelGGL(shard_finalize_k, dim3(blocks), dim3(wps * 64), 0, s,
-                     dst, dst_stride, headers, raw_size, block_len, nshards);
+                     dst, dst_stride, bids, vuids, raw_size, block_len,
+                     nshards);
 }
 
 void launch_shard_parse(const uint8_t *img, size_t stride, int64_t raw_size,

@@ -220,7 +220,10 @@ class Encoder:
         bm = (ctypes.c_uint64 * nwords)()
         check(lib().gfrs_verify_batch(self._ctx, p, ln, stride, ns, bm),
               "verify_batch")
-        return [bool(bm[s // 64] >> (s % 64) & 1) for s in range(ns)]
+        import numpy as np
+        bmn = np.ctypeslib.as_array(bm)
+        bits = np.unpackbits(bmn.view(np.uint8), bitorder="little")[:ns]
+        return bits.astype(bool).tolist()
 
     def reconstruct_batch(self, batch, bad_idx, data_only=False):
         p, ln, stride, ns = self._base(batch)
@@ -260,7 +263,10 @@ class Encoder:
         check(lib().gfrs_reconstruct_verify_batch(self._ctx, p, ln, stride,
                                                   ns, bad, len(bad_idx), bm),
               "reconstruct_verify")
-        return [bool(bm[s // 64] >> (s % 64) & 1) for s in range(ns)]
+        import numpy as np
+        bmn = np.ctypeslib.as_array(bm)
+        bits = np.unpackbits(bmn.view(np.uint8), bitorder="little")[:ns]
+        return bits.astype(bool).tolist()
 
     def repair_batch(self, batch, bad_idx, disk_dst, bids, vuids,
                      block_len=65536):
@@ -270,17 +276,23 @@ class Encoder:
         disk_dst: [nstripes*len(bad_idx), disk_size] device tensor;
         bids/vuids: flat per (stripe, bad) row-major.
         Returns the per-stripe verify-fail list."""
+        import numpy as np
         p, ln, stride, ns = self._base(batch)
         nb = len(bad_idx)
         bad = (ctypes.c_int32 * nb)(*bad_idx)
-        ab = (ctypes.c_uint64 * (ns * nb))(*bids)
-        av = (ctypes.c_uint64 * (ns * nb))(*vuids)
+        # large batches: keep the id arrays and fail bitmap vectorized
+        abn = np.ascontiguousarray(bids, dtype=np.uint64)
+        avn = np.ascontiguousarray(vuids, dtype=np.uint64)
+        ab = abn.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64))
+        av = avn.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64))
         nwords = (ns + 63) // 64
-        bm = (ctypes.c_uint64 * nwords)()
+        bmn = np.zeros(nwords, dtype=np.uint64)
+        bm = bmn.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64))
         check(lib().gfrs_repair_batch(self._ctx, p, ln, stride, ns, bad, nb,
                                       disk_dst.data_ptr(), disk_dst.stride(0),
                                       block_len, ab, av, bm), "repair_batch")
-        return [bool(bm[s // 64] >> (s % 64) & 1) for s in range(ns)]
+        bits = np.unpackbits(bmn.view(np.uint8), bitorder="little")[:ns]
+        return bits.astype(bool).tolist()
 
     def synchronize(self):
         check(lib().gfrs_synchronize(self._ctx), "synchronize")

@@ -507,14 +507,19 @@ def test_lrc_reconstruct_verify(oracle, dev, bad):
     assert fails[1] and sum(fails) == 1, (bad, fails)
 
 
-def test_repair_batch_small_legacy_path(oracle, dev):
-    """Shards below the fused-repair gate take the legacy
-    reconstruct_verify + shard_write path; images stay bit-exact."""
+@pytest.mark.parametrize("name,slen", [
+    ("EC6P3", 3000),     # wave-per-stripe small repair kernel
+    ("EC6P3", 1000),     # NI=1
+    ("EC15P12", 3000),   # m > 4: legacy reconstruct_verify + shard_write
+])
+def test_repair_batch_small_paths(oracle, dev, name, slen):
+    """Small-shard repair via the wave-per-stripe fused kernel, and the
+    legacy fallback for m > 4; images stay bit-exact either way."""
     import torch
     from cubefs_amd import codemode, ec, shard
-    t = codemode.get_tactic("EC6P3")
+    t = codemode.get_tactic(name)
     enc = ec.Encoder(t)
-    ns, slen = 6, 3000
+    ns = 6
     rng = np.random.default_rng(505)
     arr = rng.integers(0, 256, (ns, t.total, slen), dtype=np.uint8)
     batch = torch.from_numpy(arr).to(dev)
