@@ -1031,31 +1031,8 @@ int64_t gfrs_sized_decode(gfrs_ctx *ctx, void *dst, const void *framed,
 
 /* ---------------- blobnode shard images ---------------- */
 
-/* host-side CRC32-IEEE for the 28 header bytes (shard.go:258-260) */
-static uint32_t host_crc32(const uint8_t *p, size_t n) {
-  static uint32_t tab[256];
-  static bool init = false;
-  if (!init) {
-    for (uint32_t i = 0; i < 256; i++) {
-      uint32_t c = i;
-      for (int j = 0; j < 8; j++) c = (c & 1) ? (c >> 1) ^ 0xEDB88320u : c >> 1;
-      tab[i] = c;
-    }
-    init = true;
-  }
-  uint32_t c = 0xFFFFFFFFu;
-  for (size_t i = 0; i < n; i++) c = tab[(c ^ p[i]) & 0xFF] ^ (c >> 8);
-  return ~c;
-}
-
-static void put_be32(uint8_t *p, uint32_t v) {
-  p[0] = uint8_t(v >> 24); p[1] = uint8_t(v >> 16);
-  p[2] = uint8_t(v >> 8); p[3] = uint8_t(v);
-}
-static void put_be64(uint8_t *p, uint64_t v) {
-  put_be32(p, uint32_t(v >> 32));
-  put_be32(p + 4, uint32_t(v));
-}
+/* (the 32 B headers - magic, BE ids, CRC - are built on device by
+ * shard_finalize_k; see shard.go:241-261 for the format) */
 
 int64_t gfrs_shard_disk_size(int64_t size, int64_t block_len) {
   int64_t body = gfrs_crc32b_encode_size(size, block_len);
