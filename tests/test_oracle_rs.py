@@ -181,3 +181,26 @@ def test_buffer_sizes(oracle):
     assert ss == 2048 and eds == 12288 and es == 18432
     ss, _, _ = oracle.buffer_sizes(6, 3, 0, 0, 100)
     assert ss == 17  # ceil(100/6), Align0
+
+
+def test_rs_encode_mt_matches_scalar(oracle):
+    """The chunked OpenMP/AVX2 baseline path is bit-identical to the
+    scalar oracle across ragged lengths and stripe counts."""
+    import numpy as np
+    rng = np.random.default_rng(77)
+    for k, m, slen, ns in [(6, 3, 200_000, 3), (6, 3, 131_072, 1),
+                           (12, 4, 50_001, 5), (4, 2, 131_073, 2)]:
+        stripes = []
+        ref = []
+        for _ in range(ns):
+            st = [rng.integers(0, 256, slen, dtype=np.uint8)
+                  for _ in range(k)]
+            st += [np.zeros(slen, np.uint8) for _ in range(m)]
+            stripes.append(st)
+            r = [x.copy() for x in st]
+            oracle.rs_encode(k, m, r)
+            ref.append(r)
+        oracle.rs_encode_mt(k, m, stripes)
+        for s in range(ns):
+            for i in range(k + m):
+                assert np.array_equal(stripes[s][i], ref[s][i]), (k, m, s, i)
