@@ -26,9 +26,24 @@
 #include <set>
 #include <string>
 #include <algorithm>
+#include <array>
 #include <vector>
 
 namespace gfrs {
+
+/* 256-bit shard bitmask + tag: collision-free plan-cache key for every
+ * engine tactic_valid admits (n+m <= 256). */
+typedef std::array<uint64_t, 5> PlanKey;
+
+static inline PlanKey plan_key_tag(uint64_t tag) {
+  PlanKey k{};
+  k[4] = tag;
+  return k;
+}
+
+static inline void plan_key_set(PlanKey &k, int i) {
+  k[size_t(i) >> 6] |= 1ull << (unsigned(i) & 63u);
+}
 
 static thread_local std::string g_err;
 
@@ -157,8 +172,13 @@ struct gfrs_ctx_impl {
                                         n data shards (locals are linear in
                                         the data through the parity rows) */
   bool fused_lrc_ok = false;
-  std::map<uint64_t, DevPlan *> dec_cache; /* missing-bitmask → data decode */
-  std::map<uint64_t, DevPlan *> par_cache; /* missing-bitmask → parity rows */
+  /* Plan-cache key: full 256-bit missing/present bitmask (words 0..3) plus
+   * an engine/namespace tag (word 4).  tactic_valid admits engines up to
+   * n+m=256 shards, so a packed-u64 key would silently collide distinct
+   * missing sets above 58 shards (and 1ull<<i is UB at i>=64); the wide
+   * key makes every reachable engine collision-free. */
+  std::map<PlanKey, DevPlan *> dec_cache; /* missing-bitmask → data decode */
+  std::map<PlanKey, DevPlan *> par_cache; /* missing-bitmask → parity rows */
 
   DevBuf ptr_buf;   /* pointer tables for pointer-mode launches */
   DevBuf fail_buf;  /* verify flags / crc bad blocks */
@@ -213,6 +233,18 @@ static bool tactic_valid(const gfrs_tactic *t) {
 }  // namespace gfrs
 
 using namespace gfrs;
+
+/* unlocked bodies of the public batch entry points (defined below);
+ * already-locked callers reuse these instead of re-locking */
+static int encode_batch_impl(gfrs_ctx_impl *c, void *base, size_t shard_len,
+                             size_t stripe_stride, int nstripes);
+static int verify_batch_impl(gfrs_ctx_impl *c, const void *base,
+                             size_t shard_len, size_t stripe_stride,
+                             int nstripes, uint64_t *fail_bitmap);
+static int crc32b_encode_batch_impl(gfrs_ctx_impl *c, void *dst,
+                                    size_t dst_stride, const void *src,
+                                    size_t src_stride, int64_t n,
+                                    int64_t block_len, int nshards);
 
 extern "C" {
 
@@ -416,14 +448,13 @@ static int get_decode_plan(gfrs_ctx_impl *c, int k, int m,
                            const std::vector<int> &engine_idx /* global ids */,
                            const std::vector<uint8_t> &present,
                            DevPlan **out_plan, int tag) {
-  /* key = missing bitmask | engine tag (global=1, az-local=2+az, ...):
+  /* key = missing bitmask + engine tag (global=1, az-local=2+az, ...):
    * global and local engines can share k, first index AND mask, so the
    * tag is load-bearing (a collision here once wrote a local parity row
    * over a global one — caught by the randomized fuzz test) */
-  uint64_t key = 0;
+  PlanKey key = plan_key_tag(uint64_t(tag));
   for (int i = 0; i < k + m; i++)
-    if (!present[i]) key |= 1ull << i;
-  key = (key << 6) | uint64_t(tag & 63);
+    if (!present[i]) plan_key_set(key, i);
   auto it = c->dec_cache.find(key);
   if (it != c->dec_cache.end()) {
     *out_plan = it->second;
@@ -464,10 +495,9 @@ static int get_parity_plan(gfrs_ctx_impl *c, int k, int m,
                            const std::vector<int> &engine_idx,
                            const std::vector<uint8_t> &present,
                            DevPlan **out_plan, int tag) {
-  uint64_t key = 0;
+  PlanKey key = plan_key_tag(uint64_t(tag));
   for (int i = 0; i < k + m; i++)
-    if (!present[i]) key |= 1ull << i;
-  key = (key << 6) | uint64_t(tag & 63);
+    if (!present[i]) plan_key_set(key, i);
   auto it = c->par_cache.find(key);
   if (it != c->par_cache.end()) {
     *out_plan = it->second;
@@ -618,7 +648,7 @@ int gfrs_encode(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
     uint8_t *dev = (uint8_t *)c->stage_dev.p;
     HIP_TRY(hipMemcpyAsync(dev, pin, size_t(c->t.n) * shard_len,
                            hipMemcpyHostToDevice, c->stream));
-    rc = gfrs_encode_batch(ctx, dev, shard_len, tot, 1);
+    rc = encode_batch_impl(c, dev, shard_len, tot, 1);
     if (rc != GFRS_OK) return rc;
     HIP_TRY(hipMemcpyAsync(pin + size_t(c->t.n) * shard_len,
                            dev + size_t(c->t.n) * shard_len,
@@ -637,8 +667,8 @@ int gfrs_encode(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
     contig = (const uint8_t *)shards[i] ==
              (const uint8_t *)shards[0] + size_t(i) * shard_len;
   if (contig) {
-    rc = gfrs_encode_batch(ctx, shards[0], shard_len,
-                           size_t(nshards) * shard_len, 1);
+    rc = encode_batch_impl(c, shards[0], shard_len,
+                          size_t(nshards) * shard_len, 1);
     if (rc != GFRS_OK) return rc;
     HIP_TRY(hipStreamSynchronize(c->stream));
     return GFRS_OK;
@@ -658,9 +688,10 @@ int gfrs_encode(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
   return GFRS_OK;
 }
 
-int gfrs_encode_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
-                      size_t stripe_stride, int nstripes) {
-  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+/* unlocked bodies: public entry points take c->mu (the reference encoder
+ * is share-safe behind its limiter, encoder.go:29,115) */
+static int encode_batch_impl(gfrs_ctx_impl *c, void *base, size_t shard_len,
+                             size_t stripe_stride, int nstripes) {
   StreamGuard g(c);
   launch_rs_apply_strided((uint64_t)base, stripe_stride,
                           (const int32_t *)c->enc_plan.in_idx.p,
@@ -676,6 +707,13 @@ int gfrs_encode_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) return hip_fail("encode_batch launch", e);
   return GFRS_OK;
+}
+
+int gfrs_encode_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
+                      size_t stripe_stride, int nstripes) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  std::lock_guard<std::mutex> lk(c->mu);
+  return encode_batch_impl(c, base, shard_len, stripe_stride, nstripes);
 }
 
 int gfrs_verify(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
@@ -744,7 +782,7 @@ int gfrs_verify(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
     HIP_TRY(hipMemcpyAsync(c->stage_dev.p, pin, tot, hipMemcpyHostToDevice,
                            c->stream));
     uint64_t fb = 0;
-    rc = gfrs_verify_batch(ctx, c->stage_dev.p, shard_len, tot, 1, &fb);
+    rc = verify_batch_impl(c, c->stage_dev.p, shard_len, tot, 1, &fb);
     if (rc != GFRS_OK) return rc;
     *ok = fb == 0;
     return GFRS_OK;
@@ -770,10 +808,9 @@ int gfrs_verify(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
   return GFRS_OK;
 }
 
-int gfrs_verify_batch(gfrs_ctx *ctx, const void *base, size_t shard_len,
-                      size_t stripe_stride, int nstripes,
-                      uint64_t *fail_bitmap) {
-  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+static int verify_batch_impl(gfrs_ctx_impl *c, const void *base,
+                             size_t shard_len, size_t stripe_stride,
+                             int nstripes, uint64_t *fail_bitmap) {
   StreamGuard g(c);
   int rc;
   if ((rc = c->fail_buf.ensure(size_t(nstripes) * 4)) != GFRS_OK) return rc;
@@ -801,6 +838,15 @@ int gfrs_verify_batch(gfrs_ctx *ctx, const void *base, size_t shard_len,
       if (fails[s]) fail_bitmap[s / 64] |= 1ull << (s % 64);
   }
   return GFRS_OK;
+}
+
+int gfrs_verify_batch(gfrs_ctx *ctx, const void *base, size_t shard_len,
+                      size_t stripe_stride, int nstripes,
+                      uint64_t *fail_bitmap) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  std::lock_guard<std::mutex> lk(c->mu);
+  return verify_batch_impl(c, base, shard_len, stripe_stride, nstripes,
+                           fail_bitmap);
 }
 
 int gfrs_reconstruct(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
@@ -847,8 +893,14 @@ int gfrs_reconstruct(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
     if (rc != GFRS_OK) return rc;
     HIP_TRY(hipMemcpyAsync(pin, dev, tot, hipMemcpyDeviceToHost, c->stream));
     HIP_TRY(hipStreamSynchronize(c->stream));
+    /* copy back only shards actually rebuilt: with data_only the engine
+     * leaves missing parity untouched (ReconstructData semantics,
+     * reedsolomon.go:1441-1444) — overwriting the caller's buffer with
+     * stale staging bytes would be wrong */
+    const int ndata = local_form ? c->local_n : t.n;
     for (int i = 0; i < nshards; i++)
-      if (!present[i]) memcpy(shards[i], pin + size_t(i) * shard_len, shard_len);
+      if (!present[i] && (!data_only || i < ndata))
+        memcpy(shards[i], pin + size_t(i) * shard_len, shard_len);
     return GFRS_OK;
   }
 
@@ -896,6 +948,7 @@ int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
   const gfrs_tactic &t = c->t;
   if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
   if (nstripes <= 0 || shard_len == 0) return GFRS_ERR_INVALID_SHARDS;
+  std::lock_guard<std::mutex> lk(c->mu);
   StreamGuard g(c);
   /* the fused kernel gives a workgroup a whole 64 KiB frame; below a few
    * frames per shard the two-kernel composition (whose rs_apply packs
@@ -942,10 +995,10 @@ int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
   /* fallback composition: needs the contiguous ec.Buffer batch layout */
   if (stripe_stride != size_t(c->total) * shard_len)
     return GFRS_ERR_UNSUPPORTED;
-  int rc = gfrs_encode_batch(ctx, base, shard_len, stripe_stride, nstripes);
+  int rc = encode_batch_impl(c, base, shard_len, stripe_stride, nstripes);
   if (rc != GFRS_OK) return rc;
-  return gfrs_crc32b_encode_batch(ctx, framed, framed_stride, base,
-                                  shard_len, int64_t(shard_len), block_len,
+  return crc32b_encode_batch_impl(c, framed, framed_stride, base, shard_len,
+                                  int64_t(shard_len), block_len,
                                   nstripes * c->total);
 }
 
@@ -1128,7 +1181,8 @@ int gfrs_encode_idx(gfrs_ctx *ctx, const void *data_shard, int idx,
   if ((rc = upload_ptrs(c, ptrs.data(), int(ptrs.size()))) != GFRS_OK)
     return rc;
   /* per-idx plan cached in dec_cache keyed off a synthetic mask */
-  uint64_t key = (uint64_t(idx) << 6) | 62; /* EncodeIdx namespace */
+  PlanKey key = plan_key_tag(62); /* EncodeIdx namespace */
+  plan_key_set(key, idx);
   DevPlan *p;
   auto it = c->dec_cache.find(key);
   if (it != c->dec_cache.end()) {
@@ -1180,9 +1234,8 @@ static int lrc_mixed_reconstruct_verify(gfrs_ctx_impl *cc, void *base,
   }
   std::lock_guard<std::mutex> lk(cc->mu);
   StreamGuard g(cc);
-  uint64_t key = 0;
-  for (int b : badv) key |= 1ull << b;
-  key = (key << 6) | 60; /* LRC mixed reconstruct+verify namespace */
+  PlanKey key = plan_key_tag(60); /* LRC mixed reconstruct+verify */
+  for (int b : badv) plan_key_set(key, b);
   DevPlan *plan = nullptr;
   uint32_t cmp_mask = 0;
   {
@@ -1339,10 +1392,9 @@ int gfrs_reconstruct_verify_batch(gfrs_ctx *ctx, void *base,
   std::lock_guard<std::mutex> lk(c->mu);
   StreamGuard g(c);
 
-  uint64_t key = 0;
+  PlanKey key = plan_key_tag(63); /* reconstruct+verify namespace */
   for (int i = 0; i < k + m; i++)
-    if (!present[i]) key |= 1ull << i;
-  key = (key << 6) | 63; /* reconstruct+verify namespace */
+    if (!present[i]) plan_key_set(key, i);
   DevPlan *plan = nullptr;
   uint32_t cmp_mask = 0;
   {
@@ -1483,9 +1535,8 @@ int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
       }
       std::lock_guard<std::mutex> lk(cc->mu);
       StreamGuard g(cc);
-      uint64_t key = 0;
-      for (int b : badv) key |= 1ull << b; /* incl. local-parity bads */
-      key = (key << 6) | 61; /* fused-repair namespace */
+      PlanKey key = plan_key_tag(61); /* fused-repair namespace */
+      for (int b : badv) plan_key_set(key, b); /* incl. local-parity bads */
       DevPlan *plan = nullptr;
       {
         auto it = cc->dec_cache.find(key);
@@ -1678,10 +1729,10 @@ int64_t gfrs_crc32b_decode_size(int64_t size, int64_t block_len) {
   return size - 4 * ((size + block_len - 1) / block_len);
 }
 
-int gfrs_crc32b_encode_batch(gfrs_ctx *ctx, void *dst, size_t dst_stride,
-                             const void *src, size_t src_stride, int64_t n,
-                             int64_t block_len, int nshards) {
-  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+static int crc32b_encode_batch_impl(gfrs_ctx_impl *c, void *dst,
+                                    size_t dst_stride, const void *src,
+                                    size_t src_stride, int64_t n,
+                                    int64_t block_len, int nshards) {
   if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
   if (n <= 0 || nshards <= 0) return GFRS_ERR_INVALID_SHARDS;
   StreamGuard g(c);
@@ -1692,11 +1743,21 @@ int gfrs_crc32b_encode_batch(gfrs_ctx *ctx, void *dst, size_t dst_stride,
   return GFRS_OK;
 }
 
+int gfrs_crc32b_encode_batch(gfrs_ctx *ctx, void *dst, size_t dst_stride,
+                             const void *src, size_t src_stride, int64_t n,
+                             int64_t block_len, int nshards) {
+  auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  std::lock_guard<std::mutex> lk(c->mu);
+  return crc32b_encode_batch_impl(c, dst, dst_stride, src, src_stride, n,
+                                  block_len, nshards);
+}
+
 int64_t gfrs_crc32b_encode(gfrs_ctx *ctx, void *dst, const void *src,
                            int64_t n, int64_t block_len) {
-  int rc = gfrs_crc32b_encode_batch(ctx, dst, 0, src, 0, n, block_len, 1);
-  if (rc != GFRS_OK) return rc;
   auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
+  std::lock_guard<std::mutex> lk(c->mu);
+  int rc = crc32b_encode_batch_impl(c, dst, 0, src, 0, n, block_len, 1);
+  if (rc != GFRS_OK) return rc;
   StreamGuard g(c);
   HIP_TRY(hipStreamSynchronize(c->stream));
   return gfrs_crc32b_encode_size(n, block_len);

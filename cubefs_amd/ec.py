@@ -50,7 +50,11 @@ class Encoder:
     def __init__(self, tactic, device=-1, enable_verify=False):
         if isinstance(tactic, (str, int)):
             tactic = codemode.get_tactic(tactic)
-        if not tactic.is_valid() or tactic.is_replicate():
+        # Replicate tactics (M == 0) are valid per the reference
+        # (codemode.Tactic.IsValid, encoder.go:78): Encode/Verify become
+        # no-ops with zero parity and Reconstruct can only succeed when
+        # nothing is missing — all handled by the C layer (m==0 paths).
+        if not tactic.is_valid():
             raise GfrsError(-1, "invalid code mode for EC encoder")
         self.tactic = tactic
         self.enable_verify = enable_verify
