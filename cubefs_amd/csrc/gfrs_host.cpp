@@ -130,12 +130,29 @@ struct DevPlan {
     const GfTables &t = gft();
     k = rowk >= 0 ? rowk : int(in.size());
     nout = int(out.size());
-    std::vector<uint8_t> tb(size_t(nout) * k * 32);
+    /* two table regions per coefficient:
+     *  [0, nout*k*32)          lo|hi nibble tables (galois_amd64.go:37-52
+     *                          semantics) — rs_apply / staged kernels
+     *  [nout*k*32, nout*k*64)  3-way linear split A|B|C: GF(2^8) multiply
+     *                          is GF(2)-linear, so mul_c(b) =
+     *                          A[b&7] ^ B[(b>>3)&7] ^ C[b>>6] with
+     *                          A[i]=mul(c,i), B[i]=mul(c,8i), C[i]=mul(c,64i)
+     *                          — lets the register-CRC fused kernels do a
+     *                          16-entry lookup as 3 v_perm + 1 xor3 instead
+     *                          of 4 v_perm + 2 blends + masks */
+    const size_t lin0 = size_t(nout) * k * 32;
+    std::vector<uint8_t> tb(size_t(nout) * k * 64, 0);
     for (int r = 0; r < nout; r++)
       for (int c = 0; c < k; c++) {
         uint8_t coef = rows[size_t(r) * k + c];
         memcpy(&tb[(size_t(r) * k + c) * 32], t.lo[coef], 16);
         memcpy(&tb[(size_t(r) * k + c) * 32 + 16], t.hi[coef], 16);
+        uint8_t *lt = &tb[lin0 + (size_t(r) * k + c) * 32];
+        for (int i = 0; i < 8; i++) {
+          lt[i] = t.mul[coef][i];
+          lt[8 + i] = t.mul[coef][8 * i];
+        }
+        for (int i = 0; i < 4; i++) lt[16 + i] = t.mul[coef][64 * i];
       }
     int rc;
     if ((rc = in_idx.ensure(in.size() * 4)) != GFRS_OK) return rc;

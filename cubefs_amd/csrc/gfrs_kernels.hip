@@ -115,6 +115,56 @@ GFRS_DEV void gfmac16(uint4 &acc, const uint4 v, const uint4 tlo,
 }
 
 /* ------------------------------------------------------------------ */
+/* 3-way linear GF multiply (A|B|C split) on packed u32                 */
+/* ------------------------------------------------------------------ */
+
+/* GF(2^8) multiply-by-constant is GF(2)-linear, so it splits over bit
+ * groups of the operand: mul_c(b) = A[b&7] ^ B[(b>>3)&7] ^ C[b>>6] with
+ * A[i] = mul(c,i), B[i] = mul(c,8i), C[i] = mul(c,64i).  Each group is a
+ * <=8-entry byte table — exactly what one v_perm_b32 can look up for 4
+ * packed bytes — so a 4-byte multiply is 3 v_perm + 1 bitop3(xor3)
+ * instead of the nibble path's 4 v_perm + 2 blends, and the selectors
+ * (shared across all output rows) cost 5 VALU instead of ~9 incl. two
+ * v_mul_lo mask broadcasts. */
+GFRS_DEV uint32_t xor3_fwd(uint32_t a, uint32_t b, uint32_t c) {
+  return __builtin_amdgcn_bitop3_b32(a, b, c, 0x96); /* a ^ b ^ c */
+}
+
+struct LinTab {
+  uint32_t a0, a1, b0, b1, cc;
+};
+
+GFRS_DEV LinTab lintab_load(const uint8_t *ctab, int idx) {
+  const uint32_t *tp =
+      reinterpret_cast<const uint32_t *>(ctab + size_t(idx) * 32);
+  return LinTab{tp[0], tp[1], tp[2], tp[3], tp[4]};
+}
+
+GFRS_DEV uint32_t gfmul4_lin(uint32_t s012, uint32_t s345, uint32_t s67,
+                             const LinTab &t) {
+  return xor3_fwd(__builtin_amdgcn_perm(t.a1, t.a0, s012),
+                  __builtin_amdgcn_perm(t.b1, t.b0, s345),
+                  __builtin_amdgcn_perm(t.cc, t.cc, s67));
+}
+
+/* acc[r][comp] ^= mul_{row r}(w) for one packed dword w */
+template <int GM>
+GFRS_DEV void gfmac4_lin_rows(uint4 (&acc)[GM][4], int i, int d, uint32_t w,
+                              const LinTab (&lt)[GM]) {
+  const uint32_t s012 = w & 0x07070707u;
+  const uint32_t s345 = (w >> 3) & 0x07070707u;
+  const uint32_t s67 = (w >> 6) & 0x03030303u;
+#pragma unroll
+  for (int r = 0; r < GM; r++)
+    (&acc[r][i].x)[d] ^= gfmul4_lin(s012, s345, s67, lt[r]);
+}
+
+/* byte-path form for tails/prologues (same tables) */
+GFRS_DEV uint8_t gfmul1_lin(const uint8_t *tt, uint8_t b) {
+  return tt[b & 7] ^ tt[8 + ((b >> 3) & 7)] ^ tt[16 + (b >> 6)];
+}
+
+/* ------------------------------------------------------------------ */
 /* rs_apply / rs_verify                                                 */
 /* ------------------------------------------------------------------ */
 
@@ -1329,8 +1379,6 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
       gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
   __syncthreads();
 
-  const uint4 *ltab = reinterpret_cast<const uint4 *>(ctab);
-
   int64_t fr0 = blockIdx.x, frN = total_frames, frS = gridDim.x;
   if (MAP != 0) {
     const int64_t nper = (total_frames + gridDim.x - 1) / gridDim.x;
@@ -1412,6 +1460,13 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
         const uint8_t *src = sbase + size_t(c) * shard_len + p0 + r0;
         uint8_t *fdst = dst + (stripe * (k + GM) + c) * dst_stride +
                         f * block_len + CRC_LEN + r0;
+        /* A|B|C tables for this unit's GM rows, hoisted to VGPRs for the
+         * whole 64-B piece run (the nibble path re-read 2 ds_read_b128
+         * per piece per row) */
+        LinTab lt[GM];
+        if (SKEL != 1 && SKEL != 4)
+#pragma unroll
+          for (int r = 0; r < GM; r++) lt[r] = lintab_load(ctab, r * k + c);
         uint4 vcur[PIPE ? 4 : 1];
         if (PIPE == 2) {
 #pragma unroll
@@ -1458,13 +1513,13 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
                      : *reinterpret_cast<const uint4 *>(src + off);
             if (SKEL != 1 && SKEL != 4) {
 #pragma unroll
-              for (int r = 0; r < GM; r++) {
-                const int t2 = (r * k + c) * 2;
-                gfmac16<false>(acc[r][i], v, ltab[t2], ltab[t2 + 1]);
-              }
+              for (int d = 0; d < 4; d++)
+                gfmac4_lin_rows<GM>(acc, i, d, (&v.x)[d], lt);
             }
-            uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
-            dw[0] = v.x; dw[1] = v.y; dw[2] = v.z; dw[3] = v.w;
+            /* frame payload sits at +4 mod 16; the hardware takes
+             * dword-aligned dwordx4 (the loads at p0 = f*65532 already
+             * run that way), so one store instead of four */
+            *reinterpret_cast<uint4 *>(fdst + off) = v;
             if (SKEL != 1 && SKEL != 3) t = shift4k(t, stab) ^ crc16_reg(v, tab);
           }
         }
@@ -1534,9 +1589,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
         for (int i = 0; i < 4; i++) {
           const int off = i * 4096 + lane16i;
           if (off + 16 <= rbi) {
-            uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
-            dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
-            dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
+            *reinterpret_cast<uint4 *>(fdst + off) = acc[r][i];
             if (SKEL != 1 && SKEL != 3)
               t = shift4k(t, stab) ^ crc16_reg(acc[r][i], tab);
           }
@@ -1549,8 +1602,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
             uint8_t pv = 0;
             for (int c2 = 0; c2 < k; c2++) {
               const uint8_t b = tailb[c2 * 16 + (p - t0)];
-              const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
-              pv ^= tt[b & 0x0F] ^ tt[16 + (b >> 4)];
+              pv ^= gfmul1_lin(ctab + size_t(r * k + c2) * 32, b);
             }
             fdst[p] = pv;
             part ^= gf2_mulmod_d(x8tab[rbi - 1 - p], tab[0][pv]);
@@ -2231,6 +2283,9 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
   const int64_t fps = (int64_t(shard_len) + 65531) / 65532;
   const int64_t total = fps * nstripes;
   const int grid = fused_grid(total, fps);
+  /* register-CRC (reg_k) variants use the A|B|C linear-split region that
+   * DevPlan::upload lays after the nibble tables */
+  const uint8_t *ltabs = tabs + size_t(m) * size_t(k) * 32;
   /* occupancy variant: GFRS_EF = NBUF*10 + waves-per-SIMD bound (16 KiB
    * pass), or a 3-digit NBUF*100 + WPS*10 + NI form for the 8 KiB-pass
    * (NI=2) geometry.  Measured @256 stripes RS(6+3): 14 -> 13.7 ms
@@ -2254,16 +2309,16 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
     switch (m) {
       case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 3, 0, 0, 2>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps); break;
+          stripe_stride, shard_len, k, ltabs, total, fps); break;
       case 2: hipLaunchKernelGGL((rs_encode_frame_reg_k<2, 3, 0, 0, 2>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps); break;
+          stripe_stride, shard_len, k, ltabs, total, fps); break;
       case 3: hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 0, 2>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps); break;
+          stripe_stride, shard_len, k, ltabs, total, fps); break;
       default: hipLaunchKernelGGL((rs_encode_frame_reg_k<4, 3, 0, 0, 2>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps);
+          stripe_stride, shard_len, k, ltabs, total, fps);
     }
     return;
   }
@@ -2272,16 +2327,16 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
     switch (m) {
       case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 4, 0, 0, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps); break;
+          stripe_stride, shard_len, k, ltabs, total, fps); break;
       case 2: hipLaunchKernelGGL((rs_encode_frame_reg_k<2, 4, 0, 0, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps); break;
+          stripe_stride, shard_len, k, ltabs, total, fps); break;
       case 3: hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 4, 0, 0, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps); break;
+          stripe_stride, shard_len, k, ltabs, total, fps); break;
       default: hipLaunchKernelGGL((rs_encode_frame_reg_k<4, 4, 0, 0, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps);
+          stripe_stride, shard_len, k, ltabs, total, fps);
     }
     return;
   }
@@ -2290,22 +2345,22 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
     switch (m) {
       case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 3, 0, 0, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps); break;
+          stripe_stride, shard_len, k, ltabs, total, fps); break;
       case 2: hipLaunchKernelGGL((rs_encode_frame_reg_k<2, 3, 0, 0, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps); break;
+          stripe_stride, shard_len, k, ltabs, total, fps); break;
       case 3: hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 0, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps); break;
+          stripe_stride, shard_len, k, ltabs, total, fps); break;
       default: hipLaunchKernelGGL((rs_encode_frame_reg_k<4, 3, 0, 0, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps);
+          stripe_stride, shard_len, k, ltabs, total, fps);
     }
     return;
   }
   if (var == 74 || var == 75) {
     const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
-#define GFRS_EFR_GO(G, W, P) hipLaunchKernelGGL((rs_encode_frame_reg_k<G, W, P>), dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base, stripe_stride, shard_len, k, tabs, total, fps)
+#define GFRS_EFR_GO(G, W, P) hipLaunchKernelGGL((rs_encode_frame_reg_k<G, W, P>), dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base, stripe_stride, shard_len, k, ltabs, total, fps)
 #define GFRS_EFR_SW(W, P) switch (m) { case 1: GFRS_EFR_GO(1, W, P); break; case 2: GFRS_EFR_GO(2, W, P); break; case 3: GFRS_EFR_GO(3, W, P); break; default: GFRS_EFR_GO(4, W, P); }
     static const int map = []() {
       const char *e = getenv("GFRS_EF_MAP");
@@ -2321,25 +2376,25 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
       const int lds5 = 12288 + EF_RED + 64 + 256 + m * k * 32;
       if (rabl == 5) { hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 2, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds5, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps); return; }
+          stripe_stride, shard_len, k, ltabs, total, fps); return; }
       if (rabl == 6) { hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 3, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds5, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps); return; }
+          stripe_stride, shard_len, k, ltabs, total, fps); return; }
       if (rabl == 7) { hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 4, 1>),
           dim3(grid), dim3(CRC_BLOCKT), lds5, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, tabs, total, fps); return; }
+          stripe_stride, shard_len, k, ltabs, total, fps); return; }
     }
     if (rabl == 4) { /* memory-floor skeleton (diagnostic only) */
       switch (m) {
         case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 4, 1, 1>),
             dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-            stripe_stride, shard_len, k, tabs, total, fps); break;
+            stripe_stride, shard_len, k, ltabs, total, fps); break;
         case 3: hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 4, 1, 1>),
             dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-            stripe_stride, shard_len, k, tabs, total, fps); break;
+            stripe_stride, shard_len, k, ltabs, total, fps); break;
         default: hipLaunchKernelGGL((rs_encode_frame_reg_k<4, 4, 1, 1>),
             dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-            stripe_stride, shard_len, k, tabs, total, fps);
+            stripe_stride, shard_len, k, ltabs, total, fps);
       }
       return;
     }
