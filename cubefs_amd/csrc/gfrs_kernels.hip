@@ -130,6 +130,27 @@ GFRS_DEV uint32_t xor3_fwd(uint32_t a, uint32_t b, uint32_t c) {
   return __builtin_amdgcn_bitop3_b32(a, b, c, 0x96); /* a ^ b ^ c */
 }
 
+/* Frame-payload 16-B store with a selectable cache policy.  The streams
+ * are write-once (never re-read by the producer), so the interesting
+ * policies are: 0 plain (line stays dirty in the XCD L2), 1 nontemporal,
+ * 2 sc1 write-through (line dropped from L2 — frees L2 for the read
+ * streams).  Policy choice is measured, not assumed; see profiles/. */
+template <int ST>
+GFRS_DEV void fstore16(uint8_t *p, const uint4 v) {
+  if (ST == 1) {
+    u32x4 x = {v.x, v.y, v.z, v.w};
+    __builtin_nontemporal_store(x, reinterpret_cast<u32x4 *>(p));
+  } else if (ST == 2) {
+    u32x4 x = {v.x, v.y, v.z, v.w};
+    asm volatile("global_store_dwordx4 %0, %1, off sc0 sc1"
+                 :
+                 : "v"(p), "v"(x)
+                 : "memory");
+  } else {
+    *reinterpret_cast<uint4 *>(p) = v;
+  }
+}
+
 struct LinTab {
   uint32_t a0, a1, b0, b1, cc;
 };
@@ -1330,7 +1351,8 @@ GFRS_DEV uint32_t shift4k(uint32_t c, const uint32_t (*stab)[256]) {
  * issued before unit c's MAC/CRC/store consume their values, so the
  * compute of each unit overlaps the memory latency of the next.  Costs
  * ~32 VGPRs of double-buffer; pair with WPS=3. */
-template <int GM, int WPS, int MAP = 0, int SKEL = 0, int PIPE = 0>
+template <int GM, int WPS, int MAP = 0, int SKEL = 0, int PIPE = 0,
+          int ST = 0>
 __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     uint8_t *__restrict__ dst, size_t dst_stride, uint64_t base,
     uint64_t stripe_stride, size_t shard_len, int k,
@@ -1519,7 +1541,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
             /* frame payload sits at +4 mod 16; the hardware takes
              * dword-aligned dwordx4 (the loads at p0 = f*65532 already
              * run that way), so one store instead of four */
-            *reinterpret_cast<uint4 *>(fdst + off) = v;
+            fstore16<ST>(fdst + off, v);
             if (SKEL != 1 && SKEL != 3) t = shift4k(t, stab) ^ crc16_reg(v, tab);
           }
         }
@@ -1589,7 +1611,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
         for (int i = 0; i < 4; i++) {
           const int off = i * 4096 + lane16i;
           if (off + 16 <= rbi) {
-            *reinterpret_cast<uint4 *>(fdst + off) = acc[r][i];
+            fstore16<ST>(fdst + off, acc[r][i]);
             if (SKEL != 1 && SKEL != 3)
               t = shift4k(t, stab) ^ crc16_reg(acc[r][i], tab);
           }
@@ -2299,10 +2321,47 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
     switch (v) {
       case 13: case 14: case 23: case 24:
       case 74: case 75: case 76: case 77: case 78:
+      case 86: case 87: case 96: case 97:
       case 142: case 152: case 162: return v;
       default: return 14;
     }
   }();
+  /* 8x/9x = store-policy variants of the lookahead pipeline:
+   * 86/87 nontemporal @3/4 waves, 96/97 sc1 write-through @3/4 waves */
+  if (var == 86 || var == 87 || var == 96 || var == 97) {
+    const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
+#define GFRS_STP_GO(G, W, S)                                                hipLaunchKernelGGL((rs_encode_frame_reg_k<G, W, 0, 0, 1, S>),                                dim3(grid), dim3(CRC_BLOCKT), lds, s, dst,                                dst_stride, base, stripe_stride, shard_len, k,                            ltabs, total, fps)
+#define GFRS_STP_SW(W, S)                                                   switch (m) {                                                                case 1: GFRS_STP_GO(1, W, S); break;                                      case 2: GFRS_STP_GO(2, W, S); break;                                      case 3: GFRS_STP_GO(3, W, S); break;                                      default: GFRS_STP_GO(4, W, S);                                          }
+    if (var == 86) { GFRS_STP_SW(3, 1) }
+    else if (var == 87) { GFRS_STP_SW(4, 1) }
+    else if (var == 96) { GFRS_STP_SW(3, 2) }
+    else { GFRS_STP_SW(4, 2) }
+#undef GFRS_STP_SW
+#undef GFRS_STP_GO
+    return;
+  }
+  /* frame->block mapping experiment on the 4-wave lookahead pipeline
+   * (GFRS_EF=77 + GFRS_EF_MAP=1/2, GM=3 instantiations only) */
+  if (var == 77 && m == 3) {
+    static const int map77 = []() {
+      const char *e = getenv("GFRS_EF_MAP");
+      return e ? atoi(e) : 0;
+    }();
+    if (map77 == 1 || map77 == 2) {
+      const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
+      if (map77 == 1)
+        hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 4, 1, 0, 1>),
+                           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst,
+                           dst_stride, base, stripe_stride, shard_len, k,
+                           ltabs, total, fps);
+      else
+        hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 4, 2, 0, 1>),
+                           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst,
+                           dst_stride, base, stripe_stride, shard_len, k,
+                           ltabs, total, fps);
+      return;
+    }
+  }
   /* 7x = register-CRC kernel (no stage): crc tabs + shift tabs + red */
   if (var == 78) { /* two-unit-deep lookahead at 3 waves/SIMD */
     const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;

@@ -59,6 +59,69 @@ __global__ void k_mix21(uint4 *__restrict__ d, const uint4 *__restrict__ s0,
   }
 }
 
+/* fused encode+frame shape: 6 read streams, 9 write streams (6 framed
+ * copies + 3 parity combos).  MISAL adds the production +4 frame-payload
+ * byte offset to every store address; ST: 0 plain, 1 nontemporal, 2 sc1
+ * (write-through, drops the line from the XCD L2). */
+struct MixPtrs {
+  const uint4 *r[6];
+  uint8_t *w[9];
+};
+
+template <int ST, int MISAL>
+__global__ void k_mix69(MixPtrs p, size_t n16) {
+  for (size_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n16;
+       i += size_t(gridDim.x) * blockDim.x) {
+    uint4 v[6];
+#pragma unroll
+    for (int c = 0; c < 6; c++) v[c] = p.r[c][i];
+    uint4 par0{v[0].x ^ v[1].x, v[0].y ^ v[1].y, v[0].z ^ v[1].z,
+               v[0].w ^ v[1].w};
+    uint4 par1{v[2].x ^ v[3].x, v[2].y ^ v[3].y, v[2].z ^ v[3].z,
+               v[2].w ^ v[3].w};
+    uint4 par2{v[4].x ^ v[5].x, v[4].y ^ v[5].y, v[4].z ^ v[5].z,
+               v[4].w ^ v[5].w};
+    const uint4 out[9] = {v[0], v[1], v[2], v[3], v[4],
+                          v[5], par0, par1, par2};
+#pragma unroll
+    for (int c = 0; c < 9; c++) {
+      uint8_t *dst = p.w[c] + i * 16 + (MISAL ? 4 : 0);
+      if (ST == 1) {
+        u32x4 x = {out[c].x, out[c].y, out[c].z, out[c].w};
+        __builtin_nontemporal_store(x, reinterpret_cast<u32x4 *>(dst));
+      } else if (ST == 2) {
+        u32x4 x = {out[c].x, out[c].y, out[c].z, out[c].w};
+        asm volatile("global_store_dwordx4 %0, %1, off sc0 sc1"
+                     :
+                     : "v"(dst), "v"(x)
+                     : "memory");
+      } else {
+        *reinterpret_cast<uint4 *>(dst) = out[c];
+      }
+    }
+  }
+}
+
+template <int ST, int MISAL>
+static void run_mix69(const char *name, MixPtrs p, size_t n16, size_t bytes) {
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  dim3 grid(2048), blk(256);
+  const int reps = 6;
+  hipLaunchKernelGGL((k_mix69<ST, MISAL>), grid, blk, 0, 0, p, n16);
+  hipDeviceSynchronize();
+  hipEventRecord(e0);
+  for (int r = 0; r < reps; r++)
+    hipLaunchKernelGGL((k_mix69<ST, MISAL>), grid, blk, 0, 0, p, n16);
+  hipEventRecord(e1);
+  hipEventSynchronize(e1);
+  float ms;
+  hipEventElapsedTime(&ms, e0, e1);
+  printf("%-14s %8.1f GB/s (moved, 6r:9w)\n", name,
+         15.0 * bytes / (ms / reps / 1e3) / 1e9);
+}
+
 int main() {
   const size_t bytes = size_t(8) << 30;
   const size_t n16 = bytes / 16;
@@ -135,6 +198,30 @@ int main() {
     float ms; hipEventElapsedTime(&ms, e0, e1);
     printf("%-14s %8.1f GB/s (moved, 2r:1w)\n", "mix21",
            3.0 * bytes / (ms / reps / 1e3) / 1e9);
+  }
+  {
+    /* 15 streams of 1 GiB: the fused encode+frame mix */
+    const size_t sb = size_t(1) << 30;
+    const size_t sn16 = sb / 16;
+    MixPtrs p{};
+    for (int c = 0; c < 6; c++) {
+      uint4 *q;
+      hipMalloc(&q, sb);
+      hipMemset(q, 17 + c, sb);
+      p.r[c] = q;
+    }
+    for (int c = 0; c < 9; c++) {
+      uint8_t *q;
+      hipMalloc(&q, sb + 16);
+      hipMemset(q, 0, sb + 16);
+      p.w[c] = q;
+    }
+    run_mix69<0, 0>("mix69", p, sn16 - 1, sb);
+    run_mix69<0, 1>("mix69+4", p, sn16 - 1, sb);
+    run_mix69<1, 0>("mix69_nt", p, sn16 - 1, sb);
+    run_mix69<1, 1>("mix69_nt+4", p, sn16 - 1, sb);
+    run_mix69<2, 0>("mix69_sc1", p, sn16 - 1, sb);
+    run_mix69<2, 1>("mix69_sc1+4", p, sn16 - 1, sb);
   }
   return 0;
 }
