@@ -1325,13 +1325,6 @@ GFRS_DEV uint32_t crc16_reg(const uint4 q, const uint32_t (*tab)[256]) {
   return c;
 }
 
-/* one slice-by-4 CRC step over a dword */
-GFRS_DEV uint32_t crc_dw(uint32_t c, uint32_t w, const uint32_t (*tab)[256]) {
-  c ^= w;
-  return tab[3][c & 0xFF] ^ tab[2][(c >> 8) & 0xFF] ^
-         tab[1][(c >> 16) & 0xFF] ^ tab[0][c >> 24];
-}
-
 GFRS_DEV uint32_t shift4k(uint32_t c, const uint32_t (*stab)[256]) {
   return xor3(stab[0][c & 0xFF], stab[1][(c >> 8) & 0xFF],
               stab[2][(c >> 16) & 0xFF]) ^
@@ -1402,17 +1395,10 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
    * the chained value times x^(8*4096), i.e. one shift4k.  One register
    * instead of an op_full[4] array. */
   constexpr uint32_t INV16K = 0x479933FCu;
-  /* Aligned-store split: payload = 12-B prologue + 4095 aligned 16-B
-   * pieces (65532 - 12 = 4095*16 exactly).  The piece grid starts at
-   * payload offset 12, so every main-loop store lands 16-B aligned
-   * (frame base is 64 KiB aligned; header 4 B + prologue 12 B = 16 B,
-   * written as ONE aligned uint4 in the epilogue).  Probe: +4-misaligned
-   * stores cost ~16% of the 6r:9w mix ceiling (profiles/ r2 bwprobe). */
   const uint32_t op_first =
-      x8n_d(uint64_t(payload_full - 12 - (3 * 4096 + lane16 + 16)));
+      x8n_d(uint64_t(payload_full - (3 * 4096 + lane16 + 16)));
   const uint32_t it_full =
       gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
-  const uint32_t p12_full = x8n_d(uint64_t(payload_full - 12));
   __syncthreads();
 
   int64_t fr0 = blockIdx.x, frN = total_frames, frS = gridDim.x;
@@ -1437,8 +1423,6 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
                                       : int64_t(0))) {
     const int64_t p0 = f * payload_full;
     const int64_t payload = i64min(payload_full, int64_t(shard_len) - p0);
-    const int64_t pfx = i64min(int64_t(12), payload); /* prologue bytes */
-    const int64_t payload_eff = payload - pfx;        /* piece-grid bytes */
     const uint8_t *sbase = as_global(base + stripe * stripe_stride);
 
     uint4 acc[GM][4];
@@ -1446,13 +1430,13 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     uint4 vnext2[PIPE == 2 ? 4 : 1];
     if (PIPE && fr == fr0) { /* later frames are prefetched by the
                                 previous frame's last pass */
-      const int rb0 = int(i64min(int64_t(EF_PASS), payload_eff));
+      const int rb0 = int(i64min(int64_t(EF_PASS), payload));
 #pragma unroll
       for (int i = 0; i < 4; i++) {
         const int off = i * 4096 + lane16i;
         vnext[PIPE ? i : 0] =
             off + 16 <= rb0
-                ? *reinterpret_cast<const uint4 *>(sbase + p0 + 12 + off)
+                ? *reinterpret_cast<const uint4 *>(sbase + p0 + off)
                 : uint4{0, 0, 0, 0};
       }
       if (PIPE == 2) {
@@ -1461,7 +1445,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
           const int off = i * 4096 + lane16i;
           vnext2[PIPE == 2 ? i : 0] =
               off + 16 <= rb0 ? *reinterpret_cast<const uint4 *>(
-                                    sbase + shard_len + p0 + 12 + off)
+                                    sbase + shard_len + p0 + off)
                               : uint4{0, 0, 0, 0};
         }
       }
@@ -1472,7 +1456,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     uint32_t op_chain = op_first;
     for (int h = 0; h < EF_PASSES; h++) {
       const int64_t r0 = int64_t(h) * EF_PASS;
-      const int64_t rbytes = i64min(int64_t(EF_PASS), payload_eff - r0);
+      const int64_t rbytes = i64min(int64_t(EF_PASS), payload - r0);
       if (rbytes <= 0) break;
 #pragma unroll
       for (int r = 0; r < GM; r++)
@@ -1490,14 +1474,14 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
           if (int64_t(i) * 4096 + lane16 + 16 <= rbytes) np = i + 1;
         const int64_t end =
             np ? r0 + int64_t(np - 1) * 4096 + lane16 + 16 : r0;
-        op = x8n_d(uint64_t(payload_eff - end));
+        op = x8n_d(uint64_t(payload - end));
       }
 
       const int rbi = int(rbytes);
       for (int c = 0; c < k; c++) {
-        const uint8_t *src = sbase + size_t(c) * shard_len + p0 + 12 + r0;
+        const uint8_t *src = sbase + size_t(c) * shard_len + p0 + r0;
         uint8_t *fdst = dst + (stripe * (k + GM) + c) * dst_stride +
-                        f * block_len + 16 + r0; /* header+prologue = 16 */
+                        f * block_len + CRC_LEN + r0;
         /* A|B|C tables for this unit's GM rows, hoisted to VGPRs for the
          * whole 64-B piece run (the nibble path re-read 2 ds_read_b128
          * per piece per row) */
@@ -1513,8 +1497,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
             vnext[PIPE ? i : 0] = vnext2[PIPE == 2 ? i : 0];
           }
           if (c + 2 < k) {
-            const uint8_t *nsrc =
-                sbase + size_t(c + 2) * shard_len + p0 + 12 + r0;
+            const uint8_t *nsrc = sbase + size_t(c + 2) * shard_len + p0 + r0;
 #pragma unroll
             for (int i = 0; i < 4; i++) {
               const int off = i * 4096 + lane16i;
@@ -1528,8 +1511,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
 #pragma unroll
           for (int i = 0; i < 4; i++) vcur[PIPE ? i : 0] = vnext[PIPE ? i : 0];
           if (c + 1 < k) {
-            const uint8_t *nsrc =
-                sbase + size_t(c + 1) * shard_len + p0 + 12 + r0;
+            const uint8_t *nsrc = sbase + size_t(c + 1) * shard_len + p0 + r0;
 #pragma unroll
             for (int i = 0; i < 4; i++) {
               const int off = i * 4096 + lane16i;
@@ -1556,12 +1538,15 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
               for (int d = 0; d < 4; d++)
                 gfmac4_lin_rows<GM>(acc, i, d, (&v.x)[d], lt);
             }
-            fstore16<ST>(fdst + off, v); /* 16-B aligned by the split */
+            /* frame payload sits at +4 mod 16; the hardware takes
+             * dword-aligned dwordx4 (the loads at p0 = f*65532 already
+             * run that way), so one store instead of four */
+            fstore16<ST>(fdst + off, v);
             if (SKEL != 1 && SKEL != 3) t = shift4k(t, stab) ^ crc16_reg(v, tab);
           }
         }
         uint32_t part = SKEL == 2 ? t : (t ? gf2_mulmod_d(op, t) : 0);
-        if (rbi & 15) {
+        if (rbytes < EF_PASS) {
           /* lane-parallel tail: one byte per lane, folded by its own
            * position operator (the serial thread-0 loop dominated tiny
            * last frames: 9 units of dependent global loads) */
@@ -1581,18 +1566,18 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
         if ((threadIdx.x & 63) == 0)
           red[(threadIdx.x >> 6) * 16 + c] ^= part;
       }
-      if (rbi & 15) __syncthreads(); /* tailb visible to all */
+      if (rbytes < EF_PASS) __syncthreads(); /* tailb visible to all */
       if (PIPE) { /* next pass's (or next frame's) unit-0 loads fly
                      during the parity rows and the frame epilogue */
         int64_t r0n = r0 + EF_PASS;
-        int64_t rbn = i64min(int64_t(EF_PASS), payload_eff - r0n);
-        const uint8_t *nbase = sbase + p0 + 12;
+        int64_t rbn = i64min(int64_t(EF_PASS), payload - r0n);
+        const uint8_t *nbase = sbase + p0;
         if (rbn <= 0 && fr + frS < frN) {
           const int64_t fr2 = fr + frS;
           const int64_t st2 = fr2 / frames_per_shard;
           const int64_t p02 = (fr2 - st2 * frames_per_shard) * payload_full;
-          rbn = i64min(int64_t(EF_PASS), int64_t(shard_len) - p02 - 12);
-          nbase = as_global(base + st2 * stripe_stride) + p02 + 12;
+          rbn = i64min(int64_t(EF_PASS), int64_t(shard_len) - p02);
+          nbase = as_global(base + st2 * stripe_stride) + p02;
           r0n = 0;
         }
         if (rbn > 0) {
@@ -1620,7 +1605,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
 #pragma unroll
       for (int r = 0; r < GM; r++) {
         uint8_t *fdst = dst + (stripe * (k + GM) + k + r) * dst_stride +
-                        f * block_len + 16 + r0;
+                        f * block_len + CRC_LEN + r0;
         uint32_t t = 0;
 #pragma unroll
         for (int i = 0; i < 4; i++) {
@@ -1632,7 +1617,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
           }
         }
         uint32_t part = SKEL == 2 ? t : (t ? gf2_mulmod_d(op, t) : 0);
-        if (SKEL != 1 && (rbi & 15)) {
+        if (SKEL != 1 && rbytes < EF_PASS) {
           const int t0 = (rbi / 16) * 16;
           const int p = t0 + int(threadIdx.x);
           if (p < rbi) {
@@ -1654,71 +1639,17 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     }
 
     __syncthreads();
-    { /* header + payload prologue per shard, one lane per shard
-       * (j < k+GM <= 16, all in wave 0).  Phase A: data lanes stage
-       * their shard's first pfx payload bytes in LDS; phase B: parity
-       * lanes derive theirs (parity is bytewise in the data), every
-       * lane folds the prologue into its CRC and writes
-       * [crc32 | prologue] as ONE aligned 16-B store. */
+    { /* one frame-header lane per shard (j < k+GM <= 16, all in wave 0) */
       const int j = int(threadIdx.x);
-      const int pfxi = int(pfx);
-      uint32_t pw[3] = {0, 0, 0};
-      if (j < k && pfxi) {
-        const uint8_t *sp = sbase + size_t(j) * shard_len + p0;
-        if (pfxi == 12) {
-          pw[0] = *reinterpret_cast<const uint32_t *>(sp);
-          pw[1] = *reinterpret_cast<const uint32_t *>(sp + 4);
-          pw[2] = *reinterpret_cast<const uint32_t *>(sp + 8);
-        } else {
-          for (int b = 0; b < pfxi; b++)
-            reinterpret_cast<uint8_t *>(pw)[b] = sp[b];
-        }
-        for (int b = 0; b < pfxi; b++)
-          tailb[j * 16 + b] = reinterpret_cast<const uint8_t *>(pw)[b];
-      }
-      __syncthreads();
       if (j < k + GM) {
-        if (j >= k && pfxi) {
-          for (int b = 0; b < pfxi; b++) {
-            uint8_t pv = 0;
-            for (int c2 = 0; c2 < k; c2++)
-              pv ^= gfmul1_lin(ctab + size_t((j - k) * k + c2) * 32,
-                               tailb[c2 * 16 + b]);
-            reinterpret_cast<uint8_t *>(pw)[b] = pv;
-          }
-        }
-        uint32_t pcon = 0;
-        if (SKEL != 1 && pfxi) {
-          uint32_t c12 = 0;
-          if (pfxi == 12) {
-            c12 = crc_dw(c12, pw[0], tab);
-            c12 = crc_dw(c12, pw[1], tab);
-            c12 = crc_dw(c12, pw[2], tab);
-          } else {
-            for (int b = 0; b < pfxi; b++)
-              c12 = tab[0][(c12 ^ reinterpret_cast<const uint8_t *>(pw)[b]) &
-                           0xFF] ^
-                    (c12 >> 8);
-          }
-          pcon = payload == payload_full ? gf2_mulmod_d(p12_full, c12)
-                 : payload_eff ? gf2_mulmod_d(x8n_d(uint64_t(payload_eff)), c12)
-                               : c12;
-        }
         const uint32_t it =
             payload == payload_full
                 ? it_full
                 : gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
         const uint32_t crc =
-            ~(it ^ pcon ^ red[j] ^ red[16 + j] ^ red[32 + j] ^ red[48 + j]);
-        uint8_t *fb0 =
-            dst + (stripe * (k + GM) + j) * dst_stride + f * block_len;
-        if (pfxi == 12) {
-          fstore16<ST>(fb0, uint4{crc, pw[0], pw[1], pw[2]});
-        } else {
-          *reinterpret_cast<uint32_t *>(fb0) = crc;
-          for (int b = 0; b < pfxi; b++)
-            fb0[CRC_LEN + b] = reinterpret_cast<const uint8_t *>(pw)[b];
-        }
+            ~(it ^ red[j] ^ red[16 + j] ^ red[32 + j] ^ red[48 + j]);
+        *reinterpret_cast<uint32_t *>(
+            dst + (stripe * (k + GM) + j) * dst_stride + f * block_len) = crc;
       }
     }
     __syncthreads();
@@ -2389,12 +2320,25 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
         variant edges the 4-wave squeeze (17.6 vs 17.8 ms @512) */
     switch (v) {
       case 13: case 14: case 23: case 24:
+      case 72: case 73:
       case 74: case 75: case 76: case 77: case 78:
       case 86: case 87: case 96: case 97:
       case 142: case 152: case 162: return v;
       default: return 14;
     }
   }();
+  /* 72/73 = low-occupancy forms: 2 blocks/CU (512 resident WGs chip-wide,
+   * halving the number of concurrently-touched DRAM streams) with deep
+   * (72) or single (73) load lookahead */
+  if (var == 72 || var == 73) {
+    const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
+#define GFRS_LO_GO(G, P)                                                    hipLaunchKernelGGL((rs_encode_frame_reg_k<G, 2, 0, 0, P>),                                   dim3(grid), dim3(CRC_BLOCKT), lds, s, dst,                                dst_stride, base, stripe_stride, shard_len, k,                            ltabs, total, fps)
+#define GFRS_LO_SW(P)                                                       switch (m) {                                                                case 1: GFRS_LO_GO(1, P); break;                                          case 2: GFRS_LO_GO(2, P); break;                                          case 3: GFRS_LO_GO(3, P); break;                                          default: GFRS_LO_GO(4, P);                                              }
+    if (var == 72) { GFRS_LO_SW(2) } else { GFRS_LO_SW(1) }
+#undef GFRS_LO_SW
+#undef GFRS_LO_GO
+    return;
+  }
   /* 8x/9x = store-policy variants of the lookahead pipeline:
    * 86/87 nontemporal @3/4 waves, 96/97 sc1 write-through @3/4 waves */
   if (var == 86 || var == 87 || var == 96 || var == 97) {
