@@ -102,6 +102,63 @@ __global__ void k_mix69(MixPtrs p, size_t n16) {
   }
 }
 
+/* same 6r:9w mix but with PER-BLOCK private stream segments: block b walks
+ * its own contiguous L-byte window of each stream (the production shape —
+ * every workgroup owns its own stripe's shard/image streams) instead of
+ * all blocks sharing one cursor per stream.  The delta vs k_mix69 is the
+ * price of many-stream DRAM scatter, which the fused kernel inherits from
+ * the workload (independent stripes). */
+template <int ST>
+__global__ void k_mix69_pb(MixPtrs p, size_t blk_n16) {
+  const size_t b0 = size_t(blockIdx.x) * blk_n16;
+  for (size_t j = threadIdx.x; j < blk_n16; j += blockDim.x) {
+    const size_t i = b0 + j;
+    uint4 v[6];
+#pragma unroll
+    for (int c = 0; c < 6; c++) v[c] = p.r[c][i];
+    uint4 par0{v[0].x ^ v[1].x, v[0].y ^ v[1].y, v[0].z ^ v[1].z,
+               v[0].w ^ v[1].w};
+    uint4 par1{v[2].x ^ v[3].x, v[2].y ^ v[3].y, v[2].z ^ v[3].z,
+               v[2].w ^ v[3].w};
+    uint4 par2{v[4].x ^ v[5].x, v[4].y ^ v[5].y, v[4].z ^ v[5].z,
+               v[4].w ^ v[5].w};
+    const uint4 out[9] = {v[0], v[1], v[2], v[3], v[4],
+                          v[5], par0, par1, par2};
+#pragma unroll
+    for (int c = 0; c < 9; c++) {
+      uint8_t *dst = p.w[c] + i * 16;
+      if (ST == 1) {
+        u32x4 x = {out[c].x, out[c].y, out[c].z, out[c].w};
+        __builtin_nontemporal_store(x, reinterpret_cast<u32x4 *>(dst));
+      } else {
+        *reinterpret_cast<uint4 *>(dst) = out[c];
+      }
+    }
+  }
+}
+
+template <int ST>
+static void run_mix69_pb(const char *name, MixPtrs p, size_t n16,
+                         size_t bytes) {
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  dim3 grid(2048), blk(256);
+  const size_t blk_n16 = n16 / 2048;
+  const int reps = 6;
+  hipLaunchKernelGGL((k_mix69_pb<ST>), grid, blk, 0, 0, p, blk_n16);
+  hipDeviceSynchronize();
+  hipEventRecord(e0);
+  for (int r = 0; r < reps; r++)
+    hipLaunchKernelGGL((k_mix69_pb<ST>), grid, blk, 0, 0, p, blk_n16);
+  hipEventRecord(e1);
+  hipEventSynchronize(e1);
+  float ms;
+  hipEventElapsedTime(&ms, e0, e1);
+  printf("%-14s %8.1f GB/s (moved, 6r:9w per-block)\n", name,
+         15.0 * double(blk_n16) * 2048 * 16 / (ms / reps / 1e3) / 1e9);
+}
+
 template <int ST, int MISAL>
 static void run_mix69(const char *name, MixPtrs p, size_t n16, size_t bytes) {
   hipEvent_t e0, e1;
@@ -222,6 +279,8 @@ int main() {
     run_mix69<1, 1>("mix69_nt+4", p, sn16 - 1, sb);
     run_mix69<2, 0>("mix69_sc1", p, sn16 - 1, sb);
     run_mix69<2, 1>("mix69_sc1+4", p, sn16 - 1, sb);
+    run_mix69_pb<0>("mix69_pb", p, sn16, sb);
+    run_mix69_pb<1>("mix69_pb_nt", p, sn16, sb);
   }
   return 0;
 }
