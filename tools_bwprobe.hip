@@ -108,14 +108,16 @@ __global__ void k_mix69(MixPtrs p, size_t n16) {
  * all blocks sharing one cursor per stream.  The delta vs k_mix69 is the
  * price of many-stream DRAM scatter, which the fused kernel inherits from
  * the workload (independent stripes). */
-template <int ST>
+template <int ST, int RMIS = 0, int WMIS = 0>
 __global__ void k_mix69_pb(MixPtrs p, size_t blk_n16) {
   const size_t b0 = size_t(blockIdx.x) * blk_n16;
   for (size_t j = threadIdx.x; j < blk_n16; j += blockDim.x) {
     const size_t i = b0 + j;
     uint4 v[6];
 #pragma unroll
-    for (int c = 0; c < 6; c++) v[c] = p.r[c][i];
+    for (int c = 0; c < 6; c++)
+      v[c] = *reinterpret_cast<const uint4 *>(
+          reinterpret_cast<const uint8_t *>(p.r[c] + i) + (RMIS ? 12 : 0));
     uint4 par0{v[0].x ^ v[1].x, v[0].y ^ v[1].y, v[0].z ^ v[1].z,
                v[0].w ^ v[1].w};
     uint4 par1{v[2].x ^ v[3].x, v[2].y ^ v[3].y, v[2].z ^ v[3].z,
@@ -126,7 +128,7 @@ __global__ void k_mix69_pb(MixPtrs p, size_t blk_n16) {
                           v[5], par0, par1, par2};
 #pragma unroll
     for (int c = 0; c < 9; c++) {
-      uint8_t *dst = p.w[c] + i * 16;
+      uint8_t *dst = p.w[c] + i * 16 + (WMIS ? 4 : 0);
       if (ST == 1) {
         u32x4 x = {out[c].x, out[c].y, out[c].z, out[c].w};
         __builtin_nontemporal_store(x, reinterpret_cast<u32x4 *>(dst));
@@ -137,7 +139,7 @@ __global__ void k_mix69_pb(MixPtrs p, size_t blk_n16) {
   }
 }
 
-template <int ST>
+template <int ST, int RMIS = 0, int WMIS = 0>
 static void run_mix69_pb(const char *name, MixPtrs p, size_t n16,
                          size_t bytes) {
   hipEvent_t e0, e1;
@@ -146,11 +148,13 @@ static void run_mix69_pb(const char *name, MixPtrs p, size_t n16,
   dim3 grid(2048), blk(256);
   const size_t blk_n16 = n16 / 2048;
   const int reps = 6;
-  hipLaunchKernelGGL((k_mix69_pb<ST>), grid, blk, 0, 0, p, blk_n16);
+  hipLaunchKernelGGL((k_mix69_pb<ST, RMIS, WMIS>), grid, blk, 0, 0, p,
+                     blk_n16);
   hipDeviceSynchronize();
   hipEventRecord(e0);
   for (int r = 0; r < reps; r++)
-    hipLaunchKernelGGL((k_mix69_pb<ST>), grid, blk, 0, 0, p, blk_n16);
+    hipLaunchKernelGGL((k_mix69_pb<ST, RMIS, WMIS>), grid, blk, 0, 0, p,
+                       blk_n16);
   hipEventRecord(e1);
   hipEventSynchronize(e1);
   float ms;
@@ -281,6 +285,13 @@ int main() {
     run_mix69<2, 1>("mix69_sc1+4", p, sn16 - 1, sb);
     run_mix69_pb<0>("mix69_pb", p, sn16, sb);
     run_mix69_pb<1>("mix69_pb_nt", p, sn16, sb);
+    /* production misalignment attribution (reads at +12, writes at +4):
+     * the streams are 1 GiB + 16 B so the offsets stay in bounds */
+    run_mix69_pb<0, 1, 0>("pb_rmis", p, sn16 - 1, sb);
+    run_mix69_pb<0, 0, 1>("pb_wmis", p, sn16 - 1, sb);
+    run_mix69_pb<0, 1, 1>("pb_rwmis", p, sn16 - 1, sb);
+    run_mix69_pb<1, 1, 1>("pb_rwmis_nt", p, sn16 - 1, sb);
+    run_mix69_pb<1, 0, 1>("pb_wmis_nt", p, sn16 - 1, sb);
   }
   return 0;
 }
