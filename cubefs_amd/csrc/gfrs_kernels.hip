@@ -560,6 +560,8 @@ __device__ uint32_t g_crc_tab4[8][256];
 __device__ uint32_t g_shift4k[4][256];
 /* same, for x^(8*1024): the small-shard fused kernel's piece stride */
 __device__ uint32_t g_shift1k[4][256];
+/* and x^(8*8192): the 32-B-piece dual-aligned kernel's lane stride */
+__device__ uint32_t g_shift8k[4][256];
 
 /* x^(8*2^j) mod P, reflected domain — host-filled alongside the tables. */
 __device__ uint32_t g_pow8[40];
@@ -1656,6 +1658,392 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
   }
 }
 
+
+/* ------------------------------------------------------------------ */
+/* dual-aligned fused encode+frame: 32-B lane pieces, window rotation   */
+/* ------------------------------------------------------------------ */
+
+/* The frame format forces a 4 (mod 16) relative shift between a frame's
+ * source bytes (shard offset f*65532) and its framed payload (image
+ * offset f*65536+4), so one side of the straightforward piece map is
+ * always misaligned; measured on the per-block 6r:9w stream mix that
+ * costs 14-17% of the ceiling (profiles/: aligned 5.02 TB/s, either side
+ * misaligned 4.16-4.33, the production both-misaligned 4.25).  This
+ * kernel aligns BOTH sides:
+ *  - stores: payload = 28-B prologue + 2047 x 32-B pieces (65532 - 28 =
+ *    2047*32).  header(4) + prologue(28) = 32, so piece p stores at
+ *    image offset f*65536 + 32 + 32p — 16-B aligned when the image
+ *    stride is 16-B aligned; [crc | prologue] is written as two aligned
+ *    uint4 in the epilogue.
+ *  - loads: a lane's 32-B piece needs the source window
+ *    [W + d, W + 32 + d), d = (28 + p0) mod 16 — frame-constant and
+ *    always a dword multiple.  The lane loads three ALIGNED uint4
+ *    (a 48-B private window) and selects the 8 piece dwords statically
+ *    under a per-frame uniform branch on D0 = d/4.  No cross-lane
+ *    traffic; the trailing over-read (<= 16 B) stays inside the stripe
+ *    (a data shard is always followed by more shard space, m >= 1).
+ * CRC bookkeeping follows rs_encode_frame_reg_k with lane-piece stride
+ * 8192 (x^(8*8192) Horner tables g_shift8k) and a 32-entry x8tab for
+ * sub-32-B byte tails. */
+
+/* one slice-by-4 CRC step over a dword */
+GFRS_DEV uint32_t crc_dw(uint32_t c, uint32_t w, const uint32_t (*tab)[256]) {
+  c ^= w;
+  return tab[3][c & 0xFF] ^ tab[2][(c >> 8) & 0xFF] ^
+         tab[1][(c >> 16) & 0xFF] ^ tab[0][c >> 24];
+}
+
+GFRS_DEV uint32_t shift8k(uint32_t c, const uint32_t (*stab)[256]) {
+  return xor3(stab[0][c & 0xFF], stab[1][(c >> 8) & 0xFF],
+              stab[2][(c >> 16) & 0xFF]) ^
+         stab[3][c >> 24];
+}
+
+/* raw CRC of 32 B held in two uint4 (slice-by-8, 4 steps) */
+GFRS_DEV uint32_t crc32_reg(const uint4 qa, const uint4 qb,
+                            const uint32_t (*tab)[256]) {
+  uint32_t c = 0;
+#pragma unroll
+  for (int j = 0; j < 4; j++) {
+    const uint32_t w0 =
+        (j == 0 ? qa.x : j == 1 ? qa.z : j == 2 ? qb.x : qb.z) ^ c;
+    const uint32_t w1 = (j == 0 ? qa.y : j == 1 ? qa.w : j == 2 ? qb.y : qb.w);
+    c = xor3(xor3(tab[7][w0 & 0xFF], tab[6][(w0 >> 8) & 0xFF],
+                  tab[5][(w0 >> 16) & 0xFF]),
+             xor3(tab[4][w0 >> 24], tab[3][w1 & 0xFF],
+                  tab[2][(w1 >> 8) & 0xFF]),
+             tab[1][(w1 >> 16) & 0xFF] ^ tab[0][w1 >> 24]);
+  }
+  return c;
+}
+
+/* statically select the 8 piece dwords out of the lane's 12-dword
+ * aligned window (D0 = frame shift / 4) */
+template <int D0>
+GFRS_DEV void rot_pick(uint4 &h0, uint4 &h1, const uint4 w0, const uint4 w1,
+                       const uint4 w2) {
+  const uint32_t a[12] = {w0.x, w0.y, w0.z, w0.w, w1.x, w1.y,
+                          w1.z, w1.w, w2.x, w2.y, w2.z, w2.w};
+  h0 = uint4{a[D0], a[D0 + 1], a[D0 + 2], a[D0 + 3]};
+  h1 = uint4{a[D0 + 4], a[D0 + 5], a[D0 + 6], a[D0 + 7]};
+}
+
+struct RotArgs {
+  uint8_t *dst;
+  const uint8_t *sbase;
+  size_t dst_stride, shard_len;
+  int64_t stripe, f, p0, payload, payload_eff;
+  int k;
+  const uint8_t *ctab;
+  const uint32_t (*tab)[256];
+  const uint32_t (*stab)[256]; /* g_shift8k */
+  uint32_t *red;
+  const uint32_t *x8tab;
+  uint8_t *tailb;
+  uint32_t op_first;
+};
+
+template <int D0, int GM, int ST>
+GFRS_DEV void rot_passes(const RotArgs &A) {
+  constexpr int EF_PASS = 16384;
+  constexpr int EF_PASSES = 4;
+  constexpr int64_t block_len = 65536;
+  constexpr int64_t payload_full = block_len - CRC_LEN;
+  constexpr uint32_t INV16K = 0x479933FCu;
+  const int tid = int(threadIdx.x);
+  const int64_t lane32 = int64_t(tid) * 32;
+  const int lane32i = tid * 32;
+  const int k = A.k;
+
+  uint4 acc[GM][4]; /* [row][piece*2 + half] */
+  /* one-unit load lookahead: the next unit's six window vectors */
+  uint4 vn[2][3];
+  uint32_t op_chain = A.op_first;
+
+  for (int h = 0; h < EF_PASSES; h++) {
+    const int64_t r0 = int64_t(h) * EF_PASS;
+    const int64_t rbytes = i64min(int64_t(EF_PASS), A.payload_eff - r0);
+    if (rbytes <= 0) break;
+#pragma unroll
+    for (int r = 0; r < GM; r++)
+#pragma unroll
+      for (int q = 0; q < 4; q++) acc[r][q] = uint4{0, 0, 0, 0};
+
+    uint32_t op = op_chain;
+    if (h == EF_PASSES - 1 && tid == 255)
+      op = shift8k(op, A.stab); /* lane 255's last pass has 1 piece */
+    op_chain = gf2_mulmod_d(op_chain, INV16K);
+    if (A.payload != payload_full) {
+      int np = 0;
+#pragma unroll
+      for (int i = 0; i < 2; i++)
+        if (int64_t(i) * 8192 + lane32 + 32 <= rbytes) np = i + 1;
+      const int64_t end = np ? r0 + int64_t(np - 1) * 8192 + lane32 + 32 : r0;
+      op = x8n_d(uint64_t(A.payload_eff - end));
+    }
+
+    const int rbi = int(rbytes);
+    /* aligned window base of this pass (D0*4 = the frame shift) */
+    const int64_t wbase = A.p0 + 28 - int64_t(D0) * 4 + r0;
+    /* prime the lookahead with unit 0 */
+    {
+      const uint8_t *w = A.sbase + wbase;
+#pragma unroll
+      for (int i = 0; i < 2; i++) {
+        const int off = i * 8192 + lane32i;
+        const bool v = off + 32 <= rbi;
+        vn[i][0] = v ? *reinterpret_cast<const uint4 *>(w + off)
+                     : uint4{0, 0, 0, 0};
+        vn[i][1] = v ? *reinterpret_cast<const uint4 *>(w + off + 16)
+                     : uint4{0, 0, 0, 0};
+        vn[i][2] = (v && D0 != 0)
+                       ? *reinterpret_cast<const uint4 *>(w + off + 32)
+                       : uint4{0, 0, 0, 0};
+      }
+    }
+    for (int c = 0; c < k; c++) {
+      const uint8_t *src = A.sbase + size_t(c) * A.shard_len + A.p0 + 28 + r0;
+      uint8_t *fdst = A.dst + (A.stripe * (k + GM) + c) * A.dst_stride +
+                      A.f * block_len + 32 + r0;
+      LinTab lt[GM];
+#pragma unroll
+      for (int r = 0; r < GM; r++) lt[r] = lintab_load(A.ctab, r * k + c);
+      uint4 vc[2][3];
+#pragma unroll
+      for (int i = 0; i < 2; i++)
+#pragma unroll
+        for (int j = 0; j < 3; j++) vc[i][j] = vn[i][j];
+      if (c + 1 < k) {
+        const uint8_t *w = A.sbase + size_t(c + 1) * A.shard_len + wbase;
+#pragma unroll
+        for (int i = 0; i < 2; i++) {
+          const int off = i * 8192 + lane32i;
+          const bool v = off + 32 <= rbi;
+          vn[i][0] = v ? *reinterpret_cast<const uint4 *>(w + off)
+                       : uint4{0, 0, 0, 0};
+          vn[i][1] = v ? *reinterpret_cast<const uint4 *>(w + off + 16)
+                       : uint4{0, 0, 0, 0};
+          vn[i][2] = (v && D0 != 0)
+                         ? *reinterpret_cast<const uint4 *>(w + off + 32)
+                         : uint4{0, 0, 0, 0};
+        }
+      }
+      uint32_t t = 0;
+#pragma unroll
+      for (int i = 0; i < 2; i++) {
+        const int off = i * 8192 + lane32i;
+        if (off + 32 <= rbi) {
+          uint4 h0, h1;
+          rot_pick<D0>(h0, h1, vc[i][0], vc[i][1], vc[i][2]);
+#pragma unroll
+          for (int d = 0; d < 4; d++)
+            gfmac4_lin_rows<GM>(acc, i * 2, d, (&h0.x)[d], lt);
+#pragma unroll
+          for (int d = 0; d < 4; d++)
+            gfmac4_lin_rows<GM>(acc, i * 2 + 1, d, (&h1.x)[d], lt);
+          fstore16<ST>(fdst + off, h0);
+          fstore16<ST>(fdst + off + 16, h1);
+          t = shift8k(t, A.stab) ^ crc32_reg(h0, h1, A.tab);
+        }
+      }
+      uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
+      if (rbi & 31) {
+        /* lane-parallel byte tail (up to 31 B) */
+        const int t0 = rbi & ~31;
+        const int p = t0 + tid;
+        if (p < rbi) {
+          const uint8_t x = src[p];
+          fdst[p] = x;
+          A.tailb[c * 32 + (p - t0)] = x;
+          part ^= gf2_mulmod_d(A.x8tab[rbi - 1 - p], A.tab[0][x]);
+        }
+      }
+#pragma unroll
+      for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
+      if ((tid & 63) == 0) A.red[(tid >> 6) * 16 + c] ^= part;
+    }
+    if (rbi & 31) __syncthreads(); /* tailb visible to the parity rows */
+#pragma unroll
+    for (int r = 0; r < GM; r++) {
+      uint8_t *fdst = A.dst + (A.stripe * (k + GM) + k + r) * A.dst_stride +
+                      A.f * block_len + 32 + r0;
+      uint32_t t = 0;
+#pragma unroll
+      for (int i = 0; i < 2; i++) {
+        const int off = i * 8192 + lane32i;
+        if (off + 32 <= rbi) {
+          fstore16<ST>(fdst + off, acc[r][i * 2]);
+          fstore16<ST>(fdst + off + 16, acc[r][i * 2 + 1]);
+          t = shift8k(t, A.stab) ^
+              crc32_reg(acc[r][i * 2], acc[r][i * 2 + 1], A.tab);
+        }
+      }
+      uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
+      if (rbi & 31) {
+        const int t0 = rbi & ~31;
+        const int p = t0 + tid;
+        if (p < rbi) {
+          uint8_t pv = 0;
+          for (int c2 = 0; c2 < k; c2++)
+            pv ^= gfmul1_lin(A.ctab + size_t(r * k + c2) * 32,
+                             A.tailb[c2 * 32 + (p - t0)]);
+          fdst[p] = pv;
+          part ^= gf2_mulmod_d(A.x8tab[rbi - 1 - p], A.tab[0][pv]);
+        }
+      }
+#pragma unroll
+      for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
+      if ((tid & 63) == 0) A.red[(tid >> 6) * 16 + k + r] ^= part;
+    }
+  }
+}
+
+template <int GM, int WPS, int ST>
+__global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_rot_k(
+    uint8_t *__restrict__ dst, size_t dst_stride, uint64_t base,
+    uint64_t stripe_stride, size_t shard_len, int k,
+    const uint8_t *__restrict__ tabs /* linear A|B|C region */,
+    int64_t total_frames, int64_t frames_per_shard) {
+  constexpr int64_t block_len = 65536;
+  constexpr int64_t payload_full = block_len - CRC_LEN;
+  extern __shared__ __attribute__((aligned(16))) unsigned char smem[];
+  /* crc tabs 8K | shift8k 4K | red | x8tab(32) | tailb(16x32) | ctab */
+  uint32_t(*tab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem);
+  uint32_t(*stab)[256] = reinterpret_cast<uint32_t(*)[256]>(smem + 8192);
+  uint32_t *red = reinterpret_cast<uint32_t *>(smem + 12288);
+  uint32_t *x8tab = reinterpret_cast<uint32_t *>(smem + 12288 + EF_RED);
+  uint8_t *tailb = smem + 12288 + EF_RED + 128;
+  uint8_t *ctab = smem + 12288 + EF_RED + 128 + 512;
+  for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
+    (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
+  for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
+    (&stab[0][0])[i] = (&g_shift8k[0][0])[i];
+  for (int i = threadIdx.x; i < GM * k * 2; i += CRC_BLOCKT)
+    reinterpret_cast<uint4 *>(ctab)[i] =
+        reinterpret_cast<const uint4 *>(tabs)[i];
+  if (threadIdx.x == 0) {
+    uint32_t v = 0x80000000u;
+    for (int j = 0; j < 32; j++) {
+      x8tab[j] = v;
+      v = gf2_mulmod_d(v, g_pow8[0]);
+    }
+  }
+  const uint32_t op_first = x8n_d(
+      uint64_t(payload_full - 28 - (8192 + int64_t(threadIdx.x) * 32 + 32)));
+  const uint32_t it_full =
+      gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
+  const uint32_t p28_full = x8n_d(uint64_t(payload_full - 28));
+  __syncthreads();
+
+  int64_t stripe = int64_t(blockIdx.x) / frames_per_shard;
+  int64_t f = int64_t(blockIdx.x) - stripe * frames_per_shard;
+  const int64_t frS = gridDim.x;
+  const int64_t dstripe = frS / frames_per_shard;
+  const int64_t drem = frS - dstripe * frames_per_shard;
+  for (int64_t fr = blockIdx.x; fr < total_frames; fr += frS,
+               stripe += dstripe, f += drem,
+               (f >= frames_per_shard ? (f -= frames_per_shard, ++stripe)
+                                      : int64_t(0))) {
+    RotArgs A;
+    A.dst = dst;
+    A.sbase = as_global(base + stripe * stripe_stride);
+    A.dst_stride = dst_stride;
+    A.shard_len = shard_len;
+    A.stripe = stripe;
+    A.f = f;
+    A.p0 = f * payload_full;
+    A.payload = i64min(payload_full, int64_t(shard_len) - A.p0);
+    const int64_t pfx = i64min(int64_t(28), A.payload);
+    A.payload_eff = A.payload - pfx;
+    A.k = k;
+    A.ctab = ctab;
+    A.tab = tab;
+    A.stab = stab;
+    A.red = red;
+    A.x8tab = x8tab;
+    A.tailb = tailb;
+    A.op_first = op_first;
+
+    for (int j = threadIdx.x; j < 64; j += CRC_BLOCKT) red[j] = 0;
+    __syncthreads();
+
+    /* frame-constant source shift: (28 + p0) mod 16, always a multiple
+     * of 4; one uniform branch selects the static rotation */
+    switch (int((28 + A.p0) & 15) >> 2) {
+      case 0: rot_passes<0, GM, ST>(A); break;
+      case 1: rot_passes<1, GM, ST>(A); break;
+      case 2: rot_passes<2, GM, ST>(A); break;
+      default: rot_passes<3, GM, ST>(A); break;
+    }
+
+    __syncthreads();
+    { /* header + 28-B payload prologue, one lane per shard */
+      const int j = int(threadIdx.x);
+      const int pfxi = int(pfx);
+      uint32_t pw[7] = {0, 0, 0, 0, 0, 0, 0};
+      if (j < k && pfxi) {
+        const uint8_t *sp = A.sbase + size_t(j) * shard_len + A.p0;
+        if (pfxi == 28) {
+#pragma unroll
+          for (int w = 0; w < 7; w++)
+            pw[w] = *reinterpret_cast<const uint32_t *>(sp + 4 * w);
+        } else {
+          for (int b = 0; b < pfxi; b++)
+            reinterpret_cast<uint8_t *>(pw)[b] = sp[b];
+        }
+        for (int b = 0; b < pfxi; b++)
+          tailb[j * 32 + b] = reinterpret_cast<const uint8_t *>(pw)[b];
+      }
+      __syncthreads();
+      if (j < k + GM) {
+        if (j >= k && pfxi) {
+          for (int b = 0; b < pfxi; b++) {
+            uint8_t pv = 0;
+            for (int c2 = 0; c2 < k; c2++)
+              pv ^= gfmul1_lin(ctab + size_t((j - k) * k + c2) * 32,
+                               tailb[c2 * 32 + b]);
+            reinterpret_cast<uint8_t *>(pw)[b] = pv;
+          }
+        }
+        uint32_t pcon = 0;
+        if (pfxi) {
+          uint32_t cp = 0;
+          if (pfxi == 28) {
+#pragma unroll
+            for (int w = 0; w < 7; w++) cp = crc_dw(cp, pw[w], tab);
+          } else {
+            for (int b = 0; b < pfxi; b++)
+              cp = tab[0][(cp ^ reinterpret_cast<const uint8_t *>(pw)[b]) &
+                          0xFF] ^
+                   (cp >> 8);
+          }
+          pcon = A.payload == payload_full ? gf2_mulmod_d(p28_full, cp)
+                 : A.payload_eff
+                     ? gf2_mulmod_d(x8n_d(uint64_t(A.payload_eff)), cp)
+                     : cp;
+        }
+        const uint32_t it =
+            A.payload == payload_full
+                ? it_full
+                : gf2_mulmod_d(x8n_d(uint64_t(A.payload)), 0xFFFFFFFFu);
+        const uint32_t crc =
+            ~(it ^ pcon ^ red[j] ^ red[16 + j] ^ red[32 + j] ^ red[48 + j]);
+        uint8_t *fb0 =
+            dst + (stripe * (k + GM) + j) * dst_stride + f * block_len;
+        if (pfxi == 28) {
+          fstore16<ST>(fb0, uint4{crc, pw[0], pw[1], pw[2]});
+          fstore16<ST>(fb0 + 16, uint4{pw[3], pw[4], pw[5], pw[6]});
+        } else {
+          *reinterpret_cast<uint32_t *>(fb0) = crc;
+          for (int b = 0; b < pfxi; b++)
+            fb0[CRC_LEN + b] = reinterpret_cast<const uint8_t *>(pw)[b];
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
 /* ------------------------------------------------------------------ */
 /* fused repair: reconstruct + verify + crc32block-framed images        */
 /* ------------------------------------------------------------------ */
@@ -2323,10 +2711,25 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
       case 72: case 73:
       case 74: case 75: case 76: case 77: case 78:
       case 86: case 87: case 96: case 97:
+      case 103: case 104: case 113: case 114:
       case 142: case 152: case 162: return v;
       default: return 14;
     }
   }();
+  /* 1xy = dual-aligned 32-B-piece rotation kernel: 103/104 plain stores
+   * @3/4 waves, 113/114 nontemporal @3/4 waves */
+  if (var == 103 || var == 104 || var == 113 || var == 114) {
+    const int lds = 12288 + EF_RED + 128 + 512 + m * k * 32;
+#define GFRS_ROT_GO(G, W, S)                                                hipLaunchKernelGGL((rs_encode_frame_rot_k<G, W, S>), dim3(grid),                             dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,                          stripe_stride, shard_len, k, ltabs, total, fps)
+#define GFRS_ROT_SW(W, S)                                                   switch (m) {                                                                case 1: GFRS_ROT_GO(1, W, S); break;                                      case 2: GFRS_ROT_GO(2, W, S); break;                                      case 3: GFRS_ROT_GO(3, W, S); break;                                      default: GFRS_ROT_GO(4, W, S);                                          }
+    if (var == 103) { GFRS_ROT_SW(3, 0) }
+    else if (var == 104) { GFRS_ROT_SW(4, 0) }
+    else if (var == 113) { GFRS_ROT_SW(3, 1) }
+    else { GFRS_ROT_SW(4, 1) }
+#undef GFRS_ROT_SW
+#undef GFRS_ROT_GO
+    return;
+  }
   /* 72/73 = low-occupancy forms: 2 blocks/CU (512 resident WGs chip-wide,
    * halving the number of concurrently-touched DRAM streams) with deep
    * (72) or single (73) load lookahead */
@@ -2713,6 +3116,11 @@ int crc_device_init_current(void) {
     for (uint32_t b = 0; b < 256; b++)
       s4k[j][b] = mulmod(pow8[10], b << (8 * j)); /* x^(8*2^10) */
   if (hipMemcpyToSymbol(HIP_SYMBOL(g_shift1k), s4k, sizeof(s4k)) != hipSuccess)
+    return -100;
+  for (int j = 0; j < 4; j++)
+    for (uint32_t b = 0; b < 256; b++)
+      s4k[j][b] = mulmod(pow8[13], b << (8 * j)); /* x^(8*2^13) */
+  if (hipMemcpyToSymbol(HIP_SYMBOL(g_shift8k), s4k, sizeof(s4k)) != hipSuccess)
     return -100;
   if (hipMemcpyToSymbol(HIP_SYMBOL(g_crc_tab4), tab, sizeof(tab)) != hipSuccess)
     return -100;
