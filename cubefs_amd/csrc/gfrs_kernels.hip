@@ -1717,6 +1717,26 @@ GFRS_DEV uint32_t crc32_reg(const uint4 qa, const uint4 qb,
   return c;
 }
 
+/* load only the D0 dwords of the window tail that the rotation consumes
+ * (unused uint4 components are DCE'd, so this trims both registers and
+ * line requests) */
+typedef uint32_t u32x3_t __attribute__((ext_vector_type(3)));
+template <int D0>
+GFRS_DEV void rot_tail_load(uint4 &w2, const uint8_t *p) {
+  if (D0 == 1) {
+    w2.x = *reinterpret_cast<const uint32_t *>(p);
+  } else if (D0 == 2) {
+    const uint2 t = *reinterpret_cast<const uint2 *>(p);
+    w2.x = t.x;
+    w2.y = t.y;
+  } else if (D0 == 3) {
+    const u32x3_t t = *reinterpret_cast<const u32x3_t *>(p);
+    w2.x = t.x;
+    w2.y = t.y;
+    w2.z = t.z;
+  }
+}
+
 /* statically select the 8 piece dwords out of the lane's 12-dword
  * aligned window (D0 = frame shift / 4) */
 template <int D0>
@@ -1756,8 +1776,6 @@ GFRS_DEV void rot_passes(const RotArgs &A) {
   const int k = A.k;
 
   uint4 acc[GM][4]; /* [row][piece*2 + half] */
-  /* one-unit load lookahead: the next unit's six window vectors */
-  uint4 vn[2][3];
   uint32_t op_chain = A.op_first;
 
   for (int h = 0; h < EF_PASSES; h++) {
@@ -1785,100 +1803,96 @@ GFRS_DEV void rot_passes(const RotArgs &A) {
     const int rbi = int(rbytes);
     /* aligned window base of this pass (D0*4 = the frame shift) */
     const int64_t wbase = A.p0 + 28 - int64_t(D0) * 4 + r0;
-    /* prime the lookahead with unit 0 */
+
+    /* flat (unit, piece) pipeline with ONE rolling lookahead window —
+     * half the register footprint of a per-unit double buffer, which
+     * is what lets GM=3 fit 4 waves/SIMD spill-free */
+    uint4 cw[3], nw[3];
     {
+      const bool v = lane32i + 32 <= rbi;
       const uint8_t *w = A.sbase + wbase;
-#pragma unroll
-      for (int i = 0; i < 2; i++) {
-        const int off = i * 8192 + lane32i;
-        const bool v = off + 32 <= rbi;
-        vn[i][0] = v ? *reinterpret_cast<const uint4 *>(w + off)
-                     : uint4{0, 0, 0, 0};
-        vn[i][1] = v ? *reinterpret_cast<const uint4 *>(w + off + 16)
-                     : uint4{0, 0, 0, 0};
-        vn[i][2] = (v && D0 != 0)
-                       ? *reinterpret_cast<const uint4 *>(w + off + 32)
-                       : uint4{0, 0, 0, 0};
-      }
+      nw[0] = v ? *reinterpret_cast<const uint4 *>(w + lane32i)
+                : uint4{0, 0, 0, 0};
+      nw[1] = v ? *reinterpret_cast<const uint4 *>(w + lane32i + 16)
+                : uint4{0, 0, 0, 0};
+      nw[2] = uint4{0, 0, 0, 0};
+      if (v && D0 != 0) rot_tail_load<D0>(nw[2], w + lane32i + 32);
     }
-    for (int c = 0; c < k; c++) {
-      const uint8_t *src = A.sbase + size_t(c) * A.shard_len + A.p0 + 28 + r0;
+    LinTab lt[GM];
+    uint32_t t = 0;
+    for (int pc = 0; pc < k * 2; pc++) {
+      const int c = pc >> 1, i = pc & 1;
+      const int off = i * 8192 + lane32i;
+#pragma unroll
+      for (int j = 0; j < 3; j++) cw[j] = nw[j];
+      if (pc + 1 < k * 2) {
+        const int c2 = (pc + 1) >> 1, i2 = (pc + 1) & 1;
+        const int off2 = i2 * 8192 + lane32i;
+        const bool v = off2 + 32 <= rbi;
+        const uint8_t *w = A.sbase + size_t(c2) * A.shard_len + wbase;
+        nw[0] = v ? *reinterpret_cast<const uint4 *>(w + off2)
+                  : uint4{0, 0, 0, 0};
+        nw[1] = v ? *reinterpret_cast<const uint4 *>(w + off2 + 16)
+                  : uint4{0, 0, 0, 0};
+        nw[2] = uint4{0, 0, 0, 0};
+        if (v && D0 != 0) rot_tail_load<D0>(nw[2], w + off2 + 32);
+      }
+      if (i == 0) {
+        t = 0;
+#pragma unroll
+        for (int r = 0; r < GM; r++) lt[r] = lintab_load(A.ctab, r * k + c);
+      }
       uint8_t *fdst = A.dst + (A.stripe * (k + GM) + c) * A.dst_stride +
                       A.f * block_len + 32 + r0;
-      LinTab lt[GM];
+      if (off + 32 <= rbi) {
+        uint4 h0, h1;
+        rot_pick<D0>(h0, h1, cw[0], cw[1], cw[2]);
 #pragma unroll
-      for (int r = 0; r < GM; r++) lt[r] = lintab_load(A.ctab, r * k + c);
-      uint4 vc[2][3];
+        for (int d = 0; d < 4; d++)
+          gfmac4_lin_rows<GM>(acc, i * 2, d, (&h0.x)[d], lt);
 #pragma unroll
-      for (int i = 0; i < 2; i++)
-#pragma unroll
-        for (int j = 0; j < 3; j++) vc[i][j] = vn[i][j];
-      if (c + 1 < k) {
-        const uint8_t *w = A.sbase + size_t(c + 1) * A.shard_len + wbase;
-#pragma unroll
-        for (int i = 0; i < 2; i++) {
-          const int off = i * 8192 + lane32i;
-          const bool v = off + 32 <= rbi;
-          vn[i][0] = v ? *reinterpret_cast<const uint4 *>(w + off)
-                       : uint4{0, 0, 0, 0};
-          vn[i][1] = v ? *reinterpret_cast<const uint4 *>(w + off + 16)
-                       : uint4{0, 0, 0, 0};
-          vn[i][2] = (v && D0 != 0)
-                         ? *reinterpret_cast<const uint4 *>(w + off + 32)
-                         : uint4{0, 0, 0, 0};
-        }
+        for (int d = 0; d < 4; d++)
+          gfmac4_lin_rows<GM>(acc, i * 2 + 1, d, (&h1.x)[d], lt);
+        fstore16<ST>(fdst + off, h0);
+        fstore16<ST>(fdst + off + 16, h1);
+        t = shift8k(t, A.stab) ^ crc32_reg(h0, h1, A.tab);
       }
-      uint32_t t = 0;
-#pragma unroll
-      for (int i = 0; i < 2; i++) {
-        const int off = i * 8192 + lane32i;
-        if (off + 32 <= rbi) {
-          uint4 h0, h1;
-          rot_pick<D0>(h0, h1, vc[i][0], vc[i][1], vc[i][2]);
-#pragma unroll
-          for (int d = 0; d < 4; d++)
-            gfmac4_lin_rows<GM>(acc, i * 2, d, (&h0.x)[d], lt);
-#pragma unroll
-          for (int d = 0; d < 4; d++)
-            gfmac4_lin_rows<GM>(acc, i * 2 + 1, d, (&h1.x)[d], lt);
-          fstore16<ST>(fdst + off, h0);
-          fstore16<ST>(fdst + off + 16, h1);
-          t = shift8k(t, A.stab) ^ crc32_reg(h0, h1, A.tab);
+      if (i == 1) { /* unit done: tail bytes + fold + reduce */
+        uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
+        if (rbi & 31) {
+          const uint8_t *src =
+              A.sbase + size_t(c) * A.shard_len + A.p0 + 28 + r0;
+          const int t0 = rbi & ~31;
+          const int p = t0 + tid;
+          if (p < rbi) {
+            const uint8_t x = src[p];
+            fdst[p] = x;
+            A.tailb[c * 32 + (p - t0)] = x;
+            part ^= gf2_mulmod_d(A.x8tab[rbi - 1 - p], A.tab[0][x]);
+          }
         }
-      }
-      uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
-      if (rbi & 31) {
-        /* lane-parallel byte tail (up to 31 B) */
-        const int t0 = rbi & ~31;
-        const int p = t0 + tid;
-        if (p < rbi) {
-          const uint8_t x = src[p];
-          fdst[p] = x;
-          A.tailb[c * 32 + (p - t0)] = x;
-          part ^= gf2_mulmod_d(A.x8tab[rbi - 1 - p], A.tab[0][x]);
-        }
-      }
 #pragma unroll
-      for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
-      if ((tid & 63) == 0) A.red[(tid >> 6) * 16 + c] ^= part;
+        for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
+        if ((tid & 63) == 0) A.red[(tid >> 6) * 16 + c] ^= part;
+      }
     }
     if (rbi & 31) __syncthreads(); /* tailb visible to the parity rows */
 #pragma unroll
     for (int r = 0; r < GM; r++) {
       uint8_t *fdst = A.dst + (A.stripe * (k + GM) + k + r) * A.dst_stride +
                       A.f * block_len + 32 + r0;
-      uint32_t t = 0;
+      uint32_t tr = 0;
 #pragma unroll
       for (int i = 0; i < 2; i++) {
         const int off = i * 8192 + lane32i;
         if (off + 32 <= rbi) {
           fstore16<ST>(fdst + off, acc[r][i * 2]);
           fstore16<ST>(fdst + off + 16, acc[r][i * 2 + 1]);
-          t = shift8k(t, A.stab) ^
-              crc32_reg(acc[r][i * 2], acc[r][i * 2 + 1], A.tab);
+          tr = shift8k(tr, A.stab) ^
+               crc32_reg(acc[r][i * 2], acc[r][i * 2 + 1], A.tab);
         }
       }
-      uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
+      uint32_t part = tr ? gf2_mulmod_d(op, tr) : 0;
       if (rbi & 31) {
         const int t0 = rbi & ~31;
         const int p = t0 + tid;
