@@ -1804,39 +1804,37 @@ GFRS_DEV void rot_passes(const RotArgs &A) {
     /* aligned window base of this pass (D0*4 = the frame shift) */
     const int64_t wbase = A.p0 + 28 - int64_t(D0) * 4 + r0;
 
-    /* flat (unit, piece) pipeline with ONE rolling lookahead window —
-     * half the register footprint of a per-unit double buffer, which
-     * is what lets GM=3 fit 4 waves/SIMD spill-free */
-    uint4 cw[3], nw[3];
-    {
-      const bool v = lane32i + 32 <= rbi;
-      const uint8_t *w = A.sbase + wbase;
-      nw[0] = v ? *reinterpret_cast<const uint4 *>(w + lane32i)
-                : uint4{0, 0, 0, 0};
-      nw[1] = v ? *reinterpret_cast<const uint4 *>(w + lane32i + 16)
-                : uint4{0, 0, 0, 0};
-      nw[2] = uint4{0, 0, 0, 0};
-      if (v && D0 != 0) rot_tail_load<D0>(nw[2], w + lane32i + 32);
-    }
+    /* flat (unit, piece) pipeline with a DEPTH-2 rolling lookahead:
+     * two piece windows (6 loads) in flight — a depth-1 roll exposed
+     * the full HBM latency (measured 2.5x slower), while the per-unit
+     * double buffer spilled at 4 waves/SIMD */
+    uint4 w0[3], w1[3];
+    auto ldwin = [&](uint4 (&w)[3], int pcl) {
+      const int il = pcl & 1;
+      const int offl = il * 8192 + lane32i;
+      const bool v = offl + 32 <= rbi;
+      const uint8_t *wp = A.sbase + size_t(pcl >> 1) * A.shard_len + wbase;
+      w[0] = v ? *reinterpret_cast<const uint4 *>(wp + offl)
+               : uint4{0, 0, 0, 0};
+      w[1] = v ? *reinterpret_cast<const uint4 *>(wp + offl + 16)
+               : uint4{0, 0, 0, 0};
+      w[2] = uint4{0, 0, 0, 0};
+      if (v && D0 != 0) rot_tail_load<D0>(w[2], wp + offl + 32);
+    };
+    ldwin(w0, 0);
+    if (k * 2 > 1) ldwin(w1, 1);
     LinTab lt[GM];
     uint32_t t = 0;
     for (int pc = 0; pc < k * 2; pc++) {
       const int c = pc >> 1, i = pc & 1;
       const int off = i * 8192 + lane32i;
+      uint4 cw[3];
 #pragma unroll
-      for (int j = 0; j < 3; j++) cw[j] = nw[j];
-      if (pc + 1 < k * 2) {
-        const int c2 = (pc + 1) >> 1, i2 = (pc + 1) & 1;
-        const int off2 = i2 * 8192 + lane32i;
-        const bool v = off2 + 32 <= rbi;
-        const uint8_t *w = A.sbase + size_t(c2) * A.shard_len + wbase;
-        nw[0] = v ? *reinterpret_cast<const uint4 *>(w + off2)
-                  : uint4{0, 0, 0, 0};
-        nw[1] = v ? *reinterpret_cast<const uint4 *>(w + off2 + 16)
-                  : uint4{0, 0, 0, 0};
-        nw[2] = uint4{0, 0, 0, 0};
-        if (v && D0 != 0) rot_tail_load<D0>(nw[2], w + off2 + 32);
+      for (int j = 0; j < 3; j++) {
+        cw[j] = w0[j];
+        w0[j] = w1[j];
       }
+      if (pc + 2 < k * 2) ldwin(w1, pc + 2);
       if (i == 0) {
         t = 0;
 #pragma unroll
