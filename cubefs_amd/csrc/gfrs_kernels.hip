@@ -1764,7 +1764,11 @@ struct RotArgs {
 };
 
 template <int D0, int GM, int ST>
-GFRS_DEV void rot_passes(const RotArgs &A) {
+GFRS_DEV void rot_passes(const RotArgs A) {
+  /* NOTE: args by VALUE and every acc[]/window[] index a compile-time
+   * constant — a by-ref struct or a runtime-indexed accumulator array
+   * is materialized in scratch and the piece loop serializes behind
+   * vmcnt(0) drains (measured 3x slower). */
   constexpr int EF_PASS = 16384;
   constexpr int EF_PASSES = 4;
   constexpr int64_t block_len = 65536;
@@ -1775,7 +1779,7 @@ GFRS_DEV void rot_passes(const RotArgs &A) {
   const int lane32i = tid * 32;
   const int k = A.k;
 
-  uint4 acc[GM][4]; /* [row][piece*2 + half] */
+  uint4 acc[GM][4]; /* [row][piece*2 + half], constant-indexed */
   uint32_t op_chain = A.op_first;
 
   for (int h = 0; h < EF_PASSES; h++) {
@@ -1803,76 +1807,102 @@ GFRS_DEV void rot_passes(const RotArgs &A) {
     const int rbi = int(rbytes);
     /* aligned window base of this pass (D0*4 = the frame shift) */
     const int64_t wbase = A.p0 + 28 - int64_t(D0) * 4 + r0;
+    const bool v0 = lane32i + 32 <= rbi;
+    const bool v1 = 8192 + lane32i + 32 <= rbi;
 
-    /* flat (unit, piece) pipeline with a DEPTH-2 rolling lookahead:
-     * two piece windows (6 loads) in flight — a depth-1 roll exposed
-     * the full HBM latency (measured 2.5x slower), while the per-unit
-     * double buffer spilled at 4 waves/SIMD */
-    uint4 w0[3], w1[3];
-    auto ldwin = [&](uint4 (&w)[3], int pcl) {
-      const int il = pcl & 1;
-      const int offl = il * 8192 + lane32i;
-      const bool v = offl + 32 <= rbi;
-      const uint8_t *wp = A.sbase + size_t(pcl >> 1) * A.shard_len + wbase;
-      w[0] = v ? *reinterpret_cast<const uint4 *>(wp + offl)
-               : uint4{0, 0, 0, 0};
-      w[1] = v ? *reinterpret_cast<const uint4 *>(wp + offl + 16)
-               : uint4{0, 0, 0, 0};
-      w[2] = uint4{0, 0, 0, 0};
-      if (v && D0 != 0) rot_tail_load<D0>(w[2], wp + offl + 32);
-    };
-    ldwin(w0, 0);
-    if (k * 2 > 1) ldwin(w1, 1);
-    LinTab lt[GM];
-    uint32_t t = 0;
-    for (int pc = 0; pc < k * 2; pc++) {
-      const int c = pc >> 1, i = pc & 1;
-      const int off = i * 8192 + lane32i;
-      uint4 cw[3];
+    /* two rolling window buffers (one per piece half), each refilled for
+     * the next unit right after its consume: ~6 loads in flight, and
+     * both the buffers and the accumulator halves keep constant
+     * register indices */
+    uint4 wA[3], wB[3];
+    {
+      const uint8_t *wp = A.sbase + wbase;
+      wA[0] = v0 ? *reinterpret_cast<const uint4 *>(wp + lane32i)
+                 : uint4{0, 0, 0, 0};
+      wA[1] = v0 ? *reinterpret_cast<const uint4 *>(wp + lane32i + 16)
+                 : uint4{0, 0, 0, 0};
+      wA[2] = uint4{0, 0, 0, 0};
+      if (v0 && D0 != 0) rot_tail_load<D0>(wA[2], wp + lane32i + 32);
+      const int offb = 8192 + lane32i;
+      wB[0] = v1 ? *reinterpret_cast<const uint4 *>(wp + offb)
+                 : uint4{0, 0, 0, 0};
+      wB[1] = v1 ? *reinterpret_cast<const uint4 *>(wp + offb + 16)
+                 : uint4{0, 0, 0, 0};
+      wB[2] = uint4{0, 0, 0, 0};
+      if (v1 && D0 != 0) rot_tail_load<D0>(wB[2], wp + offb + 32);
+    }
+
+    for (int c = 0; c < k; c++) {
+      LinTab lt[GM];
 #pragma unroll
-      for (int j = 0; j < 3; j++) {
-        cw[j] = w0[j];
-        w0[j] = w1[j];
-      }
-      if (pc + 2 < k * 2) ldwin(w1, pc + 2);
-      if (i == 0) {
-        t = 0;
-#pragma unroll
-        for (int r = 0; r < GM; r++) lt[r] = lintab_load(A.ctab, r * k + c);
-      }
+      for (int r = 0; r < GM; r++) lt[r] = lintab_load(A.ctab, r * k + c);
       uint8_t *fdst = A.dst + (A.stripe * (k + GM) + c) * A.dst_stride +
                       A.f * block_len + 32 + r0;
-      if (off + 32 <= rbi) {
-        uint4 h0, h1;
-        rot_pick<D0>(h0, h1, cw[0], cw[1], cw[2]);
-#pragma unroll
-        for (int d = 0; d < 4; d++)
-          gfmac4_lin_rows<GM>(acc, i * 2, d, (&h0.x)[d], lt);
-#pragma unroll
-        for (int d = 0; d < 4; d++)
-          gfmac4_lin_rows<GM>(acc, i * 2 + 1, d, (&h1.x)[d], lt);
-        fstore16<ST>(fdst + off, h0);
-        fstore16<ST>(fdst + off + 16, h1);
-        t = shift8k(t, A.stab) ^ crc32_reg(h0, h1, A.tab);
-      }
-      if (i == 1) { /* unit done: tail bytes + fold + reduce */
-        uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
-        if (rbi & 31) {
-          const uint8_t *src =
-              A.sbase + size_t(c) * A.shard_len + A.p0 + 28 + r0;
-          const int t0 = rbi & ~31;
-          const int p = t0 + tid;
-          if (p < rbi) {
-            const uint8_t x = src[p];
-            fdst[p] = x;
-            A.tailb[c * 32 + (p - t0)] = x;
-            part ^= gf2_mulmod_d(A.x8tab[rbi - 1 - p], A.tab[0][x]);
-          }
+      const uint8_t *wnext =
+          c + 1 < k ? A.sbase + size_t(c + 1) * A.shard_len + wbase : nullptr;
+      uint32_t t = 0;
+      { /* piece half 0 */
+        uint4 cw0 = wA[0], cw1 = wA[1], cw2 = wA[2];
+        if (wnext) {
+          wA[0] = v0 ? *reinterpret_cast<const uint4 *>(wnext + lane32i)
+                     : uint4{0, 0, 0, 0};
+          wA[1] = v0 ? *reinterpret_cast<const uint4 *>(wnext + lane32i + 16)
+                     : uint4{0, 0, 0, 0};
+          if (v0 && D0 != 0) rot_tail_load<D0>(wA[2], wnext + lane32i + 32);
         }
+        if (v0) {
+          uint4 h0, h1;
+          rot_pick<D0>(h0, h1, cw0, cw1, cw2);
 #pragma unroll
-        for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
-        if ((tid & 63) == 0) A.red[(tid >> 6) * 16 + c] ^= part;
+          for (int d = 0; d < 4; d++)
+            gfmac4_lin_rows<GM>(acc, 0, d, (&h0.x)[d], lt);
+#pragma unroll
+          for (int d = 0; d < 4; d++)
+            gfmac4_lin_rows<GM>(acc, 1, d, (&h1.x)[d], lt);
+          fstore16<ST>(fdst + lane32i, h0);
+          fstore16<ST>(fdst + lane32i + 16, h1);
+          t = crc32_reg(h0, h1, A.tab);
+        }
       }
+      { /* piece half 1 */
+        uint4 cw0 = wB[0], cw1 = wB[1], cw2 = wB[2];
+        const int offb = 8192 + lane32i;
+        if (wnext) {
+          wB[0] = v1 ? *reinterpret_cast<const uint4 *>(wnext + offb)
+                     : uint4{0, 0, 0, 0};
+          wB[1] = v1 ? *reinterpret_cast<const uint4 *>(wnext + offb + 16)
+                     : uint4{0, 0, 0, 0};
+          if (v1 && D0 != 0) rot_tail_load<D0>(wB[2], wnext + offb + 32);
+        }
+        if (v1) {
+          uint4 h0, h1;
+          rot_pick<D0>(h0, h1, cw0, cw1, cw2);
+#pragma unroll
+          for (int d = 0; d < 4; d++)
+            gfmac4_lin_rows<GM>(acc, 2, d, (&h0.x)[d], lt);
+#pragma unroll
+          for (int d = 0; d < 4; d++)
+            gfmac4_lin_rows<GM>(acc, 3, d, (&h1.x)[d], lt);
+          fstore16<ST>(fdst + offb, h0);
+          fstore16<ST>(fdst + offb + 16, h1);
+          t = shift8k(t, A.stab) ^ crc32_reg(h0, h1, A.tab);
+        }
+      }
+      uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
+      if (rbi & 31) {
+        const uint8_t *src = A.sbase + size_t(c) * A.shard_len + A.p0 + 28 + r0;
+        const int t0 = rbi & ~31;
+        const int p = t0 + tid;
+        if (p < rbi) {
+          const uint8_t x = src[p];
+          fdst[p] = x;
+          A.tailb[c * 32 + (p - t0)] = x;
+          part ^= gf2_mulmod_d(A.x8tab[rbi - 1 - p], A.tab[0][x]);
+        }
+      }
+#pragma unroll
+      for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
+      if ((tid & 63) == 0) A.red[(tid >> 6) * 16 + c] ^= part;
     }
     if (rbi & 31) __syncthreads(); /* tailb visible to the parity rows */
 #pragma unroll
@@ -1880,15 +1910,15 @@ GFRS_DEV void rot_passes(const RotArgs &A) {
       uint8_t *fdst = A.dst + (A.stripe * (k + GM) + k + r) * A.dst_stride +
                       A.f * block_len + 32 + r0;
       uint32_t tr = 0;
-#pragma unroll
-      for (int i = 0; i < 2; i++) {
-        const int off = i * 8192 + lane32i;
-        if (off + 32 <= rbi) {
-          fstore16<ST>(fdst + off, acc[r][i * 2]);
-          fstore16<ST>(fdst + off + 16, acc[r][i * 2 + 1]);
-          tr = shift8k(tr, A.stab) ^
-               crc32_reg(acc[r][i * 2], acc[r][i * 2 + 1], A.tab);
-        }
+      if (v0) {
+        fstore16<ST>(fdst + lane32i, acc[r][0]);
+        fstore16<ST>(fdst + lane32i + 16, acc[r][1]);
+        tr = crc32_reg(acc[r][0], acc[r][1], A.tab);
+      }
+      if (v1) {
+        fstore16<ST>(fdst + 8192 + lane32i, acc[r][2]);
+        fstore16<ST>(fdst + 8192 + lane32i + 16, acc[r][3]);
+        tr = shift8k(tr, A.stab) ^ crc32_reg(acc[r][2], acc[r][3], A.tab);
       }
       uint32_t part = tr ? gf2_mulmod_d(op, tr) : 0;
       if (rbi & 31) {
