@@ -27,6 +27,8 @@
 #include <string>
 #include <algorithm>
 #include <array>
+#include <atomic>
+#include <memory>
 #include <vector>
 
 namespace gfrs {
@@ -169,6 +171,28 @@ struct DevPlan {
   }
 };
 
+/* One independent execution lane: own stream + own scratch, so
+ * concurrent single-stripe callers on one context (the reference shares
+ * one ec.Encoder across ~100 goroutines behind a counting semaphore,
+ * encoder.go:29,115) issue to the GPU in parallel instead of
+ * serializing on one stream. */
+struct Lane {
+  hipStream_t stream = nullptr;
+  DevBuf ptr_buf, fail_buf, stage_dev;
+  PinBuf stage_pin;
+  std::mutex mu;
+  ~Lane() {
+    if (stream) hipStreamDestroy(stream);
+  }
+};
+
+/* view over either a lane's or the context's stream+scratch */
+struct LaneView {
+  hipStream_t stream;
+  DevBuf *ptr_buf, *fail_buf, *stage_dev;
+  PinBuf *stage_pin;
+};
+
 struct gfrs_ctx_impl {
   gfrs_tactic t{};
   int total = 0;   /* n+m+l */
@@ -202,6 +226,19 @@ struct gfrs_ctx_impl {
   DevBuf stage_dev; /* host-mode staging */
   PinBuf stage_pin;
 
+  /* stream-lane pool for concurrent foreground (single-stripe) calls;
+   * empty when GFRS_LANES=0 or after gfrs_set_stream pins a caller
+   * stream (user_stream) */
+  std::vector<std::unique_ptr<Lane>> lanes;
+  std::atomic<uint32_t> lane_rr{0};
+  bool user_stream = false;
+  std::mutex cache_mu; /* plan caches (dec/par) — lanes bypass this->mu */
+
+  Lane *pick_lane() {
+    if (user_stream || lanes.empty()) return nullptr;
+    return lanes[lane_rr.fetch_add(1) % lanes.size()].get();
+  }
+
   ~gfrs_ctx_impl() {
     for (auto *p : local_enc) delete p;
     for (auto &kv : dec_cache) delete kv.second;
@@ -220,6 +257,15 @@ struct StreamGuard {
     if (prev >= 0) hipSetDevice(prev);
   }
 };
+
+static inline LaneView view_of(gfrs_ctx_impl *c) {
+  return LaneView{c->stream, &c->ptr_buf, &c->fail_buf, &c->stage_dev,
+                  &c->stage_pin};
+}
+static inline LaneView view_of(Lane *L) {
+  return LaneView{L->stream, &L->ptr_buf, &L->fail_buf, &L->stage_dev,
+                  &L->stage_pin};
+}
 
 /* local stripe global indices for one AZ (codemode.go:301-318) */
 static std::vector<int32_t> local_stripe(const gfrs_tactic &t, int az_idx) {
@@ -254,10 +300,12 @@ using namespace gfrs;
 /* unlocked bodies of the public batch entry points (defined below);
  * already-locked callers reuse these instead of re-locking */
 static int encode_batch_impl(gfrs_ctx_impl *c, void *base, size_t shard_len,
-                             size_t stripe_stride, int nstripes);
+                             size_t stripe_stride, int nstripes,
+                             hipStream_t st);
 static int verify_batch_impl(gfrs_ctx_impl *c, const void *base,
                              size_t shard_len, size_t stripe_stride,
-                             int nstripes, uint64_t *fail_bitmap);
+                             int nstripes, uint64_t *fail_bitmap,
+                             hipStream_t st, DevBuf *fail);
 static int crc32b_encode_batch_impl(gfrs_ctx_impl *c, void *dst,
                                     size_t dst_stride, const void *src,
                                     size_t src_stride, int64_t n,
@@ -315,6 +363,17 @@ gfrs_ctx *gfrs_create(const gfrs_tactic *t, int device) {
   hipSetDevice(device);
   bool ok = hipStreamCreate(&c->own_stream) == hipSuccess;
   c->stream = c->own_stream;
+  if (ok) {
+    const char *le = getenv("GFRS_LANES");
+    int nl = le ? atoi(le) : 4;
+    if (nl < 0) nl = 0;
+    if (nl > 16) nl = 16;
+    for (int i = 0; i < nl && ok; i++) {
+      auto L = std::make_unique<Lane>();
+      ok = hipStreamCreate(&L->stream) == hipSuccess;
+      if (ok) c->lanes.push_back(std::move(L));
+    }
+  }
 
   /* global encode matrix (reedsolomon.go:220-244); replicate modes have
    * no parity engine (reedsolomon.go:442 parityShards==0 early return) */
@@ -410,6 +469,9 @@ int gfrs_set_stream(gfrs_ctx *ctx, void *hip_stream) {
   std::lock_guard<std::mutex> lk(c->mu);
   c->stream = hip_stream ? reinterpret_cast<hipStream_t>(hip_stream)
                          : c->own_stream;
+  /* a caller-pinned stream implies caller-managed ordering: foreground
+   * calls then issue on that one stream instead of fanning out */
+  c->user_stream = hip_stream != nullptr;
   return GFRS_OK;
 }
 
@@ -417,6 +479,7 @@ int gfrs_synchronize(gfrs_ctx *ctx) {
   auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
   StreamGuard g(c);
   HIP_TRY(hipStreamSynchronize(c->stream));
+  for (auto &L : c->lanes) HIP_TRY(hipStreamSynchronize(L->stream));
   return GFRS_OK;
 }
 
@@ -447,14 +510,14 @@ namespace gfrs {
 
 /* Upload a pointer table for pointer-mode launches (stream-ordered, so
  * reuse of the scratch buffer is safe across calls on one stream). */
-static int upload_ptrs(gfrs_ctx_impl *c, void *const *shards, int nshards) {
-  int rc = c->ptr_buf.ensure(size_t(nshards) * 8);
+static int upload_ptrs(const LaneView &lv, void *const *shards, int nshards) {
+  int rc = lv.ptr_buf->ensure(size_t(nshards) * 8);
   if (rc != GFRS_OK) return rc;
   /* small, use pinned staging for async copy */
-  if ((rc = c->stage_pin.ensure(size_t(nshards) * 8)) != GFRS_OK) return rc;
-  memcpy(c->stage_pin.p, shards, size_t(nshards) * 8);
-  HIP_TRY(hipMemcpyAsync(c->ptr_buf.p, c->stage_pin.p, size_t(nshards) * 8,
-                         hipMemcpyHostToDevice, c->stream));
+  if ((rc = lv.stage_pin->ensure(size_t(nshards) * 8)) != GFRS_OK) return rc;
+  memcpy(lv.stage_pin->p, shards, size_t(nshards) * 8);
+  HIP_TRY(hipMemcpyAsync(lv.ptr_buf->p, lv.stage_pin->p, size_t(nshards) * 8,
+                         hipMemcpyHostToDevice, lv.stream));
   return GFRS_OK;
 }
 
@@ -464,7 +527,8 @@ static int get_decode_plan(gfrs_ctx_impl *c, int k, int m,
                            const std::vector<uint8_t> &matrix,
                            const std::vector<int> &engine_idx /* global ids */,
                            const std::vector<uint8_t> &present,
-                           DevPlan **out_plan, int tag) {
+                           DevPlan **out_plan, int tag, hipStream_t s) {
+  std::lock_guard<std::mutex> cache_lk(c->cache_mu);
   /* key = missing bitmask + engine tag (global=1, az-local=2+az, ...):
    * global and local engines can share k, first index AND mask, so the
    * tag is load-bearing (a collision here once wrote a local parity row
@@ -496,7 +560,7 @@ static int get_decode_plan(gfrs_ctx_impl *c, int k, int m,
       rows.insert(rows.end(), &inv[size_t(i) * k], &inv[size_t(i) * k + k]);
     }
   auto *p = new DevPlan();
-  int rc = p->upload(in, out, rows, c->stream);
+  int rc = p->upload(in, out, rows, s);
   if (rc != GFRS_OK) {
     delete p;
     return rc;
@@ -511,7 +575,8 @@ static int get_parity_plan(gfrs_ctx_impl *c, int k, int m,
                            const std::vector<uint8_t> &matrix,
                            const std::vector<int> &engine_idx,
                            const std::vector<uint8_t> &present,
-                           DevPlan **out_plan, int tag) {
+                           DevPlan **out_plan, int tag, hipStream_t s) {
+  std::lock_guard<std::mutex> cache_lk(c->cache_mu);
   PlanKey key = plan_key_tag(uint64_t(tag));
   for (int i = 0; i < k + m; i++)
     if (!present[i]) plan_key_set(key, i);
@@ -534,7 +599,7 @@ static int get_parity_plan(gfrs_ctx_impl *c, int k, int m,
     return GFRS_OK;
   }
   auto *p = new DevPlan();
-  int rc = p->upload(in, out, rows, c->stream);
+  int rc = p->upload(in, out, rows, s);
   if (rc != GFRS_OK) {
     delete p;
     return rc;
@@ -546,8 +611,8 @@ static int get_parity_plan(gfrs_ctx_impl *c, int k, int m,
 
 /* Run one engine's reconstruct over pointer-mode shards.
  * engine_idx maps engine-local 0..k+m-1 to global shard slots. */
-static int reconstruct_engine(gfrs_ctx_impl *c, int k, int m,
-                              const std::vector<uint8_t> &matrix,
+static int reconstruct_engine(gfrs_ctx_impl *c, const LaneView &lv, int k,
+                              int m, const std::vector<uint8_t> &matrix,
                               const std::vector<int> &engine_idx,
                               const std::vector<uint8_t> &present,
                               size_t shard_len, int nstripes, int data_only,
@@ -566,38 +631,40 @@ static int reconstruct_engine(gfrs_ctx_impl *c, int k, int m,
   int rc = GFRS_OK;
   bool have_missing_data = dpresent < k;
   if (have_missing_data) {
-    rc = get_decode_plan(c, k, m, matrix, engine_idx, present, &dec, tag);
+    rc = get_decode_plan(c, k, m, matrix, engine_idx, present, &dec, tag,
+                         lv.stream);
     if (rc != GFRS_OK) return rc;
     if (nptr_or_0 > 0)
-      launch_rs_apply(reinterpret_cast<const uint64_t *>(c->ptr_buf.p),
+      launch_rs_apply(reinterpret_cast<const uint64_t *>(lv.ptr_buf->p),
                       nptr_or_0, (const int32_t *)dec->in_idx.p, dec->k,
                       (const int32_t *)dec->out_idx.p, dec->nout,
                       (const uint8_t *)dec->tabs.p, shard_len, nstripes,
-                      c->stream);
+                      lv.stream);
     else
       launch_rs_apply_strided(strided_base, stripe_stride,
                               (const int32_t *)dec->in_idx.p, dec->k,
                               (const int32_t *)dec->out_idx.p, dec->nout,
                               (const uint8_t *)dec->tabs.p, shard_len,
-                              nstripes, c->stream);
+                              nstripes, lv.stream);
   }
   if (!data_only) {
     DevPlan *par = nullptr;
-    rc = get_parity_plan(c, k, m, matrix, engine_idx, present, &par, tag);
+    rc = get_parity_plan(c, k, m, matrix, engine_idx, present, &par, tag,
+                         lv.stream);
     if (rc != GFRS_OK) return rc;
     if (par) {
       if (nptr_or_0 > 0)
-        launch_rs_apply(reinterpret_cast<const uint64_t *>(c->ptr_buf.p),
+        launch_rs_apply(reinterpret_cast<const uint64_t *>(lv.ptr_buf->p),
                         nptr_or_0, (const int32_t *)par->in_idx.p, par->k,
                         (const int32_t *)par->out_idx.p, par->nout,
                         (const uint8_t *)par->tabs.p, shard_len, nstripes,
-                        c->stream);
+                        lv.stream);
       else
         launch_rs_apply_strided(strided_base, stripe_stride,
                                 (const int32_t *)par->in_idx.p, par->k,
                                 (const int32_t *)par->out_idx.p, par->nout,
                                 (const uint8_t *)par->tabs.p, shard_len,
-                                nstripes, c->stream);
+                                nstripes, lv.stream);
     }
   }
   return GFRS_OK;
@@ -605,14 +672,15 @@ static int reconstruct_engine(gfrs_ctx_impl *c, int k, int m,
 
 /* Core reconstruct across global + local engines.  present covers all
  * n+m+l global slots. */
-static int reconstruct_all(gfrs_ctx_impl *c, std::vector<uint8_t> &present,
-                           size_t shard_len, int nstripes, int data_only,
-                           uint64_t base, uint64_t stride, int nptr_or_0) {
+static int reconstruct_all(gfrs_ctx_impl *c, const LaneView &lv,
+                           std::vector<uint8_t> &present, size_t shard_len,
+                           int nstripes, int data_only, uint64_t base,
+                           uint64_t stride, int nptr_or_0) {
   const gfrs_tactic &t = c->t;
   std::vector<int> gidx(t.n + t.m);
   for (int i = 0; i < t.n + t.m; i++) gidx[i] = i;
   std::vector<uint8_t> gpresent(present.begin(), present.begin() + t.n + t.m);
-  int rc = reconstruct_engine(c, t.n, t.m, c->enc_matrix, gidx, gpresent,
+  int rc = reconstruct_engine(c, lv, t.n, t.m, c->enc_matrix, gidx, gpresent,
                               shard_len, nstripes, data_only, base, stride,
                               nptr_or_0, /*tag=*/1);
   if (rc != GFRS_OK) return rc;
@@ -629,9 +697,9 @@ static int reconstruct_all(gfrs_ctx_impl *c, std::vector<uint8_t> &present,
       if (!lp[i]) need = true;
     }
     if (!need) continue;
-    rc = reconstruct_engine(c, c->local_n, c->local_m, c->local_matrix, li,
-                            lp, shard_len, nstripes, /*data_only=*/0, base,
-                            stride, nptr_or_0, /*tag=*/2 + az);
+    rc = reconstruct_engine(c, lv, c->local_n, c->local_m, c->local_matrix,
+                            li, lp, shard_len, nstripes, /*data_only=*/0,
+                            base, stride, nptr_or_0, /*tag=*/2 + az);
     if (rc != GFRS_OK) return rc;
   }
   return GFRS_OK;
@@ -651,27 +719,32 @@ int gfrs_encode(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
     seterr("encode: want %d shards, got %d", c->total, nshards);
     return GFRS_ERR_INVALID_SHARDS;
   }
-  std::lock_guard<std::mutex> lk(c->mu);
+  /* foreground call: fan out over the stream-lane pool so concurrent
+   * callers on one context overlap on the GPU (the reference shares one
+   * encoder across ~100 goroutines, encoder.go:29) */
+  Lane *L = c->pick_lane();
+  std::unique_lock<std::mutex> lk(L ? L->mu : c->mu);
   StreamGuard g(c);
+  const LaneView lv = L ? view_of(L) : view_of(c);
   int rc;
   if (memloc == GFRS_MEM_HOST) {
     /* stage contiguous stripe, run strided, copy parity back */
     size_t tot = size_t(c->total) * shard_len;
-    if ((rc = c->stage_dev.ensure(tot)) != GFRS_OK) return rc;
-    if ((rc = c->stage_pin.ensure(tot)) != GFRS_OK) return rc;
-    uint8_t *pin = (uint8_t *)c->stage_pin.p;
+    if ((rc = lv.stage_dev->ensure(tot)) != GFRS_OK) return rc;
+    if ((rc = lv.stage_pin->ensure(tot)) != GFRS_OK) return rc;
+    uint8_t *pin = (uint8_t *)lv.stage_pin->p;
     for (int i = 0; i < c->t.n; i++)
       memcpy(pin + size_t(i) * shard_len, shards[i], shard_len);
-    uint8_t *dev = (uint8_t *)c->stage_dev.p;
+    uint8_t *dev = (uint8_t *)lv.stage_dev->p;
     HIP_TRY(hipMemcpyAsync(dev, pin, size_t(c->t.n) * shard_len,
-                           hipMemcpyHostToDevice, c->stream));
-    rc = encode_batch_impl(c, dev, shard_len, tot, 1);
+                           hipMemcpyHostToDevice, lv.stream));
+    rc = encode_batch_impl(c, dev, shard_len, tot, 1, lv.stream);
     if (rc != GFRS_OK) return rc;
     HIP_TRY(hipMemcpyAsync(pin + size_t(c->t.n) * shard_len,
                            dev + size_t(c->t.n) * shard_len,
                            size_t(c->t.m + c->t.l) * shard_len,
-                           hipMemcpyDeviceToHost, c->stream));
-    HIP_TRY(hipStreamSynchronize(c->stream));
+                           hipMemcpyDeviceToHost, lv.stream));
+    HIP_TRY(hipStreamSynchronize(lv.stream));
     for (int i = c->t.n; i < c->total; i++)
       memcpy(shards[i], pin + size_t(i) * shard_len, shard_len);
     return GFRS_OK;
@@ -685,42 +758,43 @@ int gfrs_encode(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
              (const uint8_t *)shards[0] + size_t(i) * shard_len;
   if (contig) {
     rc = encode_batch_impl(c, shards[0], shard_len,
-                          size_t(nshards) * shard_len, 1);
+                           size_t(nshards) * shard_len, 1, lv.stream);
     if (rc != GFRS_OK) return rc;
-    HIP_TRY(hipStreamSynchronize(c->stream));
+    HIP_TRY(hipStreamSynchronize(lv.stream));
     return GFRS_OK;
   }
-  if ((rc = upload_ptrs(c, shards, nshards)) != GFRS_OK) return rc;
-  launch_rs_apply((const uint64_t *)c->ptr_buf.p, c->total,
+  if ((rc = upload_ptrs(lv, shards, nshards)) != GFRS_OK) return rc;
+  launch_rs_apply((const uint64_t *)lv.ptr_buf->p, c->total,
                   (const int32_t *)c->enc_plan.in_idx.p, c->enc_plan.k,
                   (const int32_t *)c->enc_plan.out_idx.p, c->enc_plan.nout,
                   (const uint8_t *)c->enc_plan.tabs.p, shard_len, 1,
-                  c->stream);
+                  lv.stream);
   for (auto *lp : c->local_enc)
-    launch_rs_apply((const uint64_t *)c->ptr_buf.p, c->total,
+    launch_rs_apply((const uint64_t *)lv.ptr_buf->p, c->total,
                     (const int32_t *)lp->in_idx.p, lp->k,
                     (const int32_t *)lp->out_idx.p, lp->nout,
-                    (const uint8_t *)lp->tabs.p, shard_len, 1, c->stream);
-  HIP_TRY(hipStreamSynchronize(c->stream));
+                    (const uint8_t *)lp->tabs.p, shard_len, 1, lv.stream);
+  HIP_TRY(hipStreamSynchronize(lv.stream));
   return GFRS_OK;
 }
 
 /* unlocked bodies: public entry points take c->mu (the reference encoder
  * is share-safe behind its limiter, encoder.go:29,115) */
 static int encode_batch_impl(gfrs_ctx_impl *c, void *base, size_t shard_len,
-                             size_t stripe_stride, int nstripes) {
+                             size_t stripe_stride, int nstripes,
+                             hipStream_t st) {
   StreamGuard g(c);
   launch_rs_apply_strided((uint64_t)base, stripe_stride,
                           (const int32_t *)c->enc_plan.in_idx.p,
                           c->enc_plan.k, (const int32_t *)c->enc_plan.out_idx.p,
                           c->enc_plan.nout, (const uint8_t *)c->enc_plan.tabs.p,
-                          shard_len, nstripes, c->stream);
+                          shard_len, nstripes, st);
   for (auto *lp : c->local_enc)
     launch_rs_apply_strided((uint64_t)base, stripe_stride,
                             (const int32_t *)lp->in_idx.p, lp->k,
                             (const int32_t *)lp->out_idx.p, lp->nout,
                             (const uint8_t *)lp->tabs.p, shard_len, nstripes,
-                            c->stream);
+                            st);
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) return hip_fail("encode_batch launch", e);
   return GFRS_OK;
@@ -730,7 +804,8 @@ int gfrs_encode_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
                       size_t stripe_stride, int nstripes) {
   auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
   std::lock_guard<std::mutex> lk(c->mu);
-  return encode_batch_impl(c, base, shard_len, stripe_stride, nstripes);
+  return encode_batch_impl(c, base, shard_len, stripe_stride, nstripes,
+                           c->stream);
 }
 
 int gfrs_verify(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
@@ -742,113 +817,115 @@ int gfrs_verify(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
   }
   bool local_form = c->t.l > 0 && nshards == c->total / c->t.az_count;
   if (nshards != c->total && !local_form) return GFRS_ERR_INVALID_SHARDS;
+  Lane *L = c->pick_lane();
+  std::unique_lock<std::mutex> lk(L ? L->mu : c->mu);
+  StreamGuard g(c);
+  const LaneView lv = L ? view_of(L) : view_of(c);
   if (local_form) {
     /* one local stripe set (lrcencoder.go:94-99) */
-    std::lock_guard<std::mutex> lk(c->mu);
-    StreamGuard g(c);
     int rc;
     if (memloc == GFRS_MEM_HOST) {
       size_t tot = size_t(nshards) * shard_len;
-      if ((rc = c->stage_dev.ensure(tot)) != GFRS_OK) return rc;
-      if ((rc = c->stage_pin.ensure(tot)) != GFRS_OK) return rc;
-      uint8_t *pin = (uint8_t *)c->stage_pin.p;
+      if ((rc = lv.stage_dev->ensure(tot)) != GFRS_OK) return rc;
+      if ((rc = lv.stage_pin->ensure(tot)) != GFRS_OK) return rc;
+      uint8_t *pin = (uint8_t *)lv.stage_pin->p;
       for (int i = 0; i < nshards; i++)
         memcpy(pin + size_t(i) * shard_len, shards[i], shard_len);
-      HIP_TRY(hipMemcpyAsync(c->stage_dev.p, pin, tot, hipMemcpyHostToDevice,
-                             c->stream));
-      if ((rc = c->fail_buf.ensure(4)) != GFRS_OK) return rc;
-      HIP_TRY(hipMemsetAsync(c->fail_buf.p, 0, 4, c->stream));
-      launch_rs_verify_strided((uint64_t)c->stage_dev.p, tot,
+      HIP_TRY(hipMemcpyAsync(lv.stage_dev->p, pin, tot,
+                             hipMemcpyHostToDevice, lv.stream));
+      if ((rc = lv.fail_buf->ensure(4)) != GFRS_OK) return rc;
+      HIP_TRY(hipMemsetAsync(lv.fail_buf->p, 0, 4, lv.stream));
+      launch_rs_verify_strided((uint64_t)lv.stage_dev->p, tot,
                                (const int32_t *)c->local_plan.in_idx.p,
                                c->local_plan.k,
                                (const int32_t *)c->local_plan.out_idx.p,
                                c->local_plan.nout,
                                (const uint8_t *)c->local_plan.tabs.p,
-                               shard_len, 1, (uint32_t *)c->fail_buf.p,
-                               c->stream);
+                               shard_len, 1, (uint32_t *)lv.fail_buf->p,
+                               lv.stream);
     } else {
-      if ((rc = upload_ptrs(c, shards, nshards)) != GFRS_OK) return rc;
-      if ((rc = c->fail_buf.ensure(4)) != GFRS_OK) return rc;
-      HIP_TRY(hipMemsetAsync(c->fail_buf.p, 0, 4, c->stream));
-      launch_rs_verify((const uint64_t *)c->ptr_buf.p, nshards,
+      if ((rc = upload_ptrs(lv, shards, nshards)) != GFRS_OK) return rc;
+      if ((rc = lv.fail_buf->ensure(4)) != GFRS_OK) return rc;
+      HIP_TRY(hipMemsetAsync(lv.fail_buf->p, 0, 4, lv.stream));
+      launch_rs_verify((const uint64_t *)lv.ptr_buf->p, nshards,
                        (const int32_t *)c->local_plan.in_idx.p,
                        c->local_plan.k,
                        (const int32_t *)c->local_plan.out_idx.p,
                        c->local_plan.nout,
                        (const uint8_t *)c->local_plan.tabs.p, shard_len, 1,
-                       (uint32_t *)c->fail_buf.p, c->stream);
+                       (uint32_t *)lv.fail_buf->p, lv.stream);
     }
     uint32_t fail = 0;
-    HIP_TRY(hipMemcpyAsync(&fail, c->fail_buf.p, 4, hipMemcpyDeviceToHost,
-                           c->stream));
-    HIP_TRY(hipStreamSynchronize(c->stream));
+    HIP_TRY(hipMemcpyAsync(&fail, lv.fail_buf->p, 4, hipMemcpyDeviceToHost,
+                           lv.stream));
+    HIP_TRY(hipStreamSynchronize(lv.stream));
     *ok = fail == 0;
     return GFRS_OK;
   }
-  std::lock_guard<std::mutex> lk(c->mu);
-  StreamGuard g(c);
   int rc;
   const uint64_t *pt;
   if (memloc == GFRS_MEM_HOST) {
     size_t tot = size_t(c->total) * shard_len;
-    if ((rc = c->stage_dev.ensure(tot + 64)) != GFRS_OK) return rc;
-    if ((rc = c->stage_pin.ensure(tot)) != GFRS_OK) return rc;
-    uint8_t *pin = (uint8_t *)c->stage_pin.p;
+    if ((rc = lv.stage_dev->ensure(tot + 64)) != GFRS_OK) return rc;
+    if ((rc = lv.stage_pin->ensure(tot)) != GFRS_OK) return rc;
+    uint8_t *pin = (uint8_t *)lv.stage_pin->p;
     for (int i = 0; i < c->total; i++)
       memcpy(pin + size_t(i) * shard_len, shards[i], shard_len);
-    HIP_TRY(hipMemcpyAsync(c->stage_dev.p, pin, tot, hipMemcpyHostToDevice,
-                           c->stream));
+    HIP_TRY(hipMemcpyAsync(lv.stage_dev->p, pin, tot, hipMemcpyHostToDevice,
+                           lv.stream));
     uint64_t fb = 0;
-    rc = verify_batch_impl(c, c->stage_dev.p, shard_len, tot, 1, &fb);
+    rc = verify_batch_impl(c, lv.stage_dev->p, shard_len, tot, 1, &fb,
+                           lv.stream, lv.fail_buf);
     if (rc != GFRS_OK) return rc;
     *ok = fb == 0;
     return GFRS_OK;
   }
-  if ((rc = upload_ptrs(c, shards, nshards)) != GFRS_OK) return rc;
-  if ((rc = c->fail_buf.ensure(4)) != GFRS_OK) return rc;
-  HIP_TRY(hipMemsetAsync(c->fail_buf.p, 0, 4, c->stream));
-  pt = (const uint64_t *)c->ptr_buf.p;
+  if ((rc = upload_ptrs(lv, shards, nshards)) != GFRS_OK) return rc;
+  if ((rc = lv.fail_buf->ensure(4)) != GFRS_OK) return rc;
+  HIP_TRY(hipMemsetAsync(lv.fail_buf->p, 0, 4, lv.stream));
+  pt = (const uint64_t *)lv.ptr_buf->p;
   launch_rs_verify(pt, c->total, (const int32_t *)c->enc_plan.in_idx.p,
                    c->enc_plan.k, (const int32_t *)c->enc_plan.out_idx.p,
                    c->enc_plan.nout, (const uint8_t *)c->enc_plan.tabs.p,
-                   shard_len, 1, (uint32_t *)c->fail_buf.p, c->stream);
+                   shard_len, 1, (uint32_t *)lv.fail_buf->p, lv.stream);
   for (auto *lp : c->local_enc)
     launch_rs_verify(pt, c->total, (const int32_t *)lp->in_idx.p, lp->k,
                      (const int32_t *)lp->out_idx.p, lp->nout,
                      (const uint8_t *)lp->tabs.p, shard_len, 1,
-                     (uint32_t *)c->fail_buf.p, c->stream);
+                     (uint32_t *)lv.fail_buf->p, lv.stream);
   uint32_t fail = 0;
-  HIP_TRY(hipMemcpyAsync(&fail, c->fail_buf.p, 4, hipMemcpyDeviceToHost,
-                         c->stream));
-  HIP_TRY(hipStreamSynchronize(c->stream));
+  HIP_TRY(hipMemcpyAsync(&fail, lv.fail_buf->p, 4, hipMemcpyDeviceToHost,
+                         lv.stream));
+  HIP_TRY(hipStreamSynchronize(lv.stream));
   *ok = fail == 0;
   return GFRS_OK;
 }
 
 static int verify_batch_impl(gfrs_ctx_impl *c, const void *base,
                              size_t shard_len, size_t stripe_stride,
-                             int nstripes, uint64_t *fail_bitmap) {
+                             int nstripes, uint64_t *fail_bitmap,
+                             hipStream_t st, DevBuf *fail) {
   StreamGuard g(c);
   int rc;
-  if ((rc = c->fail_buf.ensure(size_t(nstripes) * 4)) != GFRS_OK) return rc;
-  HIP_TRY(hipMemsetAsync(c->fail_buf.p, 0, size_t(nstripes) * 4, c->stream));
+  if ((rc = fail->ensure(size_t(nstripes) * 4)) != GFRS_OK) return rc;
+  HIP_TRY(hipMemsetAsync(fail->p, 0, size_t(nstripes) * 4, st));
   launch_rs_verify_strided((uint64_t)base, stripe_stride,
                            (const int32_t *)c->enc_plan.in_idx.p,
                            c->enc_plan.k,
                            (const int32_t *)c->enc_plan.out_idx.p,
                            c->enc_plan.nout,
                            (const uint8_t *)c->enc_plan.tabs.p, shard_len,
-                           nstripes, (uint32_t *)c->fail_buf.p, c->stream);
+                           nstripes, (uint32_t *)fail->p, st);
   for (auto *lp : c->local_enc)
     launch_rs_verify_strided((uint64_t)base, stripe_stride,
                              (const int32_t *)lp->in_idx.p, lp->k,
                              (const int32_t *)lp->out_idx.p, lp->nout,
                              (const uint8_t *)lp->tabs.p, shard_len, nstripes,
-                             (uint32_t *)c->fail_buf.p, c->stream);
+                             (uint32_t *)fail->p, st);
   std::vector<uint32_t> fails(nstripes);
-  HIP_TRY(hipMemcpyAsync(fails.data(), c->fail_buf.p, size_t(nstripes) * 4,
-                         hipMemcpyDeviceToHost, c->stream));
-  HIP_TRY(hipStreamSynchronize(c->stream));
+  HIP_TRY(hipMemcpyAsync(fails.data(), fail->p, size_t(nstripes) * 4,
+                         hipMemcpyDeviceToHost, st));
+  HIP_TRY(hipStreamSynchronize(st));
   if (fail_bitmap) {
     memset(fail_bitmap, 0, ((nstripes + 63) / 64) * 8);
     for (int s = 0; s < nstripes; s++)
@@ -863,7 +940,7 @@ int gfrs_verify_batch(gfrs_ctx *ctx, const void *base, size_t shard_len,
   auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
   std::lock_guard<std::mutex> lk(c->mu);
   return verify_batch_impl(c, base, shard_len, stripe_stride, nstripes,
-                           fail_bitmap);
+                           fail_bitmap, c->stream, &c->fail_buf);
 }
 
 int gfrs_reconstruct(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
@@ -871,8 +948,10 @@ int gfrs_reconstruct(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
                      int nbad, int data_only) {
   auto *c = reinterpret_cast<gfrs_ctx_impl *>(ctx);
   const gfrs_tactic &t = c->t;
-  std::lock_guard<std::mutex> lk(c->mu);
+  Lane *L = c->pick_lane();
+  std::unique_lock<std::mutex> lk(L ? L->mu : c->mu);
   StreamGuard g(c);
+  const LaneView lv = L ? view_of(L) : view_of(c);
   int rc;
 
   if (t.m == 0) /* no parity: any missing shard is unrecoverable */
@@ -890,26 +969,26 @@ int gfrs_reconstruct(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
   if (memloc == GFRS_MEM_HOST) {
     /* stage the whole stripe contiguously and run in strided mode */
     size_t tot = size_t(nshards) * shard_len;
-    if ((rc = c->stage_dev.ensure(tot)) != GFRS_OK) return rc;
-    if ((rc = c->stage_pin.ensure(tot)) != GFRS_OK) return rc;
-    uint8_t *pin = (uint8_t *)c->stage_pin.p;
-    uint8_t *dev = (uint8_t *)c->stage_dev.p;
+    if ((rc = lv.stage_dev->ensure(tot)) != GFRS_OK) return rc;
+    if ((rc = lv.stage_pin->ensure(tot)) != GFRS_OK) return rc;
+    uint8_t *pin = (uint8_t *)lv.stage_pin->p;
+    uint8_t *dev = (uint8_t *)lv.stage_dev->p;
     for (int i = 0; i < nshards; i++)
       if (present[i]) memcpy(pin + size_t(i) * shard_len, shards[i], shard_len);
-    HIP_TRY(hipMemcpyAsync(dev, pin, tot, hipMemcpyHostToDevice, c->stream));
+    HIP_TRY(hipMemcpyAsync(dev, pin, tot, hipMemcpyHostToDevice, lv.stream));
     if (local_form) {
       std::vector<int> li(nshards);
       for (int i = 0; i < nshards; i++) li[i] = i;
-      rc = reconstruct_engine(c, c->local_n, c->local_m, c->local_matrix, li,
-                              present, shard_len, 1, data_only, (uint64_t)dev,
-                              tot, 0, /*tag=*/50);
+      rc = reconstruct_engine(c, lv, c->local_n, c->local_m, c->local_matrix,
+                              li, present, shard_len, 1, data_only,
+                              (uint64_t)dev, tot, 0, /*tag=*/50);
     } else {
-      rc = reconstruct_all(c, present, shard_len, 1, data_only, (uint64_t)dev,
-                           tot, 0);
+      rc = reconstruct_all(c, lv, present, shard_len, 1, data_only,
+                           (uint64_t)dev, tot, 0);
     }
     if (rc != GFRS_OK) return rc;
-    HIP_TRY(hipMemcpyAsync(pin, dev, tot, hipMemcpyDeviceToHost, c->stream));
-    HIP_TRY(hipStreamSynchronize(c->stream));
+    HIP_TRY(hipMemcpyAsync(pin, dev, tot, hipMemcpyDeviceToHost, lv.stream));
+    HIP_TRY(hipStreamSynchronize(lv.stream));
     /* copy back only shards actually rebuilt: with data_only the engine
      * leaves missing parity untouched (ReconstructData semantics,
      * reedsolomon.go:1441-1444) — overwriting the caller's buffer with
@@ -921,19 +1000,20 @@ int gfrs_reconstruct(gfrs_ctx *ctx, void *const *shards, size_t shard_len,
     return GFRS_OK;
   }
 
-  if ((rc = upload_ptrs(c, shards, nshards)) != GFRS_OK) return rc;
+  if ((rc = upload_ptrs(lv, shards, nshards)) != GFRS_OK) return rc;
 
   if (local_form) {
     std::vector<int> li(nshards);
     for (int i = 0; i < nshards; i++) li[i] = i;
-    rc = reconstruct_engine(c, c->local_n, c->local_m, c->local_matrix, li,
-                            present, shard_len, 1, data_only, 0, 0, nshards,
-                            /*tag=*/50);
+    rc = reconstruct_engine(c, lv, c->local_n, c->local_m, c->local_matrix,
+                            li, present, shard_len, 1, data_only, 0, 0,
+                            nshards, /*tag=*/50);
   } else {
-    rc = reconstruct_all(c, present, shard_len, 1, data_only, 0, 0, nshards);
+    rc = reconstruct_all(c, lv, present, shard_len, 1, data_only, 0, 0,
+                         nshards);
   }
   if (rc != GFRS_OK) return rc;
-  HIP_TRY(hipStreamSynchronize(c->stream));
+  HIP_TRY(hipStreamSynchronize(lv.stream));
   return GFRS_OK;
 }
 
@@ -949,8 +1029,8 @@ int gfrs_reconstruct_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
       return GFRS_ERR_INVALID_SHARDS;
     present[bad_idx[i]] = 0;
   }
-  int rc = reconstruct_all(c, present, shard_len, nstripes, data_only,
-                           (uint64_t)base, stripe_stride, 0);
+  int rc = reconstruct_all(c, view_of(c), present, shard_len, nstripes,
+                           data_only, (uint64_t)base, stripe_stride, 0);
   if (rc != GFRS_OK) return rc;
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) return hip_fail("reconstruct_batch launch", e);
@@ -1012,7 +1092,8 @@ int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
   /* fallback composition: needs the contiguous ec.Buffer batch layout */
   if (stripe_stride != size_t(c->total) * shard_len)
     return GFRS_ERR_UNSUPPORTED;
-  int rc = encode_batch_impl(c, base, shard_len, stripe_stride, nstripes);
+  int rc = encode_batch_impl(c, base, shard_len, stripe_stride, nstripes,
+                             c->stream);
   if (rc != GFRS_OK) return rc;
   return crc32b_encode_batch_impl(c, framed, framed_stride, base, shard_len,
                                   int64_t(shard_len), block_len,
@@ -1195,12 +1276,14 @@ int gfrs_encode_idx(gfrs_ctx *ctx, const void *data_shard, int idx,
   std::vector<void *> ptrs(1 + nparity);
   ptrs[0] = const_cast<void *>(data_shard);
   for (int r = 0; r < nparity; r++) ptrs[1 + r] = parity[r];
-  if ((rc = upload_ptrs(c, ptrs.data(), int(ptrs.size()))) != GFRS_OK)
+  if ((rc = upload_ptrs(view_of(c), ptrs.data(), int(ptrs.size()))) !=
+      GFRS_OK)
     return rc;
   /* per-idx plan cached in dec_cache keyed off a synthetic mask */
   PlanKey key = plan_key_tag(62); /* EncodeIdx namespace */
   plan_key_set(key, idx);
   DevPlan *p;
+  std::unique_lock<std::mutex> cache_lk(c->cache_mu);
   auto it = c->dec_cache.find(key);
   if (it != c->dec_cache.end()) {
     p = it->second;
@@ -1256,6 +1339,7 @@ static int lrc_mixed_reconstruct_verify(gfrs_ctx_impl *cc, void *base,
   DevPlan *plan = nullptr;
   uint32_t cmp_mask = 0;
   {
+    std::lock_guard<std::mutex> cache_lk(cc->cache_mu);
     auto it = cc->dec_cache.find(key);
     if (it != cc->dec_cache.end()) plan = it->second;
   }
@@ -1342,7 +1426,10 @@ static int lrc_mixed_reconstruct_verify(gfrs_ctx_impl *cc, void *base,
       delete plan;
       return rc;
     }
-    cc->dec_cache[key] = plan;
+    {
+      std::lock_guard<std::mutex> cache_lk(cc->cache_mu);
+      cc->dec_cache[key] = plan;
+    }
   }
   for (int r = nbad; r < plan->nout; r++) cmp_mask |= 1u << r;
   int rc;
@@ -1415,6 +1502,7 @@ int gfrs_reconstruct_verify_batch(gfrs_ctx *ctx, void *base,
   DevPlan *plan = nullptr;
   uint32_t cmp_mask = 0;
   {
+    std::lock_guard<std::mutex> cache_lk(c->cache_mu);
     auto it = c->dec_cache.find(key);
     if (it != c->dec_cache.end()) plan = it->second;
   }
@@ -1465,7 +1553,10 @@ int gfrs_reconstruct_verify_batch(gfrs_ctx *ctx, void *base,
       delete plan;
       return rc;
     }
-    c->dec_cache[key] = plan;
+    {
+      std::lock_guard<std::mutex> cache_lk(c->cache_mu);
+      c->dec_cache[key] = plan;
+    }
   }
   /* compare bits: present parity rows; missing (data or parity) written */
   {
@@ -1556,6 +1647,7 @@ int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
       for (int b : badv) plan_key_set(key, b); /* incl. local-parity bads */
       DevPlan *plan = nullptr;
       {
+        std::lock_guard<std::mutex> cache_lk(cc->cache_mu);
         auto it = cc->dec_cache.find(key);
         if (it != cc->dec_cache.end()) plan = it->second;
       }
@@ -1651,7 +1743,10 @@ int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
           delete plan;
           return rc2;
         }
-        cc->dec_cache[key] = plan;
+        {
+          std::lock_guard<std::mutex> cache_lk(cc->cache_mu);
+          cc->dec_cache[key] = plan;
+        }
       }
       const int gm = plan->nout;
       int rc2;
