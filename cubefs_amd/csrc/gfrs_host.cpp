@@ -1068,7 +1068,9 @@ int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
   const bool shapes_ok = block_len == 65536 && t.m >= 1 && gm_all <= 4 &&
                          t.n + gm_all <= 16 && framed_stride % 4 == 0 &&
                          (t.l == 0 || c->fused_lrc_ok);
-  if (shapes_ok && shard_len <= 4096) {
+  const bool small_ok =
+      shard_len <= 4096 || (shard_len <= 8192 && gm_all <= 3);
+  if (shapes_ok && small_ok) {
     /* MinShardSize-class shapes: wave-per-stripe fused kernel */
     const DevPlan &pl = t.l == 0 ? c->enc_plan : c->fused_lrc;
     launch_rs_encode_frame_small((uint8_t *)framed, framed_stride,

@@ -180,6 +180,18 @@ GFRS_DEV void gfmac4_lin_rows(uint4 (&acc)[GM][4], int i, int d, uint32_t w,
     (&acc[r][i].x)[d] ^= gfmul4_lin(s012, s345, s67, lt[r]);
 }
 
+/* variable-width accumulator form (acc[GM][NI]) */
+template <int GM, int NI>
+GFRS_DEV void gfmac4_lin_rows_n(uint4 (&acc)[GM][NI], int i, int d,
+                                uint32_t w, const LinTab (&lt)[GM]) {
+  const uint32_t s012 = w & 0x07070707u;
+  const uint32_t s345 = (w >> 3) & 0x07070707u;
+  const uint32_t s67 = (w >> 6) & 0x03030303u;
+#pragma unroll
+  for (int r = 0; r < GM; r++)
+    (&acc[r][i].x)[d] ^= gfmul4_lin(s012, s345, s67, lt[r]);
+}
+
 /* byte-path form for tails/prologues (same tables) */
 GFRS_DEV uint8_t gfmul1_lin(const uint8_t *tt, uint8_t b) {
   return tt[b & 7] ^ tt[8 + ((b >> 3) & 7)] ^ tt[16 + (b >> 6)];
@@ -2598,7 +2610,6 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_encode_frame_small_k(
   const int wv = int(threadIdx.x) >> 6, lane = int(threadIdx.x) & 63;
   const int lane16i = lane * 16;
   uint8_t *wtail = tailb + wv * 256;
-  const uint4 *ltab = reinterpret_cast<const uint4 *>(ctab);
   const int64_t payload = int64_t(shard_len);
   const int pli = int(payload);
   /* lane's fold operator: suffix after its LAST present piece */
@@ -2625,18 +2636,18 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_encode_frame_small_k(
       uint8_t *fdst =
           dst + (stripe * (k + GM) + c) * dst_stride + CRC_LEN;
       uint32_t t = 0;
+      LinTab lt[GM];
+#pragma unroll
+      for (int r = 0; r < GM; r++) lt[r] = lintab_load(ctab, r * k + c);
 #pragma unroll
       for (int i = 0; i < NI; i++) {
         const int off = i * 1024 + lane16i;
         if (off + 16 <= pli) {
           const uint4 v = *reinterpret_cast<const uint4 *>(src + off);
 #pragma unroll
-          for (int r = 0; r < GM; r++) {
-            const int t2 = (r * k + c) * 2;
-            gfmac16<false>(acc[r][i], v, ltab[t2], ltab[t2 + 1]);
-          }
-          uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
-          dw[0] = v.x; dw[1] = v.y; dw[2] = v.z; dw[3] = v.w;
+          for (int d = 0; d < 4; d++)
+            gfmac4_lin_rows_n<GM, NI>(acc, i, d, (&v.x)[d], lt);
+          *reinterpret_cast<uint4 *>(fdst + off) = v;
           t = shift4k(t, stab) ^ crc16_reg(v, tab); /* stab = x^(8*1024) */
         }
       }
@@ -2666,9 +2677,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_encode_frame_small_k(
       for (int i = 0; i < NI; i++) {
         const int off = i * 1024 + lane16i;
         if (off + 16 <= pli) {
-          uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
-          dw[0] = acc[r][i].x; dw[1] = acc[r][i].y;
-          dw[2] = acc[r][i].z; dw[3] = acc[r][i].w;
+          *reinterpret_cast<uint4 *>(fdst + off) = acc[r][i];
           t = shift4k(t, stab) ^ crc16_reg(acc[r][i], tab);
         }
       }
@@ -2679,8 +2688,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_encode_frame_small_k(
           uint8_t pv = 0;
           for (int c2 = 0; c2 < k; c2++) {
             const uint8_t b = wtail[c2 * 16 + (p - t0)];
-            const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
-            pv ^= tt[b & 0x0F] ^ tt[16 + (b >> 4)];
+            pv ^= gfmul1_lin(ctab + size_t(r * k + c2) * 32, b);
           }
           fdst[p] = pv;
           part ^= gf2_mulmod_d(x8tab[pli - 1 - p], tab[0][pv]);
@@ -2707,10 +2715,12 @@ void launch_rs_encode_frame_small(uint8_t *dst, size_t dst_stride,
   if (grid < 1) grid = 1;
   const int ni = int((shard_len + 1023) / 1024);
   const int lds = 12288 + 64 + 1024 + m * k * 32;
+  /* linear A|B|C table region (after the nibble tables) */
+  const uint8_t *ltabs = tabs + size_t(m) * size_t(k) * 32;
 #define GFRS_SM_GO(G, I)                                                  \
   hipLaunchKernelGGL((rs_encode_frame_small_k<G, I>), dim3(grid),         \
                      dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,     \
-                     stripe_stride, shard_len, k, tabs, nstripes)
+                     stripe_stride, shard_len, k, ltabs, nstripes)
 #define GFRS_SM_NI(G)                                                     \
   switch (ni) {                                                           \
     case 1: GFRS_SM_GO(G, 1); break;                                      \
@@ -2718,12 +2728,27 @@ void launch_rs_encode_frame_small(uint8_t *dst, size_t dst_stride,
     case 3: GFRS_SM_GO(G, 3); break;                                      \
     default: GFRS_SM_GO(G, 4);                                            \
   }
+/* 4-8 KiB shards (the old fused-kernel gap: a 256-lane workgroup gets
+ * only ~5 KiB of frame): extend the wave-per-stripe form to NI<=8 for
+ * gm<=3 (acc registers stay within the 4-wave budget) */
+#define GFRS_SM_NI8(G)                                                    \
+  switch (ni) {                                                           \
+    case 1: GFRS_SM_GO(G, 1); break;                                      \
+    case 2: GFRS_SM_GO(G, 2); break;                                      \
+    case 3: GFRS_SM_GO(G, 3); break;                                      \
+    case 4: GFRS_SM_GO(G, 4); break;                                      \
+    case 5: GFRS_SM_GO(G, 5); break;                                      \
+    case 6: GFRS_SM_GO(G, 6); break;                                      \
+    case 7: GFRS_SM_GO(G, 7); break;                                      \
+    default: GFRS_SM_GO(G, 8);                                            \
+  }
   switch (m) {
-    case 1: GFRS_SM_NI(1); break;
-    case 2: GFRS_SM_NI(2); break;
-    case 3: GFRS_SM_NI(3); break;
+    case 1: GFRS_SM_NI8(1); break;
+    case 2: GFRS_SM_NI8(2); break;
+    case 3: GFRS_SM_NI8(3); break;
     default: GFRS_SM_NI(4);
   }
+#undef GFRS_SM_NI8
 #undef GFRS_SM_NI
 #undef GFRS_SM_GO
 }

@@ -128,7 +128,10 @@ def test_fused_encode_frame(oracle, dev):
                        ("EC6P3", 2048), ("EC6P3", 1024), ("EC6P3", 1040),
                        ("EC6P3", 17), ("EC6P3", 2049), ("EC6P3", 4095),
                        ("EC6P3", 4096), ("EC12P4", 3000),
-                       ("LRC12P2L2", 2048), ("EC6P3", 5000)]:
+                       ("LRC12P2L2", 2048), ("EC6P3", 5000),
+                       # NI 5-8 wave-per-stripe forms (4-8 KiB, gm<=3)
+                       ("EC6P3", 6000), ("EC6P3", 8192), ("EC6P3", 7169),
+                       ("EC12P4", 5000)]:
         t = codemode.get_tactic(name)
         ns = 3
         rng = np.random.default_rng(slen ^ t.N)
