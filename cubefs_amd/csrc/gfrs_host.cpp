@@ -1068,8 +1068,10 @@ int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
   const bool shapes_ok = block_len == 65536 && t.m >= 1 && gm_all <= 4 &&
                          t.n + gm_all <= 16 && framed_stride % 4 == 0 &&
                          (t.l == 0 || c->fused_lrc_ok);
+  /* wave-per-stripe crossover (measured r02): small kernel wins to 6 KiB
+   * (5 KiB 1353 vs 514 GiB/s big-kernel, 6 KiB 1427), ties at 8 KiB */
   const bool small_ok =
-      shard_len <= 4096 || (shard_len <= 8192 && gm_all <= 3);
+      shard_len <= 4096 || (shard_len <= 6144 && gm_all <= 3);
   if (shapes_ok && small_ok) {
     /* MinShardSize-class shapes: wave-per-stripe fused kernel */
     const DevPlan &pl = t.l == 0 ? c->enc_plan : c->fused_lrc;

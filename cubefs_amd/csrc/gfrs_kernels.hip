@@ -2768,11 +2768,9 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
    * (NI=2) geometry.  Measured @256 stripes RS(6+3): 14 -> 13.7 ms
    * (single 20 KB stage, 4 blocks/CU) beats 23 (15.3), 13 (15.4),
    * 24 (16.5); NI=2 variants (142/152/162) target 5-6 blocks/CU. */
-  static const int var = []() {
+  static const int var_env = []() {
     const char *e = getenv("GFRS_EF");
-    const int v = e ? atoi(e) : 76; /* register-CRC + load lookahead;
-        after the serial tail/epilogue removal the spill-free 3-wave
-        variant edges the 4-wave squeeze (17.6 vs 17.8 ms @512) */
+    const int v = e ? atoi(e) : -1; /* -1 = auto by shard size */
     switch (v) {
       case 13: case 14: case 23: case 24:
       case 72: case 73:
@@ -2780,9 +2778,16 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
       case 86: case 87: case 96: case 97:
       case 103: case 104: case 113: case 114:
       case 142: case 152: case 162: return v;
-      default: return 14;
+      default: return -1;
     }
   }();
+  /* auto policy (measured r02): >= 1 MiB shards run the 3-wave
+   * register-CRC pipeline (76) whose PMC traffic is algorithmic-exact at
+   * the HBM-bound operating point; smaller shards run the 4-wave
+   * nontemporal form (87), whose extra latency hiding is worth more than
+   * its ~10% L2-thrash traffic there (64 KiB: 1010 vs 910 GiB/s) */
+  const int var =
+      var_env >= 0 ? var_env : (shard_len >= (size_t(1) << 20) ? 76 : 87);
   /* 1xy = dual-aligned 32-B-piece rotation kernel: 103/104 plain stores
    * @3/4 waves, 113/114 nontemporal @3/4 waves */
   if (var == 103 || var == 104 || var == 113 || var == 114) {
