@@ -295,6 +295,12 @@ def main():
         codec = crc32block.Codec(device=local_rank)
         lib().gfrs_set_stream(codec._ctx, cur)
 
+    if args.host_streamed:
+        # config-4 streamed queue: sources live in pinned HOST memory and
+        # stream in batches — no resident ns-sized device buffers
+        run_host_streamed(args, t, S, ns, enc, rank, world, d)
+        return
+
     # synthetic input resident in HBM (seeded per SURVEY.md §8d)
     g = torch.Generator(device=dev)
     g.manual_seed(0xB10B5703 ^ rank)
@@ -341,10 +347,6 @@ def main():
         rvuids = np.ones(ns * len(bad), dtype=np.uint64)
 
     fused = (args.workload == "encode" and with_crc and not args.no_fused)
-
-    if args.host_streamed:
-        run_host_streamed(args, t, S, ns, enc, rank, world, d)
-        return
 
     def step(events=None):
         if events:
