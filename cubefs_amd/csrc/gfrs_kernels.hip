@@ -299,10 +299,19 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
 #pragma unroll
         for (int j = 0; j < KB; j++) {
           if (c0 + j < k) {
+            LinTab lt[GM];
 #pragma unroll
-            for (int r = 0; r < GM; r++) {
-              const int t2 = (r * k + c0 + j) * 2;
-              gfmac16<PERM0>(acc[r], v[j], ltab[t2], ltab[t2 + 1]);
+            for (int r = 0; r < GM; r++)
+              lt[r] = lintab_load(smem, r * k + c0 + j);
+#pragma unroll
+            for (int d = 0; d < 4; d++) {
+              const uint32_t w = (&v[j].x)[d];
+              const uint32_t s012 = w & 0x07070707u;
+              const uint32_t s345 = (w >> 3) & 0x07070707u;
+              const uint32_t s67 = (w >> 6) & 0x03030303u;
+#pragma unroll
+              for (int r = 0; r < GM; r++)
+                (&acc[r].x)[d] ^= gfmul4_lin(s012, s345, s67, lt[r]);
             }
           }
         }
@@ -331,8 +340,7 @@ __global__ __launch_bounds__(RS_BLOCK) void rs_apply_k(
           uint8_t v = (!VERIFY && xor_acc) ? out[off + i] : uint8_t(0);
           for (int c = 0; c < k; c++) {
             const uint8_t b = addr.shard(stripe, in_idx[c], shard_len)[off + i];
-            const uint8_t *t = trow + size_t(c) * 32;
-            v ^= t[b & 0x0F] ^ t[16 + (b >> 4)];
+            v ^= gfmul1_lin(trow + size_t(c) * 32, b);
           }
           if (VERIFY || (cmp_mask >> og) & 1)
             mismatch |= (out[off + i] != v);
@@ -1162,7 +1170,6 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
   }
   __syncthreads();
 
-  const uint4 *ltab = reinterpret_cast<const uint4 *>(ctab);
 
   for (int64_t fr = blockIdx.x; fr < total_frames; fr += gridDim.x) {
     const int64_t stripe = fr / frames_per_shard;
@@ -1229,11 +1236,13 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
           if (off + 16 <= rbytes) {
             const uint4 v = *reinterpret_cast<const uint4 *>(src + off);
             if (ABL != 2) {
+              LinTab lt[GM];
 #pragma unroll
-              for (int r = 0; r < GM; r++) {
-                const int t2 = (r * k + c) * 2;
-                gfmac16<false>(acc[r][i], v, ltab[t2], ltab[t2 + 1]);
-              }
+              for (int r = 0; r < GM; r++)
+                lt[r] = lintab_load(ctab, r * k + c);
+#pragma unroll
+              for (int d = 0; d < 4; d++)
+                gfmac4_lin_rows_n<GM, NI>(acc, i, d, (&v.x)[d], lt);
             }
             if (ABL != 3) {
               uint32_t *dw = reinterpret_cast<uint32_t *>(fdst + off);
@@ -1287,8 +1296,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
             uint8_t pv = 0;
             for (int c2 = 0; c2 < k; c2++) {
               const uint8_t b = sbase[size_t(c2) * shard_len + p0 + r0 + p];
-              const uint8_t *t = ctab + size_t(r * k + c2) * 32;
-              pv ^= t[b & 0x0F] ^ t[16 + (b >> 4)];
+              pv ^= gfmul1_lin(ctab + size_t(r * k + c2) * 32, b);
             }
             fdst[p] = pv;
             stg[(p / EF_CHUNK) * EF_STRIDE + (p & (EF_CHUNK - 1))] = pv;
@@ -2158,7 +2166,6 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
       gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
   __syncthreads();
 
-  const uint4 *ltab = reinterpret_cast<const uint4 *>(ctab);
 
   int64_t stripe = int64_t(blockIdx.x) / frames_per_shard;
   int64_t f = int64_t(blockIdx.x) - stripe * frames_per_shard;
@@ -2229,15 +2236,16 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
                            : uint4{0, 0, 0, 0};
           }
         }
+        LinTab lt[GM];
+#pragma unroll
+        for (int r = 0; r < GM; r++) lt[r] = lintab_load(ctab, r * k + c);
 #pragma unroll
         for (int i = 0; i < 4; i++) {
           const int off = i * 4096 + lane16i;
           if (off + 16 <= rbi) {
 #pragma unroll
-            for (int r = 0; r < GM; r++) {
-              const int t2 = (r * k + c) * 2;
-              gfmac16<false>(acc[r][i], vcur[i], ltab[t2], ltab[t2 + 1]);
-            }
+            for (int d = 0; d < 4; d++)
+              gfmac4_lin_rows<GM>(acc, i, d, (&vcur[i].x)[d], lt);
           }
         }
       }
@@ -2298,8 +2306,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
               uint8_t pv = 0;
               for (int c2 = 0; c2 < k; c2++) {
                 const uint8_t b2 = tailb[c2 * 16 + (p - t0)];
-                const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
-                pv ^= tt[b2 & 0x0F] ^ tt[16 + (b2 >> 4)];
+                pv ^= gfmul1_lin(ctab + size_t(r * k + c2) * 32, b2);
               }
               fdst[p] = pv;
               part ^= gf2_mulmod_d(x8tab[rbi - 1 - p], tab[0][pv]);
@@ -2331,8 +2338,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
               uint8_t pv = 0;
               for (int c2 = 0; c2 < k; c2++) {
                 const uint8_t b2 = tailb[c2 * 16 + (p - t0)];
-                const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
-                pv ^= tt[b2 & 0x0F] ^ tt[16 + (b2 >> 4)];
+                pv ^= gfmul1_lin(ctab + size_t(r * k + c2) * 32, b2);
               }
               d |= uint32_t(pv ^ cshard[p]);
             }
@@ -2404,7 +2410,6 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_small_k(
   const int wv = int(threadIdx.x) >> 6, lane = int(threadIdx.x) & 63;
   const int lane16i = lane * 16;
   uint8_t *wtail = tailb + wv * 256;
-  const uint4 *ltab = reinterpret_cast<const uint4 *>(ctab);
   const int pli = int(shard_len);
   int np = 0;
 #pragma unroll
@@ -2426,16 +2431,17 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_small_k(
 
     for (int c = 0; c < k; c++) {
       const uint8_t *src = sbase + size_t(imap[c]) * shard_len;
+      LinTab lt[GM];
+#pragma unroll
+      for (int r = 0; r < GM; r++) lt[r] = lintab_load(ctab, r * k + c);
 #pragma unroll
       for (int i = 0; i < NI; i++) {
         const int off = i * 1024 + lane16i;
         if (off + 16 <= pli) {
           const uint4 v = *reinterpret_cast<const uint4 *>(src + off);
 #pragma unroll
-          for (int r = 0; r < GM; r++) {
-            const int t2 = (r * k + c) * 2;
-            gfmac16<false>(acc[r][i], v, ltab[t2], ltab[t2 + 1]);
-          }
+          for (int d = 0; d < 4; d++)
+            gfmac4_lin_rows_n<GM, NI>(acc, i, d, (&v.x)[d], lt);
         }
       }
     }
@@ -2473,8 +2479,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_small_k(
             uint8_t pv = 0;
             for (int c2 = 0; c2 < k; c2++) {
               const uint8_t b = wtail[c2 * 16 + (p - t0)];
-              const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
-              pv ^= tt[b & 0x0F] ^ tt[16 + (b >> 4)];
+              pv ^= gfmul1_lin(ctab + size_t(r * k + c2) * 32, b);
             }
             fdst[p] = pv;
             part ^= gf2_mulmod_d(x8tab[pli - 1 - p], tab[0][pv]);
@@ -2505,8 +2510,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_small_k(
             uint8_t pv = 0;
             for (int c2 = 0; c2 < k; c2++) {
               const uint8_t b = wtail[c2 * 16 + (p - t0)];
-              const uint8_t *tt = ctab + size_t(r * k + c2) * 32;
-              pv ^= tt[b & 0x0F] ^ tt[16 + (b >> 4)];
+              pv ^= gfmul1_lin(ctab + size_t(r * k + c2) * 32, b);
             }
             d2 |= uint32_t(pv ^ cshard[p]);
           }
@@ -2715,8 +2719,7 @@ void launch_rs_encode_frame_small(uint8_t *dst, size_t dst_stride,
   if (grid < 1) grid = 1;
   const int ni = int((shard_len + 1023) / 1024);
   const int lds = 12288 + 64 + 1024 + m * k * 32;
-  /* linear A|B|C table region (after the nibble tables) */
-  const uint8_t *ltabs = tabs + size_t(m) * size_t(k) * 32;
+  const uint8_t *ltabs = tabs; /* linear A|B|C layout */
 #define GFRS_SM_GO(G, I)                                                  \
   hipLaunchKernelGGL((rs_encode_frame_small_k<G, I>), dim3(grid),         \
                      dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,     \
@@ -2760,9 +2763,7 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
   const int64_t fps = (int64_t(shard_len) + 65531) / 65532;
   const int64_t total = fps * nstripes;
   const int grid = fused_grid(total, fps);
-  /* register-CRC (reg_k) variants use the A|B|C linear-split region that
-   * DevPlan::upload lays after the nibble tables */
-  const uint8_t *ltabs = tabs + size_t(m) * size_t(k) * 32;
+  const uint8_t *ltabs = tabs; /* linear A|B|C layout (DevPlan::upload) */
   /* occupancy variant: GFRS_EF = NBUF*10 + waves-per-SIMD bound (16 KiB
    * pass), or a 3-digit NBUF*100 + WPS*10 + NI form for the 8 KiB-pass
    * (NI=2) geometry.  Measured @256 stripes RS(6+3): 14 -> 13.7 ms
