@@ -618,6 +618,28 @@ GFRS_DEV uint32_t xor3(uint32_t a, uint32_t b, uint32_t c) {
   return __builtin_amdgcn_bitop3_b32(a, b, c, 0x96); /* a ^ b ^ c */
 }
 
+/* raw CRC of 16 bytes held in a register uint4 (slice-by-8, 2 steps) */
+GFRS_DEV uint32_t crc16_reg(const uint4 q, const uint32_t (*tab)[256]) {
+  uint32_t c = 0;
+#pragma unroll
+  for (int j = 0; j < 2; j++) {
+    const uint32_t w0 = (j == 0 ? q.x : q.z) ^ c;
+    const uint32_t w1 = (j == 0 ? q.y : q.w);
+    c = xor3(xor3(tab[7][w0 & 0xFF], tab[6][(w0 >> 8) & 0xFF],
+                  tab[5][(w0 >> 16) & 0xFF]),
+             xor3(tab[4][w0 >> 24], tab[3][w1 & 0xFF],
+                  tab[2][(w1 >> 8) & 0xFF]),
+             tab[1][(w1 >> 16) & 0xFF] ^ tab[0][w1 >> 24]);
+  }
+  return c;
+}
+
+GFRS_DEV uint32_t shift4k(uint32_t c, const uint32_t (*stab)[256]) {
+  return xor3(stab[0][c & 0xFF], stab[1][(c >> 8) & 0xFF],
+              stab[2][(c >> 16) & 0xFF]) ^
+         stab[3][c >> 24];
+}
+
 /* Raw (no init/final complement) CRC update of a chunk, 4 B at a time via
  * LDS slice-by-4 tables, byte tail scalar.  When dst != nullptr the chunk
  * is simultaneously copied there (the frame/strip move fused into the CRC
@@ -1030,6 +1052,112 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
 }
 
 template <int MODE, bool TAILCRC = false>
+
+/* Register-CRC verify for 64 KiB frames: the fused encode kernel's CRC
+ * machinery with no MAC and no stores — one workgroup per frame, each
+ * lane Horner-chains its four 4096-strided uint4 pieces (g_shift4k),
+ * folds once by its position operator, wave-reduces into a 4-dword LDS
+ * slab, and lane 0 compares the 4 B LE header.  The LDS-staged verify
+ * kernel measured 2.2 TB/s (stage round-trip + per-chunk barriers on a
+ * read-only job); this form is pure streaming reads. */
+__global__ __launch_bounds__(CRC_BLOCKT, 8) void crc32b_verify_reg_k(
+    const uint8_t *__restrict__ src, size_t src_stride, int64_t n,
+    int64_t fps, int64_t total, int64_t *bad) {
+  constexpr int EF_PASS = 16384;
+  constexpr int EF_PASSES = 4;
+  constexpr int64_t block_len = 65536;
+  constexpr int64_t payload_full = block_len - CRC_LEN;
+  constexpr uint32_t INV16K = 0x479933FCu;
+  __shared__ __attribute__((aligned(16))) uint32_t tabS[8][256];
+  __shared__ uint32_t stabS[4][256];
+  __shared__ uint32_t red[4];
+  __shared__ uint32_t x8tabS[16];
+  for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
+    (&tabS[0][0])[i] = (&g_crc_tab4[0][0])[i];
+  for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
+    (&stabS[0][0])[i] = (&g_shift4k[0][0])[i];
+  if (threadIdx.x == 0) {
+    uint32_t v = 0x80000000u;
+    for (int j = 0; j < 16; j++) {
+      x8tabS[j] = v;
+      v = gf2_mulmod_d(v, g_pow8[0]);
+    }
+  }
+  const int64_t lane16 = int64_t(threadIdx.x) * 16;
+  const int lane16i = int(threadIdx.x) * 16;
+  const uint32_t op_first =
+      x8n_d(uint64_t(payload_full - (3 * 4096 + lane16 + 16)));
+  const uint32_t it_full =
+      gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
+  __syncthreads();
+
+  for (int64_t fr = blockIdx.x; fr < total; fr += gridDim.x) {
+    const int64_t shard = fr / fps;
+    const int64_t f = fr - shard * fps;
+    const int64_t p0 = f * payload_full;
+    const int64_t payload = i64min(payload_full, n - p0);
+    const uint8_t *fb = as_global(uint64_t(src) + shard * src_stride) +
+                        f * block_len;
+    const uint8_t *pb = fb + CRC_LEN;
+    if (threadIdx.x < 4) red[threadIdx.x] = 0;
+    __syncthreads();
+
+    uint32_t op_chain = op_first;
+    for (int h = 0; h < EF_PASSES; h++) {
+      const int64_t r0 = int64_t(h) * EF_PASS;
+      const int64_t rbytes = i64min(int64_t(EF_PASS), payload - r0);
+      if (rbytes <= 0) break;
+      uint32_t op = op_chain;
+      if (h == EF_PASSES - 1 && threadIdx.x == 255)
+        op = shift4k(op, stabS); /* lane 255's last pass has 3 pieces */
+      op_chain = gf2_mulmod_d(op_chain, INV16K);
+      if (payload != payload_full) {
+        int np = 0;
+#pragma unroll
+        for (int i = 0; i < 4; i++)
+          if (int64_t(i) * 4096 + lane16 + 16 <= rbytes) np = i + 1;
+        const int64_t end =
+            np ? r0 + int64_t(np - 1) * 4096 + lane16 + 16 : r0;
+        op = x8n_d(uint64_t(payload - end));
+      }
+      const int rbi = int(rbytes);
+      uint32_t t = 0;
+#pragma unroll
+      for (int i = 0; i < 4; i++) {
+        const int off = i * 4096 + lane16i;
+        if (off + 16 <= rbi) {
+          const uint4 v = *reinterpret_cast<const uint4 *>(pb + r0 + off);
+          t = shift4k(t, stabS) ^ crc16_reg(v, tabS);
+        }
+      }
+      uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
+      if (rbi & 15) {
+        const int t0 = rbi & ~15;
+        const int p = t0 + int(threadIdx.x);
+        if (p < rbi)
+          part ^= gf2_mulmod_d(x8tabS[rbi - 1 - p], tabS[0][pb[r0 + p]]);
+      }
+#pragma unroll
+      for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
+      if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] ^= part;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      const uint32_t it =
+          payload == payload_full
+              ? it_full
+              : gf2_mulmod_d(x8n_d(uint64_t(payload)), 0xFFFFFFFFu);
+      const uint32_t crc = ~(it ^ red[0] ^ red[1] ^ red[2] ^ red[3]);
+      uint32_t want;
+      __builtin_memcpy(&want, fb, 4); /* LE header */
+      if (want != crc)
+        atomicMin(reinterpret_cast<unsigned long long *>(&bad[shard]),
+                  static_cast<unsigned long long>(f));
+    }
+    __syncthreads();
+  }
+}
+
 static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
                          size_t src_stride, int64_t n, int64_t block_len,
                          int64_t fps, int64_t total, int64_t *bad,
@@ -1048,6 +1176,12 @@ static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
     const int v = e ? atoi(e) : 64;
     return (v == 32 || v == 64 || v == 128 || v == 256) ? v : 64;
   }();
+  if (MODE == 1 && !TAILCRC && block_len == 65536) {
+    const int g = fused_grid(total, fps);
+    hipLaunchKernelGGL(crc32b_verify_reg_k, dim3(g), dim3(CRC_BLOCKT), 0, s,
+                       src, src_stride, n, fps, total, bad);
+    return;
+  }
   if (block_len == 65536) {
     const bool nt = nt_enabled() && MODE != 1;
 #define GFRS_CRC_GO(NTV, CH)                                                \
@@ -1331,27 +1465,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_k(
   }
 }
 
-/* raw CRC of 16 bytes held in a register uint4 (slice-by-8, 2 steps) */
-GFRS_DEV uint32_t crc16_reg(const uint4 q, const uint32_t (*tab)[256]) {
-  uint32_t c = 0;
-#pragma unroll
-  for (int j = 0; j < 2; j++) {
-    const uint32_t w0 = (j == 0 ? q.x : q.z) ^ c;
-    const uint32_t w1 = (j == 0 ? q.y : q.w);
-    c = xor3(xor3(tab[7][w0 & 0xFF], tab[6][(w0 >> 8) & 0xFF],
-                  tab[5][(w0 >> 16) & 0xFF]),
-             xor3(tab[4][w0 >> 24], tab[3][w1 & 0xFF],
-                  tab[2][(w1 >> 8) & 0xFF]),
-             tab[1][(w1 >> 16) & 0xFF] ^ tab[0][w1 >> 24]);
-  }
-  return c;
-}
 
-GFRS_DEV uint32_t shift4k(uint32_t c, const uint32_t (*stab)[256]) {
-  return xor3(stab[0][c & 0xFF], stab[1][(c >> 8) & 0xFF],
-              stab[2][(c >> 16) & 0xFF]) ^
-         stab[3][c >> 24];
-}
 
 /* Register-CRC variant of the fused encode+frame kernel: no LDS stage at
  * all.  Lane w's four uint4 pieces of a 16 KiB pass (subtile offsets
