@@ -1051,8 +1051,6 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
   }
 }
 
-template <int MODE, bool TAILCRC = false>
-
 /* Register-CRC verify for 64 KiB frames: the fused encode kernel's CRC
  * machinery with no MAC and no stores — one workgroup per frame, each
  * lane Horner-chains its four 4096-strided uint4 pieces (g_shift4k),
@@ -1158,6 +1156,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, 8) void crc32b_verify_reg_k(
   }
 }
 
+template <int MODE, bool TAILCRC = false>
 static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
                          size_t src_stride, int64_t n, int64_t block_len,
                          int64_t fps, int64_t total, int64_t *bad,
