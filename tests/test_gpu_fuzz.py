@@ -64,3 +64,53 @@ def test_fuzz_encode_reconstruct(oracle):
         codec.encode(framed, shards[0], block_len=bl)
         want = oracle.crc32b_encode(ref[0].copy(), block_len=bl)
         assert np.array_equal(framed.cpu().numpy(), want), (it, code, slen, bl)
+
+
+def test_fuzz_encode_frame_batch(oracle):
+    """Fused encode+frame over random (codemode, shard_len, stride pad):
+    framed images bit-identical to oracle encode + oracle framing.  This
+    sweeps the kernel-selection boundaries (wave-per-stripe <= 6 KiB,
+    workgroup-per-frame above, two-kernel fallback for m+l > 4) and both
+    the padded and tight image strides."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from cubefs_amd import codemode, crc32block, ec
+    dev = torch.device("cuda:0")
+    rng = np.random.default_rng(0xF0220002)
+    modes = [1, 2, 5, 8, 9, 11, 200, 201]
+    encoders = {}
+    for it in range(16):
+        code = modes[rng.integers(0, len(modes))]
+        t = codemode.get_tactic(code)
+        # bias toward kernel boundaries: tiny, 1-8 KiB, frame edges, multi-frame
+        pick = rng.integers(0, 5)
+        if pick == 0:
+            slen = int(rng.integers(1, 4097))
+        elif pick == 1:
+            slen = int(rng.integers(4097, 8193))
+        elif pick == 2:
+            slen = int(65532 + rng.integers(-64, 65))
+        elif pick == 3:
+            slen = int(rng.integers(8193, 300001))
+        else:
+            slen = int(65532 * 2 + rng.integers(-16, 17))
+        ns = int(rng.integers(1, 4))
+        if code not in encoders:
+            encoders[code] = ec.Encoder(t)
+        enc = encoders[code]
+        arr = rng.integers(0, 256, (ns, t.total, slen), dtype=np.uint8)
+        batch = torch.from_numpy(arr.copy()).to(dev)
+        enc_sz = crc32block.encode_size(slen)
+        stride = (enc_sz + 255) // 256 * 256 if rng.integers(0, 2) else enc_sz
+        framed = torch.zeros((ns * t.total, stride), dtype=torch.uint8,
+                             device=dev)
+        enc.encode_frame_batch(framed, batch)
+        enc.synchronize()
+        got = framed.cpu().numpy()
+        for s in range(ns):
+            sh = [arr[s, i].copy() for i in range(t.total)]
+            oracle.lrc_encode(t.N, t.M, t.L, t.AZCount, sh)
+            for j in range(t.total):
+                want = oracle.crc32b_encode(sh[j])
+                assert np.array_equal(got[s * t.total + j, :enc_sz], want), \
+                    (it, code, slen, s, j)
