@@ -1058,7 +1058,9 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
  * slab, and lane 0 compares the 4 B LE header.  The LDS-staged verify
  * kernel measured 2.2 TB/s (stage round-trip + per-chunk barriers on a
  * read-only job); this form is pure streaming reads. */
+template <int MODE> /* 1 = verify, 2 = decode (framed -> raw + verify) */
 __global__ __launch_bounds__(CRC_BLOCKT, 8) void crc32b_verify_reg_k(
+    uint8_t *__restrict__ dst, size_t dst_stride,
     const uint8_t *__restrict__ src, size_t src_stride, int64_t n,
     int64_t fps, int64_t total, int64_t *bad) {
   constexpr int EF_PASS = 16384;
@@ -1097,6 +1099,10 @@ __global__ __launch_bounds__(CRC_BLOCKT, 8) void crc32b_verify_reg_k(
     const uint8_t *fb = as_global(uint64_t(src) + shard * src_stride) +
                         f * block_len;
     const uint8_t *pb = fb + CRC_LEN;
+    uint8_t *ob = MODE == 2 ? const_cast<uint8_t *>(as_global(
+                                  uint64_t(dst) + shard * dst_stride)) +
+                                  p0
+                            : nullptr;
     if (threadIdx.x < 4) red[threadIdx.x] = 0;
     __syncthreads();
 
@@ -1125,6 +1131,7 @@ __global__ __launch_bounds__(CRC_BLOCKT, 8) void crc32b_verify_reg_k(
         const int off = i * 4096 + lane16i;
         if (off + 16 <= rbi) {
           const uint4 v = *reinterpret_cast<const uint4 *>(pb + r0 + off);
+          if (MODE == 2) *reinterpret_cast<uint4 *>(ob + r0 + off) = v;
           t = shift4k(t, stabS) ^ crc16_reg(v, tabS);
         }
       }
@@ -1132,8 +1139,11 @@ __global__ __launch_bounds__(CRC_BLOCKT, 8) void crc32b_verify_reg_k(
       if (rbi & 15) {
         const int t0 = rbi & ~15;
         const int p = t0 + int(threadIdx.x);
-        if (p < rbi)
-          part ^= gf2_mulmod_d(x8tabS[rbi - 1 - p], tabS[0][pb[r0 + p]]);
+        if (p < rbi) {
+          const uint8_t x = pb[r0 + p];
+          if (MODE == 2) ob[r0 + p] = x;
+          part ^= gf2_mulmod_d(x8tabS[rbi - 1 - p], tabS[0][x]);
+        }
       }
 #pragma unroll
       for (int sh = 32; sh > 0; sh >>= 1) part ^= __shfl_xor(part, sh, 64);
@@ -1175,10 +1185,11 @@ static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
     const int v = e ? atoi(e) : 64;
     return (v == 32 || v == 64 || v == 128 || v == 256) ? v : 64;
   }();
-  if (MODE == 1 && !TAILCRC && block_len == 65536) {
+  if ((MODE == 1 || MODE == 2) && !TAILCRC && block_len == 65536) {
     const int g = fused_grid(total, fps);
-    hipLaunchKernelGGL(crc32b_verify_reg_k, dim3(g), dim3(CRC_BLOCKT), 0, s,
-                       src, src_stride, n, fps, total, bad);
+    hipLaunchKernelGGL((crc32b_verify_reg_k<MODE>), dim3(g),
+                       dim3(CRC_BLOCKT), 0, s, dst, dst_stride, src,
+                       src_stride, n, fps, total, bad);
     return;
   }
   if (block_len == 65536) {
