@@ -1064,8 +1064,13 @@ int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
                          (t.l == 0 || c->fused_lrc_ok);
   /* wave-per-stripe crossover (measured r02): small kernel wins to 6 KiB
    * (5 KiB 1353 vs 514 GiB/s big-kernel, 6 KiB 1427), ties at 8 KiB */
+  static const size_t small_max = []() {
+    const char *e = getenv("GFRS_SMALL_MAX");
+    const long v = e ? atol(e) : 0;
+    return v > 0 ? size_t(v) : size_t(6144);
+  }();
   const bool small_ok =
-      shard_len <= 4096 || (shard_len <= 6144 && gm_all <= 3);
+      shard_len <= 4096 || (shard_len <= small_max && gm_all <= 3);
   if (shapes_ok && small_ok) {
     /* MinShardSize-class shapes: wave-per-stripe fused kernel */
     const DevPlan &pl = t.l == 0 ? c->enc_plan : c->fused_lrc;
