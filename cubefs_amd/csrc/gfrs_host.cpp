@@ -1062,12 +1062,13 @@ int gfrs_encode_frame_batch(gfrs_ctx *ctx, void *framed,
   const bool shapes_ok = block_len == 65536 && t.m >= 1 && gm_all <= 4 &&
                          t.n + gm_all <= 16 && framed_stride % 4 == 0 &&
                          (t.l == 0 || c->fused_lrc_ok);
-  /* wave-per-stripe crossover (measured r02): small kernel wins to 6 KiB
-   * (5 KiB 1353 vs 514 GiB/s big-kernel, 6 KiB 1427), ties at 8 KiB */
+  /* wave-per-stripe crossover (measured r02): the small kernel wins the
+   * whole 4-8 KiB band (5K 1353 / 6K 1427 / 7K 1056 / 8K 863 vs
+   * 514-833 GiB/s for the workgroup-per-frame form) */
   static const size_t small_max = []() {
     const char *e = getenv("GFRS_SMALL_MAX");
     const long v = e ? atol(e) : 0;
-    return v > 0 ? size_t(v) : size_t(6144);
+    return v > 0 ? size_t(v) : size_t(8192);
   }();
   const bool small_ok =
       shard_len <= 4096 || (shard_len <= small_max && gm_all <= 3);
