@@ -1881,22 +1881,6 @@ GFRS_DEV void rot_tail_load(uint4 &w2, const uint8_t *p) {
   }
 }
 
-/* rotation tail via wave shuffle: the D0 dwords past a lane's 32-B
- * aligned window are the NEXT lane's first dwords; only each wave's last
- * lane (whose neighbor is in another wave) uses its own masked tail
- * load.  Cuts the rotation's load cost from 3 to 2 uint4 per piece. */
-template <int D0>
-GFRS_DEV uint4 rot_tail_shfl(const uint4 w0, const uint4 mytail) {
-  uint4 w2{0, 0, 0, 0};
-  if (D0 != 0) {
-    const bool last = (threadIdx.x & 63) == 63;
-    w2.x = last ? mytail.x : __shfl_down(w0.x, 1, 64);
-    if (D0 >= 2) w2.y = last ? mytail.y : __shfl_down(w0.y, 1, 64);
-    if (D0 >= 3) w2.z = last ? mytail.z : __shfl_down(w0.z, 1, 64);
-  }
-  return w2;
-}
-
 /* statically select the 8 piece dwords out of the lane's 12-dword
  * aligned window (D0 = frame shift / 4) */
 template <int D0>
@@ -1967,15 +1951,8 @@ GFRS_DEV void rot_passes(const RotArgs A) {
     const int rbi = int(rbytes);
     /* aligned window base of this pass (D0*4 = the frame shift) */
     const int64_t wbase = A.p0 + 28 - int64_t(D0) * 4 + r0;
-    const bool v0 = lane32i + 32 <= rbi;       /* piece compute guards */
+    const bool v0 = lane32i + 32 <= rbi;
     const bool v1 = 8192 + lane32i + 32 <= rbi;
-    /* load guards: one lane looser, so a valid lane's NEIGHBOR (whose
-     * own piece may be past the edge) still loads the real bytes its
-     * shuffle-donated rotation tail needs; the over-read stays inside
-     * the stripe (data shards are followed by more shard space) */
-    const bool l0 = lane32i <= rbi;
-    const bool l1 = 8192 + lane32i <= rbi;
-    const bool wlast = (tid & 63) == 63;
 
     /* two rolling window buffers (one per piece half), each refilled for
      * the next unit right after its consume: ~6 loads in flight, and
@@ -1984,19 +1961,19 @@ GFRS_DEV void rot_passes(const RotArgs A) {
     uint4 wA[3], wB[3];
     {
       const uint8_t *wp = A.sbase + wbase;
-      wA[0] = l0 ? *reinterpret_cast<const uint4 *>(wp + lane32i)
+      wA[0] = v0 ? *reinterpret_cast<const uint4 *>(wp + lane32i)
                  : uint4{0, 0, 0, 0};
-      wA[1] = l0 ? *reinterpret_cast<const uint4 *>(wp + lane32i + 16)
+      wA[1] = v0 ? *reinterpret_cast<const uint4 *>(wp + lane32i + 16)
                  : uint4{0, 0, 0, 0};
       wA[2] = uint4{0, 0, 0, 0};
-      if (wlast && v0 && D0 != 0) rot_tail_load<D0>(wA[2], wp + lane32i + 32);
+      if (v0 && D0 != 0) rot_tail_load<D0>(wA[2], wp + lane32i + 32);
       const int offb = 8192 + lane32i;
-      wB[0] = l1 ? *reinterpret_cast<const uint4 *>(wp + offb)
+      wB[0] = v1 ? *reinterpret_cast<const uint4 *>(wp + offb)
                  : uint4{0, 0, 0, 0};
-      wB[1] = l1 ? *reinterpret_cast<const uint4 *>(wp + offb + 16)
+      wB[1] = v1 ? *reinterpret_cast<const uint4 *>(wp + offb + 16)
                  : uint4{0, 0, 0, 0};
       wB[2] = uint4{0, 0, 0, 0};
-      if (wlast && v1 && D0 != 0) rot_tail_load<D0>(wB[2], wp + offb + 32);
+      if (v1 && D0 != 0) rot_tail_load<D0>(wB[2], wp + offb + 32);
     }
 
     for (int c = 0; c < k; c++) {
@@ -2011,17 +1988,15 @@ GFRS_DEV void rot_passes(const RotArgs A) {
       { /* piece half 0 */
         uint4 cw0 = wA[0], cw1 = wA[1], cw2 = wA[2];
         if (wnext) {
-          wA[0] = l0 ? *reinterpret_cast<const uint4 *>(wnext + lane32i)
+          wA[0] = v0 ? *reinterpret_cast<const uint4 *>(wnext + lane32i)
                      : uint4{0, 0, 0, 0};
-          wA[1] = l0 ? *reinterpret_cast<const uint4 *>(wnext + lane32i + 16)
+          wA[1] = v0 ? *reinterpret_cast<const uint4 *>(wnext + lane32i + 16)
                      : uint4{0, 0, 0, 0};
-          if (wlast && v0 && D0 != 0)
-            rot_tail_load<D0>(wA[2], wnext + lane32i + 32);
+          if (v0 && D0 != 0) rot_tail_load<D0>(wA[2], wnext + lane32i + 32);
         }
-        const uint4 t2a = rot_tail_shfl<D0>(cw0, cw2);
         if (v0) {
           uint4 h0, h1;
-          rot_pick<D0>(h0, h1, cw0, cw1, t2a);
+          rot_pick<D0>(h0, h1, cw0, cw1, cw2);
 #pragma unroll
           for (int d = 0; d < 4; d++)
             gfmac4_lin_rows<GM>(acc, 0, d, (&h0.x)[d], lt);
@@ -2037,17 +2012,15 @@ GFRS_DEV void rot_passes(const RotArgs A) {
         uint4 cw0 = wB[0], cw1 = wB[1], cw2 = wB[2];
         const int offb = 8192 + lane32i;
         if (wnext) {
-          wB[0] = l1 ? *reinterpret_cast<const uint4 *>(wnext + offb)
+          wB[0] = v1 ? *reinterpret_cast<const uint4 *>(wnext + offb)
                      : uint4{0, 0, 0, 0};
-          wB[1] = l1 ? *reinterpret_cast<const uint4 *>(wnext + offb + 16)
+          wB[1] = v1 ? *reinterpret_cast<const uint4 *>(wnext + offb + 16)
                      : uint4{0, 0, 0, 0};
-          if (wlast && v1 && D0 != 0)
-            rot_tail_load<D0>(wB[2], wnext + offb + 32);
+          if (v1 && D0 != 0) rot_tail_load<D0>(wB[2], wnext + offb + 32);
         }
-        const uint4 t2b = rot_tail_shfl<D0>(cw0, cw2);
         if (v1) {
           uint4 h0, h1;
-          rot_pick<D0>(h0, h1, cw0, cw1, t2b);
+          rot_pick<D0>(h0, h1, cw0, cw1, cw2);
 #pragma unroll
           for (int d = 0; d < 4; d++)
             gfmac4_lin_rows<GM>(acc, 2, d, (&h0.x)[d], lt);
