@@ -141,3 +141,36 @@ def test_lane_pool_mixed_ops(enc, oracle):
     for th in ths:
         th.join()
     assert not errs, errs
+
+
+def test_lane_pool_host_memory(enc, oracle):
+    """Concurrent HOST-memory callers: each lane's pinned staging buffers
+    must be isolated (a shared staging buffer would interleave stripes)."""
+    t = enc.tactic
+    slen = 32 << 10
+    errs = []
+
+    def worker(seed):
+        try:
+            rng = np.random.default_rng(seed)
+            data = rng.integers(0, 256, (t.N, slen), dtype=np.uint8)
+            sh = [data[j].copy() for j in range(t.N)]
+            sh += [np.zeros(slen, np.uint8) for _ in range(t.M)]
+            want = [x.copy() for x in sh]
+            oracle.rs_encode(t.N, t.M, want)
+            for _ in range(8):
+                for i in range(t.N, t.N + t.M):
+                    sh[i][:] = 0
+                enc.encode(sh)  # numpy -> host-memory staging path
+                for i in range(t.N + t.M):
+                    assert np.array_equal(sh[i], want[i]), i
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    ths = [threading.Thread(target=worker, args=(8000 + i,))
+           for i in range(6)]
+    for th in ths:
+        th.start()
+    for th in ths:
+        th.join()
+    assert not errs, errs
