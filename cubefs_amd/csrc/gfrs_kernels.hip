@@ -1503,7 +1503,11 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     uint8_t *__restrict__ dst, size_t dst_stride, uint64_t base,
     uint64_t stripe_stride, size_t shard_len, int k,
     const uint8_t *__restrict__ tabs /* [GM*k][32] */, int64_t total_frames,
-    int64_t frames_per_shard) {
+    int64_t frames_per_shard, int tiny_len /* folded last-frame payload:
+    when a shard's last frame carries <= 64 payload bytes (64/128/256 KiB
+    shards leave 4-16), the host drops it from the frame grid and the
+    PRECEDING frame's workgroup emits it in its epilogue — at 64 KiB
+    shards half of all workgroup iterations otherwise process ~nothing */) {
   constexpr int EF_PASS = 16384;
   constexpr int EF_PASSES = 4; /* ceil(65532 / 16384) */
   constexpr int64_t block_len = 65536;
@@ -1518,7 +1522,8 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
   /* staged tail bytes of the (up to 16) input shards: the parity tails
    * read these from LDS instead of k dependent global loads per lane */
   uint8_t *tailb = smem + 12288 + EF_RED + 64;
-  uint8_t *ctab = smem + 12288 + EF_RED + 64 + 256;
+  uint8_t *tinyb = smem + 12288 + EF_RED + 64 + 256; /* 16 x 64 */
+  uint8_t *ctab = smem + 12288 + EF_RED + 64 + 256 + 1024;
   for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
     (&tab[0][0])[i] = (&g_crc_tab4[0][0])[i];
   for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
@@ -1796,6 +1801,39 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
             ~(it ^ red[j] ^ red[16 + j] ^ red[32 + j] ^ red[48 + j]);
         *reinterpret_cast<uint32_t *>(
             dst + (stripe * (k + GM) + j) * dst_stride + f * block_len) = crc;
+      }
+    }
+    if (tiny_len && f == frames_per_shard - 1) {
+      /* emit the folded tiny last frame: one lane per shard computes
+       * (or copies) its <= 64 payload bytes, CRCs them bytewise, and
+       * writes header + payload.  No private arrays (runtime-indexed
+       * locals land in scratch). */
+      const int j = int(threadIdx.x);
+      const int64_t tp0 = (f + 1) * payload_full;
+      if (j < k) {
+        const uint8_t *sp = sbase + size_t(j) * shard_len + tp0;
+        for (int b = 0; b < tiny_len; b++) tinyb[j * 64 + b] = sp[b];
+      }
+      __syncthreads();
+      if (j < k + GM) {
+        uint8_t *tb =
+            dst + (stripe * (k + GM) + j) * dst_stride + (f + 1) * block_len;
+        uint32_t cr = 0;
+        for (int b = 0; b < tiny_len; b++) {
+          uint8_t x;
+          if (j < k) {
+            x = tinyb[j * 64 + b];
+          } else {
+            x = 0;
+            for (int c2 = 0; c2 < k; c2++)
+              x ^= gfmul1_lin(ctab + size_t((j - k) * k + c2) * 32,
+                              tinyb[c2 * 64 + b]);
+          }
+          tb[CRC_LEN + b] = x;
+          cr = tab[0][(cr ^ x) & 0xFF] ^ (cr >> 8);
+        }
+        *reinterpret_cast<uint32_t *>(tb) =
+            ~(gf2_mulmod_d(x8n_d(uint64_t(tiny_len)), 0xFFFFFFFFu) ^ cr);
       }
     }
     __syncthreads();
@@ -2888,6 +2926,14 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
   const int64_t total = fps * nstripes;
   const int grid = fused_grid(total, fps);
   const uint8_t *ltabs = tabs; /* linear A|B|C layout (DevPlan::upload) */
+  /* tiny-last-frame fold (reg_k family only): a trailing frame with
+   * <= 64 payload bytes is emitted by the previous frame's workgroup */
+  const int64_t tiny0 =
+      fps >= 2 ? int64_t(shard_len) - (fps - 1) * 65532 : int64_t(0);
+  const int tinyi = (tiny0 > 0 && tiny0 <= 64) ? int(tiny0) : 0;
+  const int64_t fps2 = tinyi ? fps - 1 : fps;
+  const int64_t total2 = fps2 * nstripes;
+  const int grid2 = tinyi ? fused_grid(total2, fps2) : grid;
   /* occupancy variant: GFRS_EF = NBUF*10 + waves-per-SIMD bound (16 KiB
    * pass), or a 3-digit NBUF*100 + WPS*10 + NI form for the 8 KiB-pass
    * (NI=2) geometry.  Measured @256 stripes RS(6+3): 14 -> 13.7 ms
@@ -2931,8 +2977,8 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
    * halving the number of concurrently-touched DRAM streams) with deep
    * (72) or single (73) load lookahead */
   if (var == 72 || var == 73) {
-    const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
-#define GFRS_LO_GO(G, P)                                                    hipLaunchKernelGGL((rs_encode_frame_reg_k<G, 2, 0, 0, P>),                                   dim3(grid), dim3(CRC_BLOCKT), lds, s, dst,                                dst_stride, base, stripe_stride, shard_len, k,                            ltabs, total, fps)
+    const int lds = 12288 + EF_RED + 64 + 256 + 1024 + m * k * 32;
+#define GFRS_LO_GO(G, P)                                                    hipLaunchKernelGGL((rs_encode_frame_reg_k<G, 2, 0, 0, P>),                                   dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst,                                dst_stride, base, stripe_stride, shard_len, k,                            ltabs, total2, fps2, tinyi)
 #define GFRS_LO_SW(P)                                                       switch (m) {                                                                case 1: GFRS_LO_GO(1, P); break;                                          case 2: GFRS_LO_GO(2, P); break;                                          case 3: GFRS_LO_GO(3, P); break;                                          default: GFRS_LO_GO(4, P);                                              }
     if (var == 72) { GFRS_LO_SW(2) } else { GFRS_LO_SW(1) }
 #undef GFRS_LO_SW
@@ -2942,8 +2988,8 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
   /* 8x/9x = store-policy variants of the lookahead pipeline:
    * 86/87 nontemporal @3/4 waves, 96/97 sc1 write-through @3/4 waves */
   if (var == 86 || var == 87 || var == 96 || var == 97) {
-    const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
-#define GFRS_STP_GO(G, W, S)                                                hipLaunchKernelGGL((rs_encode_frame_reg_k<G, W, 0, 0, 1, S>),                                dim3(grid), dim3(CRC_BLOCKT), lds, s, dst,                                dst_stride, base, stripe_stride, shard_len, k,                            ltabs, total, fps)
+    const int lds = 12288 + EF_RED + 64 + 256 + 1024 + m * k * 32;
+#define GFRS_STP_GO(G, W, S)                                                hipLaunchKernelGGL((rs_encode_frame_reg_k<G, W, 0, 0, 1, S>),                                dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst,                                dst_stride, base, stripe_stride, shard_len, k,                            ltabs, total2, fps2, tinyi)
 #define GFRS_STP_SW(W, S)                                                   switch (m) {                                                                case 1: GFRS_STP_GO(1, W, S); break;                                      case 2: GFRS_STP_GO(2, W, S); break;                                      case 3: GFRS_STP_GO(3, W, S); break;                                      default: GFRS_STP_GO(4, W, S);                                          }
     if (var == 86) { GFRS_STP_SW(3, 1) }
     else if (var == 87) { GFRS_STP_SW(4, 1) }
@@ -2961,78 +3007,78 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
       return e ? atoi(e) : 0;
     }();
     if (map77 == 1 || map77 == 2) {
-      const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
+      const int lds = 12288 + EF_RED + 64 + 256 + 1024 + m * k * 32;
       if (map77 == 1)
         hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 4, 1, 0, 1>),
-                           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst,
+                           dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst,
                            dst_stride, base, stripe_stride, shard_len, k,
-                           ltabs, total, fps);
+                           ltabs, total2, fps2, tinyi);
       else
         hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 4, 2, 0, 1>),
-                           dim3(grid), dim3(CRC_BLOCKT), lds, s, dst,
+                           dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst,
                            dst_stride, base, stripe_stride, shard_len, k,
-                           ltabs, total, fps);
+                           ltabs, total2, fps2, tinyi);
       return;
     }
   }
   /* 7x = register-CRC kernel (no stage): crc tabs + shift tabs + red */
   if (var == 78) { /* two-unit-deep lookahead at 3 waves/SIMD */
-    const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
+    const int lds = 12288 + EF_RED + 64 + 256 + 1024 + m * k * 32;
     switch (m) {
       case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 3, 0, 0, 2>),
-          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps); break;
+          dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); break;
       case 2: hipLaunchKernelGGL((rs_encode_frame_reg_k<2, 3, 0, 0, 2>),
-          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps); break;
+          dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); break;
       case 3: hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 0, 2>),
-          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps); break;
+          dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); break;
       default: hipLaunchKernelGGL((rs_encode_frame_reg_k<4, 3, 0, 0, 2>),
-          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps);
+          dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi);
     }
     return;
   }
   if (var == 77) { /* lookahead pipeline squeezed to 4 waves/SIMD */
-    const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
+    const int lds = 12288 + EF_RED + 64 + 256 + 1024 + m * k * 32;
     switch (m) {
       case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 4, 0, 0, 1>),
-          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps); break;
+          dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); break;
       case 2: hipLaunchKernelGGL((rs_encode_frame_reg_k<2, 4, 0, 0, 1>),
-          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps); break;
+          dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); break;
       case 3: hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 4, 0, 0, 1>),
-          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps); break;
+          dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); break;
       default: hipLaunchKernelGGL((rs_encode_frame_reg_k<4, 4, 0, 0, 1>),
-          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps);
+          dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi);
     }
     return;
   }
   if (var == 76) { /* one-unit-lookahead pipeline at 3 waves/SIMD */
-    const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
+    const int lds = 12288 + EF_RED + 64 + 256 + 1024 + m * k * 32;
     switch (m) {
       case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 3, 0, 0, 1>),
-          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps); break;
+          dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); break;
       case 2: hipLaunchKernelGGL((rs_encode_frame_reg_k<2, 3, 0, 0, 1>),
-          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps); break;
+          dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); break;
       case 3: hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 0, 1>),
-          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps); break;
+          dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); break;
       default: hipLaunchKernelGGL((rs_encode_frame_reg_k<4, 3, 0, 0, 1>),
-          dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps);
+          dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi);
     }
     return;
   }
   if (var == 74 || var == 75) {
-    const int lds = 12288 + EF_RED + 64 + 256 + m * k * 32;
-#define GFRS_EFR_GO(G, W, P) hipLaunchKernelGGL((rs_encode_frame_reg_k<G, W, P>), dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base, stripe_stride, shard_len, k, ltabs, total, fps)
+    const int lds = 12288 + EF_RED + 64 + 256 + 1024 + m * k * 32;
+#define GFRS_EFR_GO(G, W, P) hipLaunchKernelGGL((rs_encode_frame_reg_k<G, W, P>), dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base, stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi)
 #define GFRS_EFR_SW(W, P) switch (m) { case 1: GFRS_EFR_GO(1, W, P); break; case 2: GFRS_EFR_GO(2, W, P); break; case 3: GFRS_EFR_GO(3, W, P); break; default: GFRS_EFR_GO(4, W, P); }
     static const int map = []() {
       const char *e = getenv("GFRS_EF_MAP");
@@ -3045,28 +3091,28 @@ void launch_rs_encode_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
       return (v >= 4 && v <= 7) ? v : 0;
     }();
     if (rabl >= 5 && var == 76 && m == 3) { /* phase diagnostics */
-      const int lds5 = 12288 + EF_RED + 64 + 256 + m * k * 32;
+      const int lds5 = 12288 + EF_RED + 64 + 256 + 1024 + m * k * 32;
       if (rabl == 5) { hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 2, 1>),
-          dim3(grid), dim3(CRC_BLOCKT), lds5, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps); return; }
+          dim3(grid2), dim3(CRC_BLOCKT), lds5, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); return; }
       if (rabl == 6) { hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 3, 1>),
-          dim3(grid), dim3(CRC_BLOCKT), lds5, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps); return; }
+          dim3(grid2), dim3(CRC_BLOCKT), lds5, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); return; }
       if (rabl == 7) { hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 3, 0, 4, 1>),
-          dim3(grid), dim3(CRC_BLOCKT), lds5, s, dst, dst_stride, base,
-          stripe_stride, shard_len, k, ltabs, total, fps); return; }
+          dim3(grid2), dim3(CRC_BLOCKT), lds5, s, dst, dst_stride, base,
+          stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); return; }
     }
     if (rabl == 4) { /* memory-floor skeleton (diagnostic only) */
       switch (m) {
         case 1: hipLaunchKernelGGL((rs_encode_frame_reg_k<1, 4, 1, 1>),
-            dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-            stripe_stride, shard_len, k, ltabs, total, fps); break;
+            dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+            stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); break;
         case 3: hipLaunchKernelGGL((rs_encode_frame_reg_k<3, 4, 1, 1>),
-            dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-            stripe_stride, shard_len, k, ltabs, total, fps); break;
+            dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+            stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi); break;
         default: hipLaunchKernelGGL((rs_encode_frame_reg_k<4, 4, 1, 1>),
-            dim3(grid), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
-            stripe_stride, shard_len, k, ltabs, total, fps);
+            dim3(grid2), dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,
+            stripe_stride, shard_len, k, ltabs, total2, fps2, tinyi);
       }
       return;
     }

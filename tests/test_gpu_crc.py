@@ -131,7 +131,13 @@ def test_fused_encode_frame(oracle, dev):
                        ("LRC12P2L2", 2048), ("EC6P3", 5000),
                        # NI 5-8 wave-per-stripe forms (4-8 KiB, gm<=3)
                        ("EC6P3", 6000), ("EC6P3", 8192), ("EC6P3", 7169),
-                       ("EC12P4", 5000)]:
+                       ("EC12P4", 5000),
+                       # tiny-last-frame fold (trailing frame <= 64 B is
+                       # emitted by the preceding frame's workgroup)
+                       ("EC6P3", 65536), ("EC6P3", 65533),
+                       ("EC6P3", 65532 + 64), ("EC6P3", 65532 + 65),
+                       ("EC6P3", 2 * 65532 + 16), ("EC12P4", 65536),
+                       ("LRC12P2L2", 131072), ("EC6P3", 262144)]:
         t = codemode.get_tactic(name)
         ns = 3
         rng = np.random.default_rng(slen ^ t.N)

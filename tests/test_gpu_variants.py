@@ -27,6 +27,7 @@ codemode.extend(240, "LRC12P2L2", codemode.Tactic(12, 2, 2, 2, 14, 0, 2048))
 # prologue-only last frame (17), sub-frame odd sizes, 8 MiB-class
 cases = [("EC6P3", 300000), ("EC6P3", 65532), ("EC6P3", 17),
          ("EC6P3", 100), ("EC6P3", 5000), ("EC6P3", 1 << 20),
+         ("EC6P3", 65536), ("EC6P3", 2 * 65532 + 16),
          ("EC12P4", 200000), ("LRC12P2L2", 130000)]
 for name, slen in cases:
     t = codemode.get_tactic(name)
