@@ -2746,8 +2746,8 @@ void launch_rs_repair_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
  * (64 lanes x 16 B = 1 KiB per subtile) Horner-chained by the constant
  * x^(8*1024) byte-sliced tables.  Four independent waves per block.
  * Covers the reference's 2 KiB MinShardSize PUT shapes in one launch. */
-template <int GM, int NI>
-__global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_encode_frame_small_k(
+template <int GM, int NI, int WPS = 4>
+__global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_small_k(
     uint8_t *__restrict__ dst, size_t dst_stride, uint64_t base,
     uint64_t stripe_stride, size_t shard_len, int k,
     const uint8_t *__restrict__ tabs /* [GM*k][32] */, int64_t nstripes) {
@@ -2896,6 +2896,12 @@ void launch_rs_encode_frame_small(uint8_t *dst, size_t dst_stride,
 /* 4-8 KiB shards (the old fused-kernel gap: a 256-lane workgroup gets
  * only ~5 KiB of frame): extend the wave-per-stripe form to NI<=8 for
  * gm<=3 (acc registers stay within the 4-wave budget) */
+/* NI >= 6 instantiates at 3 waves/SIMD: the 4-wave register budget
+ * (128) spills 6-63 VGPRs for GM=3 accumulators at those widths */
+#define GFRS_SM_GO3(G, I)                                                 \
+  hipLaunchKernelGGL((rs_encode_frame_small_k<G, I, 3>), dim3(grid),      \
+                     dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,     \
+                     stripe_stride, shard_len, k, ltabs, nstripes)
 #define GFRS_SM_NI8(G)                                                    \
   switch (ni) {                                                           \
     case 1: GFRS_SM_GO(G, 1); break;                                      \
@@ -2903,9 +2909,9 @@ void launch_rs_encode_frame_small(uint8_t *dst, size_t dst_stride,
     case 3: GFRS_SM_GO(G, 3); break;                                      \
     case 4: GFRS_SM_GO(G, 4); break;                                      \
     case 5: GFRS_SM_GO(G, 5); break;                                      \
-    case 6: GFRS_SM_GO(G, 6); break;                                      \
-    case 7: GFRS_SM_GO(G, 7); break;                                      \
-    default: GFRS_SM_GO(G, 8);                                            \
+    case 6: GFRS_SM_GO3(G, 6); break;                                     \
+    case 7: GFRS_SM_GO3(G, 7); break;                                     \
+    default: GFRS_SM_GO3(G, 8);                                           \
   }
   switch (m) {
     case 1: GFRS_SM_NI8(1); break;
@@ -2914,6 +2920,7 @@ void launch_rs_encode_frame_small(uint8_t *dst, size_t dst_stride,
     default: GFRS_SM_NI(4);
   }
 #undef GFRS_SM_NI8
+#undef GFRS_SM_GO3
 #undef GFRS_SM_NI
 #undef GFRS_SM_GO
 }
