@@ -1070,12 +1070,15 @@ __global__ __launch_bounds__(CRC_BLOCKT, 8) void crc32b_verify_reg_k(
   constexpr uint32_t INV16K = 0x479933FCu;
   __shared__ __attribute__((aligned(16))) uint32_t tabS[8][256];
   __shared__ uint32_t stabS[4][256];
+  __shared__ uint32_t stab8S[4][256]; /* x^(8*8192): split-chain combine */
   __shared__ uint32_t red[4];
   __shared__ uint32_t x8tabS[16];
   for (int i = threadIdx.x; i < 2048; i += CRC_BLOCKT)
     (&tabS[0][0])[i] = (&g_crc_tab4[0][0])[i];
   for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
     (&stabS[0][0])[i] = (&g_shift4k[0][0])[i];
+  for (int i = threadIdx.x; i < 1024; i += CRC_BLOCKT)
+    (&stab8S[0][0])[i] = (&g_shift8k[0][0])[i];
   if (threadIdx.x == 0) {
     uint32_t v = 0x80000000u;
     for (int j = 0; j < 16; j++) {
@@ -1125,16 +1128,33 @@ __global__ __launch_bounds__(CRC_BLOCKT, 8) void crc32b_verify_reg_k(
         op = x8n_d(uint64_t(payload - end));
       }
       const int rbi = int(rbytes);
-      uint32_t t = 0;
+      /* two independent Horner sub-chains (pieces 0-1 and 2-3) double
+       * the ILP of the LDS-gather chain; the combine shift depends on
+       * how many pieces landed in the second chain */
+      uint32_t tA = 0, tB = 0;
+      int nB = 0;
 #pragma unroll
-      for (int i = 0; i < 4; i++) {
+      for (int i = 0; i < 2; i++) {
         const int off = i * 4096 + lane16i;
         if (off + 16 <= rbi) {
           const uint4 v = *reinterpret_cast<const uint4 *>(pb + r0 + off);
           if (MODE == 2) *reinterpret_cast<uint4 *>(ob + r0 + off) = v;
-          t = shift4k(t, stabS) ^ crc16_reg(v, tabS);
+          tA = shift4k(tA, stabS) ^ crc16_reg(v, tabS);
         }
       }
+#pragma unroll
+      for (int i = 2; i < 4; i++) {
+        const int off = i * 4096 + lane16i;
+        if (off + 16 <= rbi) {
+          const uint4 v = *reinterpret_cast<const uint4 *>(pb + r0 + off);
+          if (MODE == 2) *reinterpret_cast<uint4 *>(ob + r0 + off) = v;
+          tB = shift4k(tB, stabS) ^ crc16_reg(v, tabS);
+          nB++;
+        }
+      }
+      const uint32_t t = nB == 2   ? shift4k(tA, stab8S) ^ tB
+                         : nB == 1 ? shift4k(tA, stabS) ^ tB
+                                   : tA;
       uint32_t part = t ? gf2_mulmod_d(op, t) : 0;
       if (rbi & 15) {
         const int t0 = rbi & ~15;
