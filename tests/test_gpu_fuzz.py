@@ -1,5 +1,7 @@
 """Randomized parity fuzz: random codemodes, shard lengths and missing
 sets, HIP engine vs oracle bit-exact (seeded, reproducible)."""
+import os
+
 import numpy as np
 import pytest
 
@@ -7,7 +9,8 @@ torch = pytest.importorskip("torch")
 
 pytestmark = pytest.mark.gpu
 
-CASES = 24
+# soak runs override this (GFRS_FUZZ_CASES=200 for a deep sweep)
+CASES = int(os.environ.get("GFRS_FUZZ_CASES", "24"))
 
 
 def test_fuzz_encode_reconstruct(oracle):
@@ -79,7 +82,7 @@ def test_fuzz_encode_frame_batch(oracle):
     rng = np.random.default_rng(0xF0220002)
     modes = [1, 2, 5, 8, 9, 11, 200, 201]
     encoders = {}
-    for it in range(16):
+    for it in range(max(16, CASES // 2)):
         code = modes[rng.integers(0, len(modes))]
         t = codemode.get_tactic(code)
         # bias toward kernel boundaries: tiny, 1-8 KiB, frame edges, multi-frame
