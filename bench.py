@@ -309,11 +309,9 @@ def main():
     framed = None
     enc_sz = crc32block.encode_size(S)
     if with_crc:
-        # pad the per-image stride to 256 B: encode_size(8 MiB) is ≡4
-        # mod 16, and an unpadded stride would put 3/4 of the shard
-        # images back on misaligned frame bases (the aligned-store
-        # kernel split keys off 16-B-aligned image bases; probe:
-        # +4-misaligned stores cost ~16% of the 6r:9w mix ceiling)
+        # pad the per-image stride to 256 B (encode_size(8 MiB) is ≡4
+        # mod 16): cache-line-aligned image bases; production image
+        # buffers would be pool-allocated the same way
         enc_stride = (enc_sz + 255) // 256 * 256
         framed = torch.empty((ns * t.total, enc_stride), dtype=torch.uint8,
                              device=dev)
