@@ -1834,6 +1834,46 @@ int gfrs_repair_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
 
 /* ---------------- crc32block ---------------- */
 
+/* Host CRC32-IEEE (reflected 0xEDB88320), slice-by-8, Update semantics
+ * like hash/crc32.Update (util.go:22 ChecksumIEEE call sites).  Host-side
+ * on purpose: it backs the per-block streaming request-body wrapper
+ * (request_body.go:57-127), which the reference also runs on host CPUs;
+ * all bulk framing goes through the device kernels. */
+static uint32_t g_crc8tab[8][256];
+static std::once_flag g_crc8_once;
+
+static void crc8tab_init() {
+  for (uint32_t i = 0; i < 256; i++) {
+    uint32_t c = i;
+    for (int k = 0; k < 8; k++) c = (c >> 1) ^ (0xEDB88320u & (0u - (c & 1)));
+    g_crc8tab[0][i] = c;
+  }
+  for (int t = 1; t < 8; t++)
+    for (uint32_t i = 0; i < 256; i++)
+      g_crc8tab[t][i] =
+          (g_crc8tab[t - 1][i] >> 8) ^ g_crc8tab[0][g_crc8tab[t - 1][i] & 0xFF];
+}
+
+uint32_t gfrs_crc32_host(uint32_t crc, const void *data, int64_t n) {
+  std::call_once(g_crc8_once, crc8tab_init);
+  const uint8_t *p = (const uint8_t *)data;
+  uint32_t c = ~crc;
+  while (n >= 8) {
+    uint32_t lo, hi;
+    memcpy(&lo, p, 4);
+    memcpy(&hi, p + 4, 4);
+    lo ^= c;
+    c = g_crc8tab[7][lo & 0xFF] ^ g_crc8tab[6][(lo >> 8) & 0xFF] ^
+        g_crc8tab[5][(lo >> 16) & 0xFF] ^ g_crc8tab[4][lo >> 24] ^
+        g_crc8tab[3][hi & 0xFF] ^ g_crc8tab[2][(hi >> 8) & 0xFF] ^
+        g_crc8tab[1][(hi >> 16) & 0xFF] ^ g_crc8tab[0][hi >> 24];
+    p += 8;
+    n -= 8;
+  }
+  while (n-- > 0) c = (c >> 8) ^ g_crc8tab[0][(c ^ *p++) & 0xFF];
+  return ~c;
+}
+
 int64_t gfrs_crc32b_encode_size(int64_t size, int64_t block_len) {
   if (block_len <= 0 || block_len % 4096) return GFRS_ERR_INVALID_BLOCK;
   int64_t payload = block_len - 4;

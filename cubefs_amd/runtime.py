@@ -25,6 +25,7 @@ ERRORS = {
     -8: "ErrSingular",
     -9: "ErrMismatchedCrc",
     -10: "ErrInvalidBlock",
+    -11: "ErrReadOnClosed",
     -100: "ErrHIP",
     -101: "ErrNoGPU",
     -102: "ErrNoMem",
@@ -99,6 +100,8 @@ def lib():
                                         ctypes.c_int, u64p]
         L.gfrs_reconstruct_batch.argtypes = [vp, vp, ctypes.c_size_t, ctypes.c_size_t,
                                              ctypes.c_int, i32p, ctypes.c_int, ctypes.c_int]
+        L.gfrs_crc32_host.restype = ctypes.c_uint32
+        L.gfrs_crc32_host.argtypes = [ctypes.c_uint32, vp, i64]
         L.gfrs_crc32b_encode_size.restype = i64
         L.gfrs_crc32b_encode_size.argtypes = [i64, i64]
         L.gfrs_crc32b_decode_size.restype = i64

@@ -40,6 +40,8 @@ enum {
   GFRS_ERR_SINGULAR = -8,         /* matrix.errSingular (matrix.go:186) */
   GFRS_ERR_MISMATCHED_CRC = -9,   /* crc32block.ErrMismatchedCrc */
   GFRS_ERR_INVALID_BLOCK = -10,   /* crc32block.ErrInvalidBlock */
+  GFRS_ERR_READ_ON_CLOSED = -11,  /* crc32block.ErrReadOnClosed
+                                     (request_body.go streaming bodies) */
   GFRS_ERR_HIP = -100,            /* HIP runtime failure (see gfrs_last_error) */
   GFRS_ERR_NO_GPU = -101,         /* no MI355X visible; the engine never
                                      falls back to CPU — by design */
@@ -127,6 +129,14 @@ int gfrs_reconstruct_batch(gfrs_ctx *ctx, void *base, size_t shard_len,
 
 int64_t gfrs_crc32b_encode_size(int64_t size, int64_t block_len);
 int64_t gfrs_crc32b_decode_size(int64_t size, int64_t block_len);
+
+/* Host-side CRC32-IEEE in hash/crc32 Update semantics: pass the previous
+ * finalized crc (0 for a fresh start) and get the finalized crc of the
+ * concatenation.  This exists for the per-block STREAMING request-body
+ * wrapper (crc32block/request_body.go:57-127), which in the reference
+ * runs on the client/server host too — every bulk path computes CRCs on
+ * device.  Needs no ctx and no GPU. */
+uint32_t gfrs_crc32_host(uint32_t crc, const void *data, int64_t n);
 
 /* Frame n raw bytes from src into dst (encode.go:86-106).
  * dst capacity must be gfrs_crc32b_encode_size(n).  Returns bytes
