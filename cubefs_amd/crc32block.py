@@ -10,6 +10,31 @@ import ctypes
 from .runtime import GfrsError, Tactic, check, lib
 
 DEFAULT_BLOCK = 64 * 1024
+_CRC_LEN = 4
+
+
+def crc32_host(data, crc=0):
+    """Host CRC32-IEEE in hash/crc32.Update semantics (gfrs_crc32_host):
+    backs the host-side streaming wrappers below, exactly as the
+    reference's streaming forms run Go stdlib crc32 on the host."""
+    if not data:
+        return crc
+    buf = (ctypes.c_char * len(data)).from_buffer_copy(data)
+    return lib().gfrs_crc32_host(crc, buf, len(data))
+
+
+def _read_full_or_to_end(rc, want):
+    """readFullOrToEnd (util.go:105-117): fill up to `want` bytes,
+    stopping early only at EOF."""
+    parts = []
+    n = 0
+    while n < want:
+        chunk = rc.read(want - n)
+        if not chunk:
+            break
+        parts.append(chunk)
+        n += len(chunk)
+    return b"".join(parts)
 
 
 def encode_size(size, block_len=DEFAULT_BLOCK):
@@ -121,3 +146,138 @@ class Codec:
 
     def synchronize(self):
         check(lib().gfrs_synchronize(self._ctx), "synchronize")
+
+
+def encode_to(reader, limit_size, writer, block_len=DEFAULT_BLOCK):
+    """Encoder.Encode (encode.go:47-57): frame exactly limit_size bytes
+    from `reader` into `writer`; a short source is an error (the
+    reference's io.ReadFull → ReaderError).  Returns bytes written.
+    Host-side streaming plumbing; bulk framing is Codec.encode."""
+    if block_len <= 0 or block_len % 4096:
+        raise GfrsError(-10, "block_len must be a positive 4096-multiple")
+    payload_len = block_len - _CRC_LEN
+    remain = limit_size
+    written = 0
+    while remain > 0:
+        need = min(remain, payload_len)
+        payload = _read_full_or_to_end(reader, need)
+        if len(payload) < need:
+            raise GfrsError(-7, "short source: want %d got %d"
+                            % (need, len(payload)))
+        crc = crc32_host(payload)
+        writer.write(crc.to_bytes(4, "little") + payload)
+        written += _CRC_LEN + len(payload)
+        remain -= len(payload)
+    return written
+
+
+class _BlockReader:
+    """blockReader (decode.go:54-108): framed stream -> payload stream
+    with the remaining payload size known, so the final short block is
+    read exactly (io.ReadFull semantics: a truncated journal errors)."""
+
+    def __init__(self, reader, limit, block_len):
+        self._r = reader
+        self._remain = limit
+        self._block_len = block_len
+        self._block = b""
+        self._i = 0
+
+    def read(self, size=-1):
+        out = []
+        got = 0
+        while (size < 0 or got < size) and not (
+                self._i == len(self._block) and self._remain == 0):
+            if self._i == len(self._block):
+                self._next_block()
+            take = len(self._block) - self._i
+            if size >= 0:
+                take = min(take, size - got)
+            out.append(self._block[self._i:self._i + take])
+            self._i += take
+            self._remain -= take
+            got += take
+        return b"".join(out)
+
+    def _next_block(self):
+        payload_len = self._block_len - _CRC_LEN
+        want = self._block_len
+        if self._remain < payload_len:
+            want = self._remain + _CRC_LEN
+        raw = _read_full_or_to_end(self._r, want)
+        if len(raw) < want:
+            raise GfrsError(-7, "truncated journal: want %d got %d"
+                            % (want, len(raw)))
+        if crc32_host(raw[_CRC_LEN:]) != int.from_bytes(raw[:_CRC_LEN],
+                                                        "little"):
+            raise GfrsError(-9, "mismatched checksum")
+        self._block = raw[_CRC_LEN:]
+        self._i = 0
+
+
+class _RangeReader:
+    """rangeReader (decode.go:46-119): skip into the first block, then
+    limit to the requested span."""
+
+    def __init__(self, reader, skip, limit):
+        self._r = reader
+        self._skip = skip
+        self._limit = limit
+
+    def read(self, size=-1):
+        if self._skip:
+            self._r.read(self._skip)
+            self._skip = 0
+        if size < 0 or size > self._limit:
+            size = self._limit
+        out = self._r.read(size)
+        self._limit -= len(out)
+        return out
+
+
+class Decoder:
+    """Decoder (decode.go:32-135): random-access ranged decode over a
+    crc32block-framed journal — the datanode range-GET shape.  `read_at`
+    is pread semantics: read_at(offset, n) -> bytes (a callable, or any
+    object with a .read_at method or seek+read file API)."""
+
+    def __init__(self, read_at, off, size, block_len=DEFAULT_BLOCK):
+        if block_len <= 0 or block_len % 4096:
+            raise GfrsError(-10, "block_len must be a positive 4096-multiple")
+        if callable(read_at):
+            self._read_at = read_at
+        elif hasattr(read_at, "read_at"):
+            self._read_at = read_at.read_at
+        else:  # seekable file object
+            def _pread(o, n, _f=read_at):
+                _f.seek(o)
+                return _f.read(n)
+            self._read_at = _pread
+        self._off = off
+        self._limit = size       # decoded (payload) size of the journal
+        self._block_len = block_len
+
+    def reader(self, from_, to):
+        """Decoder.Reader (decode.go:121-145): a file-like yielding
+        payload bytes [from_, to) of the decoded journal, touching only
+        the blocks that overlap the range."""
+        payload_len = self._block_len - _CRC_LEN
+        block_off = (from_ // payload_len) * self._block_len
+        encoded_size = encode_size(self._limit, self._block_len) - block_off
+
+        class _Section:
+            def __init__(s):
+                s.pos = 0
+
+            def read(s, n):
+                n = min(n, encoded_size - s.pos)
+                if n <= 0:
+                    return b""
+                out = self._read_at(self._off + block_off + s.pos, n)
+                s.pos += len(out)
+                return out
+
+        payload = _BlockReader(_Section(),
+                               decode_size(encoded_size, self._block_len),
+                               self._block_len)
+        return _RangeReader(payload, from_ % payload_len, to - from_)

@@ -21,33 +21,15 @@ Semantics kept from the reference:
     (request_body.go:141-146); rc=None gives a size-only body
     (request_body.go:155-163).
 """
-import ctypes
-
-from .crc32block import DEFAULT_BLOCK, decode_size, encode_size
-from .runtime import GfrsError, lib
+from .crc32block import (DEFAULT_BLOCK, _read_full_or_to_end, crc32_host,
+                         decode_size, encode_size)
+from .runtime import GfrsError
 
 _CRC_LEN = 4
 
 
 def _crc32(data):
-    if not data:
-        return 0
-    buf = (ctypes.c_char * len(data)).from_buffer_copy(data)
-    return lib().gfrs_crc32_host(0, buf, len(data))
-
-
-def _read_full_or_to_end(rc, want):
-    """readFullOrToEnd (util.go:105-117): fill up to `want` bytes, stopping
-    early only at EOF."""
-    parts = []
-    n = 0
-    while n < want:
-        chunk = rc.read(want - n)
-        if not chunk:
-            break
-        parts.append(chunk)
-        n += len(chunk)
-    return b"".join(parts)
+    return crc32_host(data)
 
 
 class _RequestBody:
