@@ -318,6 +318,8 @@ def main():
 
     flat = batch.view(ns * t.total, S)
 
+    ev_stream = None  # non-default stream the timing events live on, if any
+
     if args.workload == "reconstruct":
         # BASELINE configs[2]: parity must exist, frames pre-built; the
         # step recomputes shard bad_idx for every stripe and CRC-verifies
@@ -327,6 +329,19 @@ def main():
         if with_crc:
             codec.encode_batch(framed, flat)
             codec.synchronize()
+            # the repair read path CRC-checks the fetched frames while the
+            # EC rebuild runs — disjoint buffers, no data dependence (the
+            # reference does the same with per-shard goroutines during the
+            # download phase, worker_slice_recover.go:127-210) — so each
+            # leg gets its own NON-NULL HIP stream (the torch default
+            # stream is the legacy null stream, which serializes against
+            # every other stream) and the two grids co-schedule;
+            # torch.cuda.synchronize at the timed-region edges covers both
+            rec_stream = torch.cuda.Stream(dev)
+            ver_stream = torch.cuda.Stream(dev)
+            lib().gfrs_set_stream(enc._ctx, rec_stream.cuda_stream)
+            lib().gfrs_set_stream(codec._ctx, ver_stream.cuda_stream)
+            ev_stream = rec_stream  # HIP events must sit on the kernel's stream
         bad = [args.bad_idx]
 
     if args.workload == "repair":
@@ -348,7 +363,10 @@ def main():
 
     def step(events=None):
         if events:
-            events[0].record()
+            if ev_stream is not None:
+                events[0].record(ev_stream)
+            else:
+                events[0].record()
         if fused:
             # single-pass: parity + framed images, data read once
             enc.encode_frame_batch(framed, batch)
@@ -365,11 +383,23 @@ def main():
             if events:
                 events[1].record()
         else:
-            enc.reconstruct_batch(batch, bad)
-            if events:
-                events[1].record()
+            # optional split of the rebuild into several queued launches
+            # (measured: within noise of one launch once both legs sit on
+            # non-null streams — co-residence comes from the two HW
+            # queues, not from chunking)
+            ch = int(os.environ.get("GFRS_BENCH_RECON_CHUNK", "0")) or ns
+            for lo in range(0, ns, ch):
+                enc.reconstruct_batch(batch[lo:lo + ch], bad)
             if with_crc:
+                # blocks until the verify stream drains; the trailing
+                # event on the rebuild stream then closes the window over
+                # BOTH overlapped legs (max of the two streams' ends)
                 codec.verify_batch(framed[:, :enc_sz])
+            if events:
+                if ev_stream is not None:
+                    events[1].record(ev_stream)
+                else:
+                    events[1].record()
 
     # warmup
     for _ in range(args.warmup):
@@ -426,7 +456,13 @@ def main():
             alg_bytes = float((t.N + t.M - len(bad)) * S +
                               len(bad) * (S + 4 * fps)) * ns
         elif args.workload != "encode":
+            # reconstruct kernel reads k shards + writes nbad; the CRC
+            # verify leg (overlapped on its own stream) reads all k+m
+            # framed images — the events window spans both legs, so the
+            # roofline covers both legs' algorithmic bytes
             alg_bytes = float((t.N + len(bad)) * S * ns)
+            if with_crc:
+                alg_bytes += float(t.total * (S + 4 * fps) * ns)
         elif fused:
             alg_bytes = float((t.N + t.total) * S + 4 * fps * t.total) * ns
         else:
