@@ -358,30 +358,39 @@ def main():
                            device=dev)
         rbids = np.arange(ns * len(bad), dtype=np.uint64)
         rvuids = np.ones(ns * len(bad), dtype=np.uint64)
+        # non-null stream so the library's finalize-pipeline aux stream
+        # can overlap (the legacy null stream serializes against it)
+        torch.cuda.synchronize(dev)
+        rep_stream = torch.cuda.Stream(dev)
+        lib().gfrs_set_stream(enc._ctx, rep_stream.cuda_stream)
+        ev_stream = rep_stream
 
     fused = (args.workload == "encode" and with_crc and not args.no_fused)
 
+    def _record(ev):
+        if ev_stream is not None:
+            ev.record(ev_stream)
+        else:
+            ev.record()
+
     def step(events=None):
         if events:
-            if ev_stream is not None:
-                events[0].record(ev_stream)
-            else:
-                events[0].record()
+            _record(events[0])
         if fused:
             # single-pass: parity + framed images, data read once
             enc.encode_frame_batch(framed, batch)
             if events:
-                events[1].record()
+                _record(events[1])
         elif args.workload == "encode":
             enc.encode_batch(batch)
             if events:
-                events[1].record()
+                _record(events[1])
             if with_crc:
                 codec.encode_batch(framed, flat)
         elif args.workload == "repair":
             enc.repair_batch(batch, bad, imgs, rbids, rvuids)
             if events:
-                events[1].record()
+                _record(events[1])
         else:
             # optional split of the rebuild into several queued launches
             # (measured: within noise of one launch once both legs sit on
@@ -396,10 +405,7 @@ def main():
                 # BOTH overlapped legs (max of the two streams' ends)
                 codec.verify_batch(framed[:, :enc_sz])
             if events:
-                if ev_stream is not None:
-                    events[1].record(ev_stream)
-                else:
-                    events[1].record()
+                _record(events[1])
 
     # warmup
     for _ in range(args.warmup):
