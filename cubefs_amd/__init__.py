@@ -8,10 +8,17 @@ from scratch as hand-written HIP/CDNA4 kernels behind the reference's own
 blobstore/common/ec.Encoder boundary (C ABI: include/gfrs.h).
 
 Modules:
-  codemode   — CodeMode/Tactic registry (blobstore/common/codemode)
-  ec         — Encoder surface (blobstore/common/ec)
-  crc32block — shard-frame checksum codec (blobstore/common/crc32block)
-  runtime    — ctypes binding of libgfrs.so (no CPU fallback)
+  codemode     — CodeMode/Tactic registry (blobstore/common/codemode)
+  ec           — Encoder surface (blobstore/common/ec)
+  crc32block   — shard-frame checksum codec + ranged journal Decoder
+                 (blobstore/common/crc32block block/encode/decode/util.go)
+  sized_stream — streaming sized/partial rpc2-body coder
+                 (sized_coder.go / sized_coder_block.go)
+  request_body — streaming HTTP-body encoder/decoder (request_body.go)
+  buffer       — ec.Buffer size math (buf.go)
+  shard        — blobnode on-disk shard image codec (core/shard.go)
+  dist         — stripe-queue partitioning + peer shard staging
+  runtime      — ctypes binding of libgfrs.so (no CPU fallback)
 """
 from . import codemode  # noqa: F401
 
