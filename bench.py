@@ -230,7 +230,8 @@ def run_host_streamed(args, t, S, ns, enc, rank, world, d):
     dist.barrier()
     if d is not None:
         import torch.distributed as td
-        te = torch.tensor([el], device="cuda")
+        tdev = "cuda" if td.get_backend() == "nccl" else "cpu"
+        te = torch.tensor([el], device=tdev)
         td.all_reduce(te, op=td.ReduceOp.MAX)
         el = float(te.item())
     src_b = t.N * S * B * nbatches * args.steps * world
@@ -274,8 +275,11 @@ def main():
     rank, world = dist.env_rank_world()
     d = dist.init_process_group()
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    torch.cuda.set_device(local_rank)
-    dev = torch.device("cuda", local_rank)
+    # modulo only matters under GFRS_DIST_BACKEND=gloo smoke runs where
+    # several ranks share one visible GPU; on a real node it is identity
+    local_dev = local_rank % max(1, torch.cuda.device_count())
+    torch.cuda.set_device(local_dev)
+    dev = torch.device("cuda", local_dev)
 
     if args.codemode == "LRC12P2L2":
         codemode.extend(240, "LRC12P2L2",
@@ -285,14 +289,14 @@ def main():
     ns = args.stripes
     with_crc = not args.no_crc
 
-    enc = ec.Encoder(t, device=local_rank)
+    enc = ec.Encoder(t, device=local_dev)
     # run everything on torch's current stream so HIP events bracket the
     # kernels (torch.cuda.Event sees only torch's stream)
     cur = torch.cuda.current_stream(dev).cuda_stream
     lib().gfrs_set_stream(enc._ctx, cur)
     codec = None
     if with_crc:
-        codec = crc32block.Codec(device=local_rank)
+        codec = crc32block.Codec(device=local_dev)
         lib().gfrs_set_stream(codec._ctx, cur)
 
     if args.host_streamed:
@@ -424,8 +428,10 @@ def main():
 
     # MAX over ranks
     if d is not None:
-        te = torch.tensor([elapsed], device=dev)
         import torch.distributed as td
+        # gloo (CPU smoke of the multi-rank path) needs host tensors
+        tdev = dev if td.get_backend() == "nccl" else "cpu"
+        te = torch.tensor([elapsed], device=tdev)
         td.all_reduce(te, op=td.ReduceOp.MAX)
         elapsed = float(te.item())
 

@@ -32,7 +32,10 @@ def init_process_group(backend=None):
     if world == 1:
         return None
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # GFRS_DIST_BACKEND=gloo lets the multi-rank path be exercised
+        # with several ranks sharing one GPU (RCCL forbids that)
+        backend = os.environ.get("GFRS_DIST_BACKEND") or (
+            "nccl" if torch.cuda.is_available() else "gloo")
     if not dist.is_initialized():
         dist.init_process_group(backend=backend)
     if backend == "nccl":
