@@ -2302,8 +2302,8 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_rot_k(
  * from colpack, 4 bits per row); rows nw.. are surviving-parity checks
  * (compared against the shard itself, mismatch -> fail[stripe]).
  * Same op-chain CRC machinery as rs_encode_frame_reg_k. */
-template <int GM>
-__global__ __launch_bounds__(CRC_BLOCKT, 4) void rs_repair_frame_k(
+template <int GM, int WPS = 4>
+__global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_repair_frame_k(
     uint8_t *__restrict__ dst /* first image body (+32 into the image) */,
     size_t dst_stride /* whole-image stride */, uint64_t base,
     uint64_t stripe_stride, size_t shard_len, int k,
@@ -2750,13 +2750,27 @@ void launch_rs_repair_frame(uint8_t *dst, size_t dst_stride, uint64_t base,
   const int64_t total = fps * nstripes;
   const int grid = fused_grid(total, fps);
   const int lds = 12288 + EF_RED + 64 + 256 + gm * k * 32;
-#define GFRS_RP_GO(G)                                                       hipLaunchKernelGGL((rs_repair_frame_k<G>), dim3(grid),                                       dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,                          stripe_stride, shard_len, k, imap, tabs, nw,                              colpack, fail, total, fps)
+  /* GFRS_RP_WAVES: 3 or 4 waves/SIMD (0/unset = size policy).  Measured
+   * @512x4MiB bad=[2,6]: 3-wave 5.565 ms vs 4-wave 5.649; @256x8MiB
+   * 6.457 vs 6.515 (gpurun_out/r3_rpw.log) — same direction as the
+   * encode kernel's traffic-exact 3-wave form, so >=1 MiB shards
+   * default to 3 waves. */
+  static const int waves_env = []() {
+    const char *e = getenv("GFRS_RP_WAVES");
+    const int v = e ? atoi(e) : 0;
+    return (v == 3 || v == 4) ? v : 0;
+  }();
+  const int waves =
+      waves_env ? waves_env : (shard_len >= size_t(1) << 20 ? 3 : 4);
+#define GFRS_RP_GO(G, W)                                                    hipLaunchKernelGGL((rs_repair_frame_k<G, W>), dim3(grid),                                    dim3(CRC_BLOCKT), lds, s, dst, dst_stride, base,                          stripe_stride, shard_len, k, imap, tabs, nw,                              colpack, fail, total, fps)
+#define GFRS_RP_SEL(G)                                                      do {                                                                        if (waves == 3) GFRS_RP_GO(G, 3);                                         else GFRS_RP_GO(G, 4);                                                  } while (0)
   switch (gm) {
-    case 1: GFRS_RP_GO(1); break;
-    case 2: GFRS_RP_GO(2); break;
-    case 3: GFRS_RP_GO(3); break;
-    default: GFRS_RP_GO(4);
+    case 1: GFRS_RP_SEL(1); break;
+    case 2: GFRS_RP_SEL(2); break;
+    case 3: GFRS_RP_SEL(3); break;
+    default: GFRS_RP_SEL(4);
   }
+#undef GFRS_RP_SEL
 #undef GFRS_RP_GO
 }
 
