@@ -71,21 +71,29 @@ def test_section_decoder_stops_at_block_edges():
     assert b"".join(parts) == wb
 
 
-def test_partial_coder_roundtrip():
-    """Nonzero stableSize: the stream opens mid-block with the 512-B
-    head pad (sized_coder.go:166-170) and still round-trips."""
-    for stable, actual in [(100, 5000), (PAYLOAD - 7, 10000),
-                           (PAYLOAD + 513, 3), (512, PAYLOAD * 2)]:
-        wb = _rand(actual, stable)
+def test_partial_coder_roundtrip(oracle):
+    """TestSizedCoderPartial (sized_coder_test.go:463-509): nonzero
+    stableSize opens the stream mid-block with the 512-B head pad
+    (sized_coder.go:166-170); the emitted length equals
+    PartialEncodeSize(actual, stable) — pinned here against the oracle's
+    independent C implementation — and the stream round-trips."""
+    size = (1 << 20) + 17
+    cases = [(0, size), (size - 1, 1), (817374, 1999), (11223344, 2000),
+             (100, 5000), (PAYLOAD - 7, 10000), (PAYLOAD + 513, 3),
+             (512, PAYLOAD * 2)]
+    rng = np.random.default_rng(0x51AB1E)
+    cases += [(int(rng.integers(0, 1 << 20)), int(rng.integers(1, size)))
+              for _ in range(30)]
+    for stable, actual in cases:
+        wb = _rand(actual, stable & 0xFFFF)
         enc = ss.new_partial_encoder(io.BytesIO(wb), actual, stable)
         framed = enc.read()
-        _, tail = ss._partial_tail(actual, stable, ss.DEFAULT_BLOCK)
-        padhead = (stable % PAYLOAD) % 512
+        want_total, _ = oracle.partial_encode_size(actual, stable)
         nx0 = (stable % PAYLOAD) & ~511
-        first_cap = PAYLOAD - nx0 - padhead
-        ncells = 1 + max(0, -(-(actual - first_cap) // PAYLOAD))
-        assert len(framed) == padhead + actual + 4 * ncells + tail, \
-            (stable, actual)
+        # PartialEncodeSize counts the stable part already on the wire
+        # (the `- part` in util.go:78): the fresh stream emits total
+        assert len(framed) == want_total, (stable, actual)
+        assert (nx0 + len(framed)) % 512 == 0  # transport alignment
         dec = ss.new_partial_decoder(io.BytesIO(framed), actual, stable)
         assert dec.read() == wb, (stable, actual)
 
