@@ -33,6 +33,7 @@ _ALIGN = 512  # rpc2 transport.Alignment
 
 MODE_ENCODE = 1
 MODE_DECODE = 4
+MODE_LOAD = 5
 MODE_BLOCK_ENCODE = 11
 MODE_BLOCK_DECODE = 12
 
@@ -56,9 +57,13 @@ class SizedCoder:
                  section):
         if block_len <= 0 or block_len % 4096:
             raise GfrsError(-10, "block_len must be a positive 4096-multiple")
-        if mode not in (MODE_ENCODE, MODE_DECODE, MODE_BLOCK_ENCODE,
-                        MODE_BLOCK_DECODE):
+        if mode not in (MODE_ENCODE, MODE_DECODE, MODE_LOAD,
+                        MODE_BLOCK_ENCODE, MODE_BLOCK_DECODE):
+            # ModeAppend/Check/Fix are WriterTo-side transport machinery
+            # (sized_coder.go:446-) and stay in the Go shim
             raise GfrsError(-103, "unsupported streaming mode %d" % mode)
+        if mode == MODE_LOAD and not section:
+            raise GfrsError(-103, "load mode requires section")
         payload = block_len - _CRC_LEN
         _, tail = _partial_tail(actual_size, stable_size, block_len)
         self._rc = rc
@@ -141,6 +146,47 @@ class SizedCoder:
                 self._padtail = 0
         return data
 
+    def _produce_load(self):
+        """decodeLoad (sized_coder.go:349-413): yields [head-pad ‖ data]
+        per block — the pads the rpc2 server keeps in its aligned
+        buffers — stripping and checking the CRC cell and discarding the
+        final tail pad.  (The Go form additionally demands a 512-aligned
+        caller buffer of exactly one block; a Python bytes return has no
+        such constraint.)"""
+        if self._remain <= 0:
+            self._eof = True
+            return b""
+        want = self._payload + _CRC_LEN - self._nx
+        extra = want - self._padhead - _CRC_LEN - self._padtail - self._remain
+        last = extra >= 0
+        if last:
+            want -= extra
+        raw = _read_full_or_to_end(self._rc, want)
+        if len(raw) < want:
+            raise GfrsError(-7, "short load block: want %d got %d"
+                            % (want, len(raw)))
+        out_head = b""
+        if self._padhead:
+            out_head = raw[:self._padhead]
+            self._nx += self._padhead
+            raw = raw[self._padhead:]
+            self._padhead = 0
+        n = len(raw) - _CRC_LEN
+        if last:
+            n -= self._padtail
+            self._padtail = 0
+        cell = raw[n:n + _CRC_LEN]
+        data = raw[:n]
+        self._crc = crc32_host(data, self._crc)
+        self._nx += n
+        self._remain -= n
+        if self._nx == self._payload or self._remain == 0:
+            if int.from_bytes(cell, "big") != self._crc:
+                raise GfrsError(-9, "mismatched checksum")
+            self._crc = 0
+            self._nx = 0
+        return out_head + data
+
     def _produce_block(self, decode):
         if self._remain <= 0:
             self._eof = True
@@ -179,6 +225,10 @@ class SizedCoder:
                     elif self._mode == MODE_DECODE:
                         self._buf = self._produce_decode()
                         # a completed interior block is a section edge
+                        boundary = (self._section and self._nx == 0
+                                    and self._remain > 0)
+                    elif self._mode == MODE_LOAD:
+                        self._buf = self._produce_load()
                         boundary = (self._section and self._nx == 0
                                     and self._remain > 0)
                     else:

@@ -175,3 +175,48 @@ def test_invalid_mode_and_block():
         ss.new_sized_coder(io.BytesIO(b""), 0, block_len=1000)
     with pytest.raises(GfrsError, match="Unsupported"):
         ss.new_sized_coder(io.BytesIO(b""), 0, mode=2)
+
+
+def test_load_mode():
+    """ModeLoad (sized_coder.go:349-413): per-block [head-pad ‖ data]
+    with CRC cells stripped+checked and the tail pad discarded; requires
+    section, one block per read."""
+    size = PAYLOAD * 2 + 333
+    wb = _rand(size, 21)
+    enc = ss.new_sized_encoder(io.BytesIO(wb), size)
+    ld = ss.new_sized_coder(enc, size, mode=ss.MODE_LOAD, section=True)
+    parts = []
+    while True:
+        c = ld.read()
+        if not c:
+            break
+        parts.append(c)
+    assert [len(p) for p in parts] == [PAYLOAD, PAYLOAD, 333]
+    assert b"".join(parts) == wb
+
+    # nonzero stable: the first returned block carries the head pad
+    stable, actual = PAYLOAD - 7, 10000
+    wb = _rand(actual, 22)
+    framed = ss.new_partial_encoder(io.BytesIO(wb), actual, stable).read()
+    padhead = (stable % PAYLOAD) % 512
+    ld = ss.new_sized_coder(io.BytesIO(framed), actual, stable,
+                            mode=ss.MODE_LOAD, section=True)
+    out = bytearray()
+    while True:
+        c = ld.read()
+        if not c:
+            break
+        out += c
+    assert bytes(out[padhead:]) == wb
+
+    # corruption caught
+    bad = bytearray(ss.new_sized_encoder(io.BytesIO(wb), actual).read())
+    bad[100] ^= 1
+    ld = ss.new_sized_coder(io.BytesIO(bytes(bad)), actual,
+                            mode=ss.MODE_LOAD, section=True)
+    with pytest.raises(GfrsError, match="MismatchedCrc"):
+        while ld.read():
+            pass
+    with pytest.raises(GfrsError, match="Unsupported"):
+        ss.new_sized_coder(io.BytesIO(b""), 0, mode=ss.MODE_LOAD,
+                           section=False)
