@@ -182,14 +182,21 @@ class _BlockReader:
         self._block_len = block_len
         self._block = b""
         self._i = 0
+        self._err = None
 
     def read(self, size=-1):
+        if self._err is not None:
+            raise self._err  # sticky, like blockReader.err (decode.go:58)
         out = []
         got = 0
         while (size < 0 or got < size) and not (
                 self._i == len(self._block) and self._remain == 0):
             if self._i == len(self._block):
-                self._next_block()
+                try:
+                    self._next_block()
+                except GfrsError as e:
+                    self._err = e
+                    raise
             take = len(self._block) - self._i
             if size >= 0:
                 take = min(take, size - got)

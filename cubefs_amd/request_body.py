@@ -63,10 +63,11 @@ class _RequestBody:
                     if not self._next_block():
                         break  # clean EOF
                 except GfrsError as e:
+                    # errors are sticky (reference r.err, request_body.go:46)
+                    self._err = e
                     if got:
                         # reference Read returns (n, nil) and surfaces the
                         # error on the next call (request_body.go:64-68)
-                        self._err = e
                         break
                     raise
             take = len(self._block) - self._off

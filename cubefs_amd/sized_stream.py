@@ -236,8 +236,8 @@ class SizedCoder:
                             self._mode == MODE_BLOCK_DECODE)
                         boundary = self._section and self._remain > 0
                 except GfrsError as e:
+                    self._err = e  # sticky, like the reference's r.err
                     if got:
-                        self._err = e
                         break
                     raise
                 self._off = 0
