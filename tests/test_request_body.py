@@ -130,3 +130,17 @@ def test_host_crc_matches_oracle(oracle):
     c = lib().gfrs_crc32_host(c, buf[2000:], 3000)
     assert c == oracle.crc32(a)
     assert request_body._crc32(b"123456789") == 0xCBF43926
+
+
+def test_errors_are_sticky():
+    """After a decode error every later read raises the same error
+    (requestBody.err, request_body.go:46,58-60)."""
+    wb = _rand(1 << 12, 31)
+    enc = request_body.body_encoder(io.BytesIO(wb))
+    framed = bytearray(enc.read())
+    framed[10] ^= 1
+    dec = request_body.body_decoder(io.BytesIO(bytes(framed)))
+    with pytest.raises(GfrsError, match="MismatchedCrc"):
+        dec.read()
+    with pytest.raises(GfrsError, match="MismatchedCrc"):
+        dec.read(1)
