@@ -417,6 +417,7 @@ def test_repair_batch_lrc_images(oracle, dev, bad):
     of the original shards."""
     import torch
     from cubefs_amd import codemode, ec, shard
+    codemode.extend(240, "LRC12P2L2", codemode.Tactic(12, 2, 2, 2, 14, 0, 2048))
     t = codemode.get_tactic("LRC12P2L2")
     enc = ec.Encoder(t)
     ns, slen = 3, 100_000
@@ -449,6 +450,7 @@ def test_repair_batch_lrc_local_check_detects(oracle, dev):
     the surviving LOCAL parities do: corruption is still detected."""
     import torch
     from cubefs_amd import codemode, ec, shard
+    codemode.extend(240, "LRC12P2L2", codemode.Tactic(12, 2, 2, 2, 14, 0, 2048))
     t = codemode.get_tactic("LRC12P2L2")
     enc = ec.Encoder(t)
     ns, slen = 4, 80_000
@@ -477,6 +479,7 @@ def test_lrc_reconstruct_verify(oracle, dev, bad):
     and corruption detected through global AND local check equations."""
     import torch
     from cubefs_amd import codemode, ec
+    codemode.extend(240, "LRC12P2L2", codemode.Tactic(12, 2, 2, 2, 14, 0, 2048))
     t = codemode.get_tactic("LRC12P2L2")
     enc = ec.Encoder(t)
     ns, slen = 4, 120_000
