@@ -1058,8 +1058,17 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
  * slab, and lane 0 compares the 4 B LE header.  The LDS-staged verify
  * kernel measured 2.2 TB/s (stage round-trip + per-chunk barriers on a
  * read-only job); this form is pure streaming reads. */
-template <int MODE> /* 1 = verify, 2 = decode (framed -> raw + verify) */
-__global__ __launch_bounds__(CRC_BLOCKT, 8) void crc32b_verify_reg_k(
+/* PIPE=1 adds the fused kernel's one-pass load lookahead (pass h+1's four
+ * uint4 pieces issued before pass h's Horner chains consume theirs).
+ * Costs 16 VGPRs of buffer, so PIPE variants instantiate at WPS < 8 —
+ * and MEASURED SLOWER for exactly that reason (reconstruct workload
+ * 8.38 ms plain @8 waves vs 8.73 @6 / 8.77 @4, gpurun_out/r3_vrfy_ab.log):
+ * 8-way wave interleaving already hides the chain latency, so the
+ * occupancy give-back buys nothing here (unlike the fused encode kernel,
+ * whose MAC work left it at 3-4 waves anyway).  Kept as a measured
+ * variant (GFRS_VRFY=16/14); default is plain @8. */
+template <int MODE, int PIPE = 0, int WPS = 8>
+__global__ __launch_bounds__(CRC_BLOCKT, WPS) void crc32b_verify_reg_k(
     uint8_t *__restrict__ dst, size_t dst_stride,
     const uint8_t *__restrict__ src, size_t src_stride, int64_t n,
     int64_t fps, int64_t total, int64_t *bad) {
@@ -1109,11 +1118,38 @@ __global__ __launch_bounds__(CRC_BLOCKT, 8) void crc32b_verify_reg_k(
     if (threadIdx.x < 4) red[threadIdx.x] = 0;
     __syncthreads();
 
+    uint4 vbuf[PIPE ? 4 : 1];
+    if (PIPE) {
+      const int rb0 = int(i64min(int64_t(EF_PASS), payload));
+#pragma unroll
+      for (int i = 0; i < (PIPE ? 4 : 1); i++) {
+        const int off = i * 4096 + lane16i;
+        vbuf[i] = off + 16 <= rb0
+                      ? *reinterpret_cast<const uint4 *>(pb + off)
+                      : uint4{0, 0, 0, 0};
+      }
+    }
     uint32_t op_chain = op_first;
     for (int h = 0; h < EF_PASSES; h++) {
       const int64_t r0 = int64_t(h) * EF_PASS;
       const int64_t rbytes = i64min(int64_t(EF_PASS), payload - r0);
       if (rbytes <= 0) break;
+      uint4 vcur[PIPE ? 4 : 1];
+      if (PIPE) {
+#pragma unroll
+        for (int i = 0; i < (PIPE ? 4 : 1); i++) vcur[i] = vbuf[i];
+        if (h + 1 < EF_PASSES) {
+          const int64_t rn0 = int64_t(h + 1) * EF_PASS;
+          const int rbn = int(i64min(int64_t(EF_PASS), payload - rn0));
+#pragma unroll
+          for (int i = 0; i < (PIPE ? 4 : 1); i++) {
+            const int off = i * 4096 + lane16i;
+            vbuf[i] = (rbn > 0 && off + 16 <= rbn)
+                          ? *reinterpret_cast<const uint4 *>(pb + rn0 + off)
+                          : uint4{0, 0, 0, 0};
+          }
+        }
+      }
       uint32_t op = op_chain;
       if (h == EF_PASSES - 1 && threadIdx.x == 255)
         op = shift4k(op, stabS); /* lane 255's last pass has 3 pieces */
@@ -1137,7 +1173,9 @@ __global__ __launch_bounds__(CRC_BLOCKT, 8) void crc32b_verify_reg_k(
       for (int i = 0; i < 2; i++) {
         const int off = i * 4096 + lane16i;
         if (off + 16 <= rbi) {
-          const uint4 v = *reinterpret_cast<const uint4 *>(pb + r0 + off);
+          const uint4 v =
+              PIPE ? vcur[PIPE ? i : 0]
+                   : *reinterpret_cast<const uint4 *>(pb + r0 + off);
           if (MODE == 2) *reinterpret_cast<uint4 *>(ob + r0 + off) = v;
           tA = shift4k(tA, stabS) ^ crc16_reg(v, tabS);
         }
@@ -1146,7 +1184,9 @@ __global__ __launch_bounds__(CRC_BLOCKT, 8) void crc32b_verify_reg_k(
       for (int i = 2; i < 4; i++) {
         const int off = i * 4096 + lane16i;
         if (off + 16 <= rbi) {
-          const uint4 v = *reinterpret_cast<const uint4 *>(pb + r0 + off);
+          const uint4 v =
+              PIPE ? vcur[PIPE ? i : 0]
+                   : *reinterpret_cast<const uint4 *>(pb + r0 + off);
           if (MODE == 2) *reinterpret_cast<uint4 *>(ob + r0 + off) = v;
           tB = shift4k(tB, stabS) ^ crc16_reg(v, tabS);
           nB++;
@@ -1207,9 +1247,25 @@ static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
   }();
   if ((MODE == 1 || MODE == 2) && !TAILCRC && block_len == 65536) {
     const int g = fused_grid(total, fps);
-    hipLaunchKernelGGL((crc32b_verify_reg_k<MODE>), dim3(g),
-                       dim3(CRC_BLOCKT), 0, s, dst, dst_stride, src,
-                       src_stride, n, fps, total, bad);
+    /* GFRS_VRFY: 0 = plain 8-wave (default), 16 = one-pass load
+     * lookahead @6 waves, 14 = lookahead @4 waves */
+    static const int vsel = []() {
+      const char *e = getenv("GFRS_VRFY");
+      const int v = e ? atoi(e) : 0;
+      return (v == 16 || v == 14) ? v : 0;
+    }();
+    if (vsel == 16)
+      hipLaunchKernelGGL((crc32b_verify_reg_k<MODE, 1, 6>), dim3(g),
+                         dim3(CRC_BLOCKT), 0, s, dst, dst_stride, src,
+                         src_stride, n, fps, total, bad);
+    else if (vsel == 14)
+      hipLaunchKernelGGL((crc32b_verify_reg_k<MODE, 1, 4>), dim3(g),
+                         dim3(CRC_BLOCKT), 0, s, dst, dst_stride, src,
+                         src_stride, n, fps, total, bad);
+    else
+      hipLaunchKernelGGL((crc32b_verify_reg_k<MODE>), dim3(g),
+                         dim3(CRC_BLOCKT), 0, s, dst, dst_stride, src,
+                         src_stride, n, fps, total, bad);
     return;
   }
   if (block_len == 65536) {
