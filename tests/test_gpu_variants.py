@@ -65,3 +65,43 @@ def test_encode_frame_variant_parity(ef):
     r = subprocess.run([sys.executable, "-c", _SCRIPT], env=env, cwd=_REPO,
                        capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, (ef, r.stdout[-2000:], r.stderr[-2000:])
+
+
+_VRFY_SCRIPT = r"""
+import numpy as np
+import torch
+from cubefs_amd import crc32block
+from cubefs_amd.runtime import GfrsError
+from oracle import pyoracle as oracle
+
+codec = crc32block.Codec()
+rng = np.random.default_rng(0x5EC1FF)
+for slen in (100000, 65532, 2 * 65532 + 16, 1 << 20, 9000):
+    src = rng.integers(0, 256, slen, dtype=np.uint8)
+    framed_np = oracle.crc32b_encode(src.copy())
+    framed = torch.from_numpy(framed_np.copy()).to("cuda:0")
+    assert codec.verify(framed) == -1, slen
+    # decode (framed -> raw + verify) round-trips
+    out = torch.zeros(slen, dtype=torch.uint8, device="cuda:0")
+    n = codec.decode(out, framed)
+    assert n == slen
+    assert np.array_equal(out.cpu().numpy(), src), slen
+    # single flipped bit in each block position band is caught
+    for pos in (5, len(framed_np) // 2, len(framed_np) - 1):
+        bad = framed_np.copy()
+        bad[pos] ^= 1
+        fb = torch.from_numpy(bad).to("cuda:0")
+        assert codec.verify(fb) == pos // 65536, (slen, pos)
+print("vrfy variant parity OK")
+"""
+
+
+@pytest.mark.parametrize("v", ["16", "14"])
+def test_verify_variant_parity(v):
+    """GFRS_VRFY load-lookahead variants of crc32b_verify_reg_k are
+    bit-identical to the default (kept in-tree as measured variants)."""
+    env = dict(os.environ, GFRS_VRFY=v)
+    r = subprocess.run([sys.executable, "-c", _VRFY_SCRIPT], env=env,
+                       cwd=_REPO, capture_output=True, text=True,
+                       timeout=300)
+    assert r.returncode == 0, (v, r.stdout[-2000:], r.stderr[-2000:])
