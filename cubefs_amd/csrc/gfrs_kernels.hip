@@ -1067,7 +1067,12 @@ __global__ __launch_bounds__(CRC_BLOCKT) void crc32b_staged_k(
  * occupancy give-back buys nothing here (unlike the fused encode kernel,
  * whose MAC work left it at 3-4 waves anyway).  Kept as a measured
  * variant (GFRS_VRFY=16/14); default is plain @8. */
-template <int MODE, int PIPE = 0, int WPS = 8>
+/* SKEL (measurement-only, wrong CRC results by design — never reachable
+ * except via GFRS_VRFY=90/91): 1 = raw XOR instead of all CRC math (the
+ * kernel-structure memory floor), 2 = CRC on pieces 0-1 only, raw XOR on
+ * 2-3 (half the LDS-gather work) — separates the LDS-pipe cost from the
+ * stream cost. */
+template <int MODE, int PIPE = 0, int WPS = 8, int SKEL = 0>
 __global__ __launch_bounds__(CRC_BLOCKT, WPS) void crc32b_verify_reg_k(
     uint8_t *__restrict__ dst, size_t dst_stride,
     const uint8_t *__restrict__ src, size_t src_stride, int64_t n,
@@ -1177,7 +1182,10 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void crc32b_verify_reg_k(
               PIPE ? vcur[PIPE ? i : 0]
                    : *reinterpret_cast<const uint4 *>(pb + r0 + off);
           if (MODE == 2) *reinterpret_cast<uint4 *>(ob + r0 + off) = v;
-          tA = shift4k(tA, stabS) ^ crc16_reg(v, tabS);
+          if (SKEL == 1)
+            tA ^= v.x ^ v.y ^ v.z ^ v.w;
+          else
+            tA = shift4k(tA, stabS) ^ crc16_reg(v, tabS);
         }
       }
 #pragma unroll
@@ -1188,7 +1196,10 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void crc32b_verify_reg_k(
               PIPE ? vcur[PIPE ? i : 0]
                    : *reinterpret_cast<const uint4 *>(pb + r0 + off);
           if (MODE == 2) *reinterpret_cast<uint4 *>(ob + r0 + off) = v;
-          tB = shift4k(tB, stabS) ^ crc16_reg(v, tabS);
+          if (SKEL >= 1)
+            tB ^= v.x ^ v.y ^ v.z ^ v.w;
+          else
+            tB = shift4k(tB, stabS) ^ crc16_reg(v, tabS);
           nB++;
         }
       }
@@ -1252,9 +1263,17 @@ static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
     static const int vsel = []() {
       const char *e = getenv("GFRS_VRFY");
       const int v = e ? atoi(e) : 0;
-      return (v == 16 || v == 14) ? v : 0;
+      return (v == 16 || v == 14 || v == 90 || v == 91) ? v : 0;
     }();
-    if (vsel == 16)
+    if (vsel == 90) /* measurement skeleton: see kernel comment */
+      hipLaunchKernelGGL((crc32b_verify_reg_k<MODE, 0, 8, 1>), dim3(g),
+                         dim3(CRC_BLOCKT), 0, s, dst, dst_stride, src,
+                         src_stride, n, fps, total, bad);
+    else if (vsel == 91)
+      hipLaunchKernelGGL((crc32b_verify_reg_k<MODE, 0, 8, 2>), dim3(g),
+                         dim3(CRC_BLOCKT), 0, s, dst, dst_stride, src,
+                         src_stride, n, fps, total, bad);
+    else if (vsel == 16)
       hipLaunchKernelGGL((crc32b_verify_reg_k<MODE, 1, 6>), dim3(g),
                          dim3(CRC_BLOCKT), 0, s, dst, dst_stride, src,
                          src_stride, n, fps, total, bad);

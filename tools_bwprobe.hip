@@ -163,6 +163,58 @@ static void run_mix69_pb(const char *name, MixPtrs p, size_t n16,
          15.0 * double(blk_n16) * 2048 * 16 / (ms / reps / 1e3) / 1e9);
 }
 
+/* verify-kernel read-pattern probe: one 256-thread workgroup per 64 KiB
+ * frame at phase MIS, 4 passes of 4x4096-strided uint4 per lane,
+ * grid-strided over frames, two block barriers per frame — the exact
+ * access shape of crc32b_verify_reg_k minus the CRC math.  Answers
+ * whether that kernel sits at its access pattern's memory ceiling. */
+template <int MIS>
+__global__ void k_vrfy_read(const uint8_t *__restrict__ a, int64_t total,
+                            uint32_t *sink) {
+  __shared__ uint32_t red[4];
+  uint32_t acc = 0;
+  const int lane16 = threadIdx.x * 16;
+  for (int64_t fr = blockIdx.x; fr < total; fr += gridDim.x) {
+    const uint8_t *pb = a + fr * 65536 + MIS;
+    if (threadIdx.x < 4) red[threadIdx.x] = 0;
+    __syncthreads();
+#pragma unroll
+    for (int h = 0; h < 4; h++) {
+#pragma unroll
+      for (int i = 0; i < 4; i++) {
+        const uint4 v = *reinterpret_cast<const uint4 *>(
+            pb + h * 16384 + i * 4096 + lane16);
+        acc ^= v.x ^ v.y ^ v.z ^ v.w;
+      }
+    }
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] ^= acc;
+    __syncthreads();
+  }
+  if (acc == 0xDEADBEEFu) *sink = red[0];
+}
+
+template <int MIS>
+static void run_vrfy_read(const char *name, const uint8_t *a, int64_t total,
+                          uint32_t *sink, int grid) {
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  const int reps = 6;
+  hipLaunchKernelGGL((k_vrfy_read<MIS>), dim3(grid), dim3(256), 0, 0, a,
+                     total, sink);
+  hipDeviceSynchronize();
+  hipEventRecord(e0);
+  for (int r = 0; r < reps; r++)
+    hipLaunchKernelGGL((k_vrfy_read<MIS>), dim3(grid), dim3(256), 0, 0, a,
+                       total, sink);
+  hipEventRecord(e1);
+  hipEventSynchronize(e1);
+  float ms;
+  hipEventElapsedTime(&ms, e0, e1);
+  printf("%-14s %8.1f GB/s (frame-grid read, grid %d)\n", name,
+         double(total) * 65536 / (ms / reps / 1e3) / 1e9, grid);
+}
+
 template <int ST, int MISAL>
 static void run_mix69(const char *name, MixPtrs p, size_t n16, size_t bytes) {
   hipEvent_t e0, e1;
@@ -292,6 +344,17 @@ int main() {
     run_mix69_pb<0, 1, 1>("pb_rwmis", p, sn16 - 1, sb);
     run_mix69_pb<1, 1, 1>("pb_rwmis_nt", p, sn16 - 1, sb);
     run_mix69_pb<1, 0, 1>("pb_wmis_nt", p, sn16 - 1, sb);
+  }
+  {
+    /* crc32b_verify_reg read-pattern: frames over the first 2 GiB of a
+     * (slack for lane 255's +4-phase overhang) */
+    const int64_t vt = (int64_t(2) << 30) / 65536;
+    uint32_t *sink = reinterpret_cast<uint32_t *>(b);
+    const uint8_t *va = reinterpret_cast<const uint8_t *>(a);
+    run_vrfy_read<4>("vrfy_rd+4", va, vt, sink, 2048);
+    run_vrfy_read<0>("vrfy_rd+0", va, vt, sink, 2048);
+    run_vrfy_read<4>("vrfy_rd+4g8", va, vt, sink, 8192);
+    run_vrfy_read<4>("vrfy_rd+4gT", va, vt, sink, int(vt));
   }
   return 0;
 }
