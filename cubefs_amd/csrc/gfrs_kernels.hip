@@ -1108,9 +1108,16 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void crc32b_verify_reg_k(
       gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
   __syncthreads();
 
-  for (int64_t fr = blockIdx.x; fr < total; fr += gridDim.x) {
-    const int64_t shard = fr / fps;
-    const int64_t f = fr - shard * fps;
+  /* track (shard, f) incrementally — one i64 division per block instead
+   * of a ~100-instruction software divide serialized ahead of every
+   * frame's first load (same fix as rs_encode_frame_reg_k) */
+  int64_t shard = int64_t(blockIdx.x) / fps;
+  int64_t f = int64_t(blockIdx.x) - shard * fps;
+  const int64_t dshard = int64_t(gridDim.x) / fps;
+  const int64_t drem = int64_t(gridDim.x) - dshard * fps;
+  for (int64_t fr = blockIdx.x; fr < total; fr += gridDim.x,
+               shard += dshard, f += drem,
+               (f >= fps ? (f -= fps, ++shard) : int64_t(0))) {
     const int64_t p0 = f * payload_full;
     const int64_t payload = i64min(payload_full, n - p0);
     const uint8_t *fb = as_global(uint64_t(src) + shard * src_stride) +
