@@ -1270,9 +1270,14 @@ static void crc_dispatch(uint8_t *dst, size_t dst_stride, const uint8_t *src,
     static const int vsel = []() {
       const char *e = getenv("GFRS_VRFY");
       const int v = e ? atoi(e) : 0;
-      return (v == 16 || v == 14 || v == 90 || v == 91) ? v : 0;
+      return (v == 16 || v == 14 || v == 18 || v == 90 || v == 91) ? v : 0;
     }();
-    if (vsel == 90) /* measurement skeleton: see kernel comment */
+    if (vsel == 18) /* lookahead at full 8 waves (base kernel is only 44
+                       VGPRs, so the 16-VGPR buffer still fits 64) */
+      hipLaunchKernelGGL((crc32b_verify_reg_k<MODE, 1, 8>), dim3(g),
+                         dim3(CRC_BLOCKT), 0, s, dst, dst_stride, src,
+                         src_stride, n, fps, total, bad);
+    else if (vsel == 90) /* measurement skeleton: see kernel comment */
       hipLaunchKernelGGL((crc32b_verify_reg_k<MODE, 0, 8, 1>), dim3(g),
                          dim3(CRC_BLOCKT), 0, s, dst, dst_stride, src,
                          src_stride, n, fps, total, bad);
