@@ -1104,6 +1104,10 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void crc32b_verify_reg_k(
   const int lane16i = int(threadIdx.x) * 16;
   const uint32_t op_first =
       x8n_d(uint64_t(payload_full - (3 * 4096 + lane16 + 16)));
+  const uint32_t op_p0 = op_first;
+  const uint32_t op_p1 = gf2_mulmod_d(op_p0, INV16K);
+  const uint32_t op_p2 = gf2_mulmod_d(op_p1, INV16K);
+  const uint32_t op_p3 = gf2_mulmod_d(op_p2, INV16K);
   const uint32_t it_full =
       gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
   __syncthreads();
@@ -1141,7 +1145,6 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void crc32b_verify_reg_k(
                       : uint4{0, 0, 0, 0};
       }
     }
-    uint32_t op_chain = op_first;
     for (int h = 0; h < EF_PASSES; h++) {
       const int64_t r0 = int64_t(h) * EF_PASS;
       const int64_t rbytes = i64min(int64_t(EF_PASS), payload - r0);
@@ -1162,10 +1165,15 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void crc32b_verify_reg_k(
           }
         }
       }
-      uint32_t op = op_chain;
+      /* full-frame per-pass operators are frame-independent: op_p0..3
+       * were chained once in the prologue, replacing a ~96-VALU mulmod
+       * sequence per pass per frame with a 3-select pick */
+      uint32_t op = h == 0   ? op_p0
+                    : h == 1 ? op_p1
+                    : h == 2 ? op_p2
+                             : op_p3;
       if (h == EF_PASSES - 1 && threadIdx.x == 255)
         op = shift4k(op, stabS); /* lane 255's last pass has 3 pieces */
-      op_chain = gf2_mulmod_d(op_chain, INV16K);
       if (payload != payload_full) {
         int np = 0;
 #pragma unroll
