@@ -1663,6 +1663,13 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
   constexpr uint32_t INV16K = 0x479933FCu;
   const uint32_t op_first =
       x8n_d(uint64_t(payload_full - (3 * 4096 + lane16 + 16)));
+  /* full-frame per-pass operators are frame-independent: chain them once
+   * here (a 3-select pick in the pass loop) instead of a ~96-VALU serial
+   * mulmod per pass per frame */
+  const uint32_t op_pA = op_first;
+  const uint32_t op_pB = gf2_mulmod_d(op_pA, INV16K);
+  const uint32_t op_pC = gf2_mulmod_d(op_pB, INV16K);
+  const uint32_t op_pD = gf2_mulmod_d(op_pC, INV16K);
   const uint32_t it_full =
       gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
   __syncthreads();
@@ -1719,7 +1726,6 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
     for (int j = threadIdx.x; j < 64; j += CRC_BLOCKT) red[j] = 0;
     __syncthreads();
 
-    uint32_t op_chain = op_first;
     for (int h = 0; h < EF_PASSES; h++) {
       const int64_t r0 = int64_t(h) * EF_PASS;
       const int64_t rbytes = i64min(int64_t(EF_PASS), payload - r0);
@@ -1729,10 +1735,12 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_encode_frame_reg_k(
 #pragma unroll
         for (int i = 0; i < 4; i++) acc[r][i] = uint4{0, 0, 0, 0};
 
-      uint32_t op = op_chain;
+      uint32_t op = h == 0   ? op_pA
+                    : h == 1 ? op_pB
+                    : h == 2 ? op_pC
+                             : op_pD;
       if (h == EF_PASSES - 1 && threadIdx.x == 255)
         op = shift4k(op, stab); /* lane 255's last pass has 3 pieces */
-      op_chain = gf2_mulmod_d(op_chain, INV16K);
       if (payload != payload_full) {
         int np = 0;
 #pragma unroll
@@ -2439,6 +2447,10 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_repair_frame_k(
   constexpr uint32_t INV16K = 0x479933FCu;
   const uint32_t op_first =
       x8n_d(uint64_t(payload_full - (3 * 4096 + lane16 + 16)));
+  const uint32_t op_pA = op_first;
+  const uint32_t op_pB = gf2_mulmod_d(op_pA, INV16K);
+  const uint32_t op_pC = gf2_mulmod_d(op_pB, INV16K);
+  const uint32_t op_pD = gf2_mulmod_d(op_pC, INV16K);
   const uint32_t it_full =
       gf2_mulmod_d(x8n_d(uint64_t(payload_full)), 0xFFFFFFFFu);
   __syncthreads();
@@ -2473,7 +2485,6 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_repair_frame_k(
     __syncthreads();
 
     uint32_t mismatch = 0;
-    uint32_t op_chain = op_first;
     for (int h = 0; h < EF_PASSES; h++) {
       const int64_t r0 = int64_t(h) * EF_PASS;
       const int64_t rbytes = i64min(int64_t(EF_PASS), payload - r0);
@@ -2483,10 +2494,12 @@ __global__ __launch_bounds__(CRC_BLOCKT, WPS) void rs_repair_frame_k(
 #pragma unroll
         for (int i = 0; i < 4; i++) acc[r][i] = uint4{0, 0, 0, 0};
 
-      uint32_t op = op_chain;
+      uint32_t op = h == 0   ? op_pA
+                    : h == 1 ? op_pB
+                    : h == 2 ? op_pC
+                             : op_pD;
       if (h == EF_PASSES - 1 && threadIdx.x == 255)
         op = shift4k(op, stab);
-      op_chain = gf2_mulmod_d(op_chain, INV16K);
       if (payload != payload_full) {
         int np = 0;
 #pragma unroll
