@@ -96,7 +96,7 @@ print("vrfy variant parity OK")
 """
 
 
-@pytest.mark.parametrize("v", ["16", "14"])
+@pytest.mark.parametrize("v", ["10", "16", "14", "18"])
 def test_verify_variant_parity(v):
     """GFRS_VRFY load-lookahead variants of crc32b_verify_reg_k are
     bit-identical to the default (kept in-tree as measured variants)."""
